@@ -1,0 +1,243 @@
+// Python bindings for the MI355X-native extension: RCCL comm core
+// (csrc/comm/rccl_comm.hip) + CDNA4 HIP kernels (csrc/kernels/kernels.hip).
+
+#include <hip/hip_runtime.h>
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include <cstdint>
+
+namespace epl {
+void register_comm(py::module& m);
+}
+
+extern "C" {
+void epl_fused_adamw(float*, unsigned short*, const void*, bool, float*,
+                     float*, int64_t, float, float, float, float, float, int,
+                     float, hipStream_t);
+void epl_lamb_phase1(const float*, const void*, bool, float*, float*, float*,
+                     const int*, float*, float*, int64_t, float, float, float,
+                     float, int, float, hipStream_t);
+void epl_lamb_phase2(float*, unsigned short*, const float*, const int*,
+                     const float*, int64_t, float, hipStream_t);
+void epl_layer_norm_fwd(void*, const void*, const void*, const void*, float*,
+                        float*, int64_t, int64_t, float, bool, hipStream_t);
+void epl_layer_norm_bwd(void*, float*, float*, const void*, const void*,
+                        const void*, const float*, const float*, int64_t,
+                        int64_t, bool, hipStream_t);
+void epl_bias_gelu_fwd(void*, const void*, const void*, int64_t, int64_t,
+                       bool, hipStream_t);
+void epl_bias_gelu_bwd(void*, float*, const void*, const void*, const void*,
+                       int64_t, int64_t, bool, hipStream_t);
+void epl_ce_rowstats(const void*, const int64_t*, float*, float*, float*,
+                     int64_t, int64_t, int64_t, int64_t, bool, hipStream_t);
+void epl_ce_backward(void*, const void*, const int64_t*, const float*,
+                     const float*, const float*, int64_t, int64_t, int64_t,
+                     int64_t, float, bool, hipStream_t);
+void epl_scale(void*, int64_t, float, bool, hipStream_t);
+void epl_f32_to_bf16(unsigned short*, const float*, int64_t, hipStream_t);
+void epl_bf16_to_f32(float*, const unsigned short*, int64_t, hipStream_t);
+void epl_sqnorm(const void*, int64_t, float*, bool, hipStream_t);
+}
+
+namespace {
+
+hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+void check(const at::Tensor& t, at::ScalarType dtype, const char* name) {
+  TORCH_CHECK(t.is_cuda(), name, " must be a device tensor");
+  TORCH_CHECK(t.is_contiguous(), name, " must be contiguous");
+  TORCH_CHECK(t.scalar_type() == dtype, name, " has dtype ", t.scalar_type(),
+              ", expected ", dtype);
+}
+
+bool is_bf16(const at::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16 || t.scalar_type() == at::kFloat,
+              "expected bf16 or fp32, got ", t.scalar_type());
+  return t.scalar_type() == at::kBFloat16;
+}
+
+void fused_adamw(at::Tensor master, c10::optional<at::Tensor> param_bf16,
+                 at::Tensor grad, at::Tensor m, at::Tensor v, double lr,
+                 double beta1, double beta2, double eps, double weight_decay,
+                 int64_t step, double inv_scale) {
+  check(master, at::kFloat, "master");
+  check(m, at::kFloat, "m");
+  check(v, at::kFloat, "v");
+  const bool gbf = is_bf16(grad);
+  unsigned short* pb = nullptr;
+  if (param_bf16.has_value()) {
+    check(*param_bf16, at::kBFloat16, "param_bf16");
+    TORCH_CHECK(param_bf16->numel() == master.numel());
+    pb = reinterpret_cast<unsigned short*>(param_bf16->data_ptr());
+  }
+  const int64_t n = master.numel();
+  TORCH_CHECK(grad.numel() == n && m.numel() == n && v.numel() == n,
+              "arena size mismatch");
+  epl_fused_adamw(master.data_ptr<float>(), pb, grad.data_ptr(), gbf,
+                  m.data_ptr<float>(), v.data_ptr<float>(), n, (float)lr,
+                  (float)beta1, (float)beta2, (float)eps, (float)weight_decay,
+                  (int)step, (float)inv_scale, cur_stream());
+}
+
+void lamb_phase1(at::Tensor master, at::Tensor grad, at::Tensor m,
+                 at::Tensor v, at::Tensor update, at::Tensor chunk_of,
+                 at::Tensor wnorm_sq, at::Tensor unorm_sq, double beta1,
+                 double beta2, double eps, double weight_decay, int64_t step,
+                 double inv_scale) {
+  check(master, at::kFloat, "master");
+  check(update, at::kFloat, "update");
+  check(chunk_of, at::kInt, "chunk_of");
+  const bool gbf = is_bf16(grad);
+  epl_lamb_phase1(master.data_ptr<float>(), grad.data_ptr(), gbf,
+                  m.data_ptr<float>(), v.data_ptr<float>(),
+                  update.data_ptr<float>(), chunk_of.data_ptr<int>(),
+                  wnorm_sq.data_ptr<float>(), unorm_sq.data_ptr<float>(),
+                  master.numel(), (float)beta1, (float)beta2, (float)eps,
+                  (float)weight_decay, (int)step, (float)inv_scale,
+                  cur_stream());
+}
+
+void lamb_phase2(at::Tensor master, c10::optional<at::Tensor> param_bf16,
+                 at::Tensor update, at::Tensor chunk_of, at::Tensor ratio,
+                 double lr) {
+  check(master, at::kFloat, "master");
+  unsigned short* pb = nullptr;
+  if (param_bf16.has_value())
+    pb = reinterpret_cast<unsigned short*>(param_bf16->data_ptr());
+  epl_lamb_phase2(master.data_ptr<float>(), pb, update.data_ptr<float>(),
+                  chunk_of.data_ptr<int>(), ratio.data_ptr<float>(),
+                  master.numel(), (float)lr, cur_stream());
+}
+
+void layer_norm_fwd(at::Tensor out, at::Tensor x, at::Tensor gamma,
+                    at::Tensor beta, at::Tensor mean, at::Tensor rstd,
+                    double eps) {
+  const bool bf16 = is_bf16(x);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  TORCH_CHECK(!bf16 || cols % 8 == 0, "bf16 LayerNorm needs cols % 8 == 0");
+  check(mean, at::kFloat, "mean");
+  check(rstd, at::kFloat, "rstd");
+  epl_layer_norm_fwd(out.data_ptr(), x.data_ptr(), gamma.data_ptr(),
+                     beta.data_ptr(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), rows, cols, (float)eps, bf16,
+                     cur_stream());
+}
+
+void layer_norm_bwd(at::Tensor dx, at::Tensor dgamma, at::Tensor dbeta,
+                    at::Tensor dy, at::Tensor x, at::Tensor gamma,
+                    at::Tensor mean, at::Tensor rstd) {
+  const bool bf16 = is_bf16(x);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  TORCH_CHECK(!bf16 || cols % 8 == 0, "bf16 LayerNorm needs cols % 8 == 0");
+  TORCH_CHECK((size_t)cols * 2 * sizeof(float) <= 128 * 1024,
+              "LayerNorm backward supports cols <= 16384");
+  check(dgamma, at::kFloat, "dgamma");
+  check(dbeta, at::kFloat, "dbeta");
+  epl_layer_norm_bwd(dx.data_ptr(), dgamma.data_ptr<float>(),
+                     dbeta.data_ptr<float>(), dy.data_ptr(), x.data_ptr(),
+                     gamma.data_ptr(), mean.data_ptr<float>(),
+                     rstd.data_ptr<float>(), rows, cols, bf16, cur_stream());
+}
+
+void bias_gelu_fwd(at::Tensor out, at::Tensor x, at::Tensor bias) {
+  const bool bf16 = is_bf16(x);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  TORCH_CHECK(!bf16 || cols % 8 == 0, "bf16 bias_gelu needs cols % 8 == 0");
+  epl_bias_gelu_fwd(out.data_ptr(), x.data_ptr(), bias.data_ptr(), rows, cols,
+                    bf16, cur_stream());
+}
+
+void bias_gelu_bwd(at::Tensor dx, at::Tensor dbias, at::Tensor dy,
+                   at::Tensor x, at::Tensor bias) {
+  const bool bf16 = is_bf16(x);
+  const int64_t cols = x.size(-1);
+  const int64_t rows = x.numel() / cols;
+  TORCH_CHECK(!bf16 || cols % 8 == 0, "bf16 bias_gelu needs cols % 8 == 0");
+  TORCH_CHECK((size_t)cols * sizeof(float) <= 128 * 1024,
+              "bias_gelu backward supports cols <= 32768");
+  check(dbias, at::kFloat, "dbias");
+  epl_bias_gelu_bwd(dx.data_ptr(), dbias.data_ptr<float>(), dy.data_ptr(),
+                    x.data_ptr(), bias.data_ptr(), rows, cols, bf16,
+                    cur_stream());
+}
+
+void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor row_max,
+                 at::Tensor row_sumexp, at::Tensor target_logit,
+                 int64_t vocab_begin, int64_t ignore_index) {
+  const bool bf16 = is_bf16(logits);
+  const int64_t cols = logits.size(-1);
+  const int64_t rows = logits.numel() / cols;
+  check(targets, at::kLong, "targets");
+  epl_ce_rowstats(logits.data_ptr(), targets.data_ptr<int64_t>(),
+                  row_max.data_ptr<float>(), row_sumexp.data_ptr<float>(),
+                  target_logit.data_ptr<float>(), rows, cols, vocab_begin,
+                  ignore_index, bf16, cur_stream());
+}
+
+void ce_backward(at::Tensor dlogits, at::Tensor logits, at::Tensor targets,
+                 at::Tensor gmax, at::Tensor gsumexp, at::Tensor dloss,
+                 int64_t vocab_begin, int64_t ignore_index, double scale) {
+  const bool bf16 = is_bf16(logits);
+  const int64_t cols = logits.size(-1);
+  const int64_t rows = logits.numel() / cols;
+  epl_ce_backward(dlogits.data_ptr(), logits.data_ptr(),
+                  targets.data_ptr<int64_t>(), gmax.data_ptr<float>(),
+                  gsumexp.data_ptr<float>(), dloss.data_ptr<float>(), rows,
+                  cols, vocab_begin, ignore_index, (float)scale, bf16,
+                  cur_stream());
+}
+
+void scale_(at::Tensor t, double s) {
+  const bool bf16 = is_bf16(t);
+  epl_scale(t.data_ptr(), t.numel(), (float)s, bf16, cur_stream());
+}
+
+void f32_to_bf16(at::Tensor dst, at::Tensor src) {
+  check(dst, at::kBFloat16, "dst");
+  check(src, at::kFloat, "src");
+  TORCH_CHECK(dst.numel() == src.numel());
+  epl_f32_to_bf16(reinterpret_cast<unsigned short*>(dst.data_ptr()),
+                  src.data_ptr<float>(), src.numel(), cur_stream());
+}
+
+void bf16_to_f32(at::Tensor dst, at::Tensor src) {
+  check(dst, at::kFloat, "dst");
+  check(src, at::kBFloat16, "src");
+  TORCH_CHECK(dst.numel() == src.numel());
+  epl_bf16_to_f32(dst.data_ptr<float>(),
+                  reinterpret_cast<unsigned short*>(src.data_ptr()),
+                  src.numel(), cur_stream());
+}
+
+void sqnorm(at::Tensor t, at::Tensor out) {
+  const bool bf16 = is_bf16(t);
+  check(out, at::kFloat, "out");
+  epl_sqnorm(t.data_ptr(), t.numel(), out.data_ptr<float>(), bf16,
+             cur_stream());
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.doc() = "MI355X-native parallel library: RCCL comm core + CDNA4 kernels";
+  epl::register_comm(m);
+  m.def("fused_adamw", &fused_adamw);
+  m.def("lamb_phase1", &lamb_phase1);
+  m.def("lamb_phase2", &lamb_phase2);
+  m.def("layer_norm_fwd", &layer_norm_fwd);
+  m.def("layer_norm_bwd", &layer_norm_bwd);
+  m.def("bias_gelu_fwd", &bias_gelu_fwd);
+  m.def("bias_gelu_bwd", &bias_gelu_bwd);
+  m.def("ce_rowstats", &ce_rowstats);
+  m.def("ce_backward", &ce_backward);
+  m.def("scale_", &scale_);
+  m.def("f32_to_bf16", &f32_to_bf16);
+  m.def("bf16_to_f32", &bf16_to_f32);
+  m.def("sqnorm", &sqnorm);
+}

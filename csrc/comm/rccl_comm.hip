@@ -1,0 +1,394 @@
+// RCCL communication core for the MI355X-native parallel library.
+//
+// Capability parity with the reference's native layer
+// (/root/reference/csrc/communicators/*.cc: handle/bootstrap ops
+// nccl_communicator.cc:25-132, collectives nccl_all_reduce.cc /
+// nccl_all_gather.cc / nccl_reduce_scatter.cc / nccl_reduce.cc /
+// nccl_broadcast.cc / nccl_all_to_all.cc, stream glue tensorflow_cuda.h),
+// redesigned for ROCm: instead of borrowing the TF compute stream through a
+// StreamExecutor hack and faking asynchrony with a thread pool, each
+// communicator owns a dedicated HIP stream; asynchronous collectives are
+// fenced against the producing torch stream with HIP events (input-ready
+// event -> comm stream, completion event -> consumer stream on join()).
+// Enqueued tensors are retained until their completion event fires so the
+// caching allocator cannot recycle them mid-flight.
+//
+// xGMI note: one ring all-reduce is bound by a single xGMI link
+// (~153 GB/s); the Python layer therefore round-robins buckets over a pool
+// of these communicators (comm/pool.py), each with its own stream, so
+// several rings progress on distinct links concurrently.
+
+#include <hip/hip_runtime.h>
+#include <rccl/rccl.h>
+
+#include <torch/extension.h>
+#include <ATen/hip/impl/HIPStreamMasqueradingAsCUDA.h>
+
+#include <deque>
+#include <mutex>
+#include <string>
+#include <unordered_map>
+#include <vector>
+
+#define HIP_CHECK(cmd)                                                        \
+  do {                                                                        \
+    hipError_t err__ = (cmd);                                                 \
+    TORCH_CHECK(err__ == hipSuccess, "HIP error: ", hipGetErrorString(err__)); \
+  } while (0)
+
+#define RCCL_CHECK(cmd)                                                       \
+  do {                                                                        \
+    ncclResult_t res__ = (cmd);                                               \
+    TORCH_CHECK(res__ == ncclSuccess, "RCCL error: ",                          \
+                ncclGetErrorString(res__));                                    \
+  } while (0)
+
+namespace epl {
+
+static ncclDataType_t to_rccl_dtype(const at::Tensor& t) {
+  switch (t.scalar_type()) {
+    case at::kFloat: return ncclFloat32;
+    case at::kHalf: return ncclFloat16;
+    case at::kBFloat16: return ncclBfloat16;
+    case at::kDouble: return ncclFloat64;
+    case at::kInt: return ncclInt32;
+    case at::kLong: return ncclInt64;
+    case at::kByte: return ncclUint8;
+    case at::kChar: return ncclInt8;
+    default:
+      TORCH_CHECK(false, "unsupported dtype for RCCL: ", t.scalar_type());
+  }
+}
+
+static ncclRedOp_t to_rccl_op(const std::string& op) {
+  if (op == "sum") return ncclSum;
+  if (op == "prod") return ncclProd;
+  if (op == "max") return ncclMax;
+  if (op == "min") return ncclMin;
+  if (op == "avg") return ncclAvg;
+  TORCH_CHECK(false, "unsupported reduce op: ", op);
+}
+
+struct PendingOp {
+  hipEvent_t done;
+  std::vector<at::Tensor> keep;
+};
+
+struct Comm {
+  ncclComm_t comm = nullptr;
+  int rank = -1;
+  int size = 0;
+  hipStream_t stream = nullptr;   // dedicated comm stream
+  hipEvent_t ready = nullptr;     // producer-side fence (reused)
+  std::deque<PendingOp> pending;  // retained tensors until completion
+  std::vector<hipEvent_t> event_pool;
+
+  hipEvent_t get_event() {
+    if (!event_pool.empty()) {
+      hipEvent_t e = event_pool.back();
+      event_pool.pop_back();
+      return e;
+    }
+    hipEvent_t e;
+    HIP_CHECK(hipEventCreateWithFlags(&e, hipEventDisableTiming));
+    return e;
+  }
+
+  void gc() {
+    while (!pending.empty()) {
+      hipError_t q = hipEventQuery(pending.front().done);
+      if (q == hipErrorNotReady) break;
+      TORCH_CHECK(q == hipSuccess, "HIP event error: ", hipGetErrorString(q));
+      event_pool.push_back(pending.front().done);
+      pending.pop_front();
+    }
+  }
+};
+
+static std::unordered_map<std::string, Comm> g_comms;
+static std::mutex g_mutex;
+
+static Comm& get_comm(const std::string& name) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  auto it = g_comms.find(name);
+  TORCH_CHECK(it != g_comms.end(), "RCCL communicator '", name,
+              "' does not exist");
+  return it->second;
+}
+
+py::bytes comm_unique_id() {
+  ncclUniqueId id;
+  RCCL_CHECK(ncclGetUniqueId(&id));
+  return py::bytes(reinterpret_cast<const char*>(&id), sizeof(id));
+}
+
+void comm_init(const std::string& name, const std::string& id_bytes,
+               int64_t rank, int64_t size) {
+  TORCH_CHECK(id_bytes.size() == sizeof(ncclUniqueId),
+              "bad unique id size ", id_bytes.size());
+  {
+    std::lock_guard<std::mutex> lk(g_mutex);
+    TORCH_CHECK(g_comms.find(name) == g_comms.end(),
+                "communicator '", name, "' already exists");
+  }
+  ncclUniqueId id;
+  std::memcpy(&id, id_bytes.data(), sizeof(id));
+  Comm c;
+  c.rank = static_cast<int>(rank);
+  c.size = static_cast<int>(size);
+  RCCL_CHECK(ncclCommInitRank(&c.comm, c.size, id, c.rank));
+  HIP_CHECK(hipStreamCreateWithFlags(&c.stream, hipStreamNonBlocking));
+  HIP_CHECK(hipEventCreateWithFlags(&c.ready, hipEventDisableTiming));
+  std::lock_guard<std::mutex> lk(g_mutex);
+  g_comms.emplace(name, std::move(c));
+}
+
+bool comm_exists(const std::string& name) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  return g_comms.find(name) != g_comms.end();
+}
+
+void comm_destroy(const std::string& name) {
+  std::lock_guard<std::mutex> lk(g_mutex);
+  auto it = g_comms.find(name);
+  if (it == g_comms.end()) return;
+  Comm& c = it->second;
+  HIP_CHECK(hipStreamSynchronize(c.stream));
+  for (auto& p : c.pending) {
+    HIP_CHECK(hipEventDestroy(p.done));
+  }
+  c.pending.clear();
+  for (auto e : c.event_pool) HIP_CHECK(hipEventDestroy(e));
+  if (c.comm) ncclCommDestroy(c.comm);
+  HIP_CHECK(hipEventDestroy(c.ready));
+  HIP_CHECK(hipStreamDestroy(c.stream));
+  g_comms.erase(it);
+}
+
+static hipStream_t current_torch_stream() {
+  return c10::hip::getCurrentHIPStreamMasqueradingAsCUDA().stream();
+}
+
+// Pick the launch stream.  async=true: the communicator's own stream, fenced
+// after the producing (current) stream; async=false: the current stream
+// (zero-latency inline collective, e.g. TP ops consumed immediately).
+static hipStream_t begin_op(Comm& c, bool async) {
+  if (!async) return current_torch_stream();
+  c.gc();
+  HIP_CHECK(hipEventRecord(c.ready, current_torch_stream()));
+  HIP_CHECK(hipStreamWaitEvent(c.stream, c.ready, 0));
+  return c.stream;
+}
+
+static void end_op(Comm& c, bool async, std::vector<at::Tensor> keep) {
+  if (!async) return;
+  PendingOp p;
+  p.done = c.get_event();
+  HIP_CHECK(hipEventRecord(p.done, c.stream));
+  p.keep = std::move(keep);
+  c.pending.push_back(std::move(p));
+}
+
+void comm_join(const std::string& name) {
+  // Make the current torch stream wait for everything enqueued on the
+  // communicator's stream (device-side fence, no host sync).
+  Comm& c = get_comm(name);
+  hipEvent_t e = c.get_event();
+  HIP_CHECK(hipEventRecord(e, c.stream));
+  HIP_CHECK(hipStreamWaitEvent(current_torch_stream(), e, 0));
+  c.event_pool.push_back(e);
+  c.gc();
+}
+
+void comm_synchronize(const std::string& name) {
+  Comm& c = get_comm(name);
+  HIP_CHECK(hipStreamSynchronize(c.stream));
+  c.gc();
+}
+
+static void check_gpu_contig(const at::Tensor& t) {
+  TORCH_CHECK(t.is_cuda(), "RCCL collectives need device tensors");
+  TORCH_CHECK(t.is_contiguous(), "RCCL collectives need contiguous tensors");
+}
+
+void all_reduce(const std::string& name, at::Tensor t, const std::string& op,
+                bool async) {
+  check_gpu_contig(t);
+  Comm& c = get_comm(name);
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclAllReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                           to_rccl_dtype(t), to_rccl_op(op), c.comm, s));
+  end_op(c, async, {t});
+}
+
+void broadcast(const std::string& name, at::Tensor t, int64_t root,
+               bool async) {
+  check_gpu_contig(t);
+  Comm& c = get_comm(name);
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclBroadcast(t.data_ptr(), t.data_ptr(), t.numel(),
+                           to_rccl_dtype(t), static_cast<int>(root), c.comm,
+                           s));
+  end_op(c, async, {t});
+}
+
+void reduce(const std::string& name, at::Tensor t, int64_t root,
+            const std::string& op, bool async) {
+  check_gpu_contig(t);
+  Comm& c = get_comm(name);
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclReduce(t.data_ptr(), t.data_ptr(), t.numel(),
+                        to_rccl_dtype(t), to_rccl_op(op),
+                        static_cast<int>(root), c.comm, s));
+  end_op(c, async, {t});
+}
+
+void all_gather(const std::string& name, at::Tensor out, at::Tensor in,
+                bool async) {
+  check_gpu_contig(out);
+  check_gpu_contig(in);
+  Comm& c = get_comm(name);
+  TORCH_CHECK(out.numel() == in.numel() * c.size,
+              "all_gather output must be size * input");
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclAllGather(in.data_ptr(), out.data_ptr(), in.numel(),
+                           to_rccl_dtype(in), c.comm, s));
+  end_op(c, async, {out, in});
+}
+
+void reduce_scatter(const std::string& name, at::Tensor out, at::Tensor in,
+                    const std::string& op, bool async) {
+  check_gpu_contig(out);
+  check_gpu_contig(in);
+  Comm& c = get_comm(name);
+  TORCH_CHECK(in.numel() == out.numel() * c.size,
+              "reduce_scatter input must be size * output");
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclReduceScatter(in.data_ptr(), out.data_ptr(), out.numel(),
+                               to_rccl_dtype(in), to_rccl_op(op), c.comm, s));
+  end_op(c, async, {out, in});
+}
+
+// Single-tensor all-to-all: dim0 divisible by comm size, equal splits
+// (reference: csrc nccl_all_to_all.cc:22-76 / tensorflow_nccl.h:185-206 —
+// there a grouped send/recv loop; RCCL exposes the same grouping).
+void all_to_all_single(const std::string& name, at::Tensor out, at::Tensor in,
+                       bool async) {
+  check_gpu_contig(out);
+  check_gpu_contig(in);
+  Comm& c = get_comm(name);
+  TORCH_CHECK(in.numel() % c.size == 0, "all_to_all input not divisible");
+  TORCH_CHECK(out.numel() == in.numel(), "all_to_all size mismatch");
+  size_t chunk = in.numel() / c.size;
+  size_t esize = in.element_size();
+  ncclDataType_t dt = to_rccl_dtype(in);
+  hipStream_t s = begin_op(c, async);
+  char* src = static_cast<char*>(in.data_ptr());
+  char* dst = static_cast<char*>(out.data_ptr());
+  RCCL_CHECK(ncclGroupStart());
+  for (int r = 0; r < c.size; ++r) {
+    RCCL_CHECK(ncclSend(src + r * chunk * esize, chunk, dt, r, c.comm, s));
+    RCCL_CHECK(ncclRecv(dst + r * chunk * esize, chunk, dt, r, c.comm, s));
+  }
+  RCCL_CHECK(ncclGroupEnd());
+  end_op(c, async, {out, in});
+}
+
+// Variable all-to-all: per-rank element counts (flattened innermost layout)
+// (reference: AllToAllv, tensorflow_nccl.h:208-265).
+void all_to_all_v(const std::string& name, at::Tensor out, at::Tensor in,
+                  std::vector<int64_t> out_counts,
+                  std::vector<int64_t> in_counts, bool async) {
+  check_gpu_contig(out);
+  check_gpu_contig(in);
+  Comm& c = get_comm(name);
+  TORCH_CHECK((int)out_counts.size() == c.size &&
+              (int)in_counts.size() == c.size,
+              "all_to_all_v counts must have one entry per rank");
+  size_t esize = in.element_size();
+  ncclDataType_t dt = to_rccl_dtype(in);
+  hipStream_t s = begin_op(c, async);
+  char* src = static_cast<char*>(in.data_ptr());
+  char* dst = static_cast<char*>(out.data_ptr());
+  RCCL_CHECK(ncclGroupStart());
+  int64_t soff = 0, roff = 0;
+  for (int r = 0; r < c.size; ++r) {
+    if (in_counts[r] > 0) {
+      RCCL_CHECK(ncclSend(src + soff * esize, in_counts[r], dt, r, c.comm, s));
+    }
+    if (out_counts[r] > 0) {
+      RCCL_CHECK(ncclRecv(dst + roff * esize, out_counts[r], dt, r, c.comm, s));
+    }
+    soff += in_counts[r];
+    roff += out_counts[r];
+  }
+  RCCL_CHECK(ncclGroupEnd());
+  end_op(c, async, {out, in});
+}
+
+void send(const std::string& name, at::Tensor t, int64_t peer, bool async) {
+  check_gpu_contig(t);
+  Comm& c = get_comm(name);
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclSend(t.data_ptr(), t.numel(), to_rccl_dtype(t),
+                      static_cast<int>(peer), c.comm, s));
+  end_op(c, async, {t});
+}
+
+void recv(const std::string& name, at::Tensor t, int64_t peer, bool async) {
+  check_gpu_contig(t);
+  Comm& c = get_comm(name);
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclRecv(t.data_ptr(), t.numel(), to_rccl_dtype(t),
+                      static_cast<int>(peer), c.comm, s));
+  end_op(c, async, {t});
+}
+
+// Batched bidirectional p2p (pipeline stage boundaries): ops is a list of
+// (is_send, tensor, peer) executed inside one nccl group so send/recv pairs
+// cannot deadlock.
+void batch_p2p(const std::string& name,
+               std::vector<std::tuple<bool, at::Tensor, int64_t>> ops,
+               bool async) {
+  Comm& c = get_comm(name);
+  hipStream_t s = begin_op(c, async);
+  std::vector<at::Tensor> keep;
+  RCCL_CHECK(ncclGroupStart());
+  for (auto& op : ops) {
+    bool is_send = std::get<0>(op);
+    at::Tensor t = std::get<1>(op);
+    int peer = static_cast<int>(std::get<2>(op));
+    check_gpu_contig(t);
+    if (is_send) {
+      RCCL_CHECK(ncclSend(t.data_ptr(), t.numel(), to_rccl_dtype(t), peer,
+                          c.comm, s));
+    } else {
+      RCCL_CHECK(ncclRecv(t.data_ptr(), t.numel(), to_rccl_dtype(t), peer,
+                          c.comm, s));
+    }
+    keep.push_back(t);
+  }
+  RCCL_CHECK(ncclGroupEnd());
+  end_op(c, async, std::move(keep));
+}
+
+void register_comm(py::module& m) {
+  m.def("comm_unique_id", &comm_unique_id);
+  m.def("comm_init", &comm_init);
+  m.def("comm_exists", &comm_exists);
+  m.def("comm_destroy", &comm_destroy);
+  m.def("comm_join", &comm_join);
+  m.def("comm_synchronize", &comm_synchronize);
+  m.def("all_reduce", &all_reduce);
+  m.def("broadcast", &broadcast);
+  m.def("reduce", &reduce);
+  m.def("all_gather", &all_gather);
+  m.def("reduce_scatter", &reduce_scatter);
+  m.def("all_to_all_single", &all_to_all_single);
+  m.def("all_to_all_v", &all_to_all_v);
+  m.def("send", &send);
+  m.def("recv", &recv);
+  m.def("batch_p2p", &batch_p2p);
+}
+
+}  // namespace epl

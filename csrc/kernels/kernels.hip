@@ -1,0 +1,843 @@
+// Hand-written CDNA4 (gfx950) kernels for the MI355X-native parallel library.
+//
+// These replace the hot-path ops the reference leaves to TF (the reference's
+// native layer, /root/reference/csrc/, contains no device kernels at all —
+// only NCCL glue).  Everything here is memory-bound streaming work, designed
+// per the CDNA4 rules: 256-thread blocks (4 waves of 64), bf16 loaded as
+// packed ushort vectors (8-16 B/lane), fp32 accumulation, grid-stride loops
+// capped near 2048 blocks so the 256-CU chip is filled without launch spam,
+// LDS used for per-block column-reduction partials (dgamma/dbeta/dbias) so
+// global atomics fire once per block, not once per row.
+//
+// Optimizer kernels operate on FLAT ARENAS: the Python engine concatenates
+// every parameter of a taskgraph into one contiguous buffer (master fp32 /
+// model bf16 / grad), so one AdamW step is ONE kernel over one flat stream —
+// no multi-tensor metadata tables, no per-parameter launches.
+
+#include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
+#include <cstdint>
+#include <cmath>
+
+#define DEV __device__ __forceinline__
+
+namespace {
+
+constexpr int kBlock = 256;
+constexpr int kMaxGrid = 4096;  // >> 256 CUs; grid-stride covers the rest
+
+DEV float bf2f(unsigned short u) {
+  unsigned int x = (unsigned int)u << 16;
+  return __uint_as_float(x);
+}
+
+DEV unsigned short f2bf(float f) {
+  // round-to-nearest-even bf16 conversion
+  unsigned int x = __float_as_uint(f);
+  unsigned int lsb = (x >> 16) & 1u;
+  x += 0x7fffu + lsb;
+  return (unsigned short)(x >> 16);
+}
+
+struct ushort8 {
+  unsigned short v[8];
+};
+struct float8 {
+  float v[8];
+};
+
+DEV float8 load_bf16x8(const unsigned short* p) {
+  const ushort8 raw = *reinterpret_cast<const ushort8*>(p);
+  float8 f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) f.v[i] = bf2f(raw.v[i]);
+  return f;
+}
+
+DEV void store_bf16x8(unsigned short* p, const float8& f) {
+  ushort8 raw;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) raw.v[i] = f2bf(f.v[i]);
+  *reinterpret_cast<ushort8*>(p) = raw;
+}
+
+DEV float8 load_f32x8(const float* p) {
+  float8 f;
+  const float4 a = *reinterpret_cast<const float4*>(p);
+  const float4 b = *reinterpret_cast<const float4*>(p + 4);
+  f.v[0] = a.x; f.v[1] = a.y; f.v[2] = a.z; f.v[3] = a.w;
+  f.v[4] = b.x; f.v[5] = b.y; f.v[6] = b.z; f.v[7] = b.w;
+  return f;
+}
+
+DEV void store_f32x8(float* p, const float8& f) {
+  *reinterpret_cast<float4*>(p) = make_float4(f.v[0], f.v[1], f.v[2], f.v[3]);
+  *reinterpret_cast<float4*>(p + 4) =
+      make_float4(f.v[4], f.v[5], f.v[6], f.v[7]);
+}
+
+inline int grid_for(int64_t work_items) {
+  int64_t blocks = (work_items + kBlock - 1) / kBlock;
+  return (int)(blocks < kMaxGrid ? (blocks > 0 ? blocks : 1) : kMaxGrid);
+}
+
+// ---- block-wide reduction (256 threads = 4 waves) ---------------------------
+DEV float block_reduce_sum(float val, float* lds /* >= 4 floats */) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    val += __shfl_down(val, off, 64);
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  if (lane == 0) lds[wave] = val;
+  __syncthreads();
+  if (wave == 0) {
+    val = lane < 4 ? lds[lane] : 0.f;
+#pragma unroll
+    for (int off = 2; off > 0; off >>= 1)
+      val += __shfl_down(val, off, 64);
+    if (lane == 0) lds[0] = val;
+  }
+  __syncthreads();
+  return lds[0];
+}
+
+DEV float block_reduce_max(float val, float* lds) {
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1)
+    val = fmaxf(val, __shfl_down(val, off, 64));
+  const int wave = threadIdx.x >> 6;
+  const int lane = threadIdx.x & 63;
+  if (lane == 0) lds[wave] = val;
+  __syncthreads();
+  if (wave == 0) {
+    val = lane < 4 ? lds[lane] : -INFINITY;
+#pragma unroll
+    for (int off = 2; off > 0; off >>= 1)
+      val = fmaxf(val, __shfl_down(val, off, 64));
+    if (lane == 0) lds[0] = val;
+  }
+  __syncthreads();
+  return lds[0];
+}
+
+// ============================================================================
+// Fused AdamW over a flat arena.
+//   master: fp32 "true" parameters (flat).
+//   param : optional bf16 working copy written back after the update.
+//   grad  : bf16 or fp32 gradients; multiplied by inv_scale (AMP unscale
+//           fused in).  Decoupled weight decay (AdamW).
+// ============================================================================
+template <bool GRAD_BF16, bool HAS_BF16_PARAM>
+__global__ void fused_adamw_kernel(
+    float* __restrict__ master, unsigned short* __restrict__ param,
+    const void* __restrict__ grad_raw, float* __restrict__ m,
+    float* __restrict__ v, int64_t n, float lr, float beta1, float beta2,
+    float eps, float weight_decay, float inv_bias1, float inv_bias2,
+    float inv_scale) {
+  const int64_t nvec = n / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int64_t base = i * 8;
+    float8 g;
+    if (GRAD_BF16)
+      g = load_bf16x8(
+          reinterpret_cast<const unsigned short*>(grad_raw) + base);
+    else
+      g = load_f32x8(reinterpret_cast<const float*>(grad_raw) + base);
+    float8 pm = load_f32x8(master + base);
+    float8 mm = load_f32x8(m + base);
+    float8 vv = load_f32x8(v + base);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float gk = g.v[k] * inv_scale;
+      float mk = beta1 * mm.v[k] + (1.f - beta1) * gk;
+      float vk = beta2 * vv.v[k] + (1.f - beta2) * gk * gk;
+      float mhat = mk * inv_bias1;
+      float vhat = vk * inv_bias2;
+      float pk = pm.v[k];
+      pk -= lr * (mhat / (sqrtf(vhat) + eps) + weight_decay * pk);
+      mm.v[k] = mk;
+      vv.v[k] = vk;
+      pm.v[k] = pk;
+    }
+    store_f32x8(m + base, mm);
+    store_f32x8(v + base, vv);
+    store_f32x8(master + base, pm);
+    if (HAS_BF16_PARAM) store_bf16x8(param + base, pm);
+  }
+  // scalar tail
+  const int64_t tail = nvec * 8;
+  for (int64_t i = tail + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float gk = (GRAD_BF16
+                    ? bf2f(reinterpret_cast<const unsigned short*>(grad_raw)[i])
+                    : reinterpret_cast<const float*>(grad_raw)[i]) *
+               inv_scale;
+    float mk = beta1 * m[i] + (1.f - beta1) * gk;
+    float vk = beta2 * v[i] + (1.f - beta2) * gk * gk;
+    float pk = master[i];
+    pk -= lr * ((mk * inv_bias1) / (sqrtf(vk * inv_bias2) + eps) +
+                weight_decay * pk);
+    m[i] = mk;
+    v[i] = vk;
+    master[i] = pk;
+    if (HAS_BF16_PARAM) param[i] = f2bf(pk);
+  }
+}
+
+// ============================================================================
+// Fused LAMB over a flat arena, phase kernels.
+//   Phase 1: update m, v; write the raw Adam update u into a scratch buffer;
+//            accumulate per-CHUNK squared norms of (param, update) so the
+//            Python side can compute per-parameter trust ratios (chunks are
+//            parameter boundaries in the flat arena).
+//   Phase 2: p -= lr * ratio[chunk] * (u + wd*p) applied per chunk.
+// Chunk ids come from a per-8-element chunk-index map built once on the host.
+// ============================================================================
+__global__ void lamb_phase1_kernel(
+    const float* __restrict__ master, const void* __restrict__ grad_raw,
+    bool grad_bf16, float* __restrict__ m, float* __restrict__ v,
+    float* __restrict__ update, const int* __restrict__ chunk_of,
+    float* __restrict__ wnorm_sq, float* __restrict__ unorm_sq, int64_t n,
+    float beta1, float beta2, float eps, float weight_decay, float inv_bias1,
+    float inv_bias2, float inv_scale) {
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float gk = (grad_bf16
+                    ? bf2f(reinterpret_cast<const unsigned short*>(grad_raw)[i])
+                    : reinterpret_cast<const float*>(grad_raw)[i]) *
+               inv_scale;
+    float mk = beta1 * m[i] + (1.f - beta1) * gk;
+    float vk = beta2 * v[i] + (1.f - beta2) * gk * gk;
+    m[i] = mk;
+    v[i] = vk;
+    float pk = master[i];
+    float u = (mk * inv_bias1) / (sqrtf(vk * inv_bias2) + eps) +
+              weight_decay * pk;
+    update[i] = u;
+    const int c = chunk_of[i >> 3];
+    atomicAdd(&wnorm_sq[c], pk * pk);
+    atomicAdd(&unorm_sq[c], u * u);
+  }
+}
+
+__global__ void lamb_phase2_kernel(float* __restrict__ master,
+                                   unsigned short* __restrict__ param,
+                                   const float* __restrict__ update,
+                                   const int* __restrict__ chunk_of,
+                                   const float* __restrict__ ratio, int64_t n,
+                                   float lr) {
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const int c = chunk_of[i >> 3];
+    float pk = master[i] - lr * ratio[c] * update[i];
+    master[i] = pk;
+    if (param != nullptr) param[i] = f2bf(pk);
+  }
+}
+
+// ============================================================================
+// LayerNorm forward: one block per row, fp32 accumulation, bf16x8 loads.
+// Saves mean and rstd for the backward.
+// ============================================================================
+template <bool BF16>
+__global__ void layer_norm_fwd_kernel(
+    void* __restrict__ out, const void* __restrict__ x,
+    const void* __restrict__ gamma, const void* __restrict__ beta,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out, int64_t rows,
+    int64_t cols, float eps) {
+  __shared__ float lds[8];
+  using T = unsigned short;  // bf16 raw
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    float sum = 0.f, sumsq = 0.f;
+    if (BF16) {
+      const T* xr = reinterpret_cast<const T*>(x) + base;
+      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+        float8 f = load_bf16x8(xr + c);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          sum += f.v[k];
+          sumsq += f.v[k] * f.v[k];
+        }
+      }
+    } else {
+      const float* xr = reinterpret_cast<const float*>(x) + base;
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        float fv = xr[c];
+        sum += fv;
+        sumsq += fv * fv;
+      }
+    }
+    sum = block_reduce_sum(sum, lds);
+    __syncthreads();
+    sumsq = block_reduce_sum(sumsq, lds);
+    const float mean = sum / cols;
+    const float var = sumsq / cols - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+    if (BF16) {
+      const T* xr = reinterpret_cast<const T*>(x) + base;
+      const T* gr = reinterpret_cast<const T*>(gamma);
+      const T* br = reinterpret_cast<const T*>(beta);
+      T* yr = reinterpret_cast<T*>(out) + base;
+      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+        float8 f = load_bf16x8(xr + c);
+        float8 g = load_bf16x8(gr + c);
+        float8 b = load_bf16x8(br + c);
+        float8 y;
+#pragma unroll
+        for (int k = 0; k < 8; ++k)
+          y.v[k] = (f.v[k] - mean) * rstd * g.v[k] + b.v[k];
+        store_bf16x8(yr + c, y);
+      }
+    } else {
+      const float* xr = reinterpret_cast<const float*>(x) + base;
+      const float* gr = reinterpret_cast<const float*>(gamma);
+      const float* br = reinterpret_cast<const float*>(beta);
+      float* yr = reinterpret_cast<float*>(out) + base;
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        yr[c] = (xr[c] - mean) * rstd * gr[c] + br[c];
+      }
+    }
+    __syncthreads();
+  }
+}
+
+// ============================================================================
+// LayerNorm backward.  Each block walks a strided set of rows: computes dx
+// and accumulates dgamma/dbeta partials in LDS (cols*2 floats), flushing to
+// global fp32 buffers with one atomicAdd per column per block at the end.
+// Requires cols*8 bytes <= 128 KiB of LDS (cols <= 16384).
+// ============================================================================
+template <bool BF16>
+__global__ void layer_norm_bwd_kernel(
+    void* __restrict__ dx, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, const void* __restrict__ dy,
+    const void* __restrict__ x, const void* __restrict__ gamma,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    int64_t rows, int64_t cols) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* part_dg = reinterpret_cast<float*>(smem);            // [cols]
+  float* part_db = part_dg + cols;                            // [cols]
+  __shared__ float lds[8];
+  using T = unsigned short;
+  for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+    part_dg[c] = 0.f;
+    part_db[c] = 0.f;
+  }
+  __syncthreads();
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    const float mean = mean_in[row];
+    const float rstd = rstd_in[row];
+    float sum_dyg = 0.f, sum_dygx = 0.f;
+    if (BF16) {
+      const T* dyr = reinterpret_cast<const T*>(dy) + base;
+      const T* xr = reinterpret_cast<const T*>(x) + base;
+      const T* gr = reinterpret_cast<const T*>(gamma);
+      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+        float8 d = load_bf16x8(dyr + c);
+        float8 f = load_bf16x8(xr + c);
+        float8 g = load_bf16x8(gr + c);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          const float xhat = (f.v[k] - mean) * rstd;
+          const float dyg = d.v[k] * g.v[k];
+          sum_dyg += dyg;
+          sum_dygx += dyg * xhat;
+          part_dg[c + k] += d.v[k] * xhat;
+          part_db[c + k] += d.v[k];
+        }
+      }
+    } else {
+      const float* dyr = reinterpret_cast<const float*>(dy) + base;
+      const float* xr = reinterpret_cast<const float*>(x) + base;
+      const float* gr = reinterpret_cast<const float*>(gamma);
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        const float xhat = (xr[c] - mean) * rstd;
+        const float dyg = dyr[c] * gr[c];
+        sum_dyg += dyg;
+        sum_dygx += dyg * xhat;
+        part_dg[c] += dyr[c] * xhat;
+        part_db[c] += dyr[c];
+      }
+    }
+    sum_dyg = block_reduce_sum(sum_dyg, lds);
+    __syncthreads();
+    sum_dygx = block_reduce_sum(sum_dygx, lds);
+    const float inv_cols = 1.f / cols;
+    if (BF16) {
+      const T* dyr = reinterpret_cast<const T*>(dy) + base;
+      const T* xr = reinterpret_cast<const T*>(x) + base;
+      const T* gr = reinterpret_cast<const T*>(gamma);
+      T* dxr = reinterpret_cast<T*>(dx) + base;
+      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+        float8 d = load_bf16x8(dyr + c);
+        float8 f = load_bf16x8(xr + c);
+        float8 g = load_bf16x8(gr + c);
+        float8 o;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          const float xhat = (f.v[k] - mean) * rstd;
+          o.v[k] = (d.v[k] * g.v[k] - (sum_dyg + xhat * sum_dygx) * inv_cols) *
+                   rstd;
+        }
+        store_bf16x8(dxr + c, o);
+      }
+    } else {
+      const float* dyr = reinterpret_cast<const float*>(dy) + base;
+      const float* xr = reinterpret_cast<const float*>(x) + base;
+      const float* gr = reinterpret_cast<const float*>(gamma);
+      float* dxr = reinterpret_cast<float*>(dx) + base;
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        const float xhat = (xr[c] - mean) * rstd;
+        dxr[c] = (dyr[c] * gr[c] - (sum_dyg + xhat * sum_dygx) * inv_cols) *
+                 rstd;
+      }
+    }
+    __syncthreads();
+  }
+  for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+    atomicAdd(&dgamma[c], part_dg[c]);
+    atomicAdd(&dbeta[c], part_db[c]);
+  }
+}
+
+// ============================================================================
+// Fused bias + GeLU (tanh approximation), forward and backward.
+// ============================================================================
+DEV float gelu_f(float x) {
+  const float k0 = 0.7978845608028654f;   // sqrt(2/pi)
+  const float k1 = 0.044715f;
+  return 0.5f * x * (1.f + tanhf(k0 * (x + k1 * x * x * x)));
+}
+
+DEV float gelu_grad_f(float x) {
+  const float k0 = 0.7978845608028654f;
+  const float k1 = 0.044715f;
+  const float t = tanhf(k0 * (x + k1 * x * x * x));
+  const float dt = (1.f - t * t) * k0 * (1.f + 3.f * k1 * x * x);
+  return 0.5f * (1.f + t) + 0.5f * x * dt;
+}
+
+template <bool BF16>
+__global__ void bias_gelu_fwd_kernel(void* __restrict__ out,
+                                     const void* __restrict__ x,
+                                     const void* __restrict__ bias,
+                                     int64_t rows, int64_t cols) {
+  using T = unsigned short;
+  const int64_t n = rows * cols;
+  if (BF16) {
+    const int64_t nvec = n / 8;
+    for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+         i += (int64_t)gridDim.x * blockDim.x) {
+      const int64_t base = i * 8;
+      float8 f = load_bf16x8(reinterpret_cast<const T*>(x) + base);
+      float8 b = load_bf16x8(reinterpret_cast<const T*>(bias) +
+                             (base % cols));
+      float8 o;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) o.v[k] = gelu_f(f.v[k] + b.v[k]);
+      store_bf16x8(reinterpret_cast<T*>(out) + base, o);
+    }
+  } else {
+    for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += (int64_t)gridDim.x * blockDim.x) {
+      const float xv = reinterpret_cast<const float*>(x)[i] +
+                       reinterpret_cast<const float*>(bias)[i % cols];
+      reinterpret_cast<float*>(out)[i] = gelu_f(xv);
+    }
+  }
+}
+
+template <bool BF16>
+__global__ void bias_gelu_bwd_kernel(void* __restrict__ dx,
+                                     float* __restrict__ dbias,
+                                     const void* __restrict__ dy,
+                                     const void* __restrict__ x,
+                                     const void* __restrict__ bias,
+                                     int64_t rows, int64_t cols) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* part_db = reinterpret_cast<float*>(smem);  // [cols]
+  using T = unsigned short;
+  for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) part_db[c] = 0.f;
+  __syncthreads();
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    if (BF16) {
+      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+        float8 d = load_bf16x8(reinterpret_cast<const T*>(dy) + base + c);
+        float8 f = load_bf16x8(reinterpret_cast<const T*>(x) + base + c);
+        float8 b = load_bf16x8(reinterpret_cast<const T*>(bias) + c);
+        float8 o;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          const float g = d.v[k] * gelu_grad_f(f.v[k] + b.v[k]);
+          o.v[k] = g;
+          part_db[c + k] += g;
+        }
+        store_bf16x8(reinterpret_cast<T*>(dx) + base + c, o);
+      }
+    } else {
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        const float g = reinterpret_cast<const float*>(dy)[base + c] *
+                        gelu_grad_f(reinterpret_cast<const float*>(x)[base + c] +
+                                    reinterpret_cast<const float*>(bias)[c]);
+        reinterpret_cast<float*>(dx)[base + c] = g;
+        part_db[c] += g;
+      }
+    }
+  }
+  __syncthreads();
+  for (int64_t c = threadIdx.x; c < cols; c += blockDim.x)
+    atomicAdd(&dbias[c], part_db[c]);
+}
+
+// ============================================================================
+// Fused softmax cross-entropy over (possibly vocab-sharded) logits.
+// Forward: per row, max + sum(exp) in one pass each; writes per-row
+// (max, sumexp, loss-contribution).  For the TP-sharded case the Python side
+// allreduces max/sumexp between the two phases (reference semantics:
+// /root/reference/epl/ops/distributed_losses.py:58-109).
+//   Phase A (local): row max and sum exp(x - rowmax), plus the local logit at
+//   the target (if the target falls in this shard's [vocab_begin, end)).
+//   Phase B (local, after reduction): dlogits = softmax - onehot.
+// ============================================================================
+template <bool BF16>
+__global__ void ce_rowstats_kernel(const void* __restrict__ logits,
+                                   const int64_t* __restrict__ targets,
+                                   float* __restrict__ row_max,
+                                   float* __restrict__ row_sumexp,
+                                   float* __restrict__ target_logit,
+                                   int64_t rows, int64_t cols,
+                                   int64_t vocab_begin, int64_t ignore_index) {
+  __shared__ float lds[8];
+  using T = unsigned short;
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    float vmax = -INFINITY;
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      const float v = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+                           : reinterpret_cast<const float*>(logits)[base + c];
+      vmax = fmaxf(vmax, v);
+    }
+    vmax = block_reduce_max(vmax, lds);
+    __syncthreads();
+    float sum = 0.f;
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      const float v = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+                           : reinterpret_cast<const float*>(logits)[base + c];
+      sum += expf(v - vmax);
+    }
+    sum = block_reduce_sum(sum, lds);
+    if (threadIdx.x == 0) {
+      row_max[row] = vmax;
+      // store the raw sum at this shard's max; Python rescales by
+      // exp(local_max - global_max) before summing across shards.
+      row_sumexp[row] = sum;
+      const int64_t tgt = targets[row];
+      float tl = 0.f;
+      if (tgt != ignore_index && tgt >= vocab_begin &&
+          tgt < vocab_begin + cols) {
+        const int64_t c = tgt - vocab_begin;
+        tl = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+                  : reinterpret_cast<const float*>(logits)[base + c];
+      }
+      target_logit[row] = tl;
+    }
+    __syncthreads();
+  }
+}
+
+template <bool BF16>
+__global__ void ce_backward_kernel(void* __restrict__ dlogits,
+                                   const void* __restrict__ logits,
+                                   const int64_t* __restrict__ targets,
+                                   const float* __restrict__ gmax,
+                                   const float* __restrict__ gsumexp,
+                                   const float* __restrict__ dloss,
+                                   int64_t rows, int64_t cols,
+                                   int64_t vocab_begin, int64_t ignore_index,
+                                   float scale) {
+  using T = unsigned short;
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    const float m = gmax[row];
+    const float inv_sum = 1.f / gsumexp[row];
+    const int64_t tgt = targets[row];
+    const float dl = (tgt == ignore_index ? 0.f : dloss[row] * scale);
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      const float v = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+                           : reinterpret_cast<const float*>(logits)[base + c];
+      float g = expf(v - m) * inv_sum;
+      if (tgt - vocab_begin == c) g -= 1.f;
+      g *= dl;
+      if (BF16)
+        reinterpret_cast<T*>(dlogits)[base + c] = f2bf(g);
+      else
+        reinterpret_cast<float*>(dlogits)[base + c] = g;
+    }
+  }
+}
+
+// ============================================================================
+// Flat-arena utilities: zero / scale / axpy / casts / sq-norm.
+// ============================================================================
+__global__ void scale_f32_kernel(float* __restrict__ p, int64_t n, float s) {
+  const int64_t nvec = n / 4;
+  float4* p4 = reinterpret_cast<float4*>(p);
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float4 v = p4[i];
+    v.x *= s; v.y *= s; v.z *= s; v.w *= s;
+    p4[i] = v;
+  }
+  for (int64_t i = nvec * 4 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    p[i] *= s;
+}
+
+__global__ void scale_bf16_kernel(unsigned short* __restrict__ p, int64_t n,
+                                  float s) {
+  const int64_t nvec = n / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    float8 v = load_bf16x8(p + i * 8);
+#pragma unroll
+    for (int k = 0; k < 8; ++k) v.v[k] *= s;
+    store_bf16x8(p + i * 8, v);
+  }
+  for (int64_t i = nvec * 8 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    p[i] = f2bf(bf2f(p[i]) * s);
+}
+
+__global__ void f32_to_bf16_kernel(unsigned short* __restrict__ dst,
+                                   const float* __restrict__ src, int64_t n) {
+  const int64_t nvec = n / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (int64_t)gridDim.x * blockDim.x)
+    store_bf16x8(dst + i * 8, load_f32x8(src + i * 8));
+  for (int64_t i = nvec * 8 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = f2bf(src[i]);
+}
+
+__global__ void bf16_to_f32_kernel(float* __restrict__ dst,
+                                   const unsigned short* __restrict__ src,
+                                   int64_t n) {
+  const int64_t nvec = n / 8;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < nvec;
+       i += (int64_t)gridDim.x * blockDim.x)
+    store_f32x8(dst + i * 8, load_bf16x8(src + i * 8));
+  for (int64_t i = nvec * 8 + blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    dst[i] = bf2f(src[i]);
+}
+
+template <bool BF16>
+__global__ void sqnorm_kernel(const void* __restrict__ p, int64_t n,
+                              float* __restrict__ out) {
+  __shared__ float lds[8];
+  using T = unsigned short;
+  float acc = 0.f;
+  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const float v = BF16 ? bf2f(reinterpret_cast<const T*>(p)[i])
+                         : reinterpret_cast<const float*>(p)[i];
+    acc += v * v;
+  }
+  acc = block_reduce_sum(acc, lds);
+  if (threadIdx.x == 0) atomicAdd(out, acc);
+}
+
+}  // namespace
+
+// ============================================================================
+// extern "C" launchers (raw pointers + stream; bound to torch in bindings.hip)
+// ============================================================================
+extern "C" {
+
+void epl_fused_adamw(float* master, unsigned short* param_bf16,
+                     const void* grad, bool grad_bf16, float* m, float* v,
+                     int64_t n, float lr, float beta1, float beta2, float eps,
+                     float weight_decay, int step, float inv_scale,
+                     hipStream_t stream) {
+  const float inv_bias1 = 1.f / (1.f - powf(beta1, (float)step));
+  const float inv_bias2 = 1.f / (1.f - powf(beta2, (float)step));
+  const int grid = grid_for((n + 7) / 8);
+  if (grad_bf16) {
+    if (param_bf16)
+      hipLaunchKernelGGL((fused_adamw_kernel<true, true>), dim3(grid),
+                         dim3(kBlock), 0, stream, master, param_bf16, grad, m,
+                         v, n, lr, beta1, beta2, eps, weight_decay, inv_bias1,
+                         inv_bias2, inv_scale);
+    else
+      hipLaunchKernelGGL((fused_adamw_kernel<true, false>), dim3(grid),
+                         dim3(kBlock), 0, stream, master, param_bf16, grad, m,
+                         v, n, lr, beta1, beta2, eps, weight_decay, inv_bias1,
+                         inv_bias2, inv_scale);
+  } else {
+    if (param_bf16)
+      hipLaunchKernelGGL((fused_adamw_kernel<false, true>), dim3(grid),
+                         dim3(kBlock), 0, stream, master, param_bf16, grad, m,
+                         v, n, lr, beta1, beta2, eps, weight_decay, inv_bias1,
+                         inv_bias2, inv_scale);
+    else
+      hipLaunchKernelGGL((fused_adamw_kernel<false, false>), dim3(grid),
+                         dim3(kBlock), 0, stream, master, param_bf16, grad, m,
+                         v, n, lr, beta1, beta2, eps, weight_decay, inv_bias1,
+                         inv_bias2, inv_scale);
+  }
+}
+
+void epl_lamb_phase1(const float* master, const void* grad, bool grad_bf16,
+                     float* m, float* v, float* update, const int* chunk_of,
+                     float* wnorm_sq, float* unorm_sq, int64_t n, float beta1,
+                     float beta2, float eps, float weight_decay, int step,
+                     float inv_scale, hipStream_t stream) {
+  const float inv_bias1 = 1.f / (1.f - powf(beta1, (float)step));
+  const float inv_bias2 = 1.f / (1.f - powf(beta2, (float)step));
+  const int grid = grid_for(n);
+  hipLaunchKernelGGL(lamb_phase1_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                     master, grad, grad_bf16, m, v, update, chunk_of, wnorm_sq,
+                     unorm_sq, n, beta1, beta2, eps, weight_decay, inv_bias1,
+                     inv_bias2, inv_scale);
+}
+
+void epl_lamb_phase2(float* master, unsigned short* param_bf16,
+                     const float* update, const int* chunk_of,
+                     const float* ratio, int64_t n, float lr,
+                     hipStream_t stream) {
+  const int grid = grid_for(n);
+  hipLaunchKernelGGL(lamb_phase2_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                     master, param_bf16, update, chunk_of, ratio, n, lr);
+}
+
+void epl_layer_norm_fwd(void* out, const void* x, const void* gamma,
+                        const void* beta, float* mean, float* rstd,
+                        int64_t rows, int64_t cols, float eps, bool bf16,
+                        hipStream_t stream) {
+  const int grid = (int)(rows < kMaxGrid ? rows : kMaxGrid);
+  if (bf16)
+    hipLaunchKernelGGL(layer_norm_fwd_kernel<true>, dim3(grid), dim3(kBlock),
+                       0, stream, out, x, gamma, beta, mean, rstd, rows, cols,
+                       eps);
+  else
+    hipLaunchKernelGGL(layer_norm_fwd_kernel<false>, dim3(grid), dim3(kBlock),
+                       0, stream, out, x, gamma, beta, mean, rstd, rows, cols,
+                       eps);
+}
+
+void epl_layer_norm_bwd(void* dx, float* dgamma, float* dbeta, const void* dy,
+                        const void* x, const void* gamma, const float* mean,
+                        const float* rstd, int64_t rows, int64_t cols,
+                        bool bf16, hipStream_t stream) {
+  int grid = (int)(rows < 1024 ? rows : 1024);
+  const size_t lds_bytes = (size_t)cols * 2 * sizeof(float);
+  if (bf16)
+    hipLaunchKernelGGL(layer_norm_bwd_kernel<true>, dim3(grid), dim3(kBlock),
+                       lds_bytes, stream, dx, dgamma, dbeta, dy, x, gamma,
+                       mean, rstd, rows, cols);
+  else
+    hipLaunchKernelGGL(layer_norm_bwd_kernel<false>, dim3(grid), dim3(kBlock),
+                       lds_bytes, stream, dx, dgamma, dbeta, dy, x, gamma,
+                       mean, rstd, rows, cols);
+}
+
+void epl_bias_gelu_fwd(void* out, const void* x, const void* bias,
+                       int64_t rows, int64_t cols, bool bf16,
+                       hipStream_t stream) {
+  const int grid = grid_for((rows * cols + 7) / 8);
+  if (bf16)
+    hipLaunchKernelGGL(bias_gelu_fwd_kernel<true>, dim3(grid), dim3(kBlock), 0,
+                       stream, out, x, bias, rows, cols);
+  else
+    hipLaunchKernelGGL(bias_gelu_fwd_kernel<false>, dim3(grid), dim3(kBlock),
+                       0, stream, out, x, bias, rows, cols);
+}
+
+void epl_bias_gelu_bwd(void* dx, float* dbias, const void* dy, const void* x,
+                       const void* bias, int64_t rows, int64_t cols, bool bf16,
+                       hipStream_t stream) {
+  int grid = (int)(rows < 1024 ? rows : 1024);
+  const size_t lds_bytes = (size_t)cols * sizeof(float);
+  if (bf16)
+    hipLaunchKernelGGL(bias_gelu_bwd_kernel<true>, dim3(grid), dim3(kBlock),
+                       lds_bytes, stream, dx, dbias, dy, x, bias, rows, cols);
+  else
+    hipLaunchKernelGGL(bias_gelu_bwd_kernel<false>, dim3(grid), dim3(kBlock),
+                       lds_bytes, stream, dx, dbias, dy, x, bias, rows, cols);
+}
+
+void epl_ce_rowstats(const void* logits, const int64_t* targets,
+                     float* row_max, float* row_sumexp, float* target_logit,
+                     int64_t rows, int64_t cols, int64_t vocab_begin,
+                     int64_t ignore_index, bool bf16, hipStream_t stream) {
+  const int grid = (int)(rows < kMaxGrid ? rows : kMaxGrid);
+  if (bf16)
+    hipLaunchKernelGGL(ce_rowstats_kernel<true>, dim3(grid), dim3(kBlock), 0,
+                       stream, logits, targets, row_max, row_sumexp,
+                       target_logit, rows, cols, vocab_begin, ignore_index);
+  else
+    hipLaunchKernelGGL(ce_rowstats_kernel<false>, dim3(grid), dim3(kBlock), 0,
+                       stream, logits, targets, row_max, row_sumexp,
+                       target_logit, rows, cols, vocab_begin, ignore_index);
+}
+
+void epl_ce_backward(void* dlogits, const void* logits, const int64_t* targets,
+                     const float* gmax, const float* gsumexp,
+                     const float* dloss, int64_t rows, int64_t cols,
+                     int64_t vocab_begin, int64_t ignore_index, float scale,
+                     bool bf16, hipStream_t stream) {
+  const int grid = (int)(rows < kMaxGrid ? rows : kMaxGrid);
+  if (bf16)
+    hipLaunchKernelGGL(ce_backward_kernel<true>, dim3(grid), dim3(kBlock), 0,
+                       stream, dlogits, logits, targets, gmax, gsumexp, dloss,
+                       rows, cols, vocab_begin, ignore_index, scale);
+  else
+    hipLaunchKernelGGL(ce_backward_kernel<false>, dim3(grid), dim3(kBlock), 0,
+                       stream, dlogits, logits, targets, gmax, gsumexp, dloss,
+                       rows, cols, vocab_begin, ignore_index, scale);
+}
+
+void epl_scale(void* p, int64_t n, float s, bool bf16, hipStream_t stream) {
+  const int grid = grid_for((n + 7) / 8);
+  if (bf16)
+    hipLaunchKernelGGL(scale_bf16_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       reinterpret_cast<unsigned short*>(p), n, s);
+  else
+    hipLaunchKernelGGL(scale_f32_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                       reinterpret_cast<float*>(p), n, s);
+}
+
+void epl_f32_to_bf16(unsigned short* dst, const float* src, int64_t n,
+                     hipStream_t stream) {
+  const int grid = grid_for((n + 7) / 8);
+  hipLaunchKernelGGL(f32_to_bf16_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                     dst, src, n);
+}
+
+void epl_bf16_to_f32(float* dst, const unsigned short* src, int64_t n,
+                     hipStream_t stream) {
+  const int grid = grid_for((n + 7) / 8);
+  hipLaunchKernelGGL(bf16_to_f32_kernel, dim3(grid), dim3(kBlock), 0, stream,
+                     dst, src, n);
+}
+
+void epl_sqnorm(const void* p, int64_t n, float* out, bool bf16,
+                hipStream_t stream) {
+  const int grid = grid_for(n) < 1024 ? grid_for(n) : 1024;
+  if (bf16)
+    hipLaunchKernelGGL(sqnorm_kernel<true>, dim3(grid), dim3(kBlock), 0,
+                       stream, p, n, out);
+  else
+    hipLaunchKernelGGL(sqnorm_kernel<false>, dim3(grid), dim3(kBlock), 0,
+                       stream, p, n, out);
+}
+
+}  // extern "C"

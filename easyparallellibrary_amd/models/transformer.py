@@ -1,0 +1,120 @@
+"""Transformer building blocks used by the BERT / GPT-2 / MoE model
+families (the reference's model zoo is external; these are the in-repo
+equivalents the benchmarks run).
+
+Hot ops ride the hand-written CDNA4 kernels: FusedLayerNorm,
+FusedBiasGelu (FFN first Linear runs bias-free); attention uses torch SDPA
+(hipBLASLt GEMMs).
+"""
+
+import math
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
+from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
+
+
+class SelfAttention(nn.Module):
+    def __init__(self, hidden, num_heads, causal=False, dropout=0.0):
+        super().__init__()
+        assert hidden % num_heads == 0
+        self.hidden = hidden
+        self.num_heads = num_heads
+        self.head_dim = hidden // num_heads
+        self.causal = causal
+        self.qkv = nn.Linear(hidden, 3 * hidden)
+        self.proj = nn.Linear(hidden, hidden)
+        self.dropout = dropout
+
+    def forward(self, x):
+        b, s, h = x.shape
+        qkv = self.qkv(x).reshape(b, s, 3, self.num_heads, self.head_dim)
+        q, k, v = qkv.unbind(dim=2)
+        q = q.transpose(1, 2)
+        k = k.transpose(1, 2)
+        v = v.transpose(1, 2)
+        o = F.scaled_dot_product_attention(
+            q, k, v, is_causal=self.causal,
+            dropout_p=self.dropout if self.training else 0.0)
+        o = o.transpose(1, 2).reshape(b, s, h)
+        return self.proj(o)
+
+
+class MLP(nn.Module):
+    """FFN with the fused bias+GeLU kernel between the two GEMMs."""
+
+    def __init__(self, hidden, ffn_hidden):
+        super().__init__()
+        self.fc1 = nn.Linear(hidden, ffn_hidden, bias=False)
+        self.act = FusedBiasGelu(ffn_hidden)
+        self.fc2 = nn.Linear(ffn_hidden, hidden)
+
+    def forward(self, x):
+        return self.fc2(self.act(self.fc1(x)))
+
+
+class Block(nn.Module):
+    """Transformer block; pre_ln=True for GPT-style, False for BERT."""
+
+    def __init__(self, hidden, num_heads, ffn_hidden, causal=False,
+                 pre_ln=False):
+        super().__init__()
+        self.pre_ln = pre_ln
+        self.ln1 = FusedLayerNorm(hidden)
+        self.attn = SelfAttention(hidden, num_heads, causal=causal)
+        self.ln2 = FusedLayerNorm(hidden)
+        self.mlp = MLP(hidden, ffn_hidden)
+
+    def forward(self, x):
+        if self.pre_ln:
+            x = x + self.attn(self.ln1(x))
+            x = x + self.mlp(self.ln2(x))
+        else:
+            x = self.ln1(x + self.attn(x))
+            x = self.ln2(x + self.mlp(x))
+        return x
+
+
+class Embeddings(nn.Module):
+    def __init__(self, vocab_size, hidden, max_positions, use_ln=True):
+        super().__init__()
+        self.tok = nn.Embedding(vocab_size, hidden)
+        self.pos = nn.Embedding(max_positions, hidden)
+        self.ln = FusedLayerNorm(hidden) if use_ln else None
+        nn.init.normal_(self.tok.weight, std=0.02)
+        nn.init.normal_(self.pos.weight, std=0.02)
+
+    def forward(self, ids):
+        s = ids.shape[1]
+        pos = torch.arange(s, device=ids.device)
+        x = self.tok(ids) + self.pos(pos)[None, :, :]
+        if self.ln is not None:
+            x = self.ln(x)
+        return x
+
+
+class LMHead(nn.Module):
+    """Projection to vocab logits (untied for simplicity)."""
+
+    def __init__(self, hidden, vocab_size, use_ln=True):
+        super().__init__()
+        self.ln = FusedLayerNorm(hidden) if use_ln else None
+        self.proj = nn.Linear(hidden, vocab_size, bias=False)
+        nn.init.normal_(self.proj.weight, std=0.02)
+
+    def forward(self, x):
+        if self.ln is not None:
+            x = self.ln(x)
+        return self.proj(x)
+
+
+def init_weights(model, std=0.02):
+    for m in model.modules():
+        if isinstance(m, nn.Linear):
+            nn.init.normal_(m.weight, std=std)
+            if m.bias is not None:
+                nn.init.zeros_(m.bias)
+    return model

@@ -1,0 +1,24 @@
+"""Replica<->Split bridging.
+
+Capability parity: /root/reference/epl/ops/bridging_layer.py:46-58
+(Replica2Split: allgather a replica-local tensor to the full batch before a
+split scope; backward reduce-scatters automatically via the autograd
+collective).
+"""
+
+from easyparallellibrary_amd.comm import functional
+
+
+def replica_to_split(x, comm):
+    """Gather the per-rank local batch into the full batch (dim 0) so a
+    split (sharded) scope sees every replica's samples."""
+    return functional.all_gather(x, comm)
+
+
+def split_to_replica(x, comm):
+    """Take this replica's slice of a full-batch tensor produced by a split
+    scope (inverse bridge; backward allgathers)."""
+    if comm.size == 1:
+        return x
+    n = x.shape[0] // comm.size
+    return x[comm.rank * n:(comm.rank + 1) * n]

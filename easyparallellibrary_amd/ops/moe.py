@@ -1,0 +1,120 @@
+"""Mixture-of-Experts with expert parallelism (all-to-all dispatch/combine).
+
+Capability parity: /root/reference/epl/parallel/hooks.py:758-794 (einsum
+hook inserting all-to-all before the first and after the third einsum
+around the expert weights) + parallel/ops.py:485-495 (alltoall with
+optional fp16 compression) + csrc nccl_all_to_all.cc.
+
+MI355X redesign: an explicit MoE layer — top-1/top-2 gating with capacity,
+dispatch as ONE equal-split all-to-all over xGMI, local expert FFN batched
+as a single bmm over the rank's experts, combine with the reverse
+all-to-all.  The a2a autograd pair (a2a <-> a2a) lives in
+comm/functional.py.
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from easyparallellibrary_amd.comm import functional
+
+
+class ExpertParallelMLP(nn.Module):
+    """num_experts split across the comm group; each rank holds
+    num_experts // world experts (reference tests use the same layout,
+    tests/split_test.py:30-90)."""
+
+    def __init__(self, hidden, ffn_hidden, num_experts, comm=None, top_k=2,
+                 capacity_factor=1.25):
+        super().__init__()
+        self.comm = comm
+        self.world = comm.size if comm is not None else 1
+        assert num_experts % self.world == 0, \
+            "num_experts must divide the split degree"
+        self.num_experts = num_experts
+        self.ffn_hidden = ffn_hidden
+        self.local_experts = num_experts // self.world
+        self.hidden = hidden
+        self.top_k = top_k
+        self.capacity_factor = capacity_factor
+        self.gate = nn.Linear(hidden, num_experts, bias=False)
+        self.w1 = nn.Parameter(
+            torch.empty(self.local_experts, hidden, ffn_hidden))
+        self.w2 = nn.Parameter(
+            torch.empty(self.local_experts, ffn_hidden, hidden))
+        nn.init.normal_(self.w1, std=0.02)
+        nn.init.normal_(self.w2, std=0.02)
+
+    def set_comm(self, comm):
+        """Called by the engine's split transform: shard the expert weights
+        constructed with comm=None (full expert set) down to this rank's
+        slice (reference: expert weights inside epl.split, hooks.py:758)."""
+        if comm is None or comm.size == 1:
+            self.comm = comm
+            return
+        assert self.world == 1, "expert weights already sharded"
+        self.comm = comm
+        self.world = comm.size
+        assert self.num_experts % self.world == 0
+        self.local_experts = self.num_experts // self.world
+        shard = max(comm.rank, 0)
+        lo = shard * self.local_experts
+        hi = lo + self.local_experts
+        with torch.no_grad():
+            self.w1 = nn.Parameter(self.w1[lo:hi].clone())
+            self.w2 = nn.Parameter(self.w2[lo:hi].clone())
+
+    def forward(self, x):
+        orig_shape = x.shape
+        x = x.reshape(-1, self.hidden)
+        n_tokens = x.shape[0]
+        logits = self.gate(x.float())
+        probs = logits.softmax(dim=-1)
+        topv, topi = probs.topk(self.top_k, dim=-1)
+        topv = topv / topv.sum(dim=-1, keepdim=True)
+
+        capacity = max(
+            1, int(self.capacity_factor * n_tokens * self.top_k /
+                   self.num_experts))
+        # dispatch tensor: [num_experts, capacity, hidden]
+        dispatched = x.new_zeros(self.num_experts, capacity, self.hidden)
+        combine_idx = []  # (expert, slot, token, weight)
+        slot_count = torch.zeros(self.num_experts, dtype=torch.long,
+                                 device=x.device)
+        flat_e = topi.reshape(-1)
+        flat_t = (torch.arange(n_tokens, device=x.device)
+                  .repeat_interleave(self.top_k))
+        flat_w = topv.reshape(-1)
+        # slot assignment per expert (ordered, capacity-dropped)
+        order = torch.argsort(flat_e, stable=True)
+        fe, ft, fw = flat_e[order], flat_t[order], flat_w[order]
+        # position within expert via segmented arange
+        counts = torch.bincount(fe, minlength=self.num_experts)
+        seg_start = torch.nn.functional.pad(counts.cumsum(0), (1, 0))[:-1]
+        pos_in_e = torch.arange(fe.numel(), device=x.device) - seg_start[fe]
+        keep = pos_in_e < capacity
+        fe, ft, fw, pos_in_e = fe[keep], ft[keep], fw[keep], pos_in_e[keep]
+        dispatched[fe, pos_in_e] = x[ft]
+
+        # all-to-all: [world, local_experts*capacity, hidden]
+        d = dispatched.reshape(self.world,
+                               self.local_experts * capacity, self.hidden)
+        if self.comm is not None and self.world > 1:
+            d = functional.all_to_all(d.contiguous(), self.comm)
+        # now d[w] = tokens sent by rank w for MY local experts
+        d = d.reshape(self.world, self.local_experts, capacity, self.hidden)
+        d = d.transpose(0, 1).reshape(self.local_experts,
+                                      self.world * capacity, self.hidden)
+        h = torch.bmm(d, self.w1)
+        h = F.gelu(h)
+        h = torch.bmm(h, self.w2)
+        h = h.reshape(self.local_experts, self.world, capacity, self.hidden)
+        h = h.transpose(0, 1).reshape(
+            self.world, self.local_experts * capacity, self.hidden)
+        if self.comm is not None and self.world > 1:
+            h = functional.all_to_all(h.contiguous(), self.comm)
+        h = h.reshape(self.num_experts, capacity, self.hidden)
+
+        out = x.new_zeros(n_tokens, self.hidden)
+        out.index_add_(0, ft, h[fe, pos_in_e] * fw.unsqueeze(-1).to(h.dtype))
+        return out.reshape(orig_shape)

@@ -1,0 +1,101 @@
+"""Split-scope module transformation.
+
+Capability parity: /root/reference/epl/parallel/hooks.py split-mode op
+replacement — dense :710-723 (-> ops/distributed_dense.py), add_weight
+row-sharding :667-707, sparse softmax :726, argmax :742, equal :797.
+
+MI355X redesign: the reference swaps TF python calls at trace time; here
+the engine rewrites the already-constructed module tree of a split
+taskgraph once at plan time: nn.Linear -> ColumnParallelLinear (weight
+shard copied out of the full layer so initialization statistics match
+single-device training exactly), nn.Embedding -> row-sharded embedding.
+The model's forward code is untouched.
+"""
+
+import torch
+import torch.nn as nn
+
+from easyparallellibrary_amd.comm import functional
+from easyparallellibrary_amd.ops.distributed_dense import (
+    ColumnParallelLinear, shard_offset, shard_size)
+from easyparallellibrary_amd.utils.logging import get_logger
+
+logger = get_logger()
+
+
+class VocabParallelEmbedding(nn.Module):
+    """Row-sharded embedding (reference add_weight dim-0 sharding,
+    hooks.py:667-707): each shard holds rows [begin, begin+local); out-of-
+    range ids embed to zero and the partial outputs are all-reduced."""
+
+    def __init__(self, num_embeddings, embedding_dim, comm, source=None):
+        super().__init__()
+        self.comm = comm
+        self.nshards = comm.size
+        self.shard = max(comm.rank, 0)
+        self.num_embeddings = num_embeddings
+        self.local_rows = shard_size(num_embeddings, self.nshards, self.shard)
+        self.begin = shard_offset(num_embeddings, self.nshards, self.shard)
+        self.weight = nn.Parameter(
+            torch.empty(self.local_rows, embedding_dim))
+        if source is not None:
+            with torch.no_grad():
+                self.weight.copy_(
+                    source.weight[self.begin:self.begin + self.local_rows])
+        else:
+            nn.init.normal_(self.weight)
+
+    def forward(self, ids):
+        in_range = (ids >= self.begin) & (ids < self.begin + self.local_rows)
+        safe = (ids - self.begin).clamp(0, self.local_rows - 1)
+        out = nn.functional.embedding(safe, self.weight)
+        out = out * in_range.unsqueeze(-1).to(out.dtype)
+        return functional.all_reduce(out, self.comm)
+
+
+def _replace_module(parent, name, new):
+    setattr(parent, name, new)
+
+
+def transform_taskgraph(tg, comm, gather_input=True):
+    """Rewrite every Linear/Embedding under the taskgraph's module roots
+    into its sharded equivalent over ``comm``."""
+    from easyparallellibrary_amd.ops.moe import ExpertParallelMLP
+    replaced = 0
+    for root_i, root in enumerate(tg.modules):
+        if isinstance(root, (nn.Linear, nn.Embedding)):
+            new = _make_sharded(root, comm, gather_input)
+            tg.modules[root_i] = new
+            replaced += 1
+            continue
+        if isinstance(root, ExpertParallelMLP):
+            root.set_comm(comm)
+            replaced += 1
+            continue
+        for parent_name, parent in root.named_modules():
+            if isinstance(parent, ExpertParallelMLP):
+                parent.set_comm(comm)
+                replaced += 1
+                continue
+            for child_name, child in list(parent.named_children()):
+                if isinstance(child, ExpertParallelMLP):
+                    child.set_comm(comm)
+                    replaced += 1
+                elif isinstance(child, (nn.Linear, nn.Embedding)):
+                    _replace_module(parent, child_name,
+                                    _make_sharded(child, comm, gather_input))
+                    replaced += 1
+    logger.info("split transform: %d module(s) sharded %d-way over %s",
+                replaced, comm.size, comm.name)
+    return replaced
+
+
+def _make_sharded(mod, comm, gather_input):
+    if isinstance(mod, nn.Linear):
+        return ColumnParallelLinear(
+            mod.in_features, mod.out_features, comm,
+            bias=mod.bias is not None, gather_input=gather_input, source=mod)
+    if isinstance(mod, nn.Embedding):
+        return VocabParallelEmbedding(
+            mod.num_embeddings, mod.embedding_dim, comm, source=mod)
+    raise TypeError(type(mod))

@@ -1,0 +1,283 @@
+"""The parallel execution engine.
+
+Capability parity: /root/reference/epl/parallel/parallel.py
+(device_replacement :120-135, do_parallelism :211-231, merge_outputs
+:233-353) + graph_editor.py (replica/micro-batch handling :389-443,
+gradient aggregation :610-725).
+
+MI355X redesign: no graph cloning.  Each rank instantiates the model once,
+keeps only the taskgraphs (pipeline stage / shard) it owns, flattens their
+parameters into device arenas (parallel/dp.py), and executes either the
+plain forward (DP / TP) or the micro-batch pipeline schedule
+(parallel/pipeline.py).  All gradient traffic is bucketed RCCL over xGMI,
+overlapped with backward; the optimizer is one fused CDNA4 kernel per
+arena (runtime/optim.py).
+"""
+
+import torch
+import torch.distributed as dist
+import torch.nn as nn
+
+from easyparallellibrary_amd import constant
+from easyparallellibrary_amd.cluster import Cluster, Layout
+from easyparallellibrary_amd.comm.backend import create_communicator
+from easyparallellibrary_amd.comm.pool import CommunicationPool
+from easyparallellibrary_amd.env import Env
+from easyparallellibrary_amd.ir.plan import Plan
+from easyparallellibrary_amd.parallel.dp import FlatParamGroup, GradReducer
+from easyparallellibrary_amd.runtime.optim import OPTIMIZERS
+from easyparallellibrary_amd.utils.logging import get_logger
+
+logger = get_logger()
+
+
+class StageModule(nn.Module):
+    """A pipeline stage: the taskgraph's maximal module roots run in
+    registration order (the model must be stage-chainable — each stage
+    output feeds the next stage, mirroring the reference's stage
+    entrance/exit dataflow, ir/taskgraph.py:134-400)."""
+
+    def __init__(self, modules):
+        super().__init__()
+        self.mods = nn.ModuleList(modules)
+
+    def forward(self, x):
+        for m in self.mods:
+            x = m(x)
+        return x
+
+
+class Engine:
+    _seq = 0
+
+    def __init__(self, model, loss_fn=None, optimizer="adamw", lr=1e-3,
+                 optimizer_kwargs=None, dtype=None):
+        Engine._seq += 1
+        self._ns = "e{}".format(Engine._seq)  # unique comm-name namespace
+        self.env = Env.get()
+        self.config = self.env.config
+        self.env.get_or_create_process_group()
+        self.cluster = Cluster()
+        self.env.cluster = self.cluster
+        self.device = self.cluster.device
+        self.loss_fn = loss_fn
+        self.dtype = dtype or torch.float32
+        self.model = model
+
+        # ---- plan -----------------------------------------------------------
+        self.plan = Plan.build(model, self.env.strategy_context)
+        colocate = self.config.cluster.colocate_split_and_replicate
+        counts = self.plan.effective_device_counts(colocate)
+        layout_counts = [c for c in counts if c > 0]
+        if not layout_counts:
+            layout_counts = [1]
+        self.layout = Layout(self.cluster.world_size, layout_counts)
+        vds = self.layout.slices()
+        # assign virtual devices; colocated split tg shares the slice of the
+        # preceding replicate tg with equal device_count
+        vi = 0
+        prev_by_count = {}
+        for tg, c in zip(self.plan.taskgraphs, counts):
+            if c == 0:
+                tg.virtual_device = prev_by_count[tg.device_count]
+            else:
+                tg.virtual_device = vds[vi]
+                vi += 1
+                if not tg.is_split:
+                    prev_by_count[tg.device_count] = tg.virtual_device
+
+        self.rank = self.cluster.rank
+        self.world_size = self.cluster.world_size
+        self.per_replica = self.layout.per_replica
+        self.replica_id = self.rank // self.per_replica
+        self.num_replicas = self.layout.num_replicas
+
+        # ---- control channel (object exchange; gloo works on GPU hosts too)
+        self._control_group = None
+        if dist.is_initialized() and dist.get_backend() != "gloo":
+            self._control_group = dist.new_group(backend="gloo")
+
+        # ---- stages ---------------------------------------------------------
+        self.stage_tgs = [tg for tg in self.plan.taskgraphs
+                          if tg.strategy_type == constant.REPLICATE]
+        self.num_stages = len(self.stage_tgs) if len(self.stage_tgs) > 1 else 1
+        self.my_stage = 0
+        if self.num_stages > 1:
+            # stage s of my replica runs on the tg's virtual-device ranks
+            self.my_stage = None
+            for s, tg in enumerate(self.stage_tgs):
+                if self.rank in tg.virtual_device.local_ranks(self.replica_id):
+                    self.my_stage = s
+            assert self.my_stage is not None, (
+                "rank {} owns no pipeline stage".format(self.rank))
+
+        # ---- TP (split) transformation --------------------------------------
+        self.tp_comm = None
+        self._owned_tgs = []
+        for tg in self.plan.taskgraphs:
+            ranks = tg.virtual_device.local_ranks(self.replica_id)
+            if self.rank in ranks:
+                self._owned_tgs.append(tg)
+        all_split_tgs = [tg for tg in self.plan.taskgraphs if tg.is_split]
+        if all_split_tgs:
+            from easyparallellibrary_amd.ops import split_transform
+            for tg in all_split_tgs:
+                for rep in range(self.num_replicas):
+                    ranks = tg.virtual_device.local_ranks(rep)
+                    mine = self.rank in ranks
+                    if not mine and (torch.cuda.is_available()
+                                     or not dist.is_initialized()):
+                        continue  # RCCL init is group-local
+                    comm = create_communicator(
+                        "{}_tp_tg{}_rep{}".format(self._ns, tg.index, rep),
+                        ranks)
+                    if mine:
+                        self.tp_comm = comm
+                        split_transform.transform_taskgraph(tg, comm)
+
+        # ---- materialize my modules on device --------------------------------
+        if self.num_stages > 1:
+            self.stage_module = StageModule(self.stage_tgs[self.my_stage].modules)
+            self._runnable = self.stage_module
+        else:
+            self._runnable = model
+        for tg in self._owned_tgs:
+            for m in tg.modules:
+                m.to(self.device, dtype=self.dtype)
+            # keep norm/embedding-adjacent fp32 buffers as-is? parameters all
+            # follow engine dtype; fp32-master lives in the arenas.
+
+        # ---- DP groups + reducers -------------------------------------------
+        # Every rank must create communicators in the same global order
+        # (gloo's new_group is collective).  Enumerate all (taskgraph,
+        # position) DP groups deterministically.
+        self.flat_groups = []
+        self.reducers = []
+        pool_n = self.config.communication.num_communicators
+        bucket_bytes = self.config.communication.bucket_bytes
+        reduce_method = self.config.communication.gradients_reduce_method
+        overlap = self.config.communication.overlap_grad_reduce
+        self._bcast_jobs = []
+        for tg in self.plan.taskgraphs:
+            vd = tg.virtual_device
+            k = len(vd.local_ranks(0))
+            if tg.is_split:
+                # shard i's copies across replicas
+                groups = [[vd.local_ranks(r)[i] for r in range(self.num_replicas)]
+                          for i in range(k)]
+            else:
+                # all ranks holding this (replicated) taskgraph
+                groups = [vd.all_ranks]
+            for gi, granks in enumerate(groups):
+                name = "{}_dp_tg{}_g{}".format(self._ns, tg.index, gi)
+                if self.rank not in granks:
+                    if not torch.cuda.is_available() and dist.is_initialized():
+                        # gloo group creation is collective
+                        CommunicationPool(name, granks, pool_n)
+                        create_communicator(name + "_b", granks)
+                    continue
+                pool = CommunicationPool(name, granks, pool_n)
+                bcomm = create_communicator(name + "_b", granks)
+                params = [p for p in tg.parameters() if p.requires_grad]
+                if not params:
+                    continue
+                fg = FlatParamGroup(params, self.device,
+                                    model_dtype=self.dtype)
+                reducer = GradReducer(fg, pool, bucket_bytes,
+                                      reduce_method=reduce_method,
+                                      overlap=overlap)
+                self.flat_groups.append(fg)
+                self.reducers.append(reducer)
+                self._bcast_jobs.append((fg, bcomm))
+
+        # ---- initial weight sync (reference: hooks.py:330-357) ---------------
+        for fg, bcomm in self._bcast_jobs:
+            if bcomm is not None and bcomm.size > 1:
+                bcomm.broadcast(fg.param_arena, root=0)
+                if fg.master_arena is not fg.param_arena:
+                    fg.master_arena.copy_(fg.param_arena.to(torch.float32))
+
+        # ---- optimizer -------------------------------------------------------
+        okw = dict(optimizer_kwargs or {})
+        okw.setdefault("lr", lr)
+        self.zero = None
+        if self.config.zero.level:
+            from easyparallellibrary_amd.runtime.zero import ZeroOptimizer
+            self.optimizer = ZeroOptimizer(
+                self, optimizer, okw, level=self.config.zero.level)
+        else:
+            self.optimizer = OPTIMIZERS[optimizer](self.flat_groups, **okw)
+
+        # ---- pipeline runtime ------------------------------------------------
+        self.pipeline = None
+        if self.num_stages > 1:
+            from easyparallellibrary_amd.parallel.pipeline import (
+                PipelineRuntime)
+            self.pipeline = PipelineRuntime(self)
+
+        self.global_step = 0
+        logger.info(
+            "Engine ready: world=%d stages=%d replicas=%d per_replica=%d "
+            "dtype=%s groups=%d", self.world_size, self.num_stages,
+            self.num_replicas, self.per_replica, self.dtype,
+            len(self.flat_groups))
+
+    # ---- helpers -------------------------------------------------------------
+    @property
+    def num_micro_batch(self):
+        return max(1, self.config.pipeline.num_micro_batch)
+
+    def _set_reducers_enabled(self, flag):
+        for r in self.reducers:
+            r.enabled = flag
+
+    def finish_grad_sync(self):
+        for r in self.reducers:
+            r.finish()
+
+    def zero_grad(self):
+        for fg in self.flat_groups:
+            fg.zero_grad()
+
+    def forward(self, x):
+        return self._runnable(x)
+
+    # ---- the training step ---------------------------------------------------
+    def train_step(self, inputs, targets):
+        """One optimizer step: micro-batch loop (pipeline or GA), overlapped
+        DP gradient reduction, fused optimizer.  Returns the local mean loss
+        tensor (on the last stage; None elsewhere for pipelines)."""
+        self.zero_grad()
+        if self.pipeline is not None:
+            loss = self.pipeline.run(inputs, targets)
+        else:
+            loss = self._train_step_simple(inputs, targets)
+        grad_scale = float(self.num_micro_batch)
+        self.optimizer.step(grad_scale=grad_scale)
+        self.global_step += 1
+        return loss
+
+    def _train_step_simple(self, inputs, targets):
+        nmb = self.num_micro_batch
+        if nmb > 1:
+            input_chunks = torch.chunk(inputs, nmb, dim=0)
+            target_chunks = torch.chunk(targets, nmb, dim=0)
+        else:
+            input_chunks = [inputs]
+            target_chunks = [targets]
+        total_loss = None
+        for i, (xc, tc) in enumerate(zip(input_chunks, target_chunks)):
+            last = (i == nmb - 1)
+            self._set_reducers_enabled(last)
+            out = self._runnable(xc)
+            loss = self.loss_fn(out, tc)
+            loss.backward()
+            total_loss = loss.detach() if total_loss is None \
+                else total_loss + loss.detach()
+        self.finish_grad_sync()
+        return total_loss / nmb
+
+    # ---- eval ----------------------------------------------------------------
+    def eval_step(self, inputs):
+        with torch.no_grad():
+            return self._runnable(inputs)

@@ -1,0 +1,247 @@
+"""Pipeline-parallel runtime: micro-batch schedules over RCCL p2p.
+
+Capability parity: /root/reference/epl/strategies/scheduler.py — the three
+schedules PreferForward (GPipe, :36-50), PreferBackward (1F1B, :53-84, the
+default) and PreferBackwardOptimizer (:87-116) — plus
+graph_editor.py:610-668 (micro-batch gradient accumulation before DP
+allreduce).
+
+MI355X redesign: the reference encodes schedules as control-dependency
+wiring between cloned per-(stage, micro-batch) subgraphs.  Here each
+schedule is an imperative loop: activations/gradients move between
+adjacent stages as RCCL send/recv over one xGMI hop (stage neighbours are
+adjacent ranks by the replica-contiguous Layout), batched into one nccl
+group per steady-state exchange so the bidirectional pair cannot deadlock.
+Gradient accumulation falls out of p.grad views into the flat arena; the
+DP allreduce hooks stay disabled until the schedule's final backward.
+"""
+
+import torch
+import torch.distributed as dist
+
+from easyparallellibrary_amd import constant
+from easyparallellibrary_amd.comm.backend import create_communicator
+
+
+class PipelineRuntime:
+    def __init__(self, engine):
+        self.engine = engine
+        self.S = engine.num_stages
+        self.s = engine.my_stage
+        replica = engine.replica_id
+        # rank of each stage within my replica (device_count=1 per stage)
+        self.stage_ranks = [
+            tg.virtual_device.local_ranks(replica)[0]
+            for tg in engine.stage_tgs
+        ]
+        self.prev_rank = self.stage_ranks[self.s - 1] if self.s > 0 else None
+        self.next_rank = (self.stage_ranks[self.s + 1]
+                          if self.s < self.S - 1 else None)
+        # one p2p communicator per replica chain
+        self.comm = create_communicator(
+            "{}_pipe_rep{}".format(engine._ns, replica), self.stage_ranks)
+        self.group_rank = {r: i for i, r in enumerate(self.stage_ranks)}
+        self.schedule = engine.config.pipeline.strategy
+        self.dtype = engine.dtype
+        self.device = engine.device
+        self._shape_cache = {}  # microbatch idx -> (shape, dtype)
+        self._shapes_known = False
+
+    # ---- shape handshake (first step only) -----------------------------------
+    def _send_shape(self, t, peer):
+        if self._shapes_known:
+            return
+        meta = [tuple(t.shape), str(t.dtype)]
+        if dist.is_initialized():
+            dist.send_object_list(meta, dst=peer,
+                                  group=self.engine._control_group)
+
+    def _recv_shape(self, i, peer):
+        if i in self._shape_cache:
+            return self._shape_cache[i]
+        meta = [None, None]
+        dist.recv_object_list(meta, src=peer,
+                              group=self.engine._control_group)
+        shape = tuple(meta[0])
+        dtype = getattr(torch, meta[1].replace("torch.", ""))
+        self._shape_cache[i] = (shape, dtype)
+        return shape, dtype
+
+    # ---- p2p primitives -------------------------------------------------------
+    def _g(self, global_rank):
+        return self.group_rank[global_rank]
+
+    def recv_forward(self, i):
+        shape, dtype = self._recv_shape(i, self.prev_rank)
+        t = torch.empty(shape, dtype=dtype, device=self.device)
+        self.comm.recv(t, self._g(self.prev_rank))
+        return t
+
+    def send_forward(self, t, i):
+        self._send_shape(t, self.next_rank)
+        self.comm.send(t.contiguous(), self._g(self.next_rank))
+
+    def recv_backward(self, i):
+        shape, dtype = self._shape_cache_out[i]
+        t = torch.empty(shape, dtype=dtype, device=self.device)
+        self.comm.recv(t, self._g(self.next_rank))
+        return t
+
+    def send_backward(self, g, i):
+        self.comm.send(g.contiguous(), self._g(self.prev_rank))
+
+    def send_forward_recv_backward(self, t, i_send, i_recv):
+        self._send_shape(t, self.next_rank)
+        shape, dtype = self._shape_cache_out[i_recv]
+        g = torch.empty(shape, dtype=dtype, device=self.device)
+        self.comm.batch_p2p([
+            (True, t.contiguous(), self._g(self.next_rank)),
+            (False, g, self._g(self.next_rank)),
+        ])
+        return g
+
+    def send_backward_recv_forward(self, g, i_recv):
+        shape, dtype = self._recv_shape(i_recv, self.prev_rank)
+        t = torch.empty(shape, dtype=dtype, device=self.device)
+        self.comm.batch_p2p([
+            (True, g.contiguous(), self._g(self.prev_rank)),
+            (False, t, self._g(self.prev_rank)),
+        ])
+        return t
+
+    # ---- forward/backward wrappers -------------------------------------------
+    def _forward(self, inp, target):
+        out = self.engine.stage_module(inp)
+        loss = None
+        if self.s == self.S - 1:
+            loss = self.engine.loss_fn(out, target)
+        return out, loss
+
+    def _backward(self, inp, out, loss, grad_out, is_last_backward):
+        self.engine._set_reducers_enabled(is_last_backward)
+        if loss is not None:
+            loss.backward()
+        else:
+            out.backward(grad_out)
+        if inp is not None and inp.requires_grad:
+            g = inp.grad
+            inp.grad = None
+            return g
+        return None
+
+    # ---- schedules ------------------------------------------------------------
+    def run(self, inputs, targets):
+        M = self.engine.num_micro_batch
+        input_chunks = (torch.chunk(inputs, M, dim=0)
+                        if self.s == 0 else [None] * M)
+        target_chunks = (torch.chunk(targets, M, dim=0)
+                         if self.s == self.S - 1 else [None] * M)
+        self._shape_cache_out = {}
+        self.engine._set_reducers_enabled(False)
+        if self.schedule == constant.SCHEDULER_PREFER_FORWARD:
+            loss = self._run_gpipe(M, input_chunks, target_chunks)
+        else:
+            loss = self._run_1f1b(M, input_chunks, target_chunks)
+        self._shapes_known = True
+        self.engine.finish_grad_sync()
+        return loss
+
+    def _get_input(self, i, input_chunks):
+        if self.s == 0:
+            x = input_chunks[i].to(self.device)
+            return x
+        x = self.recv_forward(i)
+        x.requires_grad_(True)
+        return x
+
+    def _emit_forward(self, out, i):
+        """Track output meta for matching backward recv."""
+        self._shape_cache_out[i] = (tuple(out.shape), out.dtype)
+        if self.s < self.S - 1:
+            self.send_forward(out, i)
+
+    def _run_gpipe(self, M, input_chunks, target_chunks):
+        """PreferForward: all forwards, then all backwards
+        (reference: scheduler.py:36-50)."""
+        saved = []
+        total_loss = None
+        for i in range(M):
+            x = self._get_input(i, input_chunks)
+            out, loss = self._forward(x, target_chunks[i])
+            self._emit_forward(out, i)
+            saved.append((x, out, loss))
+            if loss is not None:
+                total_loss = loss.detach() if total_loss is None \
+                    else total_loss + loss.detach()
+        for i in range(M):
+            x, out, loss = saved[i]
+            grad_out = None
+            if self.s < self.S - 1:
+                grad_out = self.recv_backward(i)
+            gin = self._backward(x, out, loss, grad_out, i == M - 1)
+            if self.s > 0:
+                self.send_backward(gin, i)
+        return None if total_loss is None else total_loss / M
+
+    def _run_1f1b(self, M, input_chunks, target_chunks):
+        """PreferBackward (1F1B) — also used for PreferBackwardOptimizer
+        (the apply overlap is inherent here: the fused optimizer launches
+        right after the last backward, overlapping the tail of the DP
+        allreduce).  Reference: scheduler.py:53-116."""
+        warmup = min(self.S - 1 - self.s, M)
+        remaining = M - warmup
+        in_q = []
+        total_loss = None
+
+        def note_loss(loss):
+            nonlocal total_loss
+            if loss is not None:
+                total_loss = (loss.detach() if total_loss is None
+                              else total_loss + loss.detach())
+
+        fi = 0  # forward microbatch index
+        bi = 0  # backward microbatch index
+        for _ in range(warmup):
+            x = self._get_input(fi, input_chunks)
+            out, loss = self._forward(x, target_chunks[fi])
+            note_loss(loss)
+            self._shape_cache_out[fi] = (tuple(out.shape), out.dtype)
+            if self.s < self.S - 1:
+                self.send_forward(out, fi)
+            in_q.append((x, out, loss))
+            fi += 1
+        x = self._get_input(fi, input_chunks) if remaining > 0 else None
+        for k in range(remaining):
+            out, loss = self._forward(x, target_chunks[fi])
+            note_loss(loss)
+            self._shape_cache_out[fi] = (tuple(out.shape), out.dtype)
+            if self.s < self.S - 1:
+                grad_out = self.send_forward_recv_backward(out, fi, bi)
+            else:
+                grad_out = None
+            in_q.append((x, out, loss))
+            fi += 1
+            ox, oout, oloss = in_q.pop(0)
+            is_last = (k == remaining - 1) and warmup == 0
+            gin = self._backward(ox, oout, oloss, grad_out, is_last)
+            bi += 1
+            if k != remaining - 1:
+                if self.s > 0:
+                    x = self.send_backward_recv_forward(gin, fi)
+                    x.requires_grad_(True)
+                else:
+                    x = self._get_input(fi, input_chunks)
+            else:
+                if self.s > 0:
+                    self.send_backward(gin, bi - 1)
+                x = None
+        for k in range(warmup):
+            grad_out = (self.recv_backward(bi)
+                        if self.s < self.S - 1 else None)
+            ox, oout, oloss = in_q.pop(0)
+            gin = self._backward(ox, oout, oloss, grad_out,
+                                 k == warmup - 1)
+            if self.s > 0:
+                self.send_backward(gin, bi)
+            bi += 1
+        return None if total_loss is None else total_loss / M
