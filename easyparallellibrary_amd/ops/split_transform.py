@@ -57,15 +57,24 @@ def _replace_module(parent, name, new):
     setattr(parent, name, new)
 
 
-def transform_taskgraph(tg, comm, gather_input=True):
+def transform_taskgraph(tg, comm, gather_input=True, model=None):
     """Rewrite every Linear/Embedding under the taskgraph's module roots
-    into its sharded equivalent over ``comm``."""
+    into its sharded equivalent over ``comm``.  ``model`` (the user's root
+    module) lets a taskgraph root that is itself a Linear be rewired into
+    its parent by dotted name."""
     from easyparallellibrary_amd.ops.moe import ExpertParallelMLP
     replaced = 0
     for root_i, root in enumerate(tg.modules):
         if isinstance(root, (nn.Linear, nn.Embedding)):
             new = _make_sharded(root, comm, gather_input)
             tg.modules[root_i] = new
+            if model is not None:
+                name = tg.module_names[root_i]
+                parts = name.split(".")
+                parent = model
+                for p in parts[:-1]:
+                    parent = getattr(parent, p)
+                setattr(parent, parts[-1], new)
             replaced += 1
             continue
         if isinstance(root, ExpertParallelMLP):

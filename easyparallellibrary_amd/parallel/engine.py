@@ -133,7 +133,8 @@ class Engine:
                         ranks)
                     if mine:
                         self.tp_comm = comm
-                        split_transform.transform_taskgraph(tg, comm)
+                        split_transform.transform_taskgraph(tg, comm,
+                                                            model=model)
 
         # ---- materialize my modules on device --------------------------------
         if self.num_stages > 1:

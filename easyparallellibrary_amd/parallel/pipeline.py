@@ -37,9 +37,22 @@ class PipelineRuntime:
         self.prev_rank = self.stage_ranks[self.s - 1] if self.s > 0 else None
         self.next_rank = (self.stage_ranks[self.s + 1]
                           if self.s < self.S - 1 else None)
-        # one p2p communicator per replica chain
-        self.comm = create_communicator(
-            "{}_pipe_rep{}".format(engine._ns, replica), self.stage_ranks)
+        # one p2p communicator per replica chain.  gloo group creation is
+        # collective over ALL ranks, so every rank walks every replica's
+        # chain in the same order and keeps its own.
+        self.comm = None
+        for rep in range(engine.num_replicas):
+            chain = [tg.virtual_device.local_ranks(rep)[0]
+                     for tg in engine.stage_tgs]
+            mine = engine.rank in chain
+            if not mine and (torch.cuda.is_available()
+                             or not dist.is_initialized()):
+                continue
+            comm = create_communicator(
+                "{}_pipe_rep{}".format(engine._ns, rep), chain)
+            if mine:
+                self.comm = comm
+        assert self.comm is not None
         self.group_rank = {r: i for i, r in enumerate(self.stage_ranks)}
         self.schedule = engine.config.pipeline.strategy
         self.dtype = engine.dtype
