@@ -20,8 +20,9 @@ void epl_lamb_phase1(const float*, const void*, bool, float*, float*, float*,
                      float, int, float, hipStream_t);
 void epl_lamb_phase2(float*, unsigned short*, const float*, const int*,
                      const float*, int64_t, float, hipStream_t);
-void epl_layer_norm_fwd(void*, const void*, const void*, const void*, float*,
-                        float*, int64_t, int64_t, float, bool, hipStream_t);
+void epl_layer_norm_fwd(void*, const void*, const void*, void*, const void*,
+                        const void*, float*, float*, int64_t, int64_t, float,
+                        bool, hipStream_t);
 void epl_layer_norm_bwd(void*, float*, float*, const void*, const void*,
                         const void*, const float*, const float*, int64_t,
                         int64_t, bool, hipStream_t);
@@ -112,7 +113,9 @@ void lamb_phase2(at::Tensor master, c10::optional<at::Tensor> param_bf16,
                   master.numel(), (float)lr, cur_stream());
 }
 
-void layer_norm_fwd(at::Tensor out, at::Tensor x, at::Tensor gamma,
+void layer_norm_fwd(at::Tensor out, at::Tensor x,
+                    c10::optional<at::Tensor> res,
+                    c10::optional<at::Tensor> sum_out, at::Tensor gamma,
                     at::Tensor beta, at::Tensor mean, at::Tensor rstd,
                     double eps) {
   const bool bf16 = is_bf16(x);
@@ -121,10 +124,19 @@ void layer_norm_fwd(at::Tensor out, at::Tensor x, at::Tensor gamma,
   TORCH_CHECK(!bf16 || cols % 8 == 0, "bf16 LayerNorm needs cols % 8 == 0");
   check(mean, at::kFloat, "mean");
   check(rstd, at::kFloat, "rstd");
-  epl_layer_norm_fwd(out.data_ptr(), x.data_ptr(), gamma.data_ptr(),
-                     beta.data_ptr(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), rows, cols, (float)eps, bf16,
-                     cur_stream());
+  const void* res_p = nullptr;
+  void* sum_p = nullptr;
+  if (res.has_value()) {
+    TORCH_CHECK(sum_out.has_value(),
+                "fused residual LayerNorm needs sum_out");
+    TORCH_CHECK(res->numel() == x.numel() && sum_out->numel() == x.numel());
+    res_p = res->data_ptr();
+    sum_p = sum_out->data_ptr();
+  }
+  epl_layer_norm_fwd(out.data_ptr(), x.data_ptr(), res_p, sum_p,
+                     gamma.data_ptr(), beta.data_ptr(),
+                     mean.data_ptr<float>(), rstd.data_ptr<float>(), rows,
+                     cols, (float)eps, bf16, cur_stream());
 }
 
 void layer_norm_bwd(at::Tensor dx, at::Tensor dgamma, at::Tensor dbeta,

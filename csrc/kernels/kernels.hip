@@ -237,15 +237,22 @@ __global__ void lamb_phase2_kernel(float* __restrict__ master,
 }
 
 // ============================================================================
-// LayerNorm forward: one block per row, fp32 accumulation, bf16x8 loads.
-// Saves mean and rstd for the backward.
+// LayerNorm forward with optional fused residual add: s = x (+ res);
+// y = (s - mean(s)) * rstd(s) * gamma + beta.  One block per row; the fp32
+// row image is staged in LDS during the statistics pass so the normalize
+// pass never re-reads HBM.  With HAS_RES the summed input s is also written
+// out (bf16/f32) — it is the tensor the backward normalizes against and,
+// for pre-LN blocks, the residual stream.  Saves mean and rstd.
 // ============================================================================
-template <bool BF16>
+template <bool BF16, bool HAS_RES>
 __global__ void layer_norm_fwd_kernel(
     void* __restrict__ out, const void* __restrict__ x,
+    const void* __restrict__ res, void* __restrict__ sum_out,
     const void* __restrict__ gamma, const void* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ rstd_out, int64_t rows,
     int64_t cols, float eps) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* srow = reinterpret_cast<float*>(smem);  // [cols] fp32 row cache
   __shared__ float lds[8];
   using T = unsigned short;  // bf16 raw
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
@@ -253,22 +260,41 @@ __global__ void layer_norm_fwd_kernel(
     float sum = 0.f, sumsq = 0.f;
     if (BF16) {
       const T* xr = reinterpret_cast<const T*>(x) + base;
-      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+      const T* rr = HAS_RES ? reinterpret_cast<const T*>(res) + base : nullptr;
+      T* sr = HAS_RES ? reinterpret_cast<T*>(sum_out) + base : nullptr;
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
         float8 f = load_bf16x8(xr + c);
+        if (HAS_RES) {
+          float8 r = load_bf16x8(rr + c);
+#pragma unroll
+          for (int k = 0; k < 8; ++k) f.v[k] += r.v[k];
+          store_bf16x8(sr + c, f);
+        }
 #pragma unroll
         for (int k = 0; k < 8; ++k) {
+          srow[c + k] = f.v[k];
           sum += f.v[k];
           sumsq += f.v[k] * f.v[k];
         }
       }
     } else {
       const float* xr = reinterpret_cast<const float*>(x) + base;
+      const float* rr =
+          HAS_RES ? reinterpret_cast<const float*>(res) + base : nullptr;
+      float* sr = HAS_RES ? reinterpret_cast<float*>(sum_out) + base : nullptr;
       for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
         float fv = xr[c];
+        if (HAS_RES) {
+          fv += rr[c];
+          sr[c] = fv;
+        }
+        srow[c] = fv;
         sum += fv;
         sumsq += fv * fv;
       }
     }
+    __syncthreads();
     sum = block_reduce_sum(sum, lds);
     __syncthreads();
     sumsq = block_reduce_sum(sumsq, lds);
@@ -280,27 +306,25 @@ __global__ void layer_norm_fwd_kernel(
       rstd_out[row] = rstd;
     }
     if (BF16) {
-      const T* xr = reinterpret_cast<const T*>(x) + base;
       const T* gr = reinterpret_cast<const T*>(gamma);
       const T* br = reinterpret_cast<const T*>(beta);
       T* yr = reinterpret_cast<T*>(out) + base;
-      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
-        float8 f = load_bf16x8(xr + c);
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
         float8 g = load_bf16x8(gr + c);
         float8 b = load_bf16x8(br + c);
         float8 y;
 #pragma unroll
         for (int k = 0; k < 8; ++k)
-          y.v[k] = (f.v[k] - mean) * rstd * g.v[k] + b.v[k];
+          y.v[k] = (srow[c + k] - mean) * rstd * g.v[k] + b.v[k];
         store_bf16x8(yr + c, y);
       }
     } else {
-      const float* xr = reinterpret_cast<const float*>(x) + base;
       const float* gr = reinterpret_cast<const float*>(gamma);
       const float* br = reinterpret_cast<const float*>(beta);
       float* yr = reinterpret_cast<float*>(out) + base;
       for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-        yr[c] = (xr[c] - mean) * rstd * gr[c] + br[c];
+        yr[c] = (srow[c] - mean) * rstd * gr[c] + br[c];
       }
     }
     __syncthreads();
@@ -313,7 +337,7 @@ __global__ void layer_norm_fwd_kernel(
 // global fp32 buffers with one atomicAdd per column per block at the end.
 // Requires cols*8 bytes <= 128 KiB of LDS (cols <= 16384).
 // ============================================================================
-template <bool BF16>
+template <bool BF16, bool CACHE>
 __global__ void layer_norm_bwd_kernel(
     void* __restrict__ dx, float* __restrict__ dgamma,
     float* __restrict__ dbeta, const void* __restrict__ dy,
@@ -323,6 +347,8 @@ __global__ void layer_norm_bwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* part_dg = reinterpret_cast<float*>(smem);            // [cols]
   float* part_db = part_dg + cols;                            // [cols]
+  float* c_xhat = CACHE ? part_db + cols : nullptr;           // [cols]
+  float* c_dyg = CACHE ? c_xhat + cols : nullptr;             // [cols]
   __shared__ float lds[8];
   using T = unsigned short;
   for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
@@ -339,7 +365,8 @@ __global__ void layer_norm_bwd_kernel(
       const T* dyr = reinterpret_cast<const T*>(dy) + base;
       const T* xr = reinterpret_cast<const T*>(x) + base;
       const T* gr = reinterpret_cast<const T*>(gamma);
-      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
         float8 d = load_bf16x8(dyr + c);
         float8 f = load_bf16x8(xr + c);
         float8 g = load_bf16x8(gr + c);
@@ -351,6 +378,10 @@ __global__ void layer_norm_bwd_kernel(
           sum_dygx += dyg * xhat;
           part_dg[c + k] += d.v[k] * xhat;
           part_db[c + k] += d.v[k];
+          if (CACHE) {
+            c_xhat[c + k] = xhat;
+            c_dyg[c + k] = dyg;
+          }
         }
       }
     } else {
@@ -364,39 +395,57 @@ __global__ void layer_norm_bwd_kernel(
         sum_dygx += dyg * xhat;
         part_dg[c] += dyr[c] * xhat;
         part_db[c] += dyr[c];
+        if (CACHE) {
+          c_xhat[c] = xhat;
+          c_dyg[c] = dyg;
+        }
       }
     }
+    __syncthreads();
     sum_dyg = block_reduce_sum(sum_dyg, lds);
     __syncthreads();
     sum_dygx = block_reduce_sum(sum_dygx, lds);
     const float inv_cols = 1.f / cols;
     if (BF16) {
+      T* dxr = reinterpret_cast<T*>(dx) + base;
       const T* dyr = reinterpret_cast<const T*>(dy) + base;
       const T* xr = reinterpret_cast<const T*>(x) + base;
       const T* gr = reinterpret_cast<const T*>(gamma);
-      T* dxr = reinterpret_cast<T*>(dx) + base;
-      for (int64_t c = threadIdx.x * 8; c < cols; c += (int64_t)blockDim.x * 8) {
-        float8 d = load_bf16x8(dyr + c);
-        float8 f = load_bf16x8(xr + c);
-        float8 g = load_bf16x8(gr + c);
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
         float8 o;
+        if (CACHE) {
 #pragma unroll
-        for (int k = 0; k < 8; ++k) {
-          const float xhat = (f.v[k] - mean) * rstd;
-          o.v[k] = (d.v[k] * g.v[k] - (sum_dyg + xhat * sum_dygx) * inv_cols) *
-                   rstd;
+          for (int k = 0; k < 8; ++k)
+            o.v[k] = (c_dyg[c + k] -
+                      (sum_dyg + c_xhat[c + k] * sum_dygx) * inv_cols) * rstd;
+        } else {
+          float8 d = load_bf16x8(dyr + c);
+          float8 f = load_bf16x8(xr + c);
+          float8 g = load_bf16x8(gr + c);
+#pragma unroll
+          for (int k = 0; k < 8; ++k) {
+            const float xhat = (f.v[k] - mean) * rstd;
+            o.v[k] = (d.v[k] * g.v[k] -
+                      (sum_dyg + xhat * sum_dygx) * inv_cols) * rstd;
+          }
         }
         store_bf16x8(dxr + c, o);
       }
     } else {
+      float* dxr = reinterpret_cast<float*>(dx) + base;
       const float* dyr = reinterpret_cast<const float*>(dy) + base;
       const float* xr = reinterpret_cast<const float*>(x) + base;
       const float* gr = reinterpret_cast<const float*>(gamma);
-      float* dxr = reinterpret_cast<float*>(dx) + base;
       for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-        const float xhat = (xr[c] - mean) * rstd;
-        dxr[c] = (dyr[c] * gr[c] - (sum_dyg + xhat * sum_dygx) * inv_cols) *
-                 rstd;
+        if (CACHE) {
+          dxr[c] = (c_dyg[c] - (sum_dyg + c_xhat[c] * sum_dygx) * inv_cols) *
+                   rstd;
+        } else {
+          const float xhat = (xr[c] - mean) * rstd;
+          dxr[c] = (dyr[c] * gr[c] - (sum_dyg + xhat * sum_dygx) * inv_cols) *
+                   rstd;
+        }
       }
     }
     __syncthreads();
@@ -517,26 +566,49 @@ __global__ void ce_rowstats_kernel(const void* __restrict__ logits,
                                    int64_t vocab_begin, int64_t ignore_index) {
   __shared__ float lds[8];
   using T = unsigned short;
+  const bool vec = BF16 && (cols % 8 == 0);
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const int64_t base = row * cols;
     float vmax = -INFINITY;
-    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-      const float v = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
-                           : reinterpret_cast<const float*>(logits)[base + c];
-      vmax = fmaxf(vmax, v);
+    if (vec) {
+      const T* lr = reinterpret_cast<const T*>(logits) + base;
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
+        float8 f = load_bf16x8(lr + c);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) vmax = fmaxf(vmax, f.v[k]);
+      }
+    } else {
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        const float v = BF16
+            ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+            : reinterpret_cast<const float*>(logits)[base + c];
+        vmax = fmaxf(vmax, v);
+      }
     }
     vmax = block_reduce_max(vmax, lds);
     __syncthreads();
     float sum = 0.f;
-    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-      const float v = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
-                           : reinterpret_cast<const float*>(logits)[base + c];
-      sum += expf(v - vmax);
+    if (vec) {
+      const T* lr = reinterpret_cast<const T*>(logits) + base;
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
+        float8 f = load_bf16x8(lr + c);
+#pragma unroll
+        for (int k = 0; k < 8; ++k) sum += expf(f.v[k] - vmax);
+      }
+    } else {
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        const float v = BF16
+            ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+            : reinterpret_cast<const float*>(logits)[base + c];
+        sum += expf(v - vmax);
+      }
     }
     sum = block_reduce_sum(sum, lds);
     if (threadIdx.x == 0) {
       row_max[row] = vmax;
-      // store the raw sum at this shard's max; Python rescales by
+      // raw sum at this shard's max; Python rescales by
       // exp(local_max - global_max) before summing across shards.
       row_sumexp[row] = sum;
       const int64_t tgt = targets[row];
@@ -564,22 +636,42 @@ __global__ void ce_backward_kernel(void* __restrict__ dlogits,
                                    int64_t vocab_begin, int64_t ignore_index,
                                    float scale) {
   using T = unsigned short;
+  const bool vec = BF16 && (cols % 8 == 0);
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const int64_t base = row * cols;
     const float m = gmax[row];
     const float inv_sum = 1.f / gsumexp[row];
     const int64_t tgt = targets[row];
     const float dl = (tgt == ignore_index ? 0.f : dloss[row] * scale);
-    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-      const float v = BF16 ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
-                           : reinterpret_cast<const float*>(logits)[base + c];
-      float g = expf(v - m) * inv_sum;
-      if (tgt - vocab_begin == c) g -= 1.f;
-      g *= dl;
-      if (BF16)
-        reinterpret_cast<T*>(dlogits)[base + c] = f2bf(g);
-      else
-        reinterpret_cast<float*>(dlogits)[base + c] = g;
+    const int64_t tc = tgt - vocab_begin;
+    if (vec) {
+      const T* lr = reinterpret_cast<const T*>(logits) + base;
+      T* dr = reinterpret_cast<T*>(dlogits) + base;
+      for (int64_t c = threadIdx.x * 8; c < cols;
+           c += (int64_t)blockDim.x * 8) {
+        float8 f = load_bf16x8(lr + c);
+        float8 o;
+#pragma unroll
+        for (int k = 0; k < 8; ++k) {
+          float g = expf(f.v[k] - m) * inv_sum;
+          if (tc == c + k) g -= 1.f;
+          o.v[k] = g * dl;
+        }
+        store_bf16x8(dr + c, o);
+      }
+    } else {
+      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+        const float v = BF16
+            ? bf2f(reinterpret_cast<const T*>(logits)[base + c])
+            : reinterpret_cast<const float*>(logits)[base + c];
+        float g = expf(v - m) * inv_sum;
+        if (tc == c) g -= 1.f;
+        g *= dl;
+        if (BF16)
+          reinterpret_cast<T*>(dlogits)[base + c] = f2bf(g);
+        else
+          reinterpret_cast<float*>(dlogits)[base + c] = g;
+      }
     }
   }
 }
@@ -718,19 +810,32 @@ void epl_lamb_phase2(float* master, unsigned short* param_bf16,
                      master, param_bf16, update, chunk_of, ratio, n, lr);
 }
 
-void epl_layer_norm_fwd(void* out, const void* x, const void* gamma,
-                        const void* beta, float* mean, float* rstd,
-                        int64_t rows, int64_t cols, float eps, bool bf16,
-                        hipStream_t stream) {
+void epl_layer_norm_fwd(void* out, const void* x, const void* res,
+                        void* sum_out, const void* gamma, const void* beta,
+                        float* mean, float* rstd, int64_t rows, int64_t cols,
+                        float eps, bool bf16, hipStream_t stream) {
   const int grid = (int)(rows < kMaxGrid ? rows : kMaxGrid);
-  if (bf16)
-    hipLaunchKernelGGL(layer_norm_fwd_kernel<true>, dim3(grid), dim3(kBlock),
-                       0, stream, out, x, gamma, beta, mean, rstd, rows, cols,
-                       eps);
-  else
-    hipLaunchKernelGGL(layer_norm_fwd_kernel<false>, dim3(grid), dim3(kBlock),
-                       0, stream, out, x, gamma, beta, mean, rstd, rows, cols,
-                       eps);
+  const size_t lds_bytes = (size_t)cols * sizeof(float);
+  const bool has_res = res != nullptr;
+  if (bf16) {
+    if (has_res)
+      hipLaunchKernelGGL((layer_norm_fwd_kernel<true, true>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, out, x, res,
+                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+    else
+      hipLaunchKernelGGL((layer_norm_fwd_kernel<true, false>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, out, x, res,
+                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+  } else {
+    if (has_res)
+      hipLaunchKernelGGL((layer_norm_fwd_kernel<false, true>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, out, x, res,
+                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+    else
+      hipLaunchKernelGGL((layer_norm_fwd_kernel<false, false>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, out, x, res,
+                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+  }
 }
 
 void epl_layer_norm_bwd(void* dx, float* dgamma, float* dbeta, const void* dy,
@@ -738,15 +843,28 @@ void epl_layer_norm_bwd(void* dx, float* dgamma, float* dbeta, const void* dy,
                         const float* rstd, int64_t rows, int64_t cols,
                         bool bf16, hipStream_t stream) {
   int grid = (int)(rows < 1024 ? rows : 1024);
-  const size_t lds_bytes = (size_t)cols * 2 * sizeof(float);
-  if (bf16)
-    hipLaunchKernelGGL(layer_norm_bwd_kernel<true>, dim3(grid), dim3(kBlock),
-                       lds_bytes, stream, dx, dgamma, dbeta, dy, x, gamma,
-                       mean, rstd, rows, cols);
-  else
-    hipLaunchKernelGGL(layer_norm_bwd_kernel<false>, dim3(grid), dim3(kBlock),
-                       lds_bytes, stream, dx, dgamma, dbeta, dy, x, gamma,
-                       mean, rstd, rows, cols);
+  const bool cache = (size_t)cols * 16 <= 128 * 1024;
+  const size_t lds_bytes =
+      (size_t)cols * (cache ? 4 : 2) * sizeof(float);
+  if (bf16) {
+    if (cache)
+      hipLaunchKernelGGL((layer_norm_bwd_kernel<true, true>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
+                         dy, x, gamma, mean, rstd, rows, cols);
+    else
+      hipLaunchKernelGGL((layer_norm_bwd_kernel<true, false>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
+                         dy, x, gamma, mean, rstd, rows, cols);
+  } else {
+    if (cache)
+      hipLaunchKernelGGL((layer_norm_bwd_kernel<false, true>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
+                         dy, x, gamma, mean, rstd, rows, cols);
+    else
+      hipLaunchKernelGGL((layer_norm_bwd_kernel<false, false>), dim3(grid),
+                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
+                         dy, x, gamma, mean, rstd, rows, cols);
+  }
 }
 
 void epl_bias_gelu_fwd(void* out, const void* x, const void* bias,

@@ -70,11 +70,14 @@ class Block(nn.Module):
 
     def forward(self, x):
         if self.pre_ln:
-            x = x + self.attn(self.ln1(x))
-            x = x + self.mlp(self.ln2(x))
-        else:
-            x = self.ln1(x + self.attn(x))
-            x = self.ln2(x + self.mlp(x))
+            # residual add fused into ln2; the summed stream s is the
+            # residual carried forward
+            a = self.attn(self.ln1(x))
+            y, s = self.ln2.forward_with_sum(a, x)
+            return s + self.mlp(y)
+        # post-LN: both residual adds fuse into the LN kernels
+        x = self.ln1(self.attn(x), residual=x)
+        x = self.ln2(self.mlp(x), residual=x)
         return x
 
 

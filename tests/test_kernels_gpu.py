@@ -56,7 +56,7 @@ def test_layer_norm_fwd_bwd(dtype):
     mean = torch.empty(rows, dtype=torch.float32, device=dev())
     rstd = torch.empty(rows, dtype=torch.float32, device=dev())
     out = torch.empty_like(x)
-    _C.layer_norm_fwd(out, x, gamma, beta, mean, rstd, 1e-5)
+    _C.layer_norm_fwd(out, x, None, None, gamma, beta, mean, rstd, 1e-5)
 
     xf = x.float()
     ref_mu = xf.mean(dim=1)
@@ -173,3 +173,29 @@ def test_rccl_single_rank_comm():
     c.synchronize()
     assert torch.allclose(t, ref)
     c.destroy()
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_fused_residual_layer_norm_gpu(dtype):
+    torch.manual_seed(9)
+    from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
+    import easyparallellibrary_amd as epl
+    epl.init()
+    ln = FusedLayerNorm(1024).to(dev(), dtype)
+    x = torch.randn(256, 1024, device=dev(), dtype=dtype, requires_grad=True)
+    r = torch.randn(256, 1024, device=dev(), dtype=dtype, requires_grad=True)
+    y = ln(x, residual=r)
+    dy = torch.randn_like(y)
+    y.backward(dy)
+
+    xf = x.detach().float().clone().requires_grad_(True)
+    rf = r.detach().float().clone().requires_grad_(True)
+    yr = torch.nn.functional.layer_norm(
+        xf + rf, (1024,), ln.weight.detach().float(),
+        ln.bias.detach().float())
+    yr.backward(dy.float())
+    torch.cuda.synchronize()
+    tol = 5e-2 if dtype == torch.bfloat16 else 1e-4
+    assert torch.allclose(y.float(), yr, atol=tol, rtol=tol)
+    assert torch.allclose(x.grad.float(), xf.grad, atol=tol, rtol=tol)
+    assert torch.allclose(r.grad.float(), rf.grad, atol=tol, rtol=tol)
