@@ -47,7 +47,8 @@ class FlatParamGroup:
     bucket k+1, and allreduce overlaps the rest of backward.
     """
 
-    def __init__(self, params, device, model_dtype=None, grad_dtype=None):
+    def __init__(self, params, device, model_dtype=None, grad_dtype=None,
+                 pad_to_multiple=1):
         self.params = [p for p in params if p.requires_grad]
         assert self.params, "empty parameter group"
         if model_dtype is None:
@@ -62,6 +63,9 @@ class FlatParamGroup:
         for p in ordered:
             offsets.append(total)
             total += _aligned(p.numel())
+        if pad_to_multiple > 1:
+            m = pad_to_multiple * _ALIGN
+            total = (total + m - 1) // m * m
         self.total = total
         self.offsets = offsets
         self.ordered = ordered
@@ -104,33 +108,56 @@ class GradReducer:
     """Bucketed overlapped gradient reduction over one FlatParamGroup."""
 
     def __init__(self, group, pool, bucket_bytes, reduce_method="mean",
-                 overlap=True, world_scale=None):
+                 overlap=True, world_scale=None, shard_owners=False):
+        """shard_owners: ZeRO-v1 mode — buckets are split at shard
+        boundaries (shard s = rank s's contiguous arena slice) and each
+        bucket is REDUCED to its owning rank instead of all-reduced
+        (reference: runtime/zero.py:178-190)."""
         self.group = group
         self.pool = pool
         self.reduce_method = reduce_method
         self.overlap = overlap
+        self.shard_owners = shard_owners
         esize = group.grad_arena.element_size()
         bucket_elems = max(_ALIGN, int(bucket_bytes) // esize)
 
-        # bucket = [start, end) over the arena, aligned to param boundaries
-        self.buckets = []
-        cur_start = 0
-        cur_params = []
-        for p, off in zip(group.ordered, group.offsets):
-            cur_params.append(p)
-            end = off + _aligned(p.numel())
-            if end - cur_start >= bucket_elems:
-                self.buckets.append((cur_start, end, cur_params))
-                cur_start = end
-                cur_params = []
-        if cur_params:
-            self.buckets.append((cur_start, group.total, cur_params))
+        world = pool.comms[0].size
+        shard = group.total // world if shard_owners and world > 1 else None
 
-        self._bucket_of_param = {}
-        for bi, (_, _, ps) in enumerate(self.buckets):
+        def owner_of(pos):
+            return None if shard is None else min(pos // shard, world - 1)
+
+        # bucket = [start, end) over the arena, aligned to param boundaries
+        # (and split at shard boundaries in ZeRO-v1 mode)
+        edges = sorted(set(
+            [off + _aligned(p.numel())
+             for p, off in zip(group.ordered, group.offsets)] +
+            ([shard * i for i in range(1, world)] if shard else []) +
+            [group.total]))
+        self.buckets = []  # (start, end, params_overlapping, owner)
+        cur_start = 0
+        for edge in edges:
+            boundary_forced = shard is not None and edge % shard == 0
+            if (edge - cur_start >= bucket_elems or boundary_forced or
+                    edge == group.total):
+                if edge > cur_start:
+                    self.buckets.append(
+                        [cur_start, edge, [], owner_of(cur_start)])
+                    cur_start = edge
+        # a parameter gates EVERY bucket its extent overlaps — a bucket
+        # must not fly before all bytes in its range are written
+        for p, off in zip(group.ordered, group.offsets):
+            lo, hi = off, off + _aligned(p.numel())
+            for b in self.buckets:
+                if lo < b[1] and hi > b[0]:
+                    b[2].append(p)
+        self.buckets = [tuple(b) for b in self.buckets]
+
+        self._buckets_of_param = {}
+        for bi, (_, _, ps, _) in enumerate(self.buckets):
             for p in ps:
-                self._bucket_of_param[id(p)] = bi
-        self._pending = [len(ps) for (_, _, ps) in self.buckets]
+                self._buckets_of_param.setdefault(id(p), []).append(bi)
+        self._pending = [len(ps) for (_, _, ps, _) in self.buckets]
         self._launched = [False] * len(self.buckets)
         self.enabled = True          # pipeline sets False until last ubatch
         self._world = (self.pool.comms[0].size if world_scale is None
@@ -149,18 +176,21 @@ class GradReducer:
         if self._launched[bi]:
             return
         self._launched[bi] = True
-        start, end, _ = self.buckets[bi]
+        start, end, _, owner = self.buckets[bi]
         comm = self.pool.next_comm()
-        comm.all_reduce(self.group.grad_arena[start:end], op=self.op,
-                        async_op=True)
+        buf = self.group.grad_arena[start:end]
+        if self.shard_owners and owner is not None and comm.size > 1:
+            comm.reduce(buf, root=owner, op=self.op, async_op=True)
+        else:
+            comm.all_reduce(buf, op=self.op, async_op=True)
 
     def _on_grad_ready(self, p):
         if not self.enabled:
             return
-        bi = self._bucket_of_param[id(p)]
-        self._pending[bi] -= 1
-        if self._pending[bi] == 0:
-            self._launch(bi)
+        for bi in self._buckets_of_param[id(p)]:
+            self._pending[bi] -= 1
+            if self._pending[bi] == 0:
+                self._launch(bi)
 
     def finish(self):
         """Launch stragglers and fence the compute stream behind the pool.
@@ -179,7 +209,7 @@ class GradReducer:
         self.reset()
 
     def reset(self):
-        self._pending = [len(ps) for (_, _, ps) in self.buckets]
+        self._pending = [len(ps) for (_, _, ps, _) in self.buckets]
         self._launched = [False] * len(self.buckets)
 
     def remove_hooks(self):

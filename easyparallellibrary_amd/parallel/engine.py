@@ -154,10 +154,12 @@ class Engine:
         # position) DP groups deterministically.
         self.flat_groups = []
         self.reducers = []
+        self._group_infos = []
         pool_n = self.config.communication.num_communicators
         bucket_bytes = self.config.communication.bucket_bytes
         reduce_method = self.config.communication.gradients_reduce_method
         overlap = self.config.communication.overlap_grad_reduce
+        zero_level = self.config.zero.level
         self._bcast_jobs = []
         for tg in self.plan.taskgraphs:
             vd = tg.virtual_device
@@ -182,13 +184,18 @@ class Engine:
                 params = [p for p in tg.parameters() if p.requires_grad]
                 if not params:
                     continue
-                fg = FlatParamGroup(params, self.device,
-                                    model_dtype=self.dtype)
-                reducer = GradReducer(fg, pool, bucket_bytes,
-                                      reduce_method=reduce_method,
-                                      overlap=overlap)
+                fg = FlatParamGroup(
+                    params, self.device, model_dtype=self.dtype,
+                    pad_to_multiple=len(granks) if zero_level else 1)
+                reducer = GradReducer(
+                    fg, pool, bucket_bytes, reduce_method=reduce_method,
+                    overlap=overlap,
+                    shard_owners=(zero_level == "v1"))
                 self.flat_groups.append(fg)
                 self.reducers.append(reducer)
+                self._group_infos.append(
+                    {"fg": fg, "pool": pool, "bcomm": bcomm,
+                     "ranks": granks, "taskgraph": tg})
                 self._bcast_jobs.append((fg, bcomm))
 
         # ---- initial weight sync (reference: hooks.py:330-357) ---------------
