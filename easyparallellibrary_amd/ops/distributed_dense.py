@@ -52,8 +52,11 @@ class ColumnParallelLinear(nn.Module):
         self.offset = shard_offset(out_features, self.nshards, self.shard)
         self.weight = nn.Parameter(
             torch.empty(self.local_out, in_features))
+        self.weight._epl_shard_dim = 0
         self.bias = nn.Parameter(torch.zeros(self.local_out)) if bias \
             else None
+        if self.bias is not None:
+            self.bias._epl_shard_dim = 0
         if source is not None:
             with torch.no_grad():
                 self.weight.copy_(
@@ -93,6 +96,7 @@ class RowParallelLinear(nn.Module):
         self.in_features = in_features
         self.out_features = out_features
         self.weight = nn.Parameter(torch.empty(out_features, self.local_in))
+        self.weight._epl_shard_dim = 1
         self.bias = nn.Parameter(torch.zeros(out_features)) if bias else None
         if source is not None:
             with torch.no_grad():

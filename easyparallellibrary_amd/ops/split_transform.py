@@ -38,6 +38,7 @@ class VocabParallelEmbedding(nn.Module):
         self.begin = shard_offset(num_embeddings, self.nshards, self.shard)
         self.weight = nn.Parameter(
             torch.empty(self.local_rows, embedding_dim))
+        self.weight._epl_shard_dim = 0
         if source is not None:
             with torch.no_grad():
                 self.weight.copy_(

@@ -63,9 +63,41 @@ class Engine:
         self.loss_fn = loss_fn
         self.dtype = dtype or torch.float32
         self.model = model
+        from easyparallellibrary_amd.runtime.amp import AmpContext
+        self.amp = AmpContext(self.config, self.device)
+        if self.amp.enabled and self.dtype != torch.float32:
+            raise ValueError(
+                "amp.level=O1 keeps fp32 parameters; pass dtype=float32 "
+                "(pure-bf16 training needs no AMP)")
 
         # ---- plan -----------------------------------------------------------
         self.plan = Plan.build(model, self.env.strategy_context)
+        # auto-stage: partition an un-staged sequential model into
+        # pipeline.num_stages stages (reference: parallel/planner.py
+        # AutoStageGenerator, invoked from the gradients hook)
+        auto_n = self.config.pipeline.num_stages
+        if auto_n > 1 and self.plan.num_stages <= 1 \
+                and not self.plan.has_split:
+            from easyparallellibrary_amd.ir.plan import TaskGraph
+            from easyparallellibrary_amd.parallel.planner import (
+                AutoStageGenerator)
+            from easyparallellibrary_amd.strategies.replicate import Replicate
+            stages = AutoStageGenerator(model, auto_n).search()
+            if stages is not None:
+                tgs = []
+                for i, mods in enumerate(stages):
+                    strat = Replicate(1, name="auto_stage_{}".format(i))
+                    strat.index = i
+                    tg = TaskGraph(i, strat)
+                    tg.modules = list(mods)
+                    tg.module_names = [
+                        "auto_stage_{}[{}]".format(i, j)
+                        for j in range(len(mods))]
+                    tgs.append(tg)
+                self.plan = Plan(tgs, None)
+                logger.info("auto-stage: partitioned model into %d stages "
+                            "(%s modules per stage)", auto_n,
+                            [len(m) for m in stages])
         colocate = self.config.cluster.colocate_split_and_replicate
         counts = self.plan.effective_device_counts(colocate)
         layout_counts = [c for c in counts if c > 0]
@@ -145,8 +177,14 @@ class Engine:
         for tg in self._owned_tgs:
             for m in tg.modules:
                 m.to(self.device, dtype=self.dtype)
-            # keep norm/embedding-adjacent fp32 buffers as-is? parameters all
-            # follow engine dtype; fp32-master lives in the arenas.
+
+        # ---- gradient checkpointing (reference: runtime/gc/) ----------------
+        gc_type = self.config.gradient_checkpoint.type
+        if gc_type:
+            from easyparallellibrary_amd.runtime.gc import (
+                apply_gradient_checkpointing)
+            n = apply_gradient_checkpointing(self._runnable, mode=gc_type)
+            logger.info("gradient checkpointing: wrapped %d module(s)", n)
 
         # ---- DP groups + reducers -------------------------------------------
         # Every rank must create communicators in the same global order
@@ -260,8 +298,11 @@ class Engine:
             loss = self.pipeline.run(inputs, targets)
         else:
             loss = self._train_step_simple(inputs, targets)
-        grad_scale = float(self.num_micro_batch)
-        self.optimizer.step(grad_scale=grad_scale)
+        found_inf = self.amp.found_inf(self.flat_groups)
+        if not found_inf:
+            grad_scale = float(self.num_micro_batch) * self.amp.loss_scale
+            self.optimizer.step(grad_scale=grad_scale)
+        self.amp.post_step(found_inf)
         self.global_step += 1
         return loss
 
@@ -277,9 +318,10 @@ class Engine:
         for i, (xc, tc) in enumerate(zip(input_chunks, target_chunks)):
             last = (i == nmb - 1)
             self._set_reducers_enabled(last)
-            out = self._runnable(xc)
-            loss = self.loss_fn(out, tc)
-            loss.backward()
+            with self.amp.autocast():
+                out = self._runnable(xc)
+                loss = self.loss_fn(out, tc)
+            self.amp.scale_loss(loss).backward()
             total_loss = loss.detach() if total_loss is None \
                 else total_loss + loss.detach()
         self.finish_grad_sync()
@@ -289,3 +331,16 @@ class Engine:
     def eval_step(self, inputs):
         with torch.no_grad():
             return self._runnable(inputs)
+
+    # ---- checkpoint ----------------------------------------------------------
+    def save_checkpoint(self, path, save_optimizer=True):
+        from easyparallellibrary_amd.runtime import saver
+        saver.save_checkpoint(self, path, save_optimizer=save_optimizer)
+
+    def load_checkpoint(self, path, load_optimizer=True, assign_map=None,
+                        strict=True):
+        from easyparallellibrary_amd.runtime import saver
+        saver.load_checkpoint(self, path, load_optimizer=load_optimizer,
+                              assign_map=assign_map, strict=strict)
+        meta = saver.ShardingLoader(path).meta
+        self.global_step = meta.get("global_step", 0)

@@ -124,16 +124,17 @@ class PipelineRuntime:
 
     # ---- forward/backward wrappers -------------------------------------------
     def _forward(self, inp, target):
-        out = self.engine.stage_module(inp)
-        loss = None
-        if self.s == self.S - 1:
-            loss = self.engine.loss_fn(out, target)
+        with self.engine.amp.autocast():
+            out = self.engine.stage_module(inp)
+            loss = None
+            if self.s == self.S - 1:
+                loss = self.engine.loss_fn(out, target)
         return out, loss
 
     def _backward(self, inp, out, loss, grad_out, is_last_backward):
         self.engine._set_reducers_enabled(is_last_backward)
         if loss is not None:
-            loss.backward()
+            self.engine.amp.scale_loss(loss).backward()
         else:
             out.backward(grad_out)
         if inp is not None and inp.requires_grad:

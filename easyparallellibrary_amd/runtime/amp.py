@@ -1,0 +1,94 @@
+"""Automatic mixed precision.
+
+Capability parity: /root/reference/epl/runtime/amp/ — O1 mixed precision
+(auto_mixed_precision.py: allow/deny-list graph rewrite) and dynamic/fixed
+loss scaling (loss_scale.py:29-82, loss_scale_tf.py DynamicLossScale).
+
+MI355X redesign: the graph-rewrite machinery is unnecessary — torch.autocast
+supplies the allow/deny op classification natively.  AMP O1 here means:
+fp32 master params (single arena), forward+loss under
+``torch.autocast('cuda', bf16|fp16)``, loss scaled before backward (fp16
+only), gradients unscaled INSIDE the fused optimizer kernel (the
+inv_scale argument), and the dynamic scale update skipping steps whose
+gradients overflowed — the reference's cond-apply (amp_update,
+loss_scale.py:44-51).
+"""
+
+import torch
+
+
+class DynamicLossScaler:
+    """Reference DynamicLossScale semantics (loss_scale_tf.py): multiply by
+    ``growth_factor`` after ``growth_interval`` consecutive finite steps,
+    multiply by ``backoff_factor`` (and skip the update) on overflow."""
+
+    def __init__(self, init_scale=2.0 ** 16, growth_factor=2.0,
+                 backoff_factor=0.5, growth_interval=2000):
+        self.scale = float(init_scale)
+        self.growth_factor = growth_factor
+        self.backoff_factor = backoff_factor
+        self.growth_interval = growth_interval
+        self._good_steps = 0
+
+    def update(self, found_inf):
+        if found_inf:
+            self.scale = max(1.0, self.scale * self.backoff_factor)
+            self._good_steps = 0
+        else:
+            self._good_steps += 1
+            if self._good_steps >= self.growth_interval:
+                self.scale *= self.growth_factor
+                self._good_steps = 0
+        return self.scale
+
+
+class FixedLossScaler:
+    def __init__(self, scale):
+        self.scale = float(scale)
+
+    def update(self, found_inf):
+        return self.scale
+
+
+class AmpContext:
+    """Engine-side AMP driver."""
+
+    def __init__(self, config, device):
+        self.enabled = config.amp.level.upper() == "O1"
+        self.dtype = (torch.float16 if config.amp.dtype == "fp16"
+                      else torch.bfloat16)
+        self.device_type = "cuda" if device.type == "cuda" else "cpu"
+        ls = config.amp.loss_scale
+        if not self.enabled or self.dtype == torch.bfloat16:
+            self.scaler = FixedLossScaler(1.0)
+        elif ls == "dynamic":
+            self.scaler = DynamicLossScaler()
+        else:
+            self.scaler = FixedLossScaler(float(ls))
+
+    @property
+    def loss_scale(self):
+        return self.scaler.scale
+
+    def autocast(self):
+        if not self.enabled:
+            return torch.autocast(self.device_type, enabled=False)
+        return torch.autocast(self.device_type, dtype=self.dtype)
+
+    def scale_loss(self, loss):
+        if self.loss_scale != 1.0:
+            return loss * self.loss_scale
+        return loss
+
+    def found_inf(self, flat_groups):
+        """One fused pass over the flat grad arenas."""
+        if not self.enabled or self.dtype == torch.bfloat16:
+            return False
+        for fg in flat_groups:
+            s = fg.grad_arena.sum(dtype=torch.float32)
+            if not torch.isfinite(s):
+                return True
+        return False
+
+    def post_step(self, found_inf):
+        self.scaler.update(found_inf)

@@ -1,0 +1,186 @@
+"""Sharded checkpoint save/restore.
+
+Capability parity: /root/reference/epl/runtime/saver.py — ShardingLoader
+(restore with name remap + begin/size slicing for resharding, :46-128)
+and MemoryEfficientBuilder (sharded, serialized shard writes, :145-207);
+plus the hook behavior that only the first constructor saves replicated
+variables while split variables are saved per shard
+(/root/reference/epl/parallel/hooks.py:542-590, shard-suffixed names from
+ops/distributed_dense.py:111-123).
+
+Layout on disk (directory):
+  meta.pt                         — world/topology/step info (rank 0)
+  tg{T}_pos{P}.pt                 — taskgraph T's params owned by
+                                    within-replica position P (replica 0
+                                    only; replicated taskgraphs save one
+                                    file from position 0).  Split shards
+                                    keep their shard index in the file AND
+                                    '{param}.shard{P}' keys.
+  opt_rank{R}.pt                  — per-rank optimizer state (ZeRO keeps
+                                    states local, reference hooks.py:340-344)
+"""
+
+import os
+
+import torch
+
+
+def _tg_params(tg):
+    out = {}
+    for name, mod in zip(tg.module_names, tg.modules):
+        prefix = name + "." if name else ""
+        for pn, p in mod.state_dict(keep_vars=True).items():
+            out[prefix + pn] = p
+    return out
+
+
+def save_checkpoint(engine, path, save_optimizer=True):
+    os.makedirs(path, exist_ok=True)
+    rank = engine.rank
+    if rank == 0:
+        torch.save({
+            "world_size": engine.world_size,
+            "per_replica": engine.per_replica,
+            "num_stages": engine.num_stages,
+            "global_step": engine.global_step,
+            "taskgraphs": [
+                {"index": tg.index, "type": tg.strategy_type,
+                 "device_count": tg.device_count,
+                 "module_names": tg.module_names}
+                for tg in engine.plan.taskgraphs],
+        }, os.path.join(path, "meta.pt"))
+    if engine.replica_id == 0:
+        for tg in engine._owned_tgs:
+            ranks = tg.virtual_device.local_ranks(0)
+            pos = ranks.index(rank)
+            if not tg.is_split and pos != 0:
+                continue  # replicated: first position saves
+            params = _tg_params(tg)
+            blob = {}
+            for k, t in params.items():
+                key = "{}.shard{}".format(k, pos) if tg.is_split else k
+                blob[key] = t.detach().to("cpu")
+            torch.save(
+                {"taskgraph": tg.index, "position": pos,
+                 "is_split": tg.is_split,
+                 "nshards": len(ranks) if tg.is_split else 1,
+                 "params": blob},
+                os.path.join(path, "tg{}_pos{}.pt".format(tg.index, pos)))
+    if save_optimizer:
+        sd = engine.optimizer.state_dict()
+        sd_cpu = _to_cpu(sd)
+        torch.save(sd_cpu, os.path.join(path, "opt_rank{}.pt".format(rank)))
+    _barrier(engine)
+
+
+def _to_cpu(obj):
+    if torch.is_tensor(obj):
+        return obj.detach().to("cpu")
+    if isinstance(obj, dict):
+        return {k: _to_cpu(v) for k, v in obj.items()}
+    if isinstance(obj, list):
+        return [_to_cpu(v) for v in obj]
+    return obj
+
+
+def _barrier(engine):
+    import torch.distributed as dist
+    if dist.is_initialized():
+        dist.barrier()
+
+
+class ShardingLoader:
+    """Restore with optional name remap and re-sharding: a split
+    taskgraph saved with N shards can restore onto M shards — tensors are
+    concatenated on their shard dim and re-sliced (begin/size semantics
+    of the reference's sharding_info, runtime/saver.py:46-128)."""
+
+    def __init__(self, path, assign_map=None):
+        self.path = path
+        self.assign_map = assign_map or {}
+        self.meta = torch.load(os.path.join(path, "meta.pt"),
+                               weights_only=False)
+
+    def _remap(self, name):
+        for old, new in self.assign_map.items():
+            if name.startswith(old):
+                return new + name[len(old):]
+        return name
+
+    def load_into(self, engine, load_optimizer=True, strict=True):
+        for tg in engine._owned_tgs:
+            ranks = tg.virtual_device.local_ranks(engine.replica_id)
+            pos = ranks.index(engine.rank)
+            if tg.is_split:
+                self._load_split_tg(tg, pos, len(ranks), strict)
+            else:
+                self._load_replicated_tg(tg, strict)
+        if load_optimizer:
+            opt_path = os.path.join(
+                self.path, "opt_rank{}.pt".format(engine.rank))
+            if os.path.exists(opt_path):
+                engine.optimizer.load_state_dict(
+                    torch.load(opt_path, weights_only=False))
+        # params were loaded into module tensors == arena views; refresh
+        # fp32 masters from the arenas
+        for fg in engine.flat_groups:
+            if fg.master_arena is not fg.param_arena:
+                fg.master_arena.copy_(fg.param_arena.to(torch.float32))
+        _barrier(engine)
+
+    def _load_replicated_tg(self, tg, strict):
+        f = os.path.join(self.path, "tg{}_pos0.pt".format(tg.index))
+        blob = torch.load(f, weights_only=False)["params"]
+        params = _tg_params(tg)
+        for k, t in params.items():
+            src = self._remap(k)
+            if src in blob:
+                t.data.copy_(blob[src].to(t.device, t.dtype))
+            elif strict:
+                raise KeyError("missing checkpoint tensor {}".format(src))
+
+    def _load_split_tg(self, tg, pos, nshards, strict):
+        # gather available shard files
+        import glob
+        files = sorted(glob.glob(os.path.join(
+            self.path, "tg{}_pos*.pt".format(tg.index))))
+        shards = [torch.load(f, weights_only=False) for f in files]
+        saved_n = shards[0]["nshards"] if shards else 0
+        params = _tg_params(tg)
+        for k, t in params.items():
+            base = self._remap(k)
+            if saved_n == nshards:
+                key = "{}.shard{}".format(base, pos)
+                blob = shards[pos]["params"]
+                if key in blob:
+                    t.data.copy_(blob[key].to(t.device, t.dtype))
+                    continue
+                if strict:
+                    raise KeyError(key)
+            else:
+                # reshard: concatenate all shards on dim 0, re-slice with
+                # the remainder-to-shard-0 policy
+                pieces = []
+                for s in shards:
+                    key = "{}.shard{}".format(base, s["position"])
+                    if key in s["params"]:
+                        pieces.append(s["params"][key])
+                if not pieces:
+                    if strict:
+                        raise KeyError(base)
+                    continue
+                dim = getattr(t, "_epl_shard_dim", 0)
+                full = torch.cat(pieces, dim=dim)
+                from easyparallellibrary_amd.ops.distributed_dense import (
+                    shard_offset, shard_size)
+                total = full.shape[dim]
+                lo = shard_offset(total, nshards, pos)
+                n = shard_size(total, nshards, pos)
+                t.data.copy_(
+                    full.narrow(dim, lo, n).to(t.device, t.dtype))
+
+
+def load_checkpoint(engine, path, load_optimizer=True, assign_map=None,
+                    strict=True):
+    ShardingLoader(path, assign_map).load_into(engine, load_optimizer,
+                                               strict)
