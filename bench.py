@@ -1,15 +1,16 @@
 #!/usr/bin/env python3
 """Benchmark harness (driver contract — see BASELINE.json).
 
-Measures the headline metric: samples/sec (whole node) for BERT-Large
-DP(+PP) on synthetic MLM data, bf16, random-init weights.
+Headline metric: samples/sec (whole node) for BERT-Large DP(+PP) on
+synthetic MLM data, bf16, random-init weights (BASELINE.json configs).
 
-  python bench.py --gpus N --steps K --warmup W [--model bert-large]
-      [--pp S] [--micro-batch M] [--batch B]
+  python bench.py --gpus N --steps K --warmup W
+      [--config bert_dp|bert_pp|bert_zero|resnet_tp|gpt2_xl|moe]
+      [--pp S] [--micro-batch M] [--batch B] [--zero v0|v1]
 
 For N>1 the driver launches this under torch.distributed.run with one rank
-per GPU (RCCL over xGMI); each rank reads RANK/LOCAL_RANK/WORLD_SIZE from
-the env.  Weak scaling: per-GPU batch is fixed as N grows.
+per GPU (RCCL over xGMI); ranks read RANK/LOCAL_RANK/WORLD_SIZE from env.
+Weak scaling: per-GPU batch fixed as N grows.
 """
 
 import argparse
@@ -25,51 +26,169 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=10)
     p.add_argument("--warmup", type=int, default=3)
-    p.add_argument("--model", default="bert-large")
-    p.add_argument("--batch", type=int, default=32,
-                   help="per-GPU micro-batch x num_micro_batch samples")
+    p.add_argument("--config", default="bert_dp",
+                   choices=["bert_dp", "bert_pp", "bert_zero", "resnet_tp",
+                            "gpt2_xl", "moe"])
+    p.add_argument("--model", default=None)
+    p.add_argument("--batch", type=int, default=0,
+                   help="per-GPU samples per step (0 = config default)")
     p.add_argument("--seq-len", type=int, default=512)
-    p.add_argument("--pp", type=int, default=1, help="pipeline stages")
-    p.add_argument("--micro-batch", type=int, default=1,
-                   help="num_micro_batch (pipeline)")
-    p.add_argument("--zero", default="", choices=["", "v0", "v1"])
+    p.add_argument("--pp", type=int, default=0, help="pipeline stages")
+    p.add_argument("--micro-batch", type=int, default=0)
+    p.add_argument("--zero", default=None, choices=[None, "", "v0", "v1"])
+    p.add_argument("--gc", default=None, choices=[None, "", "auto"])
+    p.add_argument("--offload", default=None, choices=[None, "", "v0"])
     p.add_argument("--device", default=None)
     return p.parse_args()
+
+
+def build_bert_bench(args, epl, world, on_gpu, dtype):
+    from easyparallellibrary_amd.models import bert
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    batch = args.batch or 128
+    pp = args.pp or (2 if args.config == "bert_pp" else 1)
+    nmb = args.micro_batch or (4 if pp > 1 else 1)
+    zero = args.zero if args.zero is not None else (
+        "v1" if args.config == "bert_zero" else "")
+    epl.init(epl.Config({
+        "pipeline.num_micro_batch": nmb,
+        "zero.level": zero,
+    }))
+    vocab = 30528
+    model = bert.build_bert(args.model or "bert-large", vocab_size=vocab,
+                            max_pos=max(512, args.seq_len), num_stages=pp)
+    loss_fn = ParallelCrossEntropy()
+    engine = epl.Engine(model, loss_fn=loss_fn, optimizer="adamw", lr=1e-4,
+                        dtype=dtype)
+    rank = int(os.environ.get("RANK", "0"))
+    ids, tgt = bert.synthetic_mlm_batch(batch, args.seq_len, vocab,
+                                        device=engine.device,
+                                        seed=1234 + rank)
+    par = "dp{}".format(engine.num_replicas)
+    if pp > 1:
+        par += "_pp{}".format(pp)
+    if zero:
+        par += "_zero_{}".format(zero)
+    meta = {"model": args.model or "bert-large",
+            "global_batch": batch * engine.num_replicas,
+            "seq_len": args.seq_len, "parallelism": par}
+    return engine, (ids, tgt), batch * engine.num_replicas, meta
+
+
+def build_gpt2_bench(args, epl, world, on_gpu, dtype):
+    from easyparallellibrary_amd.models import gpt2
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    batch = args.batch or 16
+    seq = args.seq_len if args.seq_len != 512 else 1024
+    pp = args.pp or (2 if world >= 2 else 1)
+    nmb = args.micro_batch or (4 if pp > 1 else 1)
+    gc = args.gc if args.gc is not None else "auto"
+    offload = args.offload if args.offload is not None else ""
+    epl.init(epl.Config({
+        "pipeline.num_micro_batch": nmb,
+        "gradient_checkpoint.type": gc,
+        "offload.level": offload,
+    }))
+    vocab = 50264
+    model = gpt2.build_gpt2(args.model or "gpt2-xl", vocab_size=vocab,
+                            max_pos=seq, num_stages=pp)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=1e-4, dtype=dtype)
+    rank = int(os.environ.get("RANK", "0"))
+    ids, tgt = gpt2.synthetic_lm_batch(batch, seq, vocab,
+                                       device=engine.device, seed=99 + rank)
+    par = "dp{}".format(engine.num_replicas)
+    if pp > 1:
+        par += "_pp{}".format(pp)
+    if gc:
+        par += "_gc"
+    if offload:
+        par += "_offload"
+    meta = {"model": args.model or "gpt2-xl",
+            "global_batch": batch * engine.num_replicas,
+            "seq_len": seq, "parallelism": par}
+    return engine, (ids, tgt), batch * engine.num_replicas, meta
+
+
+def build_resnet_bench(args, epl, world, on_gpu, dtype):
+    from easyparallellibrary_amd.models import resnet
+    from easyparallellibrary_amd.ops import bridging
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    batch = args.batch or 64
+    num_classes = 100000
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    model = resnet.build_resnet50_split_classifier(world, num_classes)
+    state = {}
+
+    def loss_fn(logits, targets):
+        comm = state["engine"].tp_comm
+        full_t = bridging.replica_to_split(targets, comm)
+        ce = ParallelCrossEntropy(comm=comm,
+                                  vocab_begin=state["head"].offset)
+        return ce(logits, full_t)
+
+    engine = epl.Engine(model, loss_fn=loss_fn, optimizer="adamw", lr=1e-3,
+                        dtype=dtype)
+    state["engine"] = engine
+    state["head"] = model.head
+    rank = int(os.environ.get("RANK", "0"))
+    x, y = resnet.synthetic_image_batch(batch, num_classes, size=224,
+                                        device=engine.device,
+                                        seed=7 + rank)
+    x = x.to(dtype)
+    meta = {"model": "resnet50-split100k", "global_batch": batch * world,
+            "seq_len": 224, "parallelism": "dp{}_tp{}".format(world, world)}
+    return engine, (x, y), batch * world, meta
+
+
+def build_moe_bench(args, epl, world, on_gpu, dtype):
+    from easyparallellibrary_amd.models.moe_transformer import (
+        build_moe_transformer)
+    from easyparallellibrary_amd.models import gpt2
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    batch = args.batch or 8
+    seq = args.seq_len if args.seq_len != 512 else 1024
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    vocab = 32000
+    model = build_moe_transformer(world=world, layers=12, hidden=1024,
+                                  heads=16, ffn=4096,
+                                  num_experts=max(8, world),
+                                  vocab_size=vocab, max_pos=seq)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=1e-4, dtype=dtype)
+    rank = int(os.environ.get("RANK", "0"))
+    ids, tgt = gpt2.synthetic_lm_batch(batch, seq, vocab,
+                                       device=engine.device, seed=55 + rank)
+    meta = {"model": "moe-transformer-12L", "global_batch": batch * world,
+            "seq_len": seq, "parallelism": "dp{}_ep{}".format(world, world)}
+    return engine, (ids, tgt), batch * world, meta
+
+
+BUILDERS = {
+    "bert_dp": build_bert_bench,
+    "bert_pp": build_bert_bench,
+    "bert_zero": build_bert_bench,
+    "gpt2_xl": build_gpt2_bench,
+    "resnet_tp": build_resnet_bench,
+    "moe": build_moe_bench,
+}
 
 
 def main():
     args = parse_args()
     import easyparallellibrary_amd as epl
-    from easyparallellibrary_amd.models import bert
-    from easyparallellibrary_amd.ops.distributed_losses import (
-        ParallelCrossEntropy)
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))
     on_gpu = torch.cuda.is_available() if args.device != "cpu" else False
     dtype = torch.bfloat16 if on_gpu else torch.float32
 
-    cfg = {
-        "pipeline.num_micro_batch": args.micro_batch,
-        "zero.level": args.zero,
-    }
-    epl.init(epl.Config(cfg))
-
-    vocab = 30528
-    model = bert.build_bert(args.model, vocab_size=vocab,
-                            max_pos=max(512, args.seq_len),
-                            num_stages=args.pp)
-    loss_fn = ParallelCrossEntropy()
-
-    def ce_loss(logits, targets):
-        return loss_fn(logits, targets)
-
-    engine = epl.Engine(model, loss_fn=ce_loss, optimizer="adamw", lr=1e-4,
-                        dtype=dtype)
-    device = engine.device
-
-    ids, targets = bert.synthetic_mlm_batch(
-        args.batch, args.seq_len, vocab, device=device, seed=1234 + rank)
+    engine, (inputs, targets), samples_per_step, meta = \
+        BUILDERS[args.config](args, epl, world, on_gpu, dtype)
 
     import torch.distributed as dist
 
@@ -80,39 +199,30 @@ def main():
             torch.cuda.synchronize()
 
     for _ in range(args.warmup):
-        engine.train_step(ids, targets)
+        engine.train_step(inputs, targets)
     barrier()
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        engine.train_step(ids, targets)
+        engine.train_step(inputs, targets)
     barrier()
     elapsed = time.perf_counter() - t0
 
     # MAX over ranks
-    et = torch.tensor([elapsed], dtype=torch.float64)
     if dist.is_initialized():
-        et_dev = et.to(device) if dist.get_backend() == "nccl" else et
+        et = torch.tensor([elapsed], dtype=torch.float64)
+        et_dev = et.to(engine.device) if dist.get_backend() == "nccl" else et
         dist.all_reduce(et_dev, op=dist.ReduceOp.MAX)
         elapsed = float(et_dev.cpu().item())
 
-    n_gpus = world if on_gpu else world
-    # whole-job samples/sec: DP replicas each consume args.batch per step
-    num_replicas = engine.num_replicas
-    samples_per_step = args.batch * num_replicas
     value = samples_per_step * args.steps / elapsed
     ms_per_step = elapsed / args.steps * 1000.0
 
     if rank == 0:
-        par = "dp{}".format(num_replicas)
-        if args.pp > 1:
-            par += "_pp{}".format(args.pp)
-        if args.zero:
-            par += "_zero_{}".format(args.zero)
         print(json.dumps({
             "metric": "samples_per_sec",
             "value": value,
             "unit": "samples/s",
-            "n_gpus": n_gpus,
+            "n_gpus": world if on_gpu else world,
             "steps": args.steps,
             "warmup": args.warmup,
             "ms_per_step": ms_per_step,
@@ -121,12 +231,7 @@ def main():
             "vs_baseline": None,
             "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
             "data": "synthetic",
-            "config": {
-                "model": args.model,
-                "global_batch": samples_per_step,
-                "seq_len": args.seq_len,
-                "parallelism": par,
-            },
+            "config": meta,
         }))
 
 

@@ -27,6 +27,10 @@ class TaskGraph:
         self.module_names = []   # maximal tagged subtree roots, in order
         self.modules = []
         self.virtual_device = None
+        # exclusive parameter list (innermost-tag assignment); when set it
+        # overrides the recursive module walk — a replicate subtree that
+        # CONTAINS a split subtree must not claim the split parameters
+        self.params_exclusive = None
 
     @property
     def strategy_type(self):
@@ -41,6 +45,10 @@ class TaskGraph:
         return self.strategy.device_count
 
     def parameters(self):
+        if self.params_exclusive is not None:
+            for p in self.params_exclusive:
+                yield p
+            return
         for m in self.modules:
             for p in m.parameters():
                 yield p
@@ -99,10 +107,14 @@ class Plan:
 
         root_strategy = hooks.strategy_of(model)
         if root_strategy is not None:
-            # whole model under one scope
+            # whole model under one scope; still descend — a nested scope
+            # (e.g. split expert weights under a default-replicate model)
+            # forms its own taskgraph
             tg = tg_of(root_strategy)
             tg.module_names.append("")
             tg.modules.append(model)
+            for child_name, child in model.named_children():
+                visit(child, child_name, root_strategy)
         else:
             start = default
             for child_name, child in model.named_children():
@@ -139,6 +151,29 @@ class Plan:
             for name, mod in sorted(uncovered):
                 fallback_tg.module_names.append(name)
                 fallback_tg.modules.append(mod)
+
+        # exclusive parameter assignment: every parameter belongs to the
+        # taskgraph of its module's INNERMOST tag (untagged -> default /
+        # fallback taskgraph)
+        assignment = {}
+        eff_default = default if default is not None else (
+            fallback_tg.strategy if fallback_tg is not None else None)
+
+        def assign(module, strategy):
+            s = hooks.strategy_of(module) or strategy
+            if s is not None:
+                for p in module.parameters(recurse=False):
+                    assignment.setdefault(id(p), (p, s))
+            for c in module.children():
+                assign(c, s)
+
+        assign(model, eff_default)
+        per_tg = {}
+        for p, s in assignment.values():
+            per_tg.setdefault(s.index, []).append(p)
+        for tg in taskgraphs:
+            if tg.strategy.index in per_tg:
+                tg.params_exclusive = per_tg[tg.strategy.index]
 
         taskgraphs = [tg for tg in taskgraphs if tg.modules]
         for i, tg in enumerate(taskgraphs):

@@ -44,6 +44,9 @@ class ExpertParallelMLP(nn.Module):
             torch.empty(self.local_experts, ffn_hidden, hidden))
         nn.init.normal_(self.w1, std=0.02)
         nn.init.normal_(self.w2, std=0.02)
+        if self.world > 1:
+            self.w1._epl_shard_dim = 0
+            self.w2._epl_shard_dim = 0
 
     def set_comm(self, comm):
         """Called by the engine's split transform: shard the expert weights
@@ -63,6 +66,8 @@ class ExpertParallelMLP(nn.Module):
         with torch.no_grad():
             self.w1 = nn.Parameter(self.w1[lo:hi].clone())
             self.w2 = nn.Parameter(self.w2[lo:hi].clone())
+        self.w1._epl_shard_dim = 0
+        self.w2._epl_shard_dim = 0
 
     def forward(self, x):
         orig_shape = x.shape

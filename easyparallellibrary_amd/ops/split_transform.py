@@ -95,6 +95,9 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
                     _replace_module(parent, child_name,
                                     _make_sharded(child, comm, gather_input))
                     replaced += 1
+    # parameters changed object identity: refresh the taskgraph's
+    # exclusive param list from the (transformed) module roots
+    tg.params_exclusive = None
     logger.info("split transform: %d module(s) sharded %d-way over %s",
                 replaced, comm.size, comm.name)
     return replaced

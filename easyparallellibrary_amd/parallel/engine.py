@@ -203,13 +203,22 @@ class Engine:
             vd = tg.virtual_device
             k = len(vd.local_ranks(0))
             if tg.is_split:
-                # shard i's copies across replicas
-                groups = [[vd.local_ranks(r)[i] for r in range(self.num_replicas)]
-                          for i in range(k)]
+                # sharded params: DP group = shard i's copies across
+                # replicas.  UN-sharded params inside a split scope (e.g.
+                # the MoE gate) are replicated on every tg rank and sync
+                # across all of them.
+                groups = [
+                    ([vd.local_ranks(r)[i] for r in range(self.num_replicas)],
+                     "sharded", i)
+                    for i in range(k)]
+                groups.append((vd.all_ranks, "replicated", k))
             else:
-                # all ranks holding this (replicated) taskgraph
-                groups = [vd.all_ranks]
-            for gi, granks in enumerate(groups):
+                groups = [(vd.all_ranks, "all", 0)]
+            my_pos = None
+            ranks_here = vd.local_ranks(self.replica_id)
+            if self.rank in ranks_here:
+                my_pos = ranks_here.index(self.rank)
+            for granks, kind, gi in groups:
                 name = "{}_dp_tg{}_g{}".format(self._ns, tg.index, gi)
                 if self.rank not in granks:
                     if not torch.cuda.is_available() and dist.is_initialized():
@@ -220,6 +229,13 @@ class Engine:
                 pool = CommunicationPool(name, granks, pool_n)
                 bcomm = create_communicator(name + "_b", granks)
                 params = [p for p in tg.parameters() if p.requires_grad]
+                if kind == "sharded":
+                    params = [p for p in params
+                              if getattr(p, "_epl_shard_dim", None)
+                              is not None]
+                elif kind == "replicated":
+                    params = [p for p in params
+                              if getattr(p, "_epl_shard_dim", None) is None]
                 if not params:
                     continue
                 fg = FlatParamGroup(
@@ -251,6 +267,12 @@ class Engine:
             from easyparallellibrary_amd.runtime.zero import ZeroOptimizer
             self.optimizer = ZeroOptimizer(
                 self, optimizer, okw, level=self.config.zero.level)
+        elif self.config.offload.level:
+            from easyparallellibrary_amd.runtime.offload import (
+                CPUOffloadAdamW)
+            if optimizer != "adamw":
+                raise NotImplementedError("offload supports adamw")
+            self.optimizer = CPUOffloadAdamW(self.flat_groups, **okw)
         else:
             self.optimizer = OPTIMIZERS[optimizer](self.flat_groups, **okw)
 
