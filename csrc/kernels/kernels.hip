@@ -88,10 +88,11 @@ DEV float block_reduce_sum(float val, float* lds /* >= 4 floats */) {
     val += __shfl_down(val, off, 64);
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
+  const int nwaves = blockDim.x >> 6;
   if (lane == 0) lds[wave] = val;
   __syncthreads();
   if (wave == 0) {
-    val = lane < 4 ? lds[lane] : 0.f;
+    val = lane < nwaves ? lds[lane] : 0.f;
 #pragma unroll
     for (int off = 2; off > 0; off >>= 1)
       val += __shfl_down(val, off, 64);
@@ -107,10 +108,11 @@ DEV float block_reduce_max(float val, float* lds) {
     val = fmaxf(val, __shfl_down(val, off, 64));
   const int wave = threadIdx.x >> 6;
   const int lane = threadIdx.x & 63;
+  const int nwaves = blockDim.x >> 6;
   if (lane == 0) lds[wave] = val;
   __syncthreads();
   if (wave == 0) {
-    val = lane < 4 ? lds[lane] : -INFINITY;
+    val = lane < nwaves ? lds[lane] : -INFINITY;
 #pragma unroll
     for (int off = 2; off > 0; off >>= 1)
       val = fmaxf(val, __shfl_down(val, off, 64));
@@ -238,63 +240,56 @@ __global__ void lamb_phase2_kernel(float* __restrict__ master,
 
 // ============================================================================
 // LayerNorm forward with optional fused residual add: s = x (+ res);
-// y = (s - mean(s)) * rstd(s) * gamma + beta.  One block per row; the fp32
-// row image is staged in LDS during the statistics pass so the normalize
-// pass never re-reads HBM.  With HAS_RES the summed input s is also written
-// out (bf16/f32) — it is the tensor the backward normalizes against and,
-// for pre-LN blocks, the residual stream.  Saves mean and rstd.
+// y = (s - mean(s)) * rstd(s) * gamma + beta.  One block per row.  The
+// bf16 fast path caches each thread's own row values in REGISTERS between
+// the statistics pass and the normalize pass (threads re-touch the same
+// columns), so HBM is read once and LDS carries only the 8-float block
+// reduction.  CHUNKS = row float8-chunks per thread; the launcher picks
+// the block size so CHUNKS <= 2 covers cols <= 4096 at 256 threads.
+// With HAS_RES the summed input s is also written out (the tensor the
+// backward normalizes against and, for pre-LN blocks, the residual
+// stream).  Saves mean and rstd.
 // ============================================================================
-template <bool BF16, bool HAS_RES>
-__global__ void layer_norm_fwd_kernel(
+template <bool HAS_RES, int CHUNKS>
+__global__ void layer_norm_fwd_bf16_kernel(
     void* __restrict__ out, const void* __restrict__ x,
     const void* __restrict__ res, void* __restrict__ sum_out,
     const void* __restrict__ gamma, const void* __restrict__ beta,
     float* __restrict__ mean_out, float* __restrict__ rstd_out, int64_t rows,
     int64_t cols, float eps) {
-  extern __shared__ __attribute__((aligned(16))) char smem[];
-  float* srow = reinterpret_cast<float*>(smem);  // [cols] fp32 row cache
   __shared__ float lds[8];
-  using T = unsigned short;  // bf16 raw
+  using T = unsigned short;
+  float8 cache[CHUNKS];
+  float8 gw[CHUNKS], bw[CHUNKS];
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k) {
+    const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+    if (c < cols) {
+      gw[k] = load_bf16x8(reinterpret_cast<const T*>(gamma) + c);
+      bw[k] = load_bf16x8(reinterpret_cast<const T*>(beta) + c);
+    }
+  }
   for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
     const int64_t base = row * cols;
     float sum = 0.f, sumsq = 0.f;
-    if (BF16) {
-      const T* xr = reinterpret_cast<const T*>(x) + base;
-      const T* rr = HAS_RES ? reinterpret_cast<const T*>(res) + base : nullptr;
-      T* sr = HAS_RES ? reinterpret_cast<T*>(sum_out) + base : nullptr;
-      for (int64_t c = threadIdx.x * 8; c < cols;
-           c += (int64_t)blockDim.x * 8) {
-        float8 f = load_bf16x8(xr + c);
-        if (HAS_RES) {
-          float8 r = load_bf16x8(rr + c);
 #pragma unroll
-          for (int k = 0; k < 8; ++k) f.v[k] += r.v[k];
-          store_bf16x8(sr + c, f);
-        }
+    for (int k = 0; k < CHUNKS; ++k) {
+      const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+      if (c >= cols) continue;
+      float8 f = load_bf16x8(reinterpret_cast<const T*>(x) + base + c);
+      if (HAS_RES) {
+        float8 r = load_bf16x8(reinterpret_cast<const T*>(res) + base + c);
 #pragma unroll
-        for (int k = 0; k < 8; ++k) {
-          srow[c + k] = f.v[k];
-          sum += f.v[k];
-          sumsq += f.v[k] * f.v[k];
-        }
+        for (int j = 0; j < 8; ++j) f.v[j] += r.v[j];
+        store_bf16x8(reinterpret_cast<T*>(sum_out) + base + c, f);
       }
-    } else {
-      const float* xr = reinterpret_cast<const float*>(x) + base;
-      const float* rr =
-          HAS_RES ? reinterpret_cast<const float*>(res) + base : nullptr;
-      float* sr = HAS_RES ? reinterpret_cast<float*>(sum_out) + base : nullptr;
-      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-        float fv = xr[c];
-        if (HAS_RES) {
-          fv += rr[c];
-          sr[c] = fv;
-        }
-        srow[c] = fv;
-        sum += fv;
-        sumsq += fv * fv;
+      cache[k] = f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        sum += f.v[j];
+        sumsq += f.v[j] * f.v[j];
       }
     }
-    __syncthreads();
     sum = block_reduce_sum(sum, lds);
     __syncthreads();
     sumsq = block_reduce_sum(sumsq, lds);
@@ -305,39 +300,157 @@ __global__ void layer_norm_fwd_kernel(
       mean_out[row] = mean;
       rstd_out[row] = rstd;
     }
-    if (BF16) {
-      const T* gr = reinterpret_cast<const T*>(gamma);
-      const T* br = reinterpret_cast<const T*>(beta);
-      T* yr = reinterpret_cast<T*>(out) + base;
-      for (int64_t c = threadIdx.x * 8; c < cols;
-           c += (int64_t)blockDim.x * 8) {
-        float8 g = load_bf16x8(gr + c);
-        float8 b = load_bf16x8(br + c);
-        float8 y;
 #pragma unroll
-        for (int k = 0; k < 8; ++k)
-          y.v[k] = (srow[c + k] - mean) * rstd * g.v[k] + b.v[k];
-        store_bf16x8(yr + c, y);
-      }
-    } else {
-      const float* gr = reinterpret_cast<const float*>(gamma);
-      const float* br = reinterpret_cast<const float*>(beta);
-      float* yr = reinterpret_cast<float*>(out) + base;
-      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-        yr[c] = (srow[c] - mean) * rstd * gr[c] + br[c];
-      }
+    for (int k = 0; k < CHUNKS; ++k) {
+      const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+      if (c >= cols) continue;
+      float8 y;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        y.v[j] = (cache[k].v[j] - mean) * rstd * gw[k].v[j] + bw[k].v[j];
+      store_bf16x8(reinterpret_cast<T*>(out) + base + c, y);
     }
     __syncthreads();
   }
 }
 
-// ============================================================================
-// LayerNorm backward.  Each block walks a strided set of rows: computes dx
-// and accumulates dgamma/dbeta partials in LDS (cols*2 floats), flushing to
-// global fp32 buffers with one atomicAdd per column per block at the end.
-// Requires cols*8 bytes <= 128 KiB of LDS (cols <= 16384).
-// ============================================================================
-template <bool BF16, bool CACHE>
+// generic (fp32 / large-cols) forward: two-pass re-read
+template <bool BF16, bool HAS_RES>
+__global__ void layer_norm_fwd_kernel(
+    void* __restrict__ out, const void* __restrict__ x,
+    const void* __restrict__ res, void* __restrict__ sum_out,
+    const void* __restrict__ gamma, const void* __restrict__ beta,
+    float* __restrict__ mean_out, float* __restrict__ rstd_out, int64_t rows,
+    int64_t cols, float eps) {
+  __shared__ float lds[8];
+  using T = unsigned short;
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    float sum = 0.f, sumsq = 0.f;
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      float fv = BF16 ? bf2f(reinterpret_cast<const T*>(x)[base + c])
+                      : reinterpret_cast<const float*>(x)[base + c];
+      if (HAS_RES) {
+        fv += BF16 ? bf2f(reinterpret_cast<const T*>(res)[base + c])
+                   : reinterpret_cast<const float*>(res)[base + c];
+        if (BF16)
+          reinterpret_cast<T*>(sum_out)[base + c] = f2bf(fv);
+        else
+          reinterpret_cast<float*>(sum_out)[base + c] = fv;
+      }
+      sum += fv;
+      sumsq += fv * fv;
+    }
+    sum = block_reduce_sum(sum, lds);
+    __syncthreads();
+    sumsq = block_reduce_sum(sumsq, lds);
+    const float mean = sum / cols;
+    const float var = sumsq / cols - mean * mean;
+    const float rstd = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean_out[row] = mean;
+      rstd_out[row] = rstd;
+    }
+    const void* nin = HAS_RES ? sum_out : x;
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      float fv = BF16 ? bf2f(reinterpret_cast<const T*>(nin)[base + c])
+                      : reinterpret_cast<const float*>(nin)[base + c];
+      float gv = BF16 ? bf2f(reinterpret_cast<const T*>(gamma)[c])
+                      : reinterpret_cast<const float*>(gamma)[c];
+      float bv = BF16 ? bf2f(reinterpret_cast<const T*>(beta)[c])
+                      : reinterpret_cast<const float*>(beta)[c];
+      float y = (fv - mean) * rstd * gv + bv;
+      if (BF16)
+        reinterpret_cast<T*>(out)[base + c] = f2bf(y);
+      else
+        reinterpret_cast<float*>(out)[base + c] = y;
+    }
+    __syncthreads();
+  }
+}
+
+// bf16 fast backward: register row cache + register dgamma/dbeta
+// accumulators (flushed once per block with atomicAdd) — zero LDS traffic
+// beyond the block reduction; HBM reads dy/x once, writes dx once.
+template <int CHUNKS>
+__global__ void layer_norm_bwd_bf16_kernel(
+    void* __restrict__ dx, float* __restrict__ dgamma,
+    float* __restrict__ dbeta, const void* __restrict__ dy,
+    const void* __restrict__ x, const void* __restrict__ gamma,
+    const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
+    int64_t rows, int64_t cols) {
+  __shared__ float lds[8];
+  using T = unsigned short;
+  float8 gw[CHUNKS], acc_dg[CHUNKS], acc_db[CHUNKS];
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k) {
+    const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      acc_dg[k].v[j] = 0.f;
+      acc_db[k].v[j] = 0.f;
+    }
+    if (c < cols)
+      gw[k] = load_bf16x8(reinterpret_cast<const T*>(gamma) + c);
+  }
+  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+    const int64_t base = row * cols;
+    const float mean = mean_in[row];
+    const float rstd = rstd_in[row];
+    float sum_dyg = 0.f, sum_dygx = 0.f;
+    float8 c_xhat[CHUNKS], c_dyg[CHUNKS], c_dy[CHUNKS];
+#pragma unroll
+    for (int k = 0; k < CHUNKS; ++k) {
+      const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+      if (c >= cols) continue;
+      float8 d = load_bf16x8(reinterpret_cast<const T*>(dy) + base + c);
+      float8 f = load_bf16x8(reinterpret_cast<const T*>(x) + base + c);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const float xhat = (f.v[j] - mean) * rstd;
+        const float dyg = d.v[j] * gw[k].v[j];
+        c_xhat[k].v[j] = xhat;
+        c_dyg[k].v[j] = dyg;
+        c_dy[k].v[j] = d.v[j];
+        sum_dyg += dyg;
+        sum_dygx += dyg * xhat;
+      }
+    }
+    __syncthreads();
+    sum_dyg = block_reduce_sum(sum_dyg, lds);
+    __syncthreads();
+    sum_dygx = block_reduce_sum(sum_dygx, lds);
+    const float inv_cols = 1.f / cols;
+#pragma unroll
+    for (int k = 0; k < CHUNKS; ++k) {
+      const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+      if (c >= cols) continue;
+      float8 o;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        o.v[j] = (c_dyg[k].v[j] -
+                  (sum_dyg + c_xhat[k].v[j] * sum_dygx) * inv_cols) * rstd;
+        acc_dg[k].v[j] += c_dy[k].v[j] * c_xhat[k].v[j];
+        acc_db[k].v[j] += c_dy[k].v[j];
+      }
+      store_bf16x8(reinterpret_cast<T*>(dx) + base + c, o);
+    }
+    __syncthreads();
+  }
+#pragma unroll
+  for (int k = 0; k < CHUNKS; ++k) {
+    const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+    if (c >= cols) continue;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      atomicAdd(&dgamma[c + j], acc_dg[k].v[j]);
+      atomicAdd(&dbeta[c + j], acc_db[k].v[j]);
+    }
+  }
+}
+
+// generic (fp32 / large-cols) backward: LDS partials + global re-read
+template <bool BF16>
 __global__ void layer_norm_bwd_kernel(
     void* __restrict__ dx, float* __restrict__ dgamma,
     float* __restrict__ dbeta, const void* __restrict__ dy,
@@ -347,8 +460,6 @@ __global__ void layer_norm_bwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   float* part_dg = reinterpret_cast<float*>(smem);            // [cols]
   float* part_db = part_dg + cols;                            // [cols]
-  float* c_xhat = CACHE ? part_db + cols : nullptr;           // [cols]
-  float* c_dyg = CACHE ? c_xhat + cols : nullptr;             // [cols]
   __shared__ float lds[8];
   using T = unsigned short;
   for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
@@ -361,92 +472,39 @@ __global__ void layer_norm_bwd_kernel(
     const float mean = mean_in[row];
     const float rstd = rstd_in[row];
     float sum_dyg = 0.f, sum_dygx = 0.f;
-    if (BF16) {
-      const T* dyr = reinterpret_cast<const T*>(dy) + base;
-      const T* xr = reinterpret_cast<const T*>(x) + base;
-      const T* gr = reinterpret_cast<const T*>(gamma);
-      for (int64_t c = threadIdx.x * 8; c < cols;
-           c += (int64_t)blockDim.x * 8) {
-        float8 d = load_bf16x8(dyr + c);
-        float8 f = load_bf16x8(xr + c);
-        float8 g = load_bf16x8(gr + c);
-#pragma unroll
-        for (int k = 0; k < 8; ++k) {
-          const float xhat = (f.v[k] - mean) * rstd;
-          const float dyg = d.v[k] * g.v[k];
-          sum_dyg += dyg;
-          sum_dygx += dyg * xhat;
-          part_dg[c + k] += d.v[k] * xhat;
-          part_db[c + k] += d.v[k];
-          if (CACHE) {
-            c_xhat[c + k] = xhat;
-            c_dyg[c + k] = dyg;
-          }
-        }
-      }
-    } else {
-      const float* dyr = reinterpret_cast<const float*>(dy) + base;
-      const float* xr = reinterpret_cast<const float*>(x) + base;
-      const float* gr = reinterpret_cast<const float*>(gamma);
-      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-        const float xhat = (xr[c] - mean) * rstd;
-        const float dyg = dyr[c] * gr[c];
-        sum_dyg += dyg;
-        sum_dygx += dyg * xhat;
-        part_dg[c] += dyr[c] * xhat;
-        part_db[c] += dyr[c];
-        if (CACHE) {
-          c_xhat[c] = xhat;
-          c_dyg[c] = dyg;
-        }
-      }
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      const float dv = BF16 ? bf2f(reinterpret_cast<const T*>(dy)[base + c])
+                            : reinterpret_cast<const float*>(dy)[base + c];
+      const float fv = BF16 ? bf2f(reinterpret_cast<const T*>(x)[base + c])
+                            : reinterpret_cast<const float*>(x)[base + c];
+      const float gv = BF16 ? bf2f(reinterpret_cast<const T*>(gamma)[c])
+                            : reinterpret_cast<const float*>(gamma)[c];
+      const float xhat = (fv - mean) * rstd;
+      const float dyg = dv * gv;
+      sum_dyg += dyg;
+      sum_dygx += dyg * xhat;
+      part_dg[c] += dv * xhat;
+      part_db[c] += dv;
     }
     __syncthreads();
     sum_dyg = block_reduce_sum(sum_dyg, lds);
     __syncthreads();
     sum_dygx = block_reduce_sum(sum_dygx, lds);
     const float inv_cols = 1.f / cols;
-    if (BF16) {
-      T* dxr = reinterpret_cast<T*>(dx) + base;
-      const T* dyr = reinterpret_cast<const T*>(dy) + base;
-      const T* xr = reinterpret_cast<const T*>(x) + base;
-      const T* gr = reinterpret_cast<const T*>(gamma);
-      for (int64_t c = threadIdx.x * 8; c < cols;
-           c += (int64_t)blockDim.x * 8) {
-        float8 o;
-        if (CACHE) {
-#pragma unroll
-          for (int k = 0; k < 8; ++k)
-            o.v[k] = (c_dyg[c + k] -
-                      (sum_dyg + c_xhat[c + k] * sum_dygx) * inv_cols) * rstd;
-        } else {
-          float8 d = load_bf16x8(dyr + c);
-          float8 f = load_bf16x8(xr + c);
-          float8 g = load_bf16x8(gr + c);
-#pragma unroll
-          for (int k = 0; k < 8; ++k) {
-            const float xhat = (f.v[k] - mean) * rstd;
-            o.v[k] = (d.v[k] * g.v[k] -
-                      (sum_dyg + xhat * sum_dygx) * inv_cols) * rstd;
-          }
-        }
-        store_bf16x8(dxr + c, o);
-      }
-    } else {
-      float* dxr = reinterpret_cast<float*>(dx) + base;
-      const float* dyr = reinterpret_cast<const float*>(dy) + base;
-      const float* xr = reinterpret_cast<const float*>(x) + base;
-      const float* gr = reinterpret_cast<const float*>(gamma);
-      for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
-        if (CACHE) {
-          dxr[c] = (c_dyg[c] - (sum_dyg + c_xhat[c] * sum_dygx) * inv_cols) *
-                   rstd;
-        } else {
-          const float xhat = (xr[c] - mean) * rstd;
-          dxr[c] = (dyr[c] * gr[c] - (sum_dyg + xhat * sum_dygx) * inv_cols) *
-                   rstd;
-        }
-      }
+    for (int64_t c = threadIdx.x; c < cols; c += blockDim.x) {
+      const float dv = BF16 ? bf2f(reinterpret_cast<const T*>(dy)[base + c])
+                            : reinterpret_cast<const float*>(dy)[base + c];
+      const float fv = BF16 ? bf2f(reinterpret_cast<const T*>(x)[base + c])
+                            : reinterpret_cast<const float*>(x)[base + c];
+      const float gv = BF16 ? bf2f(reinterpret_cast<const T*>(gamma)[c])
+                            : reinterpret_cast<const float*>(gamma)[c];
+      const float xhat = (fv - mean) * rstd;
+      const float o = (dv * gv - (sum_dyg + xhat * sum_dygx) * inv_cols) *
+                      rstd;
+      if (BF16)
+        reinterpret_cast<T*>(dx)[base + c] = f2bf(o);
+      else
+        reinterpret_cast<float*>(dx)[base + c] = o;
     }
     __syncthreads();
   }
@@ -810,31 +868,64 @@ void epl_lamb_phase2(float* master, unsigned short* param_bf16,
                      master, param_bf16, update, chunk_of, ratio, n, lr);
 }
 
+static int ln_block_for(int64_t cols) {
+  // threads so that CHUNKS = ceil(cols/8/threads) <= 2 at 256 threads
+  int64_t need = (cols + 7) / 8;          // float8 chunks in a row
+  int64_t thr = (need + 1) / 2;           // 2 chunks per thread max
+  if (thr <= 64) return 64;
+  if (thr <= 128) return 128;
+  if (thr <= 192) return 192;
+  return 256;
+}
+
 void epl_layer_norm_fwd(void* out, const void* x, const void* res,
                         void* sum_out, const void* gamma, const void* beta,
                         float* mean, float* rstd, int64_t rows, int64_t cols,
                         float eps, bool bf16, hipStream_t stream) {
   const int grid = (int)(rows < kMaxGrid ? rows : kMaxGrid);
-  const size_t lds_bytes = (size_t)cols * sizeof(float);
   const bool has_res = res != nullptr;
+  if (bf16 && cols % 8 == 0 && cols <= 2 * 256 * 8) {
+    const int block = ln_block_for(cols);
+    const bool two = cols > (int64_t)block * 8;
+    if (has_res) {
+      if (two)
+        hipLaunchKernelGGL((layer_norm_fwd_bf16_kernel<true, 2>), dim3(grid),
+                           dim3(block), 0, stream, out, x, res, sum_out,
+                           gamma, beta, mean, rstd, rows, cols, eps);
+      else
+        hipLaunchKernelGGL((layer_norm_fwd_bf16_kernel<true, 1>), dim3(grid),
+                           dim3(block), 0, stream, out, x, res, sum_out,
+                           gamma, beta, mean, rstd, rows, cols, eps);
+    } else {
+      if (two)
+        hipLaunchKernelGGL((layer_norm_fwd_bf16_kernel<false, 2>), dim3(grid),
+                           dim3(block), 0, stream, out, x, res, sum_out,
+                           gamma, beta, mean, rstd, rows, cols, eps);
+      else
+        hipLaunchKernelGGL((layer_norm_fwd_bf16_kernel<false, 1>), dim3(grid),
+                           dim3(block), 0, stream, out, x, res, sum_out,
+                           gamma, beta, mean, rstd, rows, cols, eps);
+    }
+    return;
+  }
   if (bf16) {
     if (has_res)
       hipLaunchKernelGGL((layer_norm_fwd_kernel<true, true>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, out, x, res,
-                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+                         dim3(kBlock), 0, stream, out, x, res, sum_out,
+                         gamma, beta, mean, rstd, rows, cols, eps);
     else
       hipLaunchKernelGGL((layer_norm_fwd_kernel<true, false>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, out, x, res,
-                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+                         dim3(kBlock), 0, stream, out, x, res, sum_out,
+                         gamma, beta, mean, rstd, rows, cols, eps);
   } else {
     if (has_res)
       hipLaunchKernelGGL((layer_norm_fwd_kernel<false, true>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, out, x, res,
-                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+                         dim3(kBlock), 0, stream, out, x, res, sum_out,
+                         gamma, beta, mean, rstd, rows, cols, eps);
     else
       hipLaunchKernelGGL((layer_norm_fwd_kernel<false, false>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, out, x, res,
-                         sum_out, gamma, beta, mean, rstd, rows, cols, eps);
+                         dim3(kBlock), 0, stream, out, x, res, sum_out,
+                         gamma, beta, mean, rstd, rows, cols, eps);
   }
 }
 
@@ -843,28 +934,28 @@ void epl_layer_norm_bwd(void* dx, float* dgamma, float* dbeta, const void* dy,
                         const float* rstd, int64_t rows, int64_t cols,
                         bool bf16, hipStream_t stream) {
   int grid = (int)(rows < 1024 ? rows : 1024);
-  const bool cache = (size_t)cols * 16 <= 128 * 1024;
-  const size_t lds_bytes =
-      (size_t)cols * (cache ? 4 : 2) * sizeof(float);
-  if (bf16) {
-    if (cache)
-      hipLaunchKernelGGL((layer_norm_bwd_kernel<true, true>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
-                         dy, x, gamma, mean, rstd, rows, cols);
+  if (bf16 && cols % 8 == 0 && cols <= 2 * 256 * 8) {
+    const int block = ln_block_for(cols);
+    const bool two = cols > (int64_t)block * 8;
+    if (two)
+      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<2>), dim3(grid),
+                         dim3(block), 0, stream, dx, dgamma, dbeta, dy, x,
+                         gamma, mean, rstd, rows, cols);
     else
-      hipLaunchKernelGGL((layer_norm_bwd_kernel<true, false>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
-                         dy, x, gamma, mean, rstd, rows, cols);
-  } else {
-    if (cache)
-      hipLaunchKernelGGL((layer_norm_bwd_kernel<false, true>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
-                         dy, x, gamma, mean, rstd, rows, cols);
-    else
-      hipLaunchKernelGGL((layer_norm_bwd_kernel<false, false>), dim3(grid),
-                         dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
-                         dy, x, gamma, mean, rstd, rows, cols);
+      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<1>), dim3(grid),
+                         dim3(block), 0, stream, dx, dgamma, dbeta, dy, x,
+                         gamma, mean, rstd, rows, cols);
+    return;
   }
+  const size_t lds_bytes = (size_t)cols * 2 * sizeof(float);
+  if (bf16)
+    hipLaunchKernelGGL((layer_norm_bwd_kernel<true>), dim3(grid),
+                       dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
+                       dy, x, gamma, mean, rstd, rows, cols);
+  else
+    hipLaunchKernelGGL((layer_norm_bwd_kernel<false>), dim3(grid),
+                       dim3(kBlock), lds_bytes, stream, dx, dgamma, dbeta,
+                       dy, x, gamma, mean, rstd, rows, cols);
 }
 
 void epl_bias_gelu_fwd(void* out, const void* x, const void* bias,

@@ -73,7 +73,7 @@ class ExpertParallelMLP(nn.Module):
         orig_shape = x.shape
         x = x.reshape(-1, self.hidden)
         n_tokens = x.shape[0]
-        logits = self.gate(x.float())
+        logits = self.gate(x).float()
         probs = logits.softmax(dim=-1)
         topv, topi = probs.topk(self.top_k, dim=-1)
         topv = topv / topv.sum(dim=-1, keepdim=True)
