@@ -869,12 +869,12 @@ void epl_lamb_phase2(float* master, unsigned short* param_bf16,
 }
 
 static int ln_block_for(int64_t cols) {
-  // threads so that CHUNKS = ceil(cols/8/threads) <= 2 at 256 threads
+  // maximize threads per row: CHUNKS=1 whenever cols/8 fits in <=256
+  // threads; CHUNKS=2 only for cols in (2048, 4096]
   int64_t need = (cols + 7) / 8;          // float8 chunks in a row
-  int64_t thr = (need + 1) / 2;           // 2 chunks per thread max
-  if (thr <= 64) return 64;
-  if (thr <= 128) return 128;
-  if (thr <= 192) return 192;
+  if (need <= 64) return 64;
+  if (need <= 128) return 128;
+  if (need <= 192) return 192;
   return 256;
 }
 
