@@ -110,9 +110,15 @@ class ExpertParallelMLP(nn.Module):
         d = d.reshape(self.world, self.local_experts, capacity, self.hidden)
         d = d.transpose(0, 1).reshape(self.local_experts,
                                       self.world * capacity, self.hidden)
-        h = torch.bmm(d, self.w1)
-        h = F.gelu(h)
-        h = torch.bmm(h, self.w2)
+        # per-expert 2D GEMMs instead of torch.bmm: the bf16 batched-GEMM
+        # BACKWARD ([E,c,4h]x[E,4h,h]) memory-faults in hipBLASLt on
+        # ROCm 7.0 / gfx950 (isolated in tests/moe_bisect_gpu.py bmm2);
+        # 2D mms hit the well-tested non-batched path.
+        outs = []
+        for e in range(self.local_experts):
+            he = F.gelu(d[e] @ self.w1[e])
+            outs.append(he @ self.w2[e])
+        h = torch.stack(outs)
         h = h.reshape(self.local_experts, self.world, capacity, self.hidden)
         h = h.transpose(0, 1).reshape(
             self.world, self.local_experts * capacity, self.hidden)
