@@ -354,6 +354,33 @@ class Engine:
         with torch.no_grad():
             return self._runnable(inputs)
 
+    # ---- merged outputs (reference: parallel/parallel.py:233-353) ------------
+    def all_reduce_metric(self, value, op="mean"):
+        """Merge a scalar/tensor metric across every rank (the reference's
+        merged loss/metric collections).  Returns the merged tensor."""
+        t = value if torch.is_tensor(value) else torch.tensor(
+            float(value), device=self.device)
+        t = t.detach().clone()
+        if not dist.is_initialized() or self.world_size == 1:
+            return t
+        if t.is_cuda and dist.get_backend() == "nccl":
+            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+        else:
+            tc = t.cpu()
+            dist.all_reduce(tc, op=dist.ReduceOp.SUM,
+                            group=self._control_group)
+            t = tc.to(t.device)
+        if op == "mean":
+            t = t / self.world_size
+        return t
+
+    def slice_input_files(self, files):
+        """Per-replica IO slicing (config io.slicing; reference
+        graph_editor.py:149-215)."""
+        from easyparallellibrary_amd.utils.io_slicing import slice_files
+        return slice_files(files, self.num_replicas, self.replica_id,
+                           unbalanced=self.config.io.unbalanced_io_slicing)
+
     # ---- checkpoint ----------------------------------------------------------
     def save_checkpoint(self, path, save_optimizer=True):
         from easyparallellibrary_amd.runtime import saver

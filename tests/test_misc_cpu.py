@@ -1,0 +1,74 @@
+"""IO slicing, metric merging, launcher, AMP scaler units."""
+
+import subprocess
+import sys
+
+import torch
+
+from easyparallellibrary_amd.utils.io_slicing import (slice_dataset_indices,
+                                                      slice_files)
+
+
+def test_slice_files_balanced():
+    files = [str(i) for i in range(10)]
+    parts = [slice_files(files, 4, r) for r in range(4)]
+    assert [len(p) for p in parts] == [3, 3, 2, 2]
+    assert sum(parts, []) == files
+
+
+def test_slice_indices():
+    idx = [list(slice_dataset_indices(10, 3, r)) for r in range(3)]
+    assert sum(idx, []) == list(range(10))
+
+
+def test_dynamic_loss_scaler():
+    from easyparallellibrary_amd.runtime.amp import DynamicLossScaler
+    s = DynamicLossScaler(init_scale=1024, growth_interval=2)
+    s.update(False)
+    s.update(False)
+    assert s.scale == 2048
+    s.update(True)
+    assert s.scale == 1024
+    s.update(True)
+    assert s.scale == 512
+
+
+def test_launcher_cli():
+    """2-worker launcher runs a trivial script and propagates env."""
+    script = (
+        "import os,sys;"
+        "assert os.environ['WORLD_SIZE']=='2';"
+        "print('rank', os.environ['RANK'])"
+    )
+    import tempfile
+    with tempfile.NamedTemporaryFile("w", suffix=".py", delete=False) as f:
+        f.write(script)
+        path = f.name
+    rc = subprocess.call(
+        [sys.executable, "-m", "easyparallellibrary_amd.launcher",
+         "--num_workers", "2", "--gpu_per_worker", "0", path])
+    assert rc == 0
+
+
+def test_launcher_failure_propagates():
+    import tempfile
+    with tempfile.NamedTemporaryFile("w", suffix=".py", delete=False) as f:
+        f.write("import sys; sys.exit(3)")
+        path = f.name
+    rc = subprocess.call(
+        [sys.executable, "-m", "easyparallellibrary_amd.launcher",
+         "--num_workers", "2", "--gpu_per_worker", "0", path])
+    assert rc != 0
+
+
+def test_metric_merge_single():
+    import easyparallellibrary_amd as epl
+    import torch.nn as nn
+    epl.init()
+    with epl.replicate(1):
+        m = nn.Linear(2, 2)
+    engine = epl.Engine(m, loss_fn=nn.MSELoss())
+    out = engine.all_reduce_metric(3.0)
+    assert float(out) == 3.0
+    files = engine.slice_input_files(["a", "b"])
+    assert files == ["a", "b"]
