@@ -178,8 +178,25 @@ BUILDERS = {
 }
 
 
+def _enable_tunableop():
+    """Load the pre-tuned hipBLASLt algorithm table (profiles/tunableop/)
+    so library GEMMs use the algorithms tuned on MI355X — no tuning cost
+    at run time.  Opt out with PYTORCH_TUNABLEOP_ENABLED=0."""
+    if "PYTORCH_TUNABLEOP_ENABLED" in os.environ:
+        return
+    here = os.path.dirname(os.path.abspath(__file__))
+    base = os.path.join(here, "profiles", "tunableop", "tunableop.csv")
+    if os.path.exists(os.path.join(here, "profiles", "tunableop",
+                                   "tunableop0.csv")):
+        os.environ["PYTORCH_TUNABLEOP_ENABLED"] = "1"
+        os.environ["PYTORCH_TUNABLEOP_TUNING"] = "0"
+        os.environ["PYTORCH_TUNABLEOP_FILENAME"] = base
+        os.environ.setdefault("PYTORCH_TUNABLEOP_VERBOSE", "0")
+
+
 def main():
     args = parse_args()
+    _enable_tunableop()
     import easyparallellibrary_amd as epl
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
