@@ -39,6 +39,13 @@ void epl_scale(void*, int64_t, float, bool, hipStream_t);
 void epl_f32_to_bf16(unsigned short*, const float*, int64_t, hipStream_t);
 void epl_bf16_to_f32(float*, const unsigned short*, int64_t, hipStream_t);
 void epl_sqnorm(const void*, int64_t, float*, bool, hipStream_t);
+void run_mfma_probe(const unsigned short*, const unsigned short*, float*,
+                    hipStream_t);
+void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
+                  int64_t, int64_t, float, bool, hipStream_t);
+void epl_attn_bwd(const void*, const void*, const void*, const void*,
+                  const void*, const float*, float*, void*, void*, void*,
+                  int64_t, int64_t, float, bool, hipStream_t);
 }
 
 namespace {
@@ -227,6 +234,43 @@ void bf16_to_f32(at::Tensor dst, at::Tensor src) {
                   src.numel(), cur_stream());
 }
 
+void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
+              at::Tensor lse, double scale, bool causal) {
+  check(q, at::kBFloat16, "q");
+  check(k, at::kBFloat16, "k");
+  check(v, at::kBFloat16, "v");
+  check(out, at::kBFloat16, "out");
+  check(lse, at::kFloat, "lse");
+  TORCH_CHECK(q.size(-1) == 64, "attn kernels support head_dim == 64");
+  const int64_t seq = q.size(-2);
+  const int64_t bh = q.numel() / (seq * 64);
+  epl_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+               lse.data_ptr<float>(), bh, seq, (float)scale, causal,
+               cur_stream());
+}
+
+void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
+              at::Tensor dout, at::Tensor lse, at::Tensor delta_ws,
+              at::Tensor dq, at::Tensor dk, at::Tensor dv, double scale,
+              bool causal) {
+  const int64_t seq = q.size(-2);
+  const int64_t bh = q.numel() / (seq * 64);
+  check(delta_ws, at::kFloat, "delta_ws");
+  epl_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
+               dout.data_ptr(), lse.data_ptr<float>(),
+               delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
+               dv.data_ptr(), bh, seq, (float)scale, causal, cur_stream());
+}
+
+void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {
+  check(A, at::kBFloat16, "A");
+  check(B, at::kBFloat16, "B");
+  check(D, at::kFloat, "D");
+  run_mfma_probe(reinterpret_cast<const unsigned short*>(A.data_ptr()),
+                 reinterpret_cast<const unsigned short*>(B.data_ptr()),
+                 D.data_ptr<float>(), cur_stream());
+}
+
 void sqnorm(at::Tensor t, at::Tensor out) {
   const bool bf16 = is_bf16(t);
   check(out, at::kFloat, "out");
@@ -252,4 +296,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("f32_to_bf16", &f32_to_bf16);
   m.def("bf16_to_f32", &bf16_to_f32);
   m.def("sqnorm", &sqnorm);
+  m.def("mfma_probe", &mfma_probe);
+  m.def("attn_fwd", &attn_fwd);
+  m.def("attn_bwd", &attn_bwd);
 }

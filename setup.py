@@ -20,6 +20,8 @@ ext = CUDAExtension(
         "csrc/bindings.hip",
         "csrc/comm/rccl_comm.hip",
         "csrc/kernels/kernels.hip",
+        "csrc/kernels/mfma_probe.hip",
+        "csrc/kernels/attention.hip",
     ],
     include_dirs=[os.path.join(ROCM, "include")],
     library_dirs=[os.path.join(ROCM, "lib")],
