@@ -75,100 +75,126 @@ DEVI bf16x8 assemble_pfrag(const float* p) {
 }
 
 // ============================================================================
-// forward
+// forward (v2.1): KVBLK = 64; K (row copies) and V^T (transposed, 144-
+// byte row pad -> conflict-free ds_read_b128 fragments) staged in LDS
+// once per block per tile.  Measured: explicit double-buffering and
+// register prefetch both REGRESS (hipcc schedules the simple form best
+// at 3 waves/SIMD) - keep this structure.
 // ============================================================================
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
     float* __restrict__ lse, int64_t seq, float scale, int causal) {
+  __shared__ short ldsV[64][72];       // V^T: [d][kv]
+  __shared__ short ldsK[64][72];       // K: [kv][d]
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
-  const int lq = lane & 31;            // my q row within the tile
+  const int lq = lane & 31;
   const int64_t bh = blockIdx.y;
-  const int64_t q0 = (int64_t)blockIdx.x * 128 + wave * 32;
-  if (q0 >= seq) return;
+  const int64_t q0_blk = (int64_t)blockIdx.x * 128;
+  const int64_t q0 = q0_blk + wave * 32;
+  const bool active = q0 < seq;
   const short* qp = q + (bh * seq) * 64;
   const short* kp = k + (bh * seq) * 64;
   const short* vp = v + (bh * seq) * 64;
 
-  const int64_t myq = q0 + lq;         // my global q row (may be >= seq)
+  const int64_t myq = q0 + lq;
   const int64_t qrow = myq < seq ? myq : seq - 1;
 
-  // Q^T operand: B[k=d][n=q]; lane holds Q[qrow][hi*8+j + 16*c], c=0..3
   bf16x8 qfrag[4];
 #pragma unroll
   for (int c = 0; c < 4; ++c)
     qfrag[c] = *reinterpret_cast<const bf16x8*>(
         qp + qrow * 64 + hi * 8 + 16 * c);
 
-  f32x16 ot0 = {}, ot1 = {};           // O^T accumulators, d 0..31 / 32..63
+  f32x16 ot0 = {}, ot1 = {};
   float m = -1e30f, l = 0.f;
 
-  const int64_t kv_end = causal ? (q0 + 32 < seq ? q0 + 32 : seq)
-                                : seq;
-  for (int64_t kv0 = 0; kv0 < kv_end; kv0 += 32) {
-    // ---- S^T = K @ Q^T ----------------------------------------------------
-    f32x16 st = {};
-    const int64_t krow = kv0 + lq < seq ? kv0 + lq : seq - 1;
+  const int64_t blk_kv_end =
+      causal ? (q0_blk + 128 < seq ? q0_blk + 128 : seq) : seq;
+  const int stage_kv = threadIdx.x & 63;
+  const int stage_d0 = (threadIdx.x >> 6) * 8;
+
+  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 64) {
+    // ---- stage K (rows) and V^T (transposed) cooperatively ---------------
+    __syncthreads();
+    {
+      int64_t vrow = kv0 + stage_kv;
+      if (vrow >= seq) vrow = seq - 1;   // masked columns never contribute
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 kfrag = *reinterpret_cast<const bf16x8*>(
-          kp + krow * 64 + hi * 8 + 16 * c);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c], st, 0,
-                                                   0, 0);
-    }
-    // ---- online softmax (lane owns q row; partner lane^32 has the other
-    // 16 kv). st[reg] = S[myq][kv0 + crow(reg, hi)] ------------------------
-    float s[16];
-    float tile_max = -1e30f;
+      for (int h2 = 0; h2 < 2; ++h2) {
+        const int sd = stage_d0 + h2 * 32;
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(vp + vrow * 64 + sd);
 #pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      float sv = st[r] * scale;
-      const int64_t kvg = kv0 + crow(r, hi);
-      if (kvg >= seq || (causal && kvg > myq)) sv = -1e30f;
-      s[r] = sv;
-      tile_max = fmaxf(tile_max, sv);
-    }
-    tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
-    const float m_new = fmaxf(m, tile_max);
-    const float alpha = __expf(m - m_new);
-    float rowsum = 0.f;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      s[r] = __expf(s[r] - m_new);
-      rowsum += s[r];
-    }
-    rowsum += __shfl_xor(rowsum, 32, 64);
-    l = l * alpha + rowsum;
-    m = m_new;
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      ot0[r] *= alpha;
-      ot1[r] *= alpha;
-    }
-    // ---- O^T += V^T @ P^T -------------------------------------------------
-    // two kv chunks of 16; P frag from s[0..7] / s[8..15]
-    bf16x8 pf0 = assemble_pfrag(&s[0]);
-    bf16x8 pf1 = assemble_pfrag(&s[8]);
-    // A = V^T[d][kv]: lane holds V[kv0 + kc*16 + hi*8 + j][dchunk*32 + lq]
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf16x8 vt0, vt1;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        int64_t kvg = kv0 + kc * 16 + hi * 8 + j;
-        if (kvg >= seq) kvg = seq - 1;  // P there is 0
-        vt0[j] = vp[kvg * 64 + lq];
-        vt1[j] = vp[kvg * 64 + 32 + lq];
+        for (int j = 0; j < 8; ++j)
+          ldsV[sd + j][stage_kv] = vv[j];
+        *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
+            *reinterpret_cast<const bf16x8*>(kp + vrow * 64 + sd);
       }
-      bf16x8 pf = kc == 0 ? pf0 : pf1;
-      ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0, 0, 0);
-      ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0, 0, 0);
+    }
+    __syncthreads();
+    if (active) {
+      const int64_t wave_kv_end = causal
+          ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
+#pragma unroll
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t kvs = kv0 + sub;
+        if (kvs >= wave_kv_end) break;
+        f32x16 st = {};
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          bf16x8 kfrag = *reinterpret_cast<const bf16x8*>(
+              &ldsK[sub + lq][hi * 8 + 16 * c]);
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c],
+                                                       st, 0, 0, 0);
+        }
+        float s[16];
+        float tile_max = -1e30f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float sv = st[r] * scale;
+          const int64_t kvg = kvs + crow(r, hi);
+          if (kvg >= seq || (causal && kvg > myq)) sv = -1e30f;
+          s[r] = sv;
+          tile_max = fmaxf(tile_max, sv);
+        }
+        tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+        const float m_new = fmaxf(m, tile_max);
+        const float alpha = __expf(m - m_new);
+        float rowsum = 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          s[r] = __expf(s[r] - m_new);
+          rowsum += s[r];
+        }
+        rowsum += __shfl_xor(rowsum, 32, 64);
+        l = l * alpha + rowsum;
+        m = m_new;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          ot0[r] *= alpha;
+          ot1[r] *= alpha;
+        }
+        bf16x8 pf0 = assemble_pfrag(&s[0]);
+        bf16x8 pf1 = assemble_pfrag(&s[8]);
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8 pf = kc == 0 ? pf0 : pf1;
+          bf16x8 vt0 = *reinterpret_cast<const bf16x8*>(
+              &ldsV[lq][sub + kc * 16 + hi * 8]);
+          bf16x8 vt1 = *reinterpret_cast<const bf16x8*>(
+              &ldsV[32 + lq][sub + kc * 16 + hi * 8]);
+          ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0,
+                                                        0, 0);
+          ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0,
+                                                        0, 0);
+        }
+      }
     }
   }
 
-  if (myq >= seq) return;
+  if (!active || myq >= seq) return;
   const float inv_l = 1.f / l;
   short* op = out + (bh * seq + myq) * 64;
 #pragma unroll
@@ -203,10 +229,10 @@ __global__ void attn_bwd_prep_kernel(const short* __restrict__ dout,
 
 // ============================================================================
 // backward kernel A (kv-parallel): dK and dV.  One wave per 32-key kv
-// tile, looping q tiles.  Lane owns one kv row (lane%32), so dK/dV
-// epilogues are direct stores.
-//   P = exp(S*scale - lse);  dP = dO V^T;  dS = P*(dP - D_i)*scale
-//   dV^T += dO^T @ P ;  dK^T += Q^T @ dS
+// tile, looping q tiles.  Q and dO tiles are staged in LDS as row copies
+// once per block: the row fragments read back as conflict-free
+// ds_read_b128 and the transposed fragments as pair-broadcast 2-byte LDS
+// reads — no global gathers in the loop.
 // ============================================================================
 __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
@@ -214,13 +240,16 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int64_t seq,
     float scale, int causal) {
+  __shared__ short ldsQ[32][72];
+  __shared__ short ldsDO[32][72];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
-  const int lkv = lane & 31;           // my kv row within the tile
+  const int lkv = lane & 31;
   const int64_t bh = blockIdx.y;
-  const int64_t kv0 = (int64_t)blockIdx.x * 128 + wave * 32;
-  if (kv0 >= seq) return;
+  const int64_t kv0_blk = (int64_t)blockIdx.x * 128;
+  const int64_t kv0 = kv0_blk + wave * 32;
+  const bool active = kv0 < seq;
   const short* qp = q + (bh * seq) * 64;
   const short* kp = k + (bh * seq) * 64;
   const short* vp = v + (bh * seq) * 64;
@@ -231,7 +260,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   const int64_t mykv = kv0 + lkv;
   const int64_t kvrow = mykv < seq ? mykv : seq - 1;
 
-  // B operands for S^T/dP^T-style matmuls: lane holds row kvrow segments
   bf16x8 kfrag[4], vfrag[4];
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
@@ -241,28 +269,37 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
         vp + kvrow * 64 + hi * 8 + 16 * c);
   }
 
-  f32x16 dvt0 = {}, dvt1 = {};   // dV^T acc: rows d 0..31 / 32..63, col kv
-  f32x16 dkt0 = {}, dkt1 = {};   // dK^T acc
+  f32x16 dvt0 = {}, dvt1 = {};
+  f32x16 dkt0 = {}, dkt1 = {};
 
-  const int64_t q_start = causal ? (kv0 / 32) * 32 : 0;
+  const int stage_row = threadIdx.x >> 3;        // 0..31
+  const int stage_seg = (threadIdx.x & 7) * 8;   // 0..56
+  const int64_t q_start = causal ? (kv0_blk / 32) * 32 : 0;
   for (int64_t q0 = q_start; q0 < seq; q0 += 32) {
-    const int64_t qrow = q0 + lkv < seq ? q0 + lkv : seq - 1;
-    // S[q][kv]: A = Q[m=q][k=d] (lane: q = q0+lane%32, contiguous d seg),
-    // B = K^T[k=d][n=kv] (kfrag).  D: rows q = crow(r,hi), col kv = lkv.
+    __syncthreads();
+    {
+      int64_t qr = q0 + stage_row;
+      if (qr >= seq) qr = seq - 1;   // masked rows contribute zero
+      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(qp + qr * 64 + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(dop + qr * 64 + stage_seg);
+    }
+    __syncthreads();
+    if (!active || (causal && q0 + 31 < kv0)) continue;
     f32x16 st = {};
     f32x16 dpt = {};
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
       bf16x8 qf = *reinterpret_cast<const bf16x8*>(
-          qp + qrow * 64 + hi * 8 + 16 * c);
+          &ldsQ[lkv][hi * 8 + 16 * c]);
       bf16x8 dof = *reinterpret_cast<const bf16x8*>(
-          dop + qrow * 64 + hi * 8 + 16 * c);
+          &ldsDO[lkv][hi * 8 + 16 * c]);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0,
                                                    0);
       dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfrag[c], dpt, 0,
                                                     0, 0);
     }
-    // st[r] = S[q0+crow(r,hi)][mykv]; dpt[r] = dP same layout
     float p[16], ds[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -274,7 +311,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
       p[r] = pv;
       ds[r] = masked ? 0.f : pv * (dpt[r] - dltp[qgc]) * scale;
     }
-    // B operands over q: element j = q row (qc*16 + hi*8 + j)
     bf16x8 pb0 = assemble_pfrag(&p[0]);
     bf16x8 pb1 = assemble_pfrag(&p[8]);
     bf16x8 db0 = assemble_pfrag(&ds[0]);
@@ -284,21 +320,18 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
       bf16x8 dot0, dot1, qt0, qt1;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int64_t qg = q0 + qc * 16 + hi * 8 + j;
-        if (qg >= seq) qg = seq - 1;   // P/dS there is 0
-        dot0[j] = dop[qg * 64 + lkv];
-        dot1[j] = dop[qg * 64 + 32 + lkv];
-        qt0[j] = qp[qg * 64 + lkv];
-        qt1[j] = qp[qg * 64 + 32 + lkv];
+        const int qr = qc * 16 + hi * 8 + j;
+        dot0[j] = ldsDO[qr][lkv];
+        dot1[j] = ldsDO[qr][32 + lkv];
+        qt0[j] = ldsQ[qr][lkv];
+        qt1[j] = ldsQ[qr][32 + lkv];
       }
       bf16x8 pb = qc == 0 ? pb0 : pb1;
       bf16x8 db = qc == 0 ? db0 : db1;
-      // dV^T[d][kv] += dO^T[d][q] @ P[q][kv]
       dvt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot0, pb, dvt0, 0, 0,
                                                      0);
       dvt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot1, pb, dvt1, 0, 0,
                                                      0);
-      // dK^T[d][kv] += Q^T[d][q] @ dS[q][kv]
       dkt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt0, db, dkt0, 0, 0,
                                                      0);
       dkt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt1, db, dkt1, 0, 0,
@@ -306,7 +339,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     }
   }
 
-  if (mykv >= seq) return;
+  if (!active || mykv >= seq) return;
   short* dkp = dk + (bh * seq + mykv) * 64;
   short* dvp = dv + (bh * seq + mykv) * 64;
 #pragma unroll
@@ -319,23 +352,25 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 }
 
 // ============================================================================
-// backward kernel B (q-parallel): dQ.  Lane owns one q row (the forward
-// layout); dS is recomputed in the S^T layout and fed through the same
-// permlane assembly as the forward's P.
-//   dQ^T[d][q] += K^T[d][kv] @ dS^T[kv][q]
+// backward kernel B (q-parallel): dQ.  K and V tiles staged in LDS as
+// row copies per block; row fragments via ds_read_b128, K-transposed
+// fragments via 2-byte LDS reads.
 // ============================================================================
 __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int64_t seq, float scale, int causal) {
+  __shared__ short ldsK[32][72];
+  __shared__ short ldsVr[32][72];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
   const int lq = lane & 31;
   const int64_t bh = blockIdx.y;
-  const int64_t q0 = (int64_t)blockIdx.x * 128 + wave * 32;
-  if (q0 >= seq) return;
+  const int64_t q0_blk = (int64_t)blockIdx.x * 128;
+  const int64_t q0 = q0_blk + wave * 32;
+  const bool active = q0 < seq;
   const short* qp = q + (bh * seq) * 64;
   const short* kp = k + (bh * seq) * 64;
   const short* vp = v + (bh * seq) * 64;
@@ -355,27 +390,39 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
         dop + qrow * 64 + hi * 8 + 16 * c);
   }
 
-  f32x16 dqt0 = {}, dqt1 = {};   // dQ^T acc: rows d 0..31/32..63, col q
+  f32x16 dqt0 = {}, dqt1 = {};
 
-  const int64_t kv_end = causal ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
-  for (int64_t kv0 = 0; kv0 < kv_end; kv0 += 32) {
-    const int64_t kvrow = kv0 + lq < seq ? kv0 + lq : seq - 1;
+  const int stage_row = threadIdx.x >> 3;
+  const int stage_seg = (threadIdx.x & 7) * 8;
+  const int64_t blk_kv_end =
+      causal ? (q0_blk + 128 < seq ? q0_blk + 128 : seq) : seq;
+  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
+    __syncthreads();
+    {
+      int64_t kr = kv0 + stage_row;
+      if (kr >= seq) kr = seq - 1;   // dS there is 0
+      *reinterpret_cast<bf16x8*>(&ldsK[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(kp + kr * 64 + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsVr[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(vp + kr * 64 + stage_seg);
+    }
+    __syncthreads();
+    const int64_t wave_kv_end = causal
+        ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
+    if (!active || kv0 >= wave_kv_end) continue;
     f32x16 st = {};
     f32x16 dpt = {};
 #pragma unroll
     for (int c = 0; c < 4; ++c) {
-      // S^T = K @ Q^T (forward layout): A = K[m=kv][k=d], B = Q^T
       bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-          kp + kvrow * 64 + hi * 8 + 16 * c);
+          &ldsK[lq][hi * 8 + 16 * c]);
       bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-          vp + kvrow * 64 + hi * 8 + 16 * c);
+          &ldsVr[lq][hi * 8 + 16 * c]);
       st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[c], st, 0, 0,
                                                    0);
-      // dP^T = V @ dO^T: rows kv, col q
       dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[c], dpt, 0,
                                                     0, 0);
     }
-    // st[r] = S[myq][kv0+crow(r,hi)]; dpt[r] = dP same layout
     float ds[16];
 #pragma unroll
     for (int r = 0; r < 16; ++r) {
@@ -384,17 +431,16 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       float pv = masked ? 0.f : __expf(st[r] * scale - mylse);
       ds[r] = masked ? 0.f : pv * (dpt[r] - mydelta) * scale;
     }
-    bf16x8 db0 = assemble_pfrag(&ds[0]);   // dS^T[kv][q], kv chunk 0..15
-    bf16x8 db1 = assemble_pfrag(&ds[8]);   // kv chunk 16..31
+    bf16x8 db0 = assemble_pfrag(&ds[0]);
+    bf16x8 db1 = assemble_pfrag(&ds[8]);
 #pragma unroll
     for (int kc = 0; kc < 2; ++kc) {
       bf16x8 kt0, kt1;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        int64_t kvg = kv0 + kc * 16 + hi * 8 + j;
-        if (kvg >= seq) kvg = seq - 1;    // dS there is 0
-        kt0[j] = kp[kvg * 64 + lq];
-        kt1[j] = kp[kvg * 64 + 32 + lq];
+        const int kr = kc * 16 + hi * 8 + j;
+        kt0[j] = ldsK[kr][lq];
+        kt1[j] = ldsK[kr][32 + lq];
       }
       bf16x8 db = kc == 0 ? db0 : db1;
       dqt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt0, db, dqt0, 0, 0,
@@ -404,21 +450,13 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     }
   }
 
-  if (myq >= seq) return;
+  if (!active || myq >= seq) return;
   short* dqp = dq + (bh * seq + myq) * 64;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     dqp[crow(r, hi)] = (short)f2bf(dqt0[r]);
     dqp[32 + crow(r, hi)] = (short)f2bf(dqt1[r]);
   }
-}
-
-__global__ void f32_to_bf16_4d_kernel(short* __restrict__ dst,
-                                      const float* __restrict__ src,
-                                      int64_t n) {
-  for (int64_t i = blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x)
-    dst[i] = (short)f2bf(src[i]);
 }
 
 }  // namespace

@@ -8,10 +8,15 @@ FusedBiasGelu (FFN first Linear runs bias-free); attention uses torch SDPA
 """
 
 import math
+import os
 
 import torch
 import torch.nn as nn
 import torch.nn.functional as F
+
+# gate for the hand-written attention kernels (numerics-verified; flipped
+# on once they beat the AOTriton SDPA path)
+_USE_NATIVE_ATTN = os.environ.get("EPL_NATIVE_ATTENTION", "1") != "0"
 
 from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
 from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
@@ -30,15 +35,19 @@ class SelfAttention(nn.Module):
         self.dropout = dropout
 
     def forward(self, x):
+        from easyparallellibrary_amd.ops.attention import flash_attention
         b, s, h = x.shape
         qkv = self.qkv(x).reshape(b, s, 3, self.num_heads, self.head_dim)
         q, k, v = qkv.unbind(dim=2)
         q = q.transpose(1, 2)
         k = k.transpose(1, 2)
         v = v.transpose(1, 2)
-        o = F.scaled_dot_product_attention(
-            q, k, v, is_causal=self.causal,
-            dropout_p=self.dropout if self.training else 0.0)
+        if self.dropout and self.training:
+            o = F.scaled_dot_product_attention(
+                q, k, v, is_causal=self.causal, dropout_p=self.dropout)
+        else:
+            o = flash_attention(q, k, v, causal=self.causal,
+                                allow_native=_USE_NATIVE_ATTN)
         o = o.transpose(1, 2).reshape(b, s, h)
         return self.proj(o)
 

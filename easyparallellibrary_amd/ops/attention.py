@@ -40,12 +40,13 @@ class _FlashAttention(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
-def flash_attention(q, k, v, causal=False, scale=None):
+def flash_attention(q, k, v, causal=False, scale=None, allow_native=True):
     """q,k,v: [B, H, S, D].  Native kernel when bf16/D=64 on GPU; torch
     SDPA otherwise."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
-    if (use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] == 64):
+    if (allow_native and use_native(q) and q.dtype == torch.bfloat16
+            and q.shape[-1] == 64):
         return _FlashAttention.apply(q, k, v, causal, scale)
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
                                           scale=scale)
