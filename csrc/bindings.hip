@@ -24,8 +24,9 @@ void epl_layer_norm_fwd(void*, const void*, const void*, void*, const void*,
                         const void*, float*, float*, int64_t, int64_t, float,
                         bool, hipStream_t);
 void epl_layer_norm_bwd(void*, float*, float*, const void*, const void*,
-                        const void*, const float*, const float*, int64_t,
-                        int64_t, bool, hipStream_t);
+                        const void*, const float*, const float*, float*,
+                        float*, int64_t, int64_t, int64_t, bool,
+                        hipStream_t);
 void epl_bias_gelu_fwd(void*, const void*, const void*, int64_t, int64_t,
                        bool, hipStream_t);
 void epl_bias_gelu_bwd(void*, float*, const void*, const void*, const void*,
@@ -159,10 +160,23 @@ void layer_norm_bwd(at::Tensor dx, at::Tensor dgamma, at::Tensor dbeta,
               "LayerNorm backward supports cols <= 16384");
   check(dgamma, at::kFloat, "dgamma");
   check(dbeta, at::kFloat, "dbeta");
+  float* dgp = nullptr;
+  float* dbp = nullptr;
+  at::Tensor ws;
+  const bool fast = bf16 && cols % 8 == 0 && cols <= 2048;
+  int64_t nparts = 0;
+  if (fast) {
+    const int64_t wave_rows = (rows + 3) / 4;
+    nparts = (wave_rows < 1024 ? wave_rows : 1024) * 4;
+    ws = at::empty({2, nparts, cols}, dgamma.options());
+    dgp = ws.data_ptr<float>();
+    dbp = dgp + nparts * cols;
+  }
   epl_layer_norm_bwd(dx.data_ptr(), dgamma.data_ptr<float>(),
                      dbeta.data_ptr<float>(), dy.data_ptr(), x.data_ptr(),
                      gamma.data_ptr(), mean.data_ptr<float>(),
-                     rstd.data_ptr<float>(), rows, cols, bf16, cur_stream());
+                     rstd.data_ptr<float>(), dgp, dbp, nparts, rows, cols,
+                     bf16, cur_stream());
 }
 
 void bias_gelu_fwd(at::Tensor out, at::Tensor x, at::Tensor bias) {

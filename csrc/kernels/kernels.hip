@@ -369,22 +369,26 @@ __global__ void layer_norm_fwd_kernel(
   }
 }
 
-// bf16 fast backward: register row cache + register dgamma/dbeta
-// accumulators (flushed once per block with atomicAdd) — zero LDS traffic
-// beyond the block reduction; HBM reads dy/x once, writes dx once.
-template <int CHUNKS>
-__global__ void layer_norm_bwd_bf16_kernel(
-    void* __restrict__ dx, float* __restrict__ dgamma,
-    float* __restrict__ dbeta, const void* __restrict__ dy,
+// bf16 fast backward: ONE ROW PER WAVE — no barriers at all.  Each wave
+// reduces its row with shuffles, keeps the row image in registers between
+// the reduction and the dx pass (HBM read-once/write-once), accumulates
+// dgamma/dbeta in registers across its rows and flushes once with
+// atomics.  CHUNKS = ceil(cols/8/64); fast path cols <= 1024 (CHUNKS<=2).
+template <int CHUNKS, bool CACHE>
+__global__ __launch_bounds__(256, 2) void layer_norm_bwd_bf16_kernel(
+    void* __restrict__ dx, float* __restrict__ dg_part,
+    float* __restrict__ db_part, const void* __restrict__ dy,
     const void* __restrict__ x, const void* __restrict__ gamma,
     const float* __restrict__ mean_in, const float* __restrict__ rstd_in,
     int64_t rows, int64_t cols) {
-  __shared__ float lds[8];
   using T = unsigned short;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;               // wave in block (0..3)
+  const int wpb = blockDim.x >> 6;                // waves per block
   float8 gw[CHUNKS], acc_dg[CHUNKS], acc_db[CHUNKS];
 #pragma unroll
   for (int k = 0; k < CHUNKS; ++k) {
-    const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+    const int64_t c = ((int64_t)lane + k * 64) * 8;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       acc_dg[k].v[j] = 0.f;
@@ -393,15 +397,16 @@ __global__ void layer_norm_bwd_bf16_kernel(
     if (c < cols)
       gw[k] = load_bf16x8(reinterpret_cast<const T*>(gamma) + c);
   }
-  for (int64_t row = blockIdx.x; row < rows; row += gridDim.x) {
+  for (int64_t row = (int64_t)blockIdx.x * wpb + wid; row < rows;
+       row += (int64_t)gridDim.x * wpb) {
     const int64_t base = row * cols;
     const float mean = mean_in[row];
     const float rstd = rstd_in[row];
     float sum_dyg = 0.f, sum_dygx = 0.f;
-    float8 c_xhat[CHUNKS], c_dyg[CHUNKS], c_dy[CHUNKS];
+    float8 c_xhat[CACHE ? CHUNKS : 1], c_dy[CACHE ? CHUNKS : 1];
 #pragma unroll
     for (int k = 0; k < CHUNKS; ++k) {
-      const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+      const int64_t c = ((int64_t)lane + k * 64) * 8;
       if (c >= cols) continue;
       float8 d = load_bf16x8(reinterpret_cast<const T*>(dy) + base + c);
       float8 f = load_bf16x8(reinterpret_cast<const T*>(x) + base + c);
@@ -409,44 +414,79 @@ __global__ void layer_norm_bwd_bf16_kernel(
       for (int j = 0; j < 8; ++j) {
         const float xhat = (f.v[j] - mean) * rstd;
         const float dyg = d.v[j] * gw[k].v[j];
-        c_xhat[k].v[j] = xhat;
-        c_dyg[k].v[j] = dyg;
-        c_dy[k].v[j] = d.v[j];
+        if (CACHE) {
+          c_xhat[k].v[j] = xhat;
+          c_dy[k].v[j] = d.v[j];
+        }
         sum_dyg += dyg;
         sum_dygx += dyg * xhat;
       }
     }
-    __syncthreads();
-    sum_dyg = block_reduce_sum(sum_dyg, lds);
-    __syncthreads();
-    sum_dygx = block_reduce_sum(sum_dygx, lds);
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      sum_dyg += __shfl_down(sum_dyg, off, 64);
+      sum_dygx += __shfl_down(sum_dygx, off, 64);
+    }
+    sum_dyg = __shfl(sum_dyg, 0, 64);
+    sum_dygx = __shfl(sum_dygx, 0, 64);
     const float inv_cols = 1.f / cols;
 #pragma unroll
     for (int k = 0; k < CHUNKS; ++k) {
-      const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+      const int64_t c = ((int64_t)lane + k * 64) * 8;
       if (c >= cols) continue;
+      float8 dv, xh;
+      if (CACHE) {
+        dv = c_dy[k];
+        xh = c_xhat[k];
+      } else {
+        // rows in flight fit L2: the re-reads hit cache
+        float8 d = load_bf16x8(reinterpret_cast<const T*>(dy) + base + c);
+        float8 f = load_bf16x8(reinterpret_cast<const T*>(x) + base + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          dv.v[j] = d.v[j];
+          xh.v[j] = (f.v[j] - mean) * rstd;
+        }
+      }
       float8 o;
 #pragma unroll
       for (int j = 0; j < 8; ++j) {
-        o.v[j] = (c_dyg[k].v[j] -
-                  (sum_dyg + c_xhat[k].v[j] * sum_dygx) * inv_cols) * rstd;
-        acc_dg[k].v[j] += c_dy[k].v[j] * c_xhat[k].v[j];
-        acc_db[k].v[j] += c_dy[k].v[j];
+        const float dyg = dv.v[j] * gw[k].v[j];
+        o.v[j] = (dyg - (sum_dyg + xh.v[j] * sum_dygx) * inv_cols) * rstd;
+        acc_dg[k].v[j] += dv.v[j] * xh.v[j];
+        acc_db[k].v[j] += dv.v[j];
       }
       store_bf16x8(reinterpret_cast<T*>(dx) + base + c, o);
     }
-    __syncthreads();
   }
+  // one partial row per wave — no atomics, a tiny second kernel reduces
+  const int64_t prow = ((int64_t)blockIdx.x * wpb + wid) * cols;
 #pragma unroll
   for (int k = 0; k < CHUNKS; ++k) {
-    const int64_t c = ((int64_t)threadIdx.x + k * blockDim.x) * 8;
+    const int64_t c = ((int64_t)lane + k * 64) * 8;
     if (c >= cols) continue;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      atomicAdd(&dgamma[c + j], acc_dg[k].v[j]);
-      atomicAdd(&dbeta[c + j], acc_db[k].v[j]);
-    }
+    store_f32x8(dg_part + prow + c, acc_dg[k]);
+    store_f32x8(db_part + prow + c, acc_db[k]);
   }
+}
+
+// reduce the [nparts, cols] partial buffers into dgamma/dbeta.
+// grid = (col_chunks, PSPLIT): each block sums a strided slice of the
+// partial rows for its 256-column range (coalesced row-major reads) and
+// contributes one atomicAdd per column (PSPLIT per column total).
+__global__ void ln_bwd_partial_reduce_kernel(
+    float* __restrict__ dgamma, float* __restrict__ dbeta,
+    const float* __restrict__ dg_part, const float* __restrict__ db_part,
+    int64_t nparts, int64_t cols) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  float sg = 0.f, sb = 0.f;
+  for (int64_t p = blockIdx.y; p < nparts; p += gridDim.y) {
+    sg += dg_part[p * cols + c];
+    sb += db_part[p * cols + c];
+  }
+  atomicAdd(&dgamma[c], sg);
+  atomicAdd(&dbeta[c], sb);
 }
 
 // generic (fp32 / large-cols) backward: LDS partials + global re-read
@@ -931,22 +971,31 @@ void epl_layer_norm_fwd(void* out, const void* x, const void* res,
 
 void epl_layer_norm_bwd(void* dx, float* dgamma, float* dbeta, const void* dy,
                         const void* x, const void* gamma, const float* mean,
-                        const float* rstd, int64_t rows, int64_t cols,
+                        const float* rstd, float* dg_part, float* db_part,
+                        int64_t nparts, int64_t rows, int64_t cols,
                         bool bf16, hipStream_t stream) {
-  int grid = (int)(rows < 1024 ? rows : 1024);
-  if (bf16 && cols % 8 == 0 && cols <= 2 * 256 * 8) {
-    const int block = ln_block_for(cols);
-    const bool two = cols > (int64_t)block * 8;
-    if (two)
-      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<2>), dim3(grid),
-                         dim3(block), 0, stream, dx, dgamma, dbeta, dy, x,
+  if (bf16 && cols % 8 == 0 && cols <= 2048 && dg_part != nullptr) {
+    const int64_t wave_rows = (rows + 3) / 4;
+    const int grid = (int)(wave_rows < 1024 ? wave_rows : 1024);
+    if (cols <= 512)
+      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<1, true>), dim3(grid),
+                         dim3(256), 0, stream, dx, dg_part, db_part, dy, x,
+                         gamma, mean, rstd, rows, cols);
+    else if (cols <= 1024)
+      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<2, true>), dim3(grid),
+                         dim3(256), 0, stream, dx, dg_part, db_part, dy, x,
                          gamma, mean, rstd, rows, cols);
     else
-      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<1>), dim3(grid),
-                         dim3(block), 0, stream, dx, dgamma, dbeta, dy, x,
-                         gamma, mean, rstd, rows, cols);
+      hipLaunchKernelGGL((layer_norm_bwd_bf16_kernel<4, false>),
+                         dim3(grid), dim3(256), 0, stream, dx, dg_part,
+                         db_part, dy, x, gamma, mean, rstd, rows, cols);
+    hipLaunchKernelGGL(ln_bwd_partial_reduce_kernel,
+                       dim3((unsigned)((cols + 255) / 256), 64), dim3(256),
+                       0, stream, dgamma, dbeta, dg_part, db_part,
+                       (int64_t)grid * 4, cols);
     return;
   }
+  int grid = (int)(rows < 1024 ? rows : 1024);
   const size_t lds_bytes = (size_t)cols * 2 * sizeof(float);
   if (bf16)
     hipLaunchKernelGGL((layer_norm_bwd_kernel<true>), dim3(grid),
