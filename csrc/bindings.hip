@@ -30,7 +30,8 @@ void epl_layer_norm_bwd(void*, float*, float*, const void*, const void*,
 void epl_bias_gelu_fwd(void*, const void*, const void*, int64_t, int64_t,
                        bool, hipStream_t);
 void epl_bias_gelu_bwd(void*, float*, const void*, const void*, const void*,
-                       int64_t, int64_t, bool, hipStream_t);
+                       float*, int64_t, int64_t, int64_t, bool,
+                       hipStream_t);
 void epl_ce_rowstats(const void*, const int64_t*, float*, float*, float*,
                      int64_t, int64_t, int64_t, int64_t, bool, hipStream_t);
 void epl_ce_backward(void*, const void*, const int64_t*, const float*,
@@ -197,9 +198,20 @@ void bias_gelu_bwd(at::Tensor dx, at::Tensor dbias, at::Tensor dy,
   TORCH_CHECK((size_t)cols * sizeof(float) <= 128 * 1024,
               "bias_gelu backward supports cols <= 32768");
   check(dbias, at::kFloat, "dbias");
+  float* dbp = nullptr;
+  int64_t nwaves = 0;
+  at::Tensor ws;
+  if (bf16 && cols % 512 == 0) {
+    const int64_t segs = cols / 512;
+    int64_t per_seg = rows < 1024 ? rows : 1024;
+    per_seg = (per_seg + 3) / 4 * 4;   // keep nwaves % (4*segs) clean
+    nwaves = segs * per_seg;
+    ws = at::empty({nwaves, 512}, dbias.options());
+    dbp = ws.data_ptr<float>();
+  }
   epl_bias_gelu_bwd(dx.data_ptr(), dbias.data_ptr<float>(), dy.data_ptr(),
-                    x.data_ptr(), bias.data_ptr(), rows, cols, bf16,
-                    cur_stream());
+                    x.data_ptr(), bias.data_ptr(), dbp, nwaves, rows, cols,
+                    bf16, cur_stream());
 }
 
 void ce_rowstats(at::Tensor logits, at::Tensor targets, at::Tensor row_max,

@@ -44,10 +44,13 @@ def test_fused_adamw_matches_torch():
     assert torch.allclose(pbf.float(), ref_p, atol=1e-2, rtol=1e-2)
 
 
-@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
-def test_layer_norm_fwd_bwd(dtype):
+@pytest.mark.parametrize("dtype,cols", [
+    (torch.bfloat16, 1024), (torch.float32, 1024),
+    (torch.bfloat16, 512), (torch.bfloat16, 1600),
+    (torch.bfloat16, 2048), (torch.bfloat16, 4096)])
+def test_layer_norm_fwd_bwd(dtype, cols):
     torch.manual_seed(1)
-    rows, cols = 512, 1024
+    rows = 512
     x = torch.randn(rows, cols, device=dev(), dtype=dtype)
     gamma = torch.randn(cols, device=dev(), dtype=dtype)
     beta = torch.randn(cols, device=dev(), dtype=dtype)
