@@ -43,6 +43,29 @@ def set_default_strategy(strategy):
     Env.get().strategy_context.set_default_strategy(strategy)
 
 
+class GraphKeys:
+    """Merged-output collection keys (reference: epl/ir/graph.py:40-65).
+    Values collected under GLOBAL_* keys are merged across every rank by
+    Engine.merged_collections(): MEAN -> all-reduce mean, SUM -> sum,
+    CONCAT -> all-gather."""
+    GLOBAL_MEAN_OBJECTS = "global_mean_objects"
+    GLOBAL_SUM_OBJECTS = "global_sum_objects"
+    GLOBAL_CONCAT_OBJECTS = "global_concat_objects"
+    LOCAL_MEAN_OBJECTS = "local_mean_objects"
+    LOCAL_SUM_OBJECTS = "local_sum_objects"
+    LOCAL_CONCAT_OBJECTS = "local_concat_objects"
+
+
+def add_to_collection(value, name):
+    """reference: epl.add_to_collection (epl/__init__.py:23-55)"""
+    Env.get().add_to_collection(value, name)
+
+
+def get_collection(name):
+    """reference: epl.get_collection"""
+    return Env.get().get_collection(name)
+
+
 def __getattr__(name):
     # lazy to avoid import cycles (engine imports config/env/strategies)
     if name == "Engine":
@@ -53,4 +76,5 @@ def __getattr__(name):
 __all__ = [
     "init", "set_default_strategy", "replicate", "split", "Replicate",
     "Split", "Config", "Cluster", "VirtualDevice", "Env", "Engine",
+    "GraphKeys", "add_to_collection", "get_collection",
 ]

@@ -27,6 +27,7 @@ class Env:
         self.strategy_context = StrategyContext()
         self._pg_initialized_here = False
         self._hooks_installed = False
+        self._collections = {}
 
     @classmethod
     def get(cls):
@@ -38,6 +39,15 @@ class Env:
         self.config = Config()
         self.cluster = None
         self.strategy_context.reset()
+        self._collections = {}
+
+    # ---- collections (reference: epl add_to_collection / get_collection,
+    # GraphKeys GLOBAL_/LOCAL_ CONCAT/MEAN/SUM — ir/graph.py:40-65) --------
+    def add_to_collection(self, value, name):
+        self._collections.setdefault(name, []).append(value)
+
+    def get_collection(self, name):
+        return list(self._collections.get(name, []))
 
     def init(self, config=None):
         if config is None:

@@ -22,3 +22,20 @@ def split_to_replica(x, comm):
         return x
     n = x.shape[0] // comm.size
     return x[comm.rank * n:(comm.rank + 1) * n]
+
+
+def replica_to_replica(x, comm):
+    """Bridge between two replicate scopes on the same devices: identity
+    (reference stub, bridging_layer.py Replica2Replica)."""
+    return x
+
+
+def split_to_split(x, src_comm, dst_comm):
+    """Bridge between two split scopes; equal degrees pass through
+    (reference stub, bridging_layer.py Split2Split)."""
+    if (src_comm is None) != (dst_comm is None):
+        raise ValueError("split_to_split needs both communicators")
+    if src_comm is not None and src_comm.size != dst_comm.size:
+        raise NotImplementedError(
+            "re-sharding between different split degrees")
+    return x

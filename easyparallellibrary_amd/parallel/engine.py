@@ -374,6 +374,38 @@ class Engine:
             t = t / self.world_size
         return t
 
+    def merged_collections(self):
+        """Merge the GLOBAL_* collections across ranks (reference:
+        parallel/parallel.py merge_outputs :233-353: allgather/allreduce of
+        user-collected losses/metrics)."""
+        import easyparallellibrary_amd as epl
+        out = {}
+        env = self.env
+        for name, op in ((epl.GraphKeys.GLOBAL_MEAN_OBJECTS, "mean"),
+                         (epl.GraphKeys.GLOBAL_SUM_OBJECTS, "sum")):
+            vals = env.get_collection(name)
+            if vals:
+                out[name] = [self.all_reduce_metric(v, op=op) for v in vals]
+        concat = env.get_collection(epl.GraphKeys.GLOBAL_CONCAT_OBJECTS)
+        if concat:
+            from easyparallellibrary_amd.comm import functional
+            from easyparallellibrary_amd.comm.backend import (
+                create_communicator)
+            comm = create_communicator(
+                "{}_concat".format(self._ns), list(range(self.world_size)))
+            out[epl.GraphKeys.GLOBAL_CONCAT_OBJECTS] = [
+                functional.all_gather(
+                    (v if torch.is_tensor(v) else torch.tensor(v))
+                    .detach().reshape(1, -1).contiguous(), comm)
+                for v in concat]
+        for name in (epl.GraphKeys.LOCAL_MEAN_OBJECTS,
+                     epl.GraphKeys.LOCAL_SUM_OBJECTS,
+                     epl.GraphKeys.LOCAL_CONCAT_OBJECTS):
+            vals = env.get_collection(name)
+            if vals:
+                out[name] = vals
+        return out
+
     def slice_input_files(self, files):
         """Per-replica IO slicing (config io.slicing; reference
         graph_editor.py:149-215)."""

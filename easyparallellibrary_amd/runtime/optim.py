@@ -37,15 +37,26 @@ class FusedAdamW:
     def step(self, grad_scale=1.0):
         self.step_count += 1
         inv_scale = 1.0 / grad_scale
+        from easyparallellibrary_amd.env import Env
+        napply = max(1, Env.get().config.optimizer.num_apply_group)
         for g in self.groups:
             if use_native(g.master_arena):
-                native_ext().fused_adamw(
-                    g.master_arena,
-                    g.param_arena if g.param_arena.dtype == torch.bfloat16
-                    else None,
-                    g.grad_arena, g.state["exp_avg"], g.state["exp_avg_sq"],
-                    self.lr, self.beta1, self.beta2, self.eps,
-                    self.weight_decay, self.step_count, inv_scale)
+                # grouped apply (reference: runtime/optimizer_helper.py
+                # :75-128): the flat arena makes this a chunked launch —
+                # one fused kernel per apply group
+                n = g.master_arena.numel()
+                chunk = (n + napply - 1) // napply
+                chunk = (chunk + 127) // 128 * 128
+                for lo in range(0, n, chunk):
+                    hi = min(lo + chunk, n)
+                    native_ext().fused_adamw(
+                        g.master_arena[lo:hi],
+                        g.param_arena[lo:hi]
+                        if g.param_arena.dtype == torch.bfloat16 else None,
+                        g.grad_arena[lo:hi], g.state["exp_avg"][lo:hi],
+                        g.state["exp_avg_sq"][lo:hi],
+                        self.lr, self.beta1, self.beta2, self.eps,
+                        self.weight_decay, self.step_count, inv_scale)
             else:
                 self._step_torch(g, inv_scale)
 

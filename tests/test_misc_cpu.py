@@ -72,3 +72,35 @@ def test_metric_merge_single():
     assert float(out) == 3.0
     files = engine.slice_input_files(["a", "b"])
     assert files == ["a", "b"]
+
+
+def test_collections_api():
+    import easyparallellibrary_amd as epl
+    epl.init()
+    epl.add_to_collection(1.5, epl.GraphKeys.GLOBAL_MEAN_OBJECTS)
+    epl.add_to_collection(2.5, epl.GraphKeys.GLOBAL_MEAN_OBJECTS)
+    assert epl.get_collection(epl.GraphKeys.GLOBAL_MEAN_OBJECTS) == [1.5, 2.5]
+
+    import torch.nn as nn
+    with epl.replicate(1):
+        m = nn.Linear(2, 2)
+    engine = epl.Engine(m, loss_fn=nn.MSELoss())
+    merged = engine.merged_collections()
+    vals = [float(v) for v in merged[epl.GraphKeys.GLOBAL_MEAN_OBJECTS]]
+    assert vals == [1.5, 2.5]
+
+
+def test_grouped_apply_config():
+    import torch
+    import torch.nn as nn
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"optimizer.num_apply_group": 3}))
+    torch.manual_seed(9)
+    with epl.replicate(1):
+        m = nn.Linear(8, 8)
+    engine = epl.Engine(m, loss_fn=nn.MSELoss(), lr=1e-2)
+    x = torch.randn(4, 8)
+    y = torch.randn(4, 8)
+    l0 = float(engine.train_step(x, y))
+    l1 = float(engine.train_step(x, y))
+    assert l1 < l0
