@@ -17,11 +17,10 @@
 //    from global (L2-resident tiles); V^T fragments gather columns (L2).
 // Saves per-row LSE (m + log l) for the backward.
 //
-// Backward: one kernel parallel over kv tiles; per q tile it recomputes
-// P from Q,K,LSE, accumulates dV += P^T dO and dK += dS^T Q locally, and
-// scatters dQ += dS K with fp32 global atomics (dq workspace); host casts
-// dq to bf16 afterwards.  D_i = rowsum(dO*O) is precomputed by a small
-// kernel.
+// Backward: two kernels (kv-parallel for dK/dV, q-parallel for dQ), each
+// lane owning its output row — no atomics.  D_i = rowsum(dO*O) is
+// precomputed by a small kernel.  All kernels take (batch, head, seq)
+// strides so the qkv-unbind views are consumed without copies.
 
 #include <hip/hip_runtime.h>
 #include <cstdint>

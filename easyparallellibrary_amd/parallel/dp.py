@@ -71,14 +71,18 @@ class FlatParamGroup:
         self.ordered = ordered
 
         self.param_arena = torch.zeros(total, dtype=model_dtype, device=device)
+        # grad arena starts (and its inter-parameter padding forever stays)
+        # zero — the padding is never written, so collectives see zeros
         self.grad_arena = torch.zeros(self.total, dtype=self.grad_dtype,
                                       device=device)
+        self._grad_views = {}
         for p, off in zip(ordered, offsets):
             n = p.numel()
             self.param_arena[off:off + n].copy_(
                 p.data.reshape(-1).to(model_dtype))
             p.data = self.param_arena[off:off + n].view(p.shape)
             p.grad = self.grad_arena[off:off + n].view(p.shape)
+            self._grad_views[id(p)] = p.grad
 
         if model_dtype == torch.float32:
             self.master_arena = self.param_arena
@@ -96,7 +100,30 @@ class FlatParamGroup:
         return self.offsets[i]
 
     def zero_grad(self):
-        self.grad_arena.zero_()
+        """Drop grads: p.grad=None lets autograd hand us its fresh grad
+        tensor (no += against a zeroed arena, no arena-wide fill); the
+        reducer's post-accumulate hook copies it into the arena view and
+        re-points p.grad so later micro-batches accumulate in place.
+        Requires every parameter to receive a gradient each step (our
+        engine's models do; set grad_copy_first=False otherwise)."""
+        if self.grad_copy_first:
+            for p in self.ordered:
+                p.grad = None
+        else:
+            self.grad_arena.zero_()
+
+    grad_copy_first = True
+
+    def arena_view_of(self, p):
+        return self._grad_views[id(p)]
+
+    def adopt_grad(self, p):
+        """Called from the post-accumulate hook: fold autograd's fresh
+        grad into the arena view (first touch this step) and re-point."""
+        view = self._grad_views[id(p)]
+        if p.grad is not view:
+            view.copy_(p.grad)
+            p.grad = view
 
     def sync_master_to_params(self):
         """After a master-arena update without the fused kernel's writeback."""
@@ -163,7 +190,7 @@ class GradReducer:
         self._world = (self.pool.comms[0].size if world_scale is None
                        else world_scale)
         self._hook_handles = []
-        if overlap:
+        if overlap or group.grad_copy_first:
             for p in group.ordered:
                 h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
                 self._hook_handles.append(h)
@@ -185,6 +212,8 @@ class GradReducer:
             comm.all_reduce(buf, op=self.op, async_op=True)
 
     def _on_grad_ready(self, p):
+        if self.group.grad_copy_first:
+            self.group.adopt_grad(p)
         if not self.enabled:
             return
         for bi in self._buckets_of_param[id(p)]:
