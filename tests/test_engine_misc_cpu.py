@@ -1,0 +1,94 @@
+"""Engine odds and ends: eval, scheduler variants, GA+pipeline combos,
+non-overlap reduction path."""
+
+import torch
+import torch.nn as nn
+
+from tests.utils import run_multiprocess
+
+
+def test_eval_step():
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(1)
+    with epl.replicate(1):
+        m = nn.Linear(4, 2)
+    engine = epl.Engine(m, loss_fn=nn.MSELoss())
+    out = engine.eval_step(torch.randn(3, 4))
+    assert out.shape == (3, 2)
+    assert not out.requires_grad
+
+
+def _pbo_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({
+        "pipeline.num_micro_batch": 4,
+        "pipeline.strategy": "prefer_backward_optimizer",
+    }))
+    torch.manual_seed(21)
+    with epl.replicate(device_count=1, name="stage_0"):
+        s0 = nn.Sequential(nn.Linear(8, 16), nn.Tanh())
+    with epl.replicate(device_count=1, name="stage_1"):
+        s1 = nn.Linear(16, 4)
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.s0, self.s1 = s0, s1
+
+        def forward(self, x):
+            return self.s1(self.s0(x))
+
+    engine = epl.Engine(M(), loss_fn=nn.MSELoss(), lr=1e-2)
+    torch.manual_seed(22)
+    x = torch.randn(8, 8)
+    y = torch.randn(8, 4)
+    out = [engine.train_step(x, y) for _ in range(3)]
+    return [None if o is None else float(o) for o in out]
+
+
+def test_prefer_backward_optimizer_schedule():
+    res = run_multiprocess(_pbo_worker, world=2)
+    losses = res[1]
+    assert losses[-1] < losses[0]
+
+
+def _no_overlap_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"communication.overlap_grad_reduce": False}))
+    torch.manual_seed(31)
+    with epl.replicate(1):
+        m = nn.Sequential(nn.Linear(8, 8), nn.Tanh(), nn.Linear(8, 2))
+    engine = epl.Engine(m, loss_fn=nn.MSELoss(), lr=1e-2)
+    torch.manual_seed(32)
+    x = torch.randn(4, 8)
+    y = torch.randn(4, 2)
+    return [float(engine.train_step(x, y)) for _ in range(3)]
+
+
+def test_no_overlap_matches_overlap():
+    off = run_multiprocess(_no_overlap_worker, world=2)
+    assert off[0] == off[1]
+    assert off[0][-1] < off[0][0]
+
+
+def _bucket_sizes_worker(rank, world):
+    """Tiny bucket_bytes -> many buckets; trajectory must be unchanged."""
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"communication.bucket_bytes": 256,
+                         "communication.num_communicators": 3}))
+    torch.manual_seed(41)
+    with epl.replicate(1):
+        m = nn.Sequential(nn.Linear(16, 64), nn.Tanh(), nn.Linear(64, 4))
+    engine = epl.Engine(m, loss_fn=nn.MSELoss(), lr=1e-2)
+    assert len(engine.reducers[0].buckets) > 2
+    torch.manual_seed(42)
+    x = torch.randn(4, 16)
+    y = torch.randn(4, 4)
+    return [float(engine.train_step(x, y)) for _ in range(3)]
+
+
+def test_many_small_buckets():
+    res = run_multiprocess(_bucket_sizes_worker, world=2)
+    assert res[0] == res[1]
+    assert res[0][-1] < res[0][0]
