@@ -49,7 +49,8 @@ void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
 void epl_attn_bwd(const void*, const void*, const void*, const void*,
                   const void*, const float*, float*, void*, void*, void*,
                   int64_t, int64_t, float, bool, int64_t, const int64_t*,
-                  const int64_t*, const int64_t*, hipStream_t);
+                  const int64_t*, const int64_t*, const int64_t*,
+                  hipStream_t);
 }
 
 namespace {
@@ -299,13 +300,17 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
               at::Tensor dout, at::Tensor lse, at::Tensor delta_ws,
               at::Tensor dq, at::Tensor dk, at::Tensor dv, double scale,
               bool causal) {
-  int64_t in_s[3], o_s[3], do_s[3], tmp[3];
+  int64_t in_s[3], o_s[3], do_s[3], g_s[3], tmp[3];
   attn_strides(q, in_s, "q");
   attn_strides(out, o_s, "out");
   attn_strides(dout, do_s, "dout");
-  attn_strides(dq, tmp, "dq");
-  TORCH_CHECK(dq.is_contiguous() && dk.is_contiguous() &&
-              dv.is_contiguous(), "grad outputs must be contiguous");
+  attn_strides(dq, g_s, "dq");
+  attn_strides(dk, tmp, "dk");
+  TORCH_CHECK(tmp[0] == g_s[0] && tmp[1] == g_s[1] && tmp[2] == g_s[2],
+              "dq/dk/dv must share strides");
+  attn_strides(dv, tmp, "dv");
+  TORCH_CHECK(tmp[0] == g_s[0] && tmp[1] == g_s[1] && tmp[2] == g_s[2],
+              "dq/dk/dv must share strides");
   const int64_t heads = q.size(1);
   const int64_t seq = q.size(2);
   const int64_t bh = q.size(0) * heads;
@@ -314,7 +319,7 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
                dout.data_ptr(), lse.data_ptr<float>(),
                delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                dv.data_ptr(), bh, seq, (float)scale, causal, heads, in_s,
-               o_s, do_s, cur_stream());
+               o_s, do_s, g_s, cur_stream());
 }
 
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {

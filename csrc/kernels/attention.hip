@@ -251,7 +251,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dk, short* __restrict__ dv, int64_t seq,
     float scale, int causal, int64_t heads, int64_t in_sb, int64_t in_sh,
-    int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss) {
+    int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss,
+    int64_t g_sb, int64_t g_sh, int64_t g_ss) {
   __shared__ short ldsQ[32][72];
   __shared__ short ldsDO[32][72];
   const int lane = threadIdx.x & 63;
@@ -353,8 +354,10 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
   }
 
   if (!active || mykv >= seq) return;
-  short* dkp = dk + (bh * seq + mykv) * 64;
-  short* dvp = dv + (bh * seq + mykv) * 64;
+  const int64_t goff = (bh / heads) * g_sb + (bh % heads) * g_sh +
+                       mykv * g_ss;
+  short* dkp = dk + goff;
+  short* dvp = dv + goff;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     dvp[crow(r, hi)] = (short)f2bf(dvt0[r]);
@@ -375,7 +378,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const float* __restrict__ lse, const float* __restrict__ delta,
     short* __restrict__ dq, int64_t seq, float scale, int causal,
     int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
-    int64_t do_sb, int64_t do_sh, int64_t do_ss) {
+    int64_t do_sb, int64_t do_sh, int64_t do_ss,
+    int64_t g_sb, int64_t g_sh, int64_t g_ss) {
   __shared__ short ldsK[32][72];
   __shared__ short ldsVr[32][72];
   const int lane = threadIdx.x & 63;
@@ -467,7 +471,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   }
 
   if (!active || myq >= seq) return;
-  short* dqp = dq + (bh * seq + myq) * 64;
+  short* dqp = dq + (bh / heads) * g_sb + (bh % heads) * g_sh + myq * g_ss;
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     dqp[crow(r, hi)] = (short)f2bf(dqt0[r]);
@@ -500,7 +504,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   int64_t batch_heads, int64_t seq, float scale,
                   bool causal, int64_t heads, const int64_t* in_strides,
                   const int64_t* o_strides, const int64_t* do_strides,
-                  hipStream_t stream) {
+                  const int64_t* g_strides, hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
   {
     const int64_t blocks = (rows + 3) / 4;
@@ -523,7 +527,8 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      reinterpret_cast<short*>(dv), seq, scale,
                      causal ? 1 : 0, heads, in_strides[0], in_strides[1],
                      in_strides[2], do_strides[0], do_strides[1],
-                     do_strides[2]);
+                     do_strides[2], g_strides[0], g_strides[1],
+                     g_strides[2]);
   hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),
@@ -532,7 +537,8 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      reinterpret_cast<short*>(dq), seq, scale,
                      causal ? 1 : 0, heads, in_strides[0], in_strides[1],
                      in_strides[2], do_strides[0], do_strides[1],
-                     do_strides[2]);
+                     do_strides[2], g_strides[0], g_strides[1],
+                     g_strides[2]);
 }
 
 }  // extern "C"

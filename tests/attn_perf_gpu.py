@@ -41,3 +41,30 @@ for (b, h, s, causal) in [(128, 16, 512, False), (16, 25, 1024, True)]:
         out.backward(dout)
     t = bench(sdpa_fb, iters=20)
     print(f"  sdpa f+b {t:8.1f} us  {3.5*flops_fwd/t/1e6:7.1f} TF-equiv")
+
+# strided-input case: the module path's qkv-unbind views
+print("--- strided (qkv unbind views) ---")
+for (b, h, s, causal) in [(128, 16, 512, False)]:
+    d = 64
+    scale = 1/math.sqrt(d)
+    qkv = torch.randn(b, s, 3, h, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    q, k, v = (t.transpose(1, 2) for t in qkv.unbind(dim=2))
+    dout = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16)
+    flops_fwd = 4 * b * h * s * s * d
+
+    t = bench(lambda: _FlashAttention.apply(q, k, v, causal, scale))
+    print(f"ours fwd strided {t:8.1f} us  {flops_fwd/t/1e6:7.1f} TF")
+    t = bench(lambda: F.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale))
+    print(f"sdpa fwd strided {t:8.1f} us  {flops_fwd/t/1e6:7.1f} TF")
+    def ours_fb():
+        qkv.grad = None
+        out = _FlashAttention.apply(q, k, v, causal, scale)
+        out.backward(dout)
+    t = bench(ours_fb, iters=20)
+    print(f"ours f+b strided {t:8.1f} us")
+    def sdpa_fb():
+        qkv.grad = None
+        out = F.scaled_dot_product_attention(q, k, v, is_causal=causal, scale=scale)
+        out.backward(dout)
+    t = bench(sdpa_fb, iters=20)
+    print(f"sdpa f+b strided {t:8.1f} us")
