@@ -557,16 +557,22 @@ __global__ void layer_norm_bwd_kernel(
 // ============================================================================
 // Fused bias + GeLU (tanh approximation), forward and backward.
 // ============================================================================
+DEV float fast_tanh(float u) {
+  // tanh(u) = 1 - 2/(e^2u + 1): one fast exp instead of tanhf's polynomial
+  // + range-reduction chain (bf16-accurate; numerics-tested vs torch)
+  return 1.f - 2.f / (__expf(2.f * u) + 1.f);
+}
+
 DEV float gelu_f(float x) {
   const float k0 = 0.7978845608028654f;   // sqrt(2/pi)
   const float k1 = 0.044715f;
-  return 0.5f * x * (1.f + tanhf(k0 * (x + k1 * x * x * x)));
+  return 0.5f * x * (1.f + fast_tanh(k0 * (x + k1 * x * x * x)));
 }
 
 DEV float gelu_grad_f(float x) {
   const float k0 = 0.7978845608028654f;
   const float k1 = 0.044715f;
-  const float t = tanhf(k0 * (x + k1 * x * x * x));
+  const float t = fast_tanh(k0 * (x + k1 * x * x * x));
   const float dt = (1.f - t * t) * k0 * (1.f + 3.f * k1 * x * x);
   return 0.5f * (1.f + t) + 0.5f * x * dt;
 }
