@@ -49,7 +49,7 @@ void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
 void epl_attn_bwd(const void*, const void*, const void*, const void*,
                   const void*, const float*, float*, void*, void*, void*,
                   int64_t, int64_t, float, bool, int64_t, const int64_t*,
-                  const int64_t*, const int64_t*, const int64_t*,
+                  const int64_t*, const int64_t*, const int64_t*, int,
                   hipStream_t);
 }
 
@@ -299,7 +299,7 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
 void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
               at::Tensor dout, at::Tensor lse, at::Tensor delta_ws,
               at::Tensor dq, at::Tensor dk, at::Tensor dv, double scale,
-              bool causal) {
+              bool causal, bool split_dkdv) {
   int64_t in_s[3], o_s[3], do_s[3], g_s[3], tmp[3];
   attn_strides(q, in_s, "q");
   attn_strides(out, o_s, "out");
@@ -319,7 +319,7 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
                dout.data_ptr(), lse.data_ptr<float>(),
                delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                dv.data_ptr(), bh, seq, (float)scale, causal, heads, in_s,
-               o_s, do_s, g_s, cur_stream());
+               o_s, do_s, g_s, split_dkdv ? 1 : 0, cur_stream());
 }
 
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {

@@ -368,6 +368,205 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 }
 
 // ============================================================================
+// split backward A kernels: dV-only and dK-only.  Same structure as the
+// combined kernel but half the accumulators each -> 3 waves/SIMD instead
+// of 2 (A/B-selectable via epl_attn_bwd mode).
+// ============================================================================
+__global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ dout, const float* __restrict__ lse,
+    short* __restrict__ dv, int64_t seq, float scale, int causal,
+    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
+    int64_t do_sb, int64_t do_sh, int64_t do_ss, int64_t g_sb,
+    int64_t g_sh, int64_t g_ss) {
+  __shared__ short ldsQ[32][72];
+  __shared__ short ldsDO[32][72];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int lkv = lane & 31;
+  const int64_t bh = blockIdx.y;
+  const int64_t kv0_blk = (int64_t)blockIdx.x * 128;
+  const int64_t kv0 = kv0_blk + wave * 32;
+  const bool active = kv0 < seq;
+  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
+  const short* qp = q + boff;
+  const short* kp = k + boff;
+  const short* dop = dout + (bh / heads) * do_sb + (bh % heads) * do_sh;
+  const float* lsep = lse + bh * seq;
+
+  const int64_t mykv = kv0 + lkv;
+  const int64_t kvrow = mykv < seq ? mykv : seq - 1;
+  bf16x8 kfrag[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c)
+    kfrag[c] = *reinterpret_cast<const bf16x8*>(
+        kp + kvrow * in_ss + hi * 8 + 16 * c);
+
+  f32x16 dvt0 = {}, dvt1 = {};
+  const int stage_row = threadIdx.x >> 3;
+  const int stage_seg = (threadIdx.x & 7) * 8;
+  const int64_t q_start = causal ? (kv0_blk / 32) * 32 : 0;
+  for (int64_t q0 = q_start; q0 < seq; q0 += 32) {
+    __syncthreads();
+    {
+      int64_t qr = q0 + stage_row;
+      if (qr >= seq) qr = seq - 1;
+      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+    }
+    __syncthreads();
+    if (!active || (causal && q0 + 31 < kv0)) continue;
+    f32x16 st = {};
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+          &ldsQ[lkv][hi * 8 + 16 * c]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0,
+                                                   0);
+    }
+    float p[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int64_t qg = q0 + crow(r, hi);
+      const int64_t qgc = qg < seq ? qg : seq - 1;
+      bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
+      p[r] = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
+    }
+    bf16x8 pb0 = assemble_pfrag(&p[0]);
+    bf16x8 pb1 = assemble_pfrag(&p[8]);
+#pragma unroll
+    for (int qc = 0; qc < 2; ++qc) {
+      bf16x8 dot0, dot1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int qr = qc * 16 + hi * 8 + j;
+        dot0[j] = ldsDO[qr][lkv];
+        dot1[j] = ldsDO[qr][32 + lkv];
+      }
+      bf16x8 pb = qc == 0 ? pb0 : pb1;
+      dvt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot0, pb, dvt0, 0, 0,
+                                                     0);
+      dvt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot1, pb, dvt1, 0, 0,
+                                                     0);
+    }
+  }
+  if (!active || mykv >= seq) return;
+  short* dvp = dv + (bh / heads) * g_sb + (bh % heads) * g_sh +
+               mykv * g_ss;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    dvp[crow(r, hi)] = (short)f2bf(dvt0[r]);
+    dvp[32 + crow(r, hi)] = (short)f2bf(dvt1[r]);
+  }
+}
+
+__global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, const short* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    short* __restrict__ dk, int64_t seq, float scale, int causal,
+    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
+    int64_t do_sb, int64_t do_sh, int64_t do_ss, int64_t g_sb,
+    int64_t g_sh, int64_t g_ss) {
+  __shared__ short ldsQ[32][72];
+  __shared__ short ldsDO[32][72];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int hi = lane >> 5;
+  const int lkv = lane & 31;
+  const int64_t bh = blockIdx.y;
+  const int64_t kv0_blk = (int64_t)blockIdx.x * 128;
+  const int64_t kv0 = kv0_blk + wave * 32;
+  const bool active = kv0 < seq;
+  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
+  const short* qp = q + boff;
+  const short* kp = k + boff;
+  const short* vp = v + boff;
+  const short* dop = dout + (bh / heads) * do_sb + (bh % heads) * do_sh;
+  const float* lsep = lse + bh * seq;
+  const float* dltp = delta + bh * seq;
+
+  const int64_t mykv = kv0 + lkv;
+  const int64_t kvrow = mykv < seq ? mykv : seq - 1;
+  bf16x8 kfrag[4], vfrag[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    kfrag[c] = *reinterpret_cast<const bf16x8*>(
+        kp + kvrow * in_ss + hi * 8 + 16 * c);
+    vfrag[c] = *reinterpret_cast<const bf16x8*>(
+        vp + kvrow * in_ss + hi * 8 + 16 * c);
+  }
+
+  f32x16 dkt0 = {}, dkt1 = {};
+  const int stage_row = threadIdx.x >> 3;
+  const int stage_seg = (threadIdx.x & 7) * 8;
+  const int64_t q_start = causal ? (kv0_blk / 32) * 32 : 0;
+  for (int64_t q0 = q_start; q0 < seq; q0 += 32) {
+    __syncthreads();
+    {
+      int64_t qr = q0 + stage_row;
+      if (qr >= seq) qr = seq - 1;
+      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+    }
+    __syncthreads();
+    if (!active || (causal && q0 + 31 < kv0)) continue;
+    f32x16 st = {};
+    f32x16 dpt = {};
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+          &ldsQ[lkv][hi * 8 + 16 * c]);
+      bf16x8 dof = *reinterpret_cast<const bf16x8*>(
+          &ldsDO[lkv][hi * 8 + 16 * c]);
+      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0,
+                                                   0);
+      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfrag[c], dpt, 0,
+                                                    0, 0);
+    }
+    float ds[16];
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const int64_t qg = q0 + crow(r, hi);
+      const int64_t qgc = qg < seq ? qg : seq - 1;
+      bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
+      float pv = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
+      ds[r] = masked ? 0.f : pv * (dpt[r] - dltp[qgc]) * scale;
+    }
+    bf16x8 db0 = assemble_pfrag(&ds[0]);
+    bf16x8 db1 = assemble_pfrag(&ds[8]);
+#pragma unroll
+    for (int qc = 0; qc < 2; ++qc) {
+      bf16x8 qt0, qt1;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int qr = qc * 16 + hi * 8 + j;
+        qt0[j] = ldsQ[qr][lkv];
+        qt1[j] = ldsQ[qr][32 + lkv];
+      }
+      bf16x8 db = qc == 0 ? db0 : db1;
+      dkt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt0, db, dkt0, 0, 0,
+                                                     0);
+      dkt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt1, db, dkt1, 0, 0,
+                                                     0);
+    }
+  }
+  if (!active || mykv >= seq) return;
+  short* dkp = dk + (bh / heads) * g_sb + (bh % heads) * g_sh +
+               mykv * g_ss;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    dkp[crow(r, hi)] = (short)f2bf(dkt0[r]);
+    dkp[32 + crow(r, hi)] = (short)f2bf(dkt1[r]);
+  }
+}
+
+// ============================================================================
 // backward kernel B (q-parallel): dQ.  K and V tiles staged in LDS as
 // row copies per block; row fragments via ds_read_b128, K-transposed
 // fragments via 2-byte LDS reads.
@@ -504,7 +703,8 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   int64_t batch_heads, int64_t seq, float scale,
                   bool causal, int64_t heads, const int64_t* in_strides,
                   const int64_t* o_strides, const int64_t* do_strides,
-                  const int64_t* g_strides, hipStream_t stream) {
+                  const int64_t* g_strides, int split_dkdv,
+                  hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
   {
     const int64_t blocks = (rows + 3) / 4;
@@ -518,6 +718,27 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                        o_strides[2]);
   }
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
+  if (split_dkdv) {
+    hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(q),
+                       reinterpret_cast<const short*>(k),
+                       reinterpret_cast<const short*>(dout), lse,
+                       reinterpret_cast<short*>(dv), seq, scale,
+                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                       in_strides[2], do_strides[0], do_strides[1],
+                       do_strides[2], g_strides[0], g_strides[1],
+                       g_strides[2]);
+    hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(q),
+                       reinterpret_cast<const short*>(k),
+                       reinterpret_cast<const short*>(v),
+                       reinterpret_cast<const short*>(dout), lse, delta_ws,
+                       reinterpret_cast<short*>(dk), seq, scale,
+                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                       in_strides[2], do_strides[0], do_strides[1],
+                       do_strides[2], g_strides[0], g_strides[1],
+                       g_strides[2]);
+  } else
   hipLaunchKernelGGL(attn_bwd_dkdv_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),

@@ -62,8 +62,12 @@ class _FlashAttention(torch.autograd.Function):
         dk = grad_like(orig_q)
         dv = grad_like(orig_q)
         delta = torch.empty_like(lse)
+        import os
+        # split dK/dV kernels measure ~4% faster than the combined one
+        # (3 vs 2 waves/SIMD; A/B in profiles/r01_attention_ab.txt)
+        split = os.environ.get("EPL_ATTN_BWD_SPLIT", "1") == "1"
         native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
-                              ctx.scale, ctx.causal)
+                              ctx.scale, ctx.causal, split)
         return dq, dk, dv, None, None
 
 

@@ -68,3 +68,20 @@ for (b, h, s, causal) in [(128, 16, 512, False)]:
         out.backward(dout)
     t = bench(sdpa_fb, iters=20)
     print(f"sdpa f+b strided {t:8.1f} us")
+
+import os
+print("--- bwd split A/B (contiguous) ---")
+for split in ("0", "1", "0", "1"):
+    os.environ["EPL_ATTN_BWD_SPLIT"] = split
+    b, h, s, d = 128, 16, 512, 64
+    scale = 1/math.sqrt(d)
+    q = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16, requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    dout = torch.randn_like(q)
+    def fb():
+        out = _FlashAttention.apply(q, k, v, False, scale)
+        out.backward(dout)
+    t = bench(fb, iters=20)
+    print(f"split={split}  f+b {t:8.1f} us")
+os.environ.pop("EPL_ATTN_BWD_SPLIT", None)
