@@ -135,16 +135,22 @@ class GradReducer:
     """Bucketed overlapped gradient reduction over one FlatParamGroup."""
 
     def __init__(self, group, pool, bucket_bytes, reduce_method="mean",
-                 overlap=True, world_scale=None, shard_owners=False):
+                 overlap=True, world_scale=None, shard_owners=False,
+                 compression=""):
         """shard_owners: ZeRO-v1 mode — buckets are split at shard
         boundaries (shard s = rank s's contiguous arena slice) and each
         bucket is REDUCED to its owning rank instead of all-reduced
-        (reference: runtime/zero.py:178-190)."""
+        (reference: runtime/zero.py:178-190).
+        compression: '' | 'fp16' | 'bf16' — compress fp32 gradient buckets
+        over the wire (reference: coalescing.py fp16 option).  bf16 models
+        already ship bf16 by construction; this knob matters for fp32
+        training."""
         self.group = group
         self.pool = pool
         self.reduce_method = reduce_method
         self.overlap = overlap
         self.shard_owners = shard_owners
+        self.compression = compression if             group.grad_arena.dtype == torch.float32 else ""
         esize = group.grad_arena.element_size()
         bucket_elems = max(_ALIGN, int(bucket_bytes) // esize)
 
@@ -186,6 +192,7 @@ class GradReducer:
                 self._buckets_of_param.setdefault(id(p), []).append(bi)
         self._pending = [len(ps) for (_, _, ps, _) in self.buckets]
         self._launched = [False] * len(self.buckets)
+        self._decompress = []
         self.enabled = True          # pipeline sets False until last ubatch
         self._world = (self.pool.comms[0].size if world_scale is None
                        else world_scale)
@@ -206,10 +213,19 @@ class GradReducer:
         start, end, _, owner = self.buckets[bi]
         comm = self.pool.next_comm()
         buf = self.group.grad_arena[start:end]
+        wire = buf
+        if self.compression and comm.size > 1:
+            cdt = (torch.float16 if self.compression == "fp16"
+                   else torch.bfloat16)
+            wire = buf.to(cdt)
         if self.shard_owners and owner is not None and comm.size > 1:
-            comm.reduce(buf, root=owner, op=self.op, async_op=True)
+            comm.reduce(wire, root=owner, op=self.op, async_op=True)
         else:
-            comm.all_reduce(buf, op=self.op, async_op=True)
+            comm.all_reduce(wire, op=self.op, async_op=True)
+        if wire is not buf:
+            # decompress on the comm's completion order: since async ops on
+            # one comm serialize on its stream, enqueue the copy after join
+            self._decompress.append((bi, wire, buf))
 
     def _on_grad_ready(self, p):
         if self.group.grad_copy_first:
@@ -228,6 +244,9 @@ class GradReducer:
             if not self._launched[bi]:
                 self._launch(bi)
         self.pool.join()
+        for _, wire, buf in self._decompress:
+            buf.copy_(wire)
+        self._decompress = []
         self.reset()
 
     def reduce_now(self):

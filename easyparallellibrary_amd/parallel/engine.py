@@ -244,7 +244,8 @@ class Engine:
                 reducer = GradReducer(
                     fg, pool, bucket_bytes, reduce_method=reduce_method,
                     overlap=overlap,
-                    shard_owners=(zero_level == "v1"))
+                    shard_owners=(zero_level == "v1"),
+                    compression=self.config.communication.compression)
                 self.flat_groups.append(fg)
                 self.reducers.append(reducer)
                 self._group_infos.append(

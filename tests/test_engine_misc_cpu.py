@@ -92,3 +92,22 @@ def test_many_small_buckets():
     res = run_multiprocess(_bucket_sizes_worker, world=2)
     assert res[0] == res[1]
     assert res[0][-1] < res[0][0]
+
+
+def _compress_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"communication.compression": "bf16"}))
+    torch.manual_seed(51)
+    with epl.replicate(1):
+        m = nn.Sequential(nn.Linear(8, 16), nn.Tanh(), nn.Linear(16, 2))
+    engine = epl.Engine(m, loss_fn=nn.MSELoss(), lr=1e-2)
+    torch.manual_seed(52)
+    x = torch.randn(4, 8)
+    y = torch.randn(4, 2)
+    return [float(engine.train_step(x, y)) for _ in range(3)]
+
+
+def test_grad_compression():
+    res = run_multiprocess(_compress_worker, world=2)
+    assert res[0] == res[1]
+    assert res[0][-1] < res[0][0]
