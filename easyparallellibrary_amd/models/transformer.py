@@ -7,7 +7,6 @@ FusedBiasGelu (FFN first Linear runs bias-free); attention uses torch SDPA
 (hipBLASLt GEMMs).
 """
 
-import math
 import os
 
 import torch

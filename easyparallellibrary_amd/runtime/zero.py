@@ -19,7 +19,6 @@ multiple of (dp_world x alignment), rank r owns the contiguous shard
 Optimizer-state memory per rank drops from 8 bytes/param to 8/W.
 """
 
-import torch
 
 from easyparallellibrary_amd.runtime.optim import FusedAdamW
 
