@@ -575,10 +575,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq, int64_t seq, float scale, int causal,
-    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
-    int64_t do_sb, int64_t do_sh, int64_t do_ss,
-    int64_t g_sb, int64_t g_sh, int64_t g_ss) {
+    short* __restrict__ dq, const short* __restrict__ out, int64_t seq,
+    float scale, int causal, int64_t heads, int64_t in_sb, int64_t in_sh,
+    int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss,
+    int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
+    int64_t o_ss) {
   __shared__ short ldsK[32][72];
   __shared__ short ldsVr[32][72];
   const int lane = threadIdx.x & 63;
@@ -598,16 +599,28 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int64_t myq = q0 + lq;
   const int64_t qrow = myq < seq ? myq : seq - 1;
   const float mylse = lse[bh * seq + qrow];
-  const float mydelta = delta[bh * seq + qrow];
 
+  const short* op_ = out + (bh / heads) * o_sb + (bh % heads) * o_sh;
   bf16x8 qfrag[4], dofrag[4];
+  float dsum = 0.f;
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     qfrag[c] = *reinterpret_cast<const bf16x8*>(
         qp + qrow * in_ss + hi * 8 + 16 * c);
     dofrag[c] = *reinterpret_cast<const bf16x8*>(
         dop + qrow * do_ss + hi * 8 + 16 * c);
+    bf16x8 of = *reinterpret_cast<const bf16x8*>(
+        op_ + qrow * o_ss + hi * 8 + 16 * c);
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      dsum += bf2f(dofrag[c][j]) * bf2f(of[j]);
   }
+  // D_i = rowsum(dO * O): this lane covers 32 of the 64 d-elements, the
+  // partner lane (^32) the rest
+  const float mydelta = dsum + __shfl_xor(dsum, 32, 64);
+  // publish delta for the dK kernel that follows on the same stream
+  if (myq < seq && hi == 0)
+    const_cast<float*>(delta)[bh * seq + myq] = mydelta;
 
   f32x16 dqt0 = {}, dqt1 = {};
 
@@ -706,6 +719,44 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   const int64_t* g_strides, int split_dkdv,
                   hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
+  dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
+  if (split_dkdv) {
+    // order: dV (needs no delta) -> dQ (computes + publishes delta from
+    // the dO/O rows it already loads) -> dK (consumes delta).  The prep
+    // pass disappears.
+    hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(q),
+                       reinterpret_cast<const short*>(k),
+                       reinterpret_cast<const short*>(dout), lse,
+                       reinterpret_cast<short*>(dv), seq, scale,
+                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                       in_strides[2], do_strides[0], do_strides[1],
+                       do_strides[2], g_strides[0], g_strides[1],
+                       g_strides[2]);
+    hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(q),
+                       reinterpret_cast<const short*>(k),
+                       reinterpret_cast<const short*>(v),
+                       reinterpret_cast<const short*>(dout), lse, delta_ws,
+                       reinterpret_cast<short*>(dq),
+                       reinterpret_cast<const short*>(out), seq, scale,
+                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                       in_strides[2], do_strides[0], do_strides[1],
+                       do_strides[2], g_strides[0], g_strides[1],
+                       g_strides[2], o_strides[0], o_strides[1],
+                       o_strides[2]);
+    hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0, stream,
+                       reinterpret_cast<const short*>(q),
+                       reinterpret_cast<const short*>(k),
+                       reinterpret_cast<const short*>(v),
+                       reinterpret_cast<const short*>(dout), lse, delta_ws,
+                       reinterpret_cast<short*>(dk), seq, scale,
+                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
+                       in_strides[2], do_strides[0], do_strides[1],
+                       do_strides[2], g_strides[0], g_strides[1],
+                       g_strides[2]);
+    return;
+  }
   {
     const int64_t blocks = (rows + 3) / 4;
     hipLaunchKernelGGL(attn_bwd_prep_kernel,
@@ -717,28 +768,6 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                        do_strides[2], o_strides[0], o_strides[1],
                        o_strides[2]);
   }
-  dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
-  if (split_dkdv) {
-    hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0, stream,
-                       reinterpret_cast<const short*>(q),
-                       reinterpret_cast<const short*>(k),
-                       reinterpret_cast<const short*>(dout), lse,
-                       reinterpret_cast<short*>(dv), seq, scale,
-                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
-                       in_strides[2], do_strides[0], do_strides[1],
-                       do_strides[2], g_strides[0], g_strides[1],
-                       g_strides[2]);
-    hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0, stream,
-                       reinterpret_cast<const short*>(q),
-                       reinterpret_cast<const short*>(k),
-                       reinterpret_cast<const short*>(v),
-                       reinterpret_cast<const short*>(dout), lse, delta_ws,
-                       reinterpret_cast<short*>(dk), seq, scale,
-                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
-                       in_strides[2], do_strides[0], do_strides[1],
-                       do_strides[2], g_strides[0], g_strides[1],
-                       g_strides[2]);
-  } else
   hipLaunchKernelGGL(attn_bwd_dkdv_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),
@@ -755,11 +784,13 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      reinterpret_cast<const short*>(k),
                      reinterpret_cast<const short*>(v),
                      reinterpret_cast<const short*>(dout), lse, delta_ws,
-                     reinterpret_cast<short*>(dq), seq, scale,
+                     reinterpret_cast<short*>(dq),
+                     reinterpret_cast<const short*>(out), seq, scale,
                      causal ? 1 : 0, heads, in_strides[0], in_strides[1],
                      in_strides[2], do_strides[0], do_strides[1],
                      do_strides[2], g_strides[0], g_strides[1],
-                     g_strides[2]);
+                     g_strides[2], o_strides[0], o_strides[1],
+                     o_strides[2]);
 }
 
 }  // extern "C"
