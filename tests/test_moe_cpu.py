@@ -68,3 +68,41 @@ def test_moe_ep2_matches_serial():
     ep = run_multiprocess(_moe_worker, world=2)
     for a, b in zip(serial, ep[0][0]):
         assert abs(a - b) < 1e-4, (serial, ep[0][0])
+
+
+def _moe_dp2ep2_worker(rank, world):
+    """2 replicas x EP2 on 4 ranks: expert shards sync across replicas,
+    gate syncs across all 4."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models.moe_transformer import (
+        build_moe_transformer)
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    torch.manual_seed(30)
+    model = build_moe_transformer(world=2, layers=2, hidden=32, heads=4,
+                                  ffn=64, num_experts=4, vocab_size=128,
+                                  max_pos=32)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=1e-3)
+    torch.manual_seed(31)
+    ids = torch.randint(0, 128, (4, 16))
+    tgt = torch.randint(0, 128, (4, 16))
+    losses = [float(engine.train_step(ids, tgt)) for _ in range(3)]
+    return (losses, engine.replica_id,
+            model.blocks[0].moe.gate.weight.detach().clone(),
+            model.blocks[0].moe.w1.detach().clone())
+
+
+def test_moe_dp2_ep2():
+    res = run_multiprocess(_moe_dp2ep2_worker, world=4, timeout=240)
+    losses = [r[0] for r in res]
+    assert losses[0] == losses[1] == losses[2] == losses[3]
+    # gate identical everywhere
+    for r in res[1:]:
+        assert torch.allclose(res[0][2], r[2])
+    # expert shards: same position across replicas identical, across
+    # positions different.  ranks 0,1 = replica 0 pos 0,1; 2,3 = replica 1
+    assert torch.allclose(res[0][3], res[2][3])   # pos 0 across replicas
+    assert torch.allclose(res[1][3], res[3][3])   # pos 1 across replicas
+    assert not torch.allclose(res[0][3], res[1][3])

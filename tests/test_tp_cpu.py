@@ -103,3 +103,38 @@ def test_shard_sizes_remainder():
     # remainder goes to shard 0 (reference distributed_dense.py:102-108)
     assert [shard_size(10, 3, s) for s in range(3)] == [4, 3, 3]
     assert [shard_offset(10, 3, s) for s in range(3)] == [0, 4, 7]
+
+
+def _tp_embedding_worker(rank, world):
+    """Vocab-sharded embedding inside a split scope matches the full
+    embedding."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    torch.manual_seed(60)
+    with epl.replicate(world, name="trunk"):
+        trunk = nn.Linear(8, 8)
+    with epl.split(world, name="emb"):
+        emb = nn.Embedding(11, 8)   # odd vocab: remainder shard policy
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb, self.trunk = emb, trunk
+
+        def forward(self, ids):
+            return self.trunk(torch.tanh(self.emb(ids)))
+
+    engine = epl.Engine(M(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(61)
+    ids = torch.randint(0, 11, (4, 6))
+    y = torch.randn(4, 6, 8)
+    return [float(engine.train_step(ids, y)) for _ in range(3)]
+
+
+def test_tp_embedding_sharded():
+    res = run_multiprocess(_tp_embedding_worker, world=2)
+    assert res[0] == res[1]
+    assert res[0][-1] < res[0][0]
