@@ -95,3 +95,18 @@ def test_pp2_dp2_runs():
     # last-stage ranks (1 and 3) report identical losses (same data/seed)
     assert pp[1] == pp[3]
     assert pp[1][-1] < pp[1][0]
+
+
+def test_pp2_fewer_microbatches_than_stages():
+    """M=1 < S=2: warmup clamps; still matches serial."""
+    serial = run_multiprocess(_serial_worker, world=1, args=(1,))[0]
+    pp = run_multiprocess(_pipeline_worker, world=2,
+                          args=("prefer_backward", 1))
+    assert all(abs(a - b) < 1e-5 for a, b in zip(serial, pp[1]))
+
+
+def test_pp2_odd_microbatches():
+    serial = run_multiprocess(_serial_worker, world=1, args=(3,))[0]
+    pp = run_multiprocess(_pipeline_worker, world=2,
+                          args=("prefer_backward", 3))
+    assert all(abs(a - b) < 1e-5 for a, b in zip(serial, pp[1]))
