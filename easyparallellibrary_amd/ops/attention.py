@@ -41,15 +41,17 @@ class _FlashAttention(torch.autograd.Function):
 
     @staticmethod
     def backward(ctx, dout):
+        import os
         q, k, v, out, lse = ctx.saved_tensors
         # the bwd kernels re-stage q/k/v/dout tiles every 32-row iteration;
         # strided rows (6 KB apart in the qkv views) measured +405us/call
         # vs 4 contiguization copies at ~43us each — copy for backward only
         # (forward shows no strided penalty)
         orig_q = q
-        if q.stride(2) != 64:
+        contig = os.environ.get("EPL_ATTN_BWD_CONTIG", "0") == "1"
+        if contig and q.stride(2) != 64:
             q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
-        if not _kernel_ok(dout) or dout.stride(2) != 64:
+        if not _kernel_ok(dout) or (contig and dout.stride(2) != 64):
             dout = dout.contiguous()
         # write grads in the INPUT layout: the qkv-unbind/transpose
         # backward then stacks matching-layout chunks (fast memcpy)
