@@ -104,3 +104,34 @@ def test_grouped_apply_config():
     l0 = float(engine.train_step(x, y))
     l1 = float(engine.train_step(x, y))
     assert l1 < l0
+
+
+def test_launcher_runs_dp_training():
+    """End-to-end: epl-launch spawns 2 workers that train a DP model."""
+    import os
+    import tempfile
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    script = """
+import sys
+sys.path.insert(0, {!r})
+import torch, torch.nn as nn
+import easyparallellibrary_amd as epl""".format(repo) + """
+epl.init()
+torch.manual_seed(1)
+with epl.replicate(1):
+    m = nn.Linear(4, 2)
+engine = epl.Engine(m, loss_fn=nn.MSELoss(), lr=1e-2)
+x = torch.randn(4, 4); y = torch.randn(4, 2)
+l0 = float(engine.train_step(x, y))
+l1 = float(engine.train_step(x, y))
+assert l1 < l0, (l0, l1)
+assert engine.world_size == 2
+print("worker ok", engine.rank)
+"""
+    with tempfile.NamedTemporaryFile("w", suffix=".py", delete=False) as f:
+        f.write(script)
+        path = f.name
+    rc = subprocess.call(
+        [sys.executable, "-m", "easyparallellibrary_amd.launcher",
+         "--num_workers", "2", "--gpu_per_worker", "0", path])
+    assert rc == 0
