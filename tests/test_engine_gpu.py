@@ -45,3 +45,37 @@ def test_lamb_gpu_step_finite():
     l1 = float(engine.train_step(ids, tgt))
     torch.cuda.synchronize()
     assert torch.isfinite(torch.tensor([l0, l1])).all()
+
+
+def test_checkpoint_roundtrip_gpu(tmp_path):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models import bert
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    epl.init()
+    torch.manual_seed(3)
+    model = bert.build_bert(dict(layers=1, hidden=128, heads=2, ffn=256),
+                            vocab_size=512, max_pos=64)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=1e-3, dtype=torch.bfloat16)
+    ids, tgt = bert.synthetic_mlm_batch(4, 64, 512, device=engine.device,
+                                        seed=5)
+    engine.train_step(ids, tgt)
+    engine.save_checkpoint(str(tmp_path))
+    cont = [float(engine.train_step(ids, tgt)) for _ in range(2)]
+
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    hooks.remove_hooks()
+    Env._instance = None
+    epl.init()
+    torch.manual_seed(99)  # different init; restore must override
+    model2 = bert.build_bert(dict(layers=1, hidden=128, heads=2, ffn=256),
+                             vocab_size=512, max_pos=64)
+    engine2 = epl.Engine(model2, loss_fn=ParallelCrossEntropy(),
+                         optimizer="adamw", lr=1e-3, dtype=torch.bfloat16)
+    engine2.load_checkpoint(str(tmp_path))
+    resumed = [float(engine2.train_step(ids, tgt)) for _ in range(2)]
+    torch.cuda.synchronize()
+    assert all(abs(a - b) < 1e-3 for a, b in zip(cont, resumed)), (
+        cont, resumed)

@@ -44,13 +44,13 @@ def test_fused_adamw_matches_torch():
     assert torch.allclose(pbf.float(), ref_p, atol=1e-2, rtol=1e-2)
 
 
-@pytest.mark.parametrize("dtype,cols", [
-    (torch.bfloat16, 1024), (torch.float32, 1024),
-    (torch.bfloat16, 512), (torch.bfloat16, 1600),
-    (torch.bfloat16, 2048), (torch.bfloat16, 4096)])
-def test_layer_norm_fwd_bwd(dtype, cols):
+@pytest.mark.parametrize("dtype,cols,rows", [
+    (torch.bfloat16, 1024, 512), (torch.float32, 1024, 512),
+    (torch.bfloat16, 512, 512), (torch.bfloat16, 1600, 512),
+    (torch.bfloat16, 2048, 512), (torch.bfloat16, 4096, 512),
+    (torch.bfloat16, 1024, 3), (torch.bfloat16, 1024, 8191)])
+def test_layer_norm_fwd_bwd(dtype, cols, rows):
     torch.manual_seed(1)
-    rows = 512
     x = torch.randn(rows, cols, device=dev(), dtype=dtype)
     gamma = torch.randn(cols, device=dev(), dtype=dtype)
     beta = torch.randn(cols, device=dev(), dtype=dtype)
