@@ -48,7 +48,7 @@ class FlatParamGroup:
     """
 
     def __init__(self, params, device, model_dtype=None, grad_dtype=None,
-                 pad_to_multiple=1):
+                 pad_to_multiple=1, master_shard=None):
         self.params = [p for p in params if p.requires_grad]
         assert self.params, "empty parameter group"
         if model_dtype is None:
@@ -84,8 +84,19 @@ class FlatParamGroup:
             p.grad = self.grad_arena[off:off + n].view(p.shape)
             self._grad_views[id(p)] = p.grad
 
+        # Under ZeRO the fp32 master only ever needs to cover this rank's
+        # shard — allocate just that slice (1/W of the fp32 memory) instead
+        # of materializing the full master and slicing later.
+        self.master_lo, self.master_hi = 0, total
         if model_dtype == torch.float32:
             self.master_arena = self.param_arena
+        elif master_shard is not None and master_shard[1] > 1:
+            r, w = master_shard
+            assert total % w == 0, (total, w)
+            s = total // w
+            self.master_lo, self.master_hi = r * s, (r + 1) * s
+            self.master_arena = self.param_arena[
+                self.master_lo:self.master_hi].to(torch.float32)
         else:
             self.master_arena = self.param_arena.to(torch.float32)
 
@@ -128,7 +139,14 @@ class FlatParamGroup:
     def sync_master_to_params(self):
         """After a master-arena update without the fused kernel's writeback."""
         if self.master_arena is not self.param_arena:
-            self.param_arena.copy_(self.master_arena.to(self.model_dtype))
+            self.param_arena[self.master_lo:self.master_hi].copy_(
+                self.master_arena.to(self.model_dtype))
+
+    def refresh_master(self):
+        """Re-derive the fp32 master from (possibly freshly loaded) params."""
+        if self.master_arena is not self.param_arena:
+            self.master_arena.copy_(self.param_arena[
+                self.master_lo:self.master_hi].to(torch.float32))
 
 
 class GradReducer:

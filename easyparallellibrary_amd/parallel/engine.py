@@ -240,7 +240,10 @@ class Engine:
                     continue
                 fg = FlatParamGroup(
                     params, self.device, model_dtype=self.dtype,
-                    pad_to_multiple=len(granks) if zero_level else 1)
+                    pad_to_multiple=len(granks) if zero_level else 1,
+                    master_shard=((bcomm.rank, bcomm.size)
+                                  if zero_level and bcomm is not None
+                                  else None))
                 reducer = GradReducer(
                     fg, pool, bucket_bytes, reduce_method=reduce_method,
                     overlap=overlap,
@@ -257,8 +260,7 @@ class Engine:
         for fg, bcomm in self._bcast_jobs:
             if bcomm is not None and bcomm.size > 1:
                 bcomm.broadcast(fg.param_arena, root=0)
-                if fg.master_arena is not fg.param_arena:
-                    fg.master_arena.copy_(fg.param_arena.to(torch.float32))
+                fg.refresh_master()
 
         # ---- optimizer -------------------------------------------------------
         okw = dict(optimizer_kwargs or {})

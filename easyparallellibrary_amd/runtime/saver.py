@@ -115,17 +115,19 @@ class ShardingLoader:
                 self._load_split_tg(tg, pos, len(ranks), strict)
             else:
                 self._load_replicated_tg(tg, strict)
+        # params were loaded into module tensors == arena views; refresh
+        # fp32 masters from the arenas FIRST — if optimizer state follows,
+        # it restores the exact fp32 master (higher precision than the
+        # bf16 params) and re-syncs params from it, so refreshing after
+        # the optimizer load would round the master through bf16.
+        for fg in engine.flat_groups:
+            fg.refresh_master()
         if load_optimizer:
             opt_path = os.path.join(
                 self.path, "opt_rank{}.pt".format(engine.rank))
             if os.path.exists(opt_path):
                 engine.optimizer.load_state_dict(
                     torch.load(opt_path, weights_only=False))
-        # params were loaded into module tensors == arena views; refresh
-        # fp32 masters from the arenas
-        for fg in engine.flat_groups:
-            if fg.master_arena is not fg.param_arena:
-                fg.master_arena.copy_(fg.param_arena.to(torch.float32))
         _barrier(engine)
 
     def _load_replicated_tg(self, tg, strict):

@@ -30,7 +30,12 @@ class _ShardView:
     def __init__(self, fg, lo, hi):
         self.fg = fg
         self.lo, self.hi = lo, hi
-        self.master_arena = fg.master_arena[lo:hi]
+        # the fg's master may itself already be the shard (allocated
+        # shard-sized at construction); index relative to its origin
+        assert fg.master_lo <= lo and hi <= fg.master_hi, (
+            lo, hi, fg.master_lo, fg.master_hi)
+        self.master_arena = fg.master_arena[lo - fg.master_lo:
+                                            hi - fg.master_lo]
         self.param_arena = fg.param_arena[lo:hi]
         self.grad_arena = fg.grad_arena[lo:hi]
         self.state = {}

@@ -62,6 +62,8 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
                                         seed=5)
     engine.train_step(ids, tgt)
     engine.save_checkpoint(str(tmp_path))
+    saved_params = engine.flat_groups[0].param_arena.detach().clone()
+    saved_master = engine.flat_groups[0].master_arena.detach().clone()
     cont = [float(engine.train_step(ids, tgt)) for _ in range(2)]
 
     from easyparallellibrary_amd.env import Env
@@ -75,7 +77,12 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
     engine2 = epl.Engine(model2, loss_fn=ParallelCrossEntropy(),
                          optimizer="adamw", lr=1e-3, dtype=torch.bfloat16)
     engine2.load_checkpoint(str(tmp_path))
+    # restore is exact: params and fp32 master match the snapshot
+    assert torch.equal(engine2.flat_groups[0].param_arena, saved_params)
+    assert torch.equal(engine2.flat_groups[0].master_arena, saved_master)
     resumed = [float(engine2.train_step(ids, tgt)) for _ in range(2)]
     torch.cuda.synchronize()
-    assert all(abs(a - b) < 1e-3 for a, b in zip(cont, resumed)), (
+    # loss trajectory only loosely: bf16 + atomics-based attention
+    # backward is not bit-deterministic across runs
+    assert all(abs(a - b) < 5e-2 for a, b in zip(cont, resumed)), (
         cont, resumed)
