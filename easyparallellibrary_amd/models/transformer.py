@@ -66,9 +66,17 @@ class SelfAttention(nn.Module):
         self.dropout = dropout
 
     def forward(self, x):
-        from easyparallellibrary_amd.ops.attention import flash_attention
+        from easyparallellibrary_amd.ops.attention import (
+            flash_attention, qkv_flash_attention, qkv_native_ok)
         b, s, h = x.shape
         qkv = self.qkv(x).reshape(b, s, 3, self.num_heads, self.head_dim)
+        if (_USE_NATIVE_ATTN and qkv_native_ok(qkv)
+                and not (self.dropout and self.training)):
+            # fully fused: kernels read the qkv views and write d_qkv
+            # slices directly — no unbind/stack copies at all
+            o = qkv_flash_attention(qkv, causal=self.causal)
+            o = o.transpose(1, 2).reshape(b, s, h)
+            return self.proj(o)
         if _QKV_SPLIT:
             q, k, v = _QKVSplit.apply(qkv)
         else:

@@ -73,6 +73,64 @@ class _FlashAttention(torch.autograd.Function):
         return dq, dk, dv, None, None
 
 
+class _QKVFlashAttention(torch.autograd.Function):
+    """Module-level fused path: consumes the [B,S,3,H,D] qkv projection
+    buffer directly.  Forward feeds the kernels the q/k/v strided views
+    (no unbind copies); backward allocates ONE d_qkv buffer and the
+    kernels write dq/dk/dv straight into its slices (they take arbitrary
+    shared [B,H,S] stride triplets) — no stack/CatArrayBatchedCopy, no
+    3x-oversized ``empty_strided`` transients."""
+
+    @staticmethod
+    def forward(ctx, qkv, causal, scale):
+        b, s, three, h, d = qkv.shape
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        out = torch.empty(b, s, h, d, dtype=qkv.dtype,
+                          device=qkv.device).permute(0, 2, 1, 3)
+        lse = torch.empty(b * h * s, dtype=torch.float32, device=qkv.device)
+        native_ext().attn_fwd(q, k, v, out, lse, scale, causal)
+        ctx.save_for_backward(qkv, out, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, out, lse = ctx.saved_tensors
+        q = qkv[:, :, 0].transpose(1, 2)
+        k = qkv[:, :, 1].transpose(1, 2)
+        v = qkv[:, :, 2].transpose(1, 2)
+        if not _kernel_ok(dout):
+            dout = dout.contiguous()
+        dqkv = torch.empty_like(qkv)
+        dq = dqkv[:, :, 0].transpose(1, 2)
+        dk = dqkv[:, :, 1].transpose(1, 2)
+        dv = dqkv[:, :, 2].transpose(1, 2)
+        delta = torch.empty_like(lse)
+        native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
+                              ctx.scale, ctx.causal, True)
+        return dqkv, None, None
+
+
+def qkv_flash_attention(qkv, causal=False, scale=None):
+    """qkv: [B, S, 3, H, D] (the fused projection output).  Returns
+    attention output as a [B, H, S, D] view whose memory order is
+    [B, S, H, D] (the caller's transpose+reshape is free).  Requires the
+    native kernels (bf16, D=64); callers fall back to the split-views +
+    SDPA path otherwise."""
+    if scale is None:
+        scale = qkv.shape[-1] ** -0.5
+    return _QKVFlashAttention.apply(qkv, causal, scale)
+
+
+def qkv_native_ok(qkv):
+    return (use_native(qkv) and qkv.dtype == torch.bfloat16
+            and qkv.dim() == 5 and qkv.shape[-1] == 64
+            and qkv.is_contiguous())
+
+
 def flash_attention(q, k, v, causal=False, scale=None, allow_native=True):
     """q,k,v: [B, H, S, D].  Native kernel when bf16/D=64 on GPU; torch
     SDPA otherwise."""

@@ -85,3 +85,27 @@ for split in ("0", "1", "0", "1"):
     t = bench(fb, iters=20)
     print(f"split={split}  f+b {t:8.1f} us")
 os.environ.pop("EPL_ATTN_BWD_SPLIT", None)
+
+print("--- module-level qkv paths (BERT shape, fwd+bwd) ---")
+from easyparallellibrary_amd.ops.attention import qkv_flash_attention
+b, s, h, d = 128, 512, 16, 64
+scale = 1/math.sqrt(d)
+qkv = torch.randn(b, s, 3, h, d, device="cuda", dtype=torch.bfloat16,
+                  requires_grad=True)
+dout_bshd = torch.randn(b, s, h, d, device="cuda", dtype=torch.bfloat16)
+def fused_fb():
+    qkv.grad = None
+    out = qkv_flash_attention(qkv, causal=False)
+    out.backward(dout_bshd.permute(0, 2, 1, 3))
+for _ in range(2):
+    t = bench(fused_fb, iters=20)
+    print(f"qkv fused native f+b {t:8.1f} us")
+def sdpa_split_fb():
+    qkv.grad = None
+    from easyparallellibrary_amd.models.transformer import _QKVSplit
+    q, k, v = _QKVSplit.apply(qkv)
+    out = F.scaled_dot_product_attention(q, k, v, is_causal=False, scale=scale)
+    out.backward(dout_bshd.transpose(1, 2))
+for _ in range(2):
+    t = bench(sdpa_split_fb, iters=20)
+    print(f"sdpa + qkvsplit  f+b {t:8.1f} us")

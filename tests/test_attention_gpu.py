@@ -117,3 +117,32 @@ def test_flash_attention_bwd(causal):
         err = (got.float() - want).abs().max().item()
         rel = err / want.abs().max().clamp_min(1e-6).item()
         assert err < 0.1 or rel < 5e-2, (name, err, rel)
+
+
+@pytest.mark.parametrize("causal", [False, True])
+def test_qkv_flash_attention_fused(causal):
+    """Module-level fused path: kernels read the qkv buffer views and
+    write d_qkv slices directly."""
+    torch.manual_seed(4)
+    import easyparallellibrary_amd as epl
+    epl.init()
+    from easyparallellibrary_amd.ops.attention import qkv_flash_attention
+    b, s, h, d = 2, 256, 4, 64
+    qkv = torch.randn(b, s, 3, h, d, device="cuda", dtype=torch.bfloat16,
+                      requires_grad=True)
+    out = qkv_flash_attention(qkv, causal=causal)
+    dout = torch.randn_like(out)
+    out.backward(dout)
+
+    ref_in = qkv.detach().float().requires_grad_(True)
+    qf = ref_in[:, :, 0].transpose(1, 2)
+    kf = ref_in[:, :, 1].transpose(1, 2)
+    vf = ref_in[:, :, 2].transpose(1, 2)
+    ref = ref_attention(qf, kf, vf, causal, d ** -0.5)
+    ref.backward(dout.float())
+    torch.cuda.synchronize()
+    err = (out.float() - ref).abs().max().item()
+    assert err < 3e-2, err
+    gerr = (qkv.grad.float() - ref_in.grad).abs().max().item()
+    grel = gerr / ref_in.grad.abs().max().clamp_min(1e-6).item()
+    assert gerr < 0.1 or grel < 5e-2, (gerr, grel)
