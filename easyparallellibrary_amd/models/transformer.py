@@ -13,12 +13,16 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-# Gate for the hand-written attention kernels (EPL_NATIVE_ATTENTION=1).
-# Numerics-verified (tests/test_attention_gpu.py) and within ~4% of the
-# AOTriton SDPA per-op on contiguous inputs, but the strided-qkv module
-# path measures ~6% slower end-to-end (A/B in profiles/) — SDPA stays the
-# default until the strided staging path is tuned.
-_USE_NATIVE_ATTN = os.environ.get("EPL_NATIVE_ATTENTION", "0") == "1"
+# Hand-written attention kernels (tests/test_attention_gpu.py).  The
+# fully fused qkv path (kernels read the qkv buffer views, write d_qkv
+# slices directly) beats SDPA+assembly per-op on non-causal BERT shapes
+# (2225 vs 2263 us f+b; profiles/r01_fused_qkv_ab.txt) and ties it
+# end-to-end, so it is the DEFAULT for non-causal attention.  Causal
+# shapes stay on SDPA (AOTriton's causal kernel is still ahead, 1468 vs
+# 1751 us) until the triangle-skipping rework (NOTES.md item 1b).
+# EPL_NATIVE_ATTENTION: "auto" (default) = native for non-causal only;
+# "1" = native everywhere; "0" = SDPA everywhere.
+_NATIVE_ATTN = os.environ.get("EPL_NATIVE_ATTENTION", "auto")
 
 from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
 from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
@@ -70,7 +74,9 @@ class SelfAttention(nn.Module):
             flash_attention, qkv_flash_attention, qkv_native_ok)
         b, s, h = x.shape
         qkv = self.qkv(x).reshape(b, s, 3, self.num_heads, self.head_dim)
-        if (_USE_NATIVE_ATTN and qkv_native_ok(qkv)
+        native_here = (_NATIVE_ATTN == "1"
+                       or (_NATIVE_ATTN == "auto" and not self.causal))
+        if (native_here and qkv_native_ok(qkv)
                 and not (self.dropout and self.training)):
             # fully fused: kernels read the qkv views and write d_qkv
             # slices directly — no unbind/stack copies at all
@@ -89,7 +95,7 @@ class SelfAttention(nn.Module):
                 q, k, v, is_causal=self.causal, dropout_p=self.dropout)
         else:
             o = flash_attention(q, k, v, causal=self.causal,
-                                allow_native=_USE_NATIVE_ATTN)
+                                allow_native=(_NATIVE_ATTN == "1"))
         o = o.transpose(1, 2).reshape(b, s, h)
         return self.proj(o)
 

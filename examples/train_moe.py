@@ -1,6 +1,12 @@
 #!/usr/bin/env python3
 """Expert-parallel MoE transformer (experts split across all ranks)."""
 import os
+import sys
+
+# allow running as a plain script from anywhere in the repo
+sys.path.insert(0, os.path.dirname(os.path.dirname(
+    os.path.abspath(__file__))))
+
 import torch
 import easyparallellibrary_amd as epl
 from easyparallellibrary_amd.models.moe_transformer import build_moe_transformer
