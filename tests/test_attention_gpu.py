@@ -39,7 +39,7 @@ def test_mfma_fragment_layout():
 
 
 @pytest.mark.parametrize("causal", [False, True])
-@pytest.mark.parametrize("seq", [128, 512, 200])
+@pytest.mark.parametrize("seq", [128, 512, 200, 64, 72])
 def test_flash_attention_fwd(causal, seq):
     torch.manual_seed(1)
     import easyparallellibrary_amd as epl
@@ -88,12 +88,13 @@ def test_flash_attention_strided_qkv_views():
 
 
 @pytest.mark.parametrize("causal", [False, True])
-def test_flash_attention_bwd(causal):
+@pytest.mark.parametrize("seq", [256, 64, 72])
+def test_flash_attention_bwd(causal, seq):
     torch.manual_seed(2)
     import easyparallellibrary_amd as epl
     epl.init()
     from easyparallellibrary_amd.ops.attention import flash_attention
-    b, h, seq, d = 2, 4, 256, 64
+    b, h, d = 2, 4, 64
     q = torch.randn(b, h, seq, d, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
     k = torch.randn(b, h, seq, d, device="cuda", dtype=torch.bfloat16,
@@ -120,14 +121,15 @@ def test_flash_attention_bwd(causal):
 
 
 @pytest.mark.parametrize("causal", [False, True])
-def test_qkv_flash_attention_fused(causal):
+@pytest.mark.parametrize("s", [256, 64])
+def test_qkv_flash_attention_fused(causal, s):
     """Module-level fused path: kernels read the qkv buffer views and
     write d_qkv slices directly."""
     torch.manual_seed(4)
     import easyparallellibrary_amd as epl
     epl.init()
     from easyparallellibrary_amd.ops.attention import qkv_flash_attention
-    b, s, h, d = 2, 256, 4, 64
+    b, h, d = 2, 4, 64
     qkv = torch.randn(b, s, 3, h, d, device="cuda", dtype=torch.bfloat16,
                       requires_grad=True)
     out = qkv_flash_attention(qkv, causal=causal)
