@@ -1,9 +1,13 @@
 """IO slicing, metric merging, launcher, AMP scaler units."""
 
+import os
 import subprocess
 import sys
 
+import pytest
 import torch
+
+REPO_ROOT = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 from easyparallellibrary_amd.utils.io_slicing import (slice_dataset_indices,
                                                       slice_files)
@@ -135,3 +139,18 @@ print("worker ok", engine.rank)
         [sys.executable, "-m", "easyparallellibrary_amd.launcher",
          "--num_workers", "2", "--gpu_per_worker", "0", path])
     assert rc == 0
+
+
+@pytest.mark.parametrize("script", ["train_bert_dp.py",
+                                    "train_bert_pipeline.py",
+                                    "train_moe.py"])
+def test_examples_tiny_cpu(script):
+    """Every example runs end-to-end in tiny mode on CPU."""
+    import subprocess
+    import sys
+    env = dict(os.environ, EPL_EXAMPLE_TINY="1")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO_ROOT, "examples", script)],
+        env=env, capture_output=True, text=True, timeout=240)
+    assert r.returncode == 0, r.stderr[-2000:]
+    assert "step 9" in r.stdout, r.stdout[-500:]
