@@ -29,29 +29,39 @@ class PipelineRuntime:
         self.S = engine.num_stages
         self.s = engine.my_stage
         replica = engine.replica_id
-        # rank of each stage within my replica (device_count=1 per stage)
-        self.stage_ranks = [
-            tg.virtual_device.local_ranks(replica)[0]
-            for tg in engine.stage_tgs
-        ]
+        # Stages may span k>1 ranks each (e.g. EP-MoE or TP inside a
+        # stage).  Position p of stage s exchanges activations with
+        # position p of stage s±1 — k independent pipeline chains per
+        # replica.  All stages must share one width.
+        rank_lists = [tg.virtual_device.local_ranks(replica)
+                      for tg in engine.stage_tgs]
+        widths = {len(r) for r in rank_lists}
+        if len(widths) != 1:
+            raise NotImplementedError(
+                "pipeline stages must share one device count per replica; "
+                "got widths {}".format(sorted(widths)))
+        self.width = widths.pop()
+        self.pos = rank_lists[self.s].index(engine.rank)
+        self.stage_ranks = [r[self.pos] for r in rank_lists]
         self.prev_rank = self.stage_ranks[self.s - 1] if self.s > 0 else None
         self.next_rank = (self.stage_ranks[self.s + 1]
                           if self.s < self.S - 1 else None)
-        # one p2p communicator per replica chain.  gloo group creation is
-        # collective over ALL ranks, so every rank walks every replica's
+        # one p2p communicator per (replica, position) chain.  gloo group
+        # creation is collective over ALL ranks, so every rank walks every
         # chain in the same order and keeps its own.
         self.comm = None
         for rep in range(engine.num_replicas):
-            chain = [tg.virtual_device.local_ranks(rep)[0]
-                     for tg in engine.stage_tgs]
-            mine = engine.rank in chain
-            if not mine and (torch.cuda.is_available()
-                             or not dist.is_initialized()):
-                continue
-            comm = create_communicator(
-                "{}_pipe_rep{}".format(engine._ns, rep), chain)
-            if mine:
-                self.comm = comm
+            for p in range(self.width):
+                chain = [tg.virtual_device.local_ranks(rep)[p]
+                         for tg in engine.stage_tgs]
+                mine = engine.rank in chain
+                if not mine and (torch.cuda.is_available()
+                                 or not dist.is_initialized()):
+                    continue
+                comm = create_communicator(
+                    "{}_pipe_rep{}_p{}".format(engine._ns, rep, p), chain)
+                if mine:
+                    self.comm = comm
         assert self.comm is not None
         self.group_rank = {r: i for i, r in enumerate(self.stage_ranks)}
         self.schedule = engine.config.pipeline.strategy
