@@ -13,7 +13,8 @@ SOFTWARE PIPELINE over 64 MB arena slices:
   D2H stream:  grad chunk i+1 (device fp32 convert + hipMemcpyAsync)
   CPU:         torch._fused_adamw_ on chunk i (one multithreaded pass,
                ~7x the eager op chain)
-  H2D stream:  updated bf16 params of chunk i-1 (pinned staging buffer)
+  H2D stream:  updated params of chunk i-1 (direct pinned-fp32 ->
+               device-bf16 copy; conversion runs on the GPU)
 so copies hide under compute and the optimizer wall-time approaches the
 fused-CPU-kernel time alone.  State is 12 bytes/param on host; HBM holds
 only bf16 params/grads.
@@ -49,9 +50,6 @@ class CPUOffloadAdamW:
             g.state["master_cpu"].copy_(g.master_arena.to("cpu"))
             g.state["grad_cpu"] = torch.empty(
                 g.total, dtype=torch.float32, pin_memory=self.pin)
-            # pinned bf16 staging for the H2D leg of the pipeline
-            g.state["param_cpu"] = torch.empty(
-                g.total, dtype=g.param_arena.dtype, pin_memory=self.pin)
             # free the device-side fp32 master: CPU owns the truth now
             if g.master_arena is not g.param_arena:
                 g.master_arena = g.state["master_cpu"]
@@ -85,7 +83,6 @@ class CPUOffloadAdamW:
             gc = g.state["grad_cpu"]
             m, v = g.state["exp_avg"], g.state["exp_avg_sq"]
             w = g.state["master_cpu"]
-            pc = g.state["param_cpu"]
             spans = [(lo, min(lo + _CHUNK, g.total))
                      for lo in range(0, g.total, _CHUNK)]
             if not g.grad_arena.is_cuda or self._d2h is None:
@@ -112,9 +109,11 @@ class CPUOffloadAdamW:
                 ev.synchronize()
                 self._chunk_update(w[lo:hi], gc[lo:hi], m[lo:hi],
                                    v[lo:hi], grad_scale)
-                pc[lo:hi].copy_(w[lo:hi])  # fp32 -> bf16 into pinned
+                # direct pinned-fp32 -> device-bf16 copy: the transfer
+                # stays async and the dtype conversion runs on the GPU,
+                # keeping the (DRAM-bound) CPU free for the next chunk
                 with torch.cuda.stream(self._h2d):
-                    g.param_arena[lo:hi].copy_(pc[lo:hi],
+                    g.param_arena[lo:hi].copy_(w[lo:hi],
                                                non_blocking=True)
             torch.cuda.current_stream().wait_stream(self._h2d)
 
