@@ -208,6 +208,12 @@ class GradReducer:
         for bi, (_, _, ps, _) in enumerate(self.buckets):
             for p in ps:
                 self._buckets_of_param.setdefault(id(p), []).append(bi)
+        # bucket-composition log (reference: coalescing.py:336-353)
+        logger.debug(
+            "grad buckets: %d over %.1f MB arena (%s): %s", len(self.buckets),
+            group.total * esize / 1e6,
+            "reduce-to-owner" if shard_owners else self.reduce_method,
+            [(b[1] - b[0]) * esize // 1024 for b in self.buckets])
         self._pending = [len(ps) for (_, _, ps, _) in self.buckets]
         self._launched = [False] * len(self.buckets)
         self._decompress = []
