@@ -94,7 +94,11 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int lq = lane & 31;
   const int64_t bh = blockIdx.y;
   const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
-  const int64_t q0_blk = (int64_t)blockIdx.x * 128;
+  // causal: launch the LONGEST q-blocks first (work grows with q0) so
+  // the dispatch tail is short blocks, not 128-kv-tile ones
+  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
+                            : (int64_t)blockIdx.x;
+  const int64_t q0_blk = qb * 128;
   const int64_t q0 = q0_blk + wave * 32;
   const bool active = q0 < seq;
   const short* qp = q + boff;
@@ -587,7 +591,10 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   const int hi = lane >> 5;
   const int lq = lane & 31;
   const int64_t bh = blockIdx.y;
-  const int64_t q0_blk = (int64_t)blockIdx.x * 128;
+  // causal: longest q-blocks first (see attn_fwd_kernel)
+  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
+                            : (int64_t)blockIdx.x;
+  const int64_t q0_blk = qb * 128;
   const int64_t q0 = q0_blk + wave * 32;
   const bool active = q0 < seq;
   const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
