@@ -41,6 +41,8 @@ void epl_scale(void*, int64_t, float, bool, hipStream_t);
 void epl_f32_to_bf16(unsigned short*, const float*, int64_t, hipStream_t);
 void epl_bf16_to_f32(float*, const unsigned short*, int64_t, hipStream_t);
 void epl_sqnorm(const void*, int64_t, float*, bool, hipStream_t);
+void epl_colsum(void*, const void*, float*, int64_t, int64_t, int64_t,
+                bool, hipStream_t);
 void run_mfma_probe(const unsigned short*, const unsigned short*, float*,
                     hipStream_t);
 void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
@@ -331,6 +333,19 @@ void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {
                  D.data_ptr<float>(), cur_stream());
 }
 
+void colsum(at::Tensor dy, at::Tensor db, at::Tensor partial) {
+  const bool bf16 = is_bf16(dy);
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 2,
+              "dy must be a contiguous 2-D device tensor");
+  const int64_t rows = dy.size(0), cols = dy.size(1);
+  TORCH_CHECK(cols % 8 == 0, "cols must be a multiple of 8");
+  TORCH_CHECK(db.numel() == cols && db.scalar_type() == dy.scalar_type());
+  const int64_t stripes = partial.numel() / cols;
+  TORCH_CHECK(stripes >= 1 && partial.scalar_type() == at::kFloat);
+  epl_colsum(db.data_ptr(), dy.data_ptr(), partial.data_ptr<float>(), rows,
+             cols, stripes, bf16, cur_stream());
+}
+
 void sqnorm(at::Tensor t, at::Tensor out) {
   const bool bf16 = is_bf16(t);
   check(out, at::kFloat, "out");
@@ -344,6 +359,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.doc() = "MI355X-native parallel library: RCCL comm core + CDNA4 kernels";
   epl::register_comm(m);
   m.def("fused_adamw", &fused_adamw);
+  m.def("colsum", &colsum);
   m.def("lamb_phase1", &lamb_phase1);
   m.def("lamb_phase2", &lamb_phase2);
   m.def("layer_norm_fwd", &layer_norm_fwd);

@@ -915,6 +915,57 @@ __global__ void sqnorm_kernel(const void* __restrict__ p, int64_t n,
 }  // namespace
 
 // ============================================================================
+
+// ============================================================================
+// Column sum (Linear bias gradient): db[c] = sum_r dy[r, c].
+// torch's generic reduce runs at ~2.2 TB/s on [65536, 1024] bf16; this
+// pair streams bf16x8 rows into fp32 stripe partials (no atomics in the
+// hot pass, same pattern as the LN backward) and a tiny reduce kernel
+// folds the stripes.  ~73 of these per BERT-Large step (qkv/proj/fc2
+// biases).
+// ============================================================================
+template <bool BF16>
+__global__ void colsum_partial_kernel(const void* __restrict__ dy,
+                                      float* __restrict__ partial,
+                                      int64_t rows, int64_t cols,
+                                      int64_t stripes) {
+  // grid: (ceil(cols/(256*8)), stripes); each thread owns 8 columns
+  const int64_t c = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  if (c >= cols) return;
+  const int64_t per = (rows + stripes - 1) / stripes;
+  const int64_t r0 = (int64_t)blockIdx.y * per;
+  const int64_t r1 = r0 + per < rows ? r0 + per : rows;
+  float8 acc = {};
+  for (int64_t r = r0; r < r1; ++r) {
+    if (BF16) {
+      const float8 v =
+          load_bf16x8(reinterpret_cast<const unsigned short*>(dy) +
+                      r * cols + c);
+#pragma unroll
+      for (int k = 0; k < 8; ++k) acc.v[k] += v.v[k];
+    } else {
+      const float* p = reinterpret_cast<const float*>(dy) + r * cols + c;
+#pragma unroll
+      for (int k = 0; k < 8; ++k) acc.v[k] += p[k];
+    }
+  }
+  store_f32x8(partial + (int64_t)blockIdx.y * cols + c, acc);
+}
+
+template <bool BF16>
+__global__ void colsum_reduce_kernel(void* __restrict__ db,
+                                     const float* __restrict__ partial,
+                                     int64_t stripes, int64_t cols) {
+  const int64_t c = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= cols) return;
+  float s = 0.f;
+  for (int64_t p = 0; p < stripes; ++p) s += partial[p * cols + c];
+  if (BF16)
+    reinterpret_cast<unsigned short*>(db)[c] = f2bf(s);
+  else
+    reinterpret_cast<float*>(db)[c] = s;
+}
+
 // extern "C" launchers (raw pointers + stream; bound to torch in bindings.hip)
 // ============================================================================
 extern "C" {
@@ -1176,6 +1227,26 @@ void epl_sqnorm(const void* p, int64_t n, float* out, bool bf16,
   else
     hipLaunchKernelGGL(sqnorm_kernel<false>, dim3(grid), dim3(kBlock), 0,
                        stream, p, n, out);
+}
+
+
+void epl_colsum(void* db, const void* dy, float* partial, int64_t rows,
+                int64_t cols, int64_t stripes, bool bf16,
+                hipStream_t stream) {
+  dim3 grid((unsigned)((cols / 8 + 255) / 256), (unsigned)stripes);
+  if (bf16)
+    hipLaunchKernelGGL((colsum_partial_kernel<true>), grid, dim3(256), 0,
+                       stream, dy, partial, rows, cols, stripes);
+  else
+    hipLaunchKernelGGL((colsum_partial_kernel<false>), grid, dim3(256), 0,
+                       stream, dy, partial, rows, cols, stripes);
+  dim3 rgrid((unsigned)((cols + 255) / 256));
+  if (bf16)
+    hipLaunchKernelGGL((colsum_reduce_kernel<true>), rgrid, dim3(256), 0,
+                       stream, db, partial, stripes, cols);
+  else
+    hipLaunchKernelGGL((colsum_reduce_kernel<false>), rgrid, dim3(256), 0,
+                       stream, db, partial, stripes, cols);
 }
 
 }  // extern "C"

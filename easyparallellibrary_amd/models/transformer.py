@@ -25,7 +25,13 @@ import torch.nn.functional as F
 _NATIVE_ATTN = os.environ.get("EPL_NATIVE_ATTENTION", "auto")
 
 from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
+from easyparallellibrary_amd.ops.bias_linear import FusedBiasLinear
 from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
+
+# Linear bias grads via the fused colsum kernel (EPL_FUSED_BIAS_GRAD=0
+# restores torch's reduce)
+_FBG = os.environ.get("EPL_FUSED_BIAS_GRAD", "1") == "1"
+_Linear = FusedBiasLinear if _FBG else nn.Linear
 
 # qkv unbind backward: 1 = fused slice-copy backward (below), 0 = torch
 # unbind/stack (CatArrayBatchedCopy).  A/B'd on GPU; see profiles/.
@@ -65,8 +71,8 @@ class SelfAttention(nn.Module):
         self.num_heads = num_heads
         self.head_dim = hidden // num_heads
         self.causal = causal
-        self.qkv = nn.Linear(hidden, 3 * hidden)
-        self.proj = nn.Linear(hidden, hidden)
+        self.qkv = _Linear(hidden, 3 * hidden)
+        self.proj = _Linear(hidden, hidden)
         self.dropout = dropout
 
     def forward(self, x):
@@ -107,7 +113,7 @@ class MLP(nn.Module):
         super().__init__()
         self.fc1 = nn.Linear(hidden, ffn_hidden, bias=False)
         self.act = FusedBiasGelu(ffn_hidden)
-        self.fc2 = nn.Linear(ffn_hidden, hidden)
+        self.fc2 = _Linear(ffn_hidden, hidden)
 
     def forward(self, x):
         return self.fc2(self.act(self.fc1(x)))

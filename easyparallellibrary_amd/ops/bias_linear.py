@@ -1,0 +1,62 @@
+"""Linear with a fused column-sum bias gradient.
+
+The forward is exactly ``F.linear`` (hipBLASLt GEMM with fused bias
+epilogue).  The backward replaces torch's generic reduce for ``db``
+(~2.2 TB/s on [65536, 1024] bf16) with the colsum kernel pair in
+csrc/kernels/kernels.hip (stripe partials + tiny reduce, no atomics —
+the same pattern as the LayerNorm backward).  dx / dW are the same two
+GEMMs autograd would issue.
+
+Reference parity: the reference's dense layers are plain TF ops; this is
+an MI355X-side optimization of the same computation (BERT-Large runs ~73
+bias-grad reductions per step: qkv / attention-proj / fc2).
+"""
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from easyparallellibrary_amd.ops.dispatch import native_ext, use_native
+
+_STRIPES = 256
+
+
+def fused_colsum(dy2):
+    """Column sum of a contiguous 2-D tensor via the native kernel."""
+    ext = native_ext()
+    cols = dy2.shape[1]
+    db = torch.empty(cols, dtype=dy2.dtype, device=dy2.device)
+    partial = torch.empty(_STRIPES * cols, dtype=torch.float32,
+                          device=dy2.device)
+    ext.colsum(dy2, db, partial)
+    return db
+
+
+class _LinearFusedBiasGrad(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        return F.linear(x, weight, bias)
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1])
+        if not dy2.is_contiguous():
+            dy2 = dy2.contiguous()
+        x2 = x.reshape(-1, x.shape[-1])
+        dx = (dy2 @ weight).view(x.shape)
+        dw = dy2.t() @ x2
+        db = fused_colsum(dy2)
+        return dx, dw, db
+
+
+class FusedBiasLinear(nn.Linear):
+    """nn.Linear whose bias gradient uses the fused colsum kernel on GPU
+    (falls back to standard autograd on CPU / unsupported shapes)."""
+
+    def forward(self, x):
+        if (self.bias is not None and use_native(x)
+                and self.out_features % 8 == 0):
+            return _LinearFusedBiasGrad.apply(x, self.weight, self.bias)
+        return F.linear(x, self.weight, self.bias)

@@ -202,3 +202,44 @@ def test_fused_residual_layer_norm_gpu(dtype):
     assert torch.allclose(y.float(), yr, atol=tol, rtol=tol)
     assert torch.allclose(x.grad.float(), xf.grad, atol=tol, rtol=tol)
     assert torch.allclose(r.grad.float(), rf.grad, atol=tol, rtol=tol)
+
+
+@pytest.mark.parametrize("cols,dtype", [(1024, torch.bfloat16),
+                                        (3072, torch.bfloat16),
+                                        (1024, torch.float32)])
+def test_colsum(cols, dtype):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.ops.bias_linear import fused_colsum
+    epl.init()
+    torch.manual_seed(8)
+    dy = torch.randn(4096, cols, device="cuda", dtype=dtype)
+    got = fused_colsum(dy)
+    want = dy.float().sum(dim=0)
+    torch.cuda.synchronize()
+    err = (got.float() - want).abs().max().item()
+    rel = err / want.abs().max().clamp_min(1e-6).item()
+    assert rel < 2e-2, (err, rel)
+
+
+def test_fused_bias_linear_grads():
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.ops.bias_linear import FusedBiasLinear
+    epl.init()
+    torch.manual_seed(9)
+    lin = FusedBiasLinear(64, 128).cuda().to(torch.bfloat16)
+    ref = torch.nn.Linear(64, 128).cuda().to(torch.bfloat16)
+    ref.load_state_dict(lin.state_dict())
+    x = torch.randn(8, 16, 64, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    xr = x.detach().clone().requires_grad_(True)
+    y = lin(x); yr = ref(xr)
+    assert torch.equal(y, yr)
+    dy = torch.randn_like(y)
+    y.backward(dy); yr.backward(dy)
+    torch.cuda.synchronize()
+    for got, want, name in ((x.grad, xr.grad, "dx"),
+                            (lin.weight.grad, ref.weight.grad, "dw"),
+                            (lin.bias.grad, ref.bias.grad, "db")):
+        err = (got.float() - want.float()).abs().max().item()
+        rel = err / want.float().abs().max().clamp_min(1e-6).item()
+        assert rel < 3e-2, (name, err, rel)
