@@ -1233,13 +1233,19 @@ void epl_sqnorm(const void* p, int64_t n, float* out, bool bf16,
 void epl_colsum(void* db, const void* dy, float* partial, int64_t rows,
                 int64_t cols, int64_t stripes, bool bf16,
                 hipStream_t stream) {
-  dim3 grid((unsigned)((cols / 8 + 255) / 256), (unsigned)stripes);
+  // size the block to the column count so narrow matrices (cols/8 < 256
+  // lanes) don't launch half-idle blocks
+  int64_t thr = cols / 8;
+  thr = thr < 64 ? 64 : (thr > 256 ? 256 : thr);
+  thr = (thr + 63) / 64 * 64;
+  dim3 grid((unsigned)((cols / 8 + thr - 1) / thr), (unsigned)stripes);
   if (bf16)
-    hipLaunchKernelGGL((colsum_partial_kernel<true>), grid, dim3(256), 0,
-                       stream, dy, partial, rows, cols, stripes);
+    hipLaunchKernelGGL((colsum_partial_kernel<true>), grid, dim3((unsigned)thr),
+                       0, stream, dy, partial, rows, cols, stripes);
   else
-    hipLaunchKernelGGL((colsum_partial_kernel<false>), grid, dim3(256), 0,
-                       stream, dy, partial, rows, cols, stripes);
+    hipLaunchKernelGGL((colsum_partial_kernel<false>), grid,
+                       dim3((unsigned)thr), 0, stream, dy, partial, rows,
+                       cols, stripes);
   dim3 rgrid((unsigned)((cols + 255) / 256));
   if (bf16)
     hipLaunchKernelGGL((colsum_reduce_kernel<true>), rgrid, dim3(256), 0,

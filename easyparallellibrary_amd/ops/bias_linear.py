@@ -18,15 +18,21 @@ import torch.nn.functional as F
 
 from easyparallellibrary_amd.ops.dispatch import native_ext, use_native
 
-_STRIPES = 256
+def _stripes_for(rows, cols):
+    # target ~2048 workgroups so the 256-CU chip stays full even for
+    # narrow matrices (one x-block at cols 1024)
+    blocks_x = max(1, (cols // 8 + 255) // 256)
+    s = max(64, min(1024, 2048 // blocks_x))
+    return max(1, min(s, rows // 4 or 1))
 
 
 def fused_colsum(dy2):
     """Column sum of a contiguous 2-D tensor via the native kernel."""
     ext = native_ext()
-    cols = dy2.shape[1]
+    rows, cols = dy2.shape
+    stripes = _stripes_for(rows, cols)
     db = torch.empty(cols, dtype=dy2.dtype, device=dy2.device)
-    partial = torch.empty(_STRIPES * cols, dtype=torch.float32,
+    partial = torch.empty(stripes * cols, dtype=torch.float32,
                           device=dy2.device)
     ext.colsum(dy2, db, partial)
     return db
