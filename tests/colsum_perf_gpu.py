@@ -1,5 +1,9 @@
 """colsum vs torch.sum micro-bench (script, GPU box)."""
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
