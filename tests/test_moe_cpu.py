@@ -147,3 +147,16 @@ def test_pp2_ep2_hybrid_matches_pp2():
     assert base[1][-1] < base[1][0]
     for a, b in zip(base[1], hyb[2]):
         assert abs(a - b) < 1e-4, (base[1], hyb[2])
+
+
+def test_pp2_ep2_dp2_full_3d():
+    """8 ranks: 2 pipeline stages x width-2 stages x 2 replicas — the
+    full DP x EP x PP hybrid.  Same data everywhere -> all last-stage
+    ranks agree and match the 2-rank plain-PP2 run."""
+    base = run_multiprocess(_pp_moe_worker, world=2, args=(1,))
+    hyb = run_multiprocess(_pp_moe_worker, world=8, args=(2,), timeout=420)
+    # replica 0 = ranks 0-3 (stage0: 0,1; stage1: 2,3); replica 1 = 4-7
+    last = [hyb[r] for r in (2, 3, 6, 7)]
+    assert last[0] == last[1] == last[2] == last[3]
+    for a, b in zip(base[1], last[0]):
+        assert abs(a - b) < 1e-4, (base[1], last[0])
