@@ -110,3 +110,31 @@ def test_pp2_odd_microbatches():
     pp = run_multiprocess(_pipeline_worker, world=2,
                           args=("prefer_backward", 3))
     assert all(abs(a - b) < 1e-5 for a, b in zip(serial, pp[1]))
+
+
+def _pipeline_gc_worker(rank, world, gc_type):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({
+        "pipeline.num_micro_batch": 4,
+        "gradient_checkpoint.type": gc_type,
+    }))
+    model = _build(world_stages=world)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(33)
+    x = torch.randn(8, 8)
+    y = torch.randn(8, 4)
+    out = []
+    for _ in range(3):
+        loss = engine.train_step(x, y)
+        out.append(None if loss is None else float(loss))
+    return out
+
+
+def test_pp2_with_gradient_checkpoint():
+    """Recompute inside pipeline stages: same losses as without GC."""
+    plain = run_multiprocess(_pipeline_worker, world=2,
+                             args=("prefer_backward", 4))
+    gc = run_multiprocess(_pipeline_gc_worker, world=2, args=("auto",))
+    assert all(abs(a - b) < 1e-5 for a, b in zip(plain[1], gc[1])), (
+        plain[1], gc[1])
