@@ -83,3 +83,28 @@ def test_zero_v1_with_pipeline():
     assert base[1] == base[3] and z1[1] == z1[3]
     assert all(abs(a - b) < 1e-5 for a, b in zip(base[1], z1[1])), (
         base[1], z1[1])
+
+
+def _comp_worker(rank, world, zero_level, comp):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"zero.level": zero_level,
+                         "communication.compression": comp}))
+    torch.manual_seed(17)
+    with epl.replicate(device_count=1):
+        model = nn.Sequential(nn.Linear(16, 64), nn.Tanh(),
+                              nn.Linear(64, 4))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(23)
+    x = torch.randn(8, 16)
+    y = torch.randn(8, 4)
+    return [float(engine.train_step(x, y)) for _ in range(4)]
+
+
+def test_zero_v1_with_wire_compression():
+    """bf16-compressed reduce-to-owner buckets track the fp32 wire to
+    bf16 rounding error."""
+    base = run_multiprocess(_comp_worker, world=2, args=("v1", ""))
+    comp = run_multiprocess(_comp_worker, world=2, args=("v1", "bf16"))
+    assert all(abs(a - b) < 1e-3 for a, b in zip(base[0], comp[0])), (
+        base[0], comp[0])
