@@ -121,3 +121,64 @@ def test_tp_shard_files_and_reshard():
     engine.load_checkpoint(CKPT + "_tp", load_optimizer=False)
     full = torch.cat([results[0][0], results[1][0]], dim=0)
     assert torch.allclose(engine.model.hd.weight.detach(), full, atol=1e-7)
+
+
+CKPT_PP = os.path.join(tempfile.gettempdir(), "epl_test_ckpt_pp")
+
+
+def _worker_pp_save_resume(rank, world):
+    import shutil
+    import easyparallellibrary_amd as epl
+
+    def build():
+        torch.manual_seed(88)
+        with epl.replicate(device_count=1, name="stage_0"):
+            s0 = nn.Sequential(nn.Linear(8, 16), nn.Tanh())
+        with epl.replicate(device_count=1, name="stage_1"):
+            s1 = nn.Sequential(nn.Linear(16, 16), nn.Tanh(),
+                               nn.Linear(16, 2))
+
+        class M(nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.s0, self.s1 = s0, s1
+
+            def forward(self, x):
+                return self.s1(self.s0(x))
+
+        return M()
+
+    if rank == 0 and os.path.exists(CKPT_PP):
+        shutil.rmtree(CKPT_PP)
+    epl.init(epl.Config({"pipeline.num_micro_batch": 2}))
+    engine = epl.Engine(build(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(6)
+    x = torch.randn(4, 8)
+    y = torch.randn(4, 2)
+    _train_pp = lambda e, n: [e.train_step(x, y) for _ in range(n)]
+    _train_pp(engine, 3)
+    engine.save_checkpoint(CKPT_PP)
+    cont = _train_pp(engine, 3)
+
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    hooks.remove_hooks()
+    Env._instance = None
+    epl.init(epl.Config({"pipeline.num_micro_batch": 2}))
+    engine2 = epl.Engine(build(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                         lr=1e-2)
+    engine2.load_checkpoint(CKPT_PP)
+    resumed = _train_pp(engine2, 3)
+    to_f = lambda seq: [None if l is None else float(l) for l in seq]
+    return to_f(cont), to_f(resumed)
+
+
+def test_pp2_save_resume_exact():
+    """Each rank checkpoints its own stage; resume reproduces the
+    uninterrupted loss trajectory."""
+    res = run_multiprocess(_worker_pp_save_resume, world=2, timeout=300)
+    cont, resumed = res[1]  # stage-1 rank holds the loss
+    assert all(l is not None for l in cont)
+    assert all(abs(a - b) < 1e-6 for a, b in zip(cont, resumed)), (
+        cont, resumed)
