@@ -168,6 +168,11 @@ class OptimizerConfig(BaseConfig):
     _SECTION = "optimizer"
     _DEFAULTS = {
         "num_apply_group": 1,
+        # global grad-norm clip, 0 = off.  Applied AFTER gradient
+        # aggregation (reference: communication.clip_after_allreduce,
+        # epl/config.py:96-97 — the recommended ordering is the only one
+        # implemented; the clip folds into the fused optimizer's scale)
+        "max_grad_norm": 0.0,
     }
 
 

@@ -326,10 +326,43 @@ class Engine:
         found_inf = self.amp.found_inf(self.flat_groups)
         if not found_inf:
             grad_scale = float(self.num_micro_batch) * self.amp.loss_scale
+            max_norm = self.config.optimizer.max_grad_norm
+            if max_norm:
+                gnorm = self._global_grad_norm() / grad_scale
+                if gnorm > max_norm:
+                    # shrink by gnorm/max_norm via the optimizer's fused
+                    # unscale — no extra pass over the arenas
+                    grad_scale *= gnorm / max_norm
             self.optimizer.step(grad_scale=grad_scale)
         self.amp.post_step(found_inf)
         self.global_step += 1
         return loss
+
+    def _global_grad_norm(self):
+        """Global L2 norm of the de-duplicated gradient: each arena's
+        ||g||^2 is divided by its DP-group size (the number of ranks
+        holding an identical copy) and summed over the world, so every
+        unique parameter counts exactly once across DP replicas,
+        pipeline stages and TP/EP shards (reference capability:
+        communication.clip_after_allreduce, epl/config.py:96-97)."""
+        from easyparallellibrary_amd.ops.dispatch import (native_ext,
+                                                          use_native)
+        total = torch.zeros(1, dtype=torch.float32, device=self.device)
+        for info in self._group_infos:
+            g = info["fg"].grad_arena
+            if use_native(g):
+                sq = torch.zeros(1, dtype=torch.float32, device=g.device)
+                native_ext().sqnorm(g, sq)
+            else:
+                sq = g.float().pow(2).sum().reshape(1)
+            total += sq / max(1, len(info["ranks"]))
+        if dist.is_initialized():
+            use_dev = (dist.get_backend() == "nccl"
+                       and self.device.type == "cuda")
+            t = total if use_dev else total.cpu()
+            dist.all_reduce(t)
+            total = t
+        return float(total.sqrt())
 
     def _train_step_simple(self, inputs, targets):
         nmb = self.num_micro_batch

@@ -81,14 +81,28 @@ class AmpContext:
         return loss
 
     def found_inf(self, flat_groups):
-        """One fused pass over the flat grad arenas."""
+        """One fused pass over the flat grad arenas.  The verdict is
+        agreed ACROSS RANKS (max-allreduce) so every rank skips or steps
+        together — pipeline stages hold disjoint grads, so a local-only
+        check would desynchronize the schedule."""
         if not self.enabled or self.dtype == torch.bfloat16:
             return False
+        local = False
         for fg in flat_groups:
             s = fg.grad_arena.sum(dtype=torch.float32)
             if not torch.isfinite(s):
-                return True
-        return False
+                local = True
+                break
+        import torch.distributed as dist
+        if dist.is_initialized():
+            dev = (flat_groups[0].grad_arena.device
+                   if flat_groups else torch.device("cpu"))
+            use_dev = dist.get_backend() == "nccl" and dev.type == "cuda"
+            t = torch.tensor([1.0 if local else 0.0],
+                             device=dev if use_dev else "cpu")
+            dist.all_reduce(t, op=dist.ReduceOp.MAX)
+            local = bool(t.item() > 0)
+        return local
 
     def post_step(self, found_inf):
         self.scaler.update(found_inf)

@@ -1,0 +1,85 @@
+"""Global grad-norm clipping (reference capability:
+communication.clip_after_allreduce, epl/config.py:96-97): de-duplicated
+global norm across DP/PP; clip folds into the fused optimizer scale."""
+
+import torch
+import torch.nn as nn
+
+import easyparallellibrary_amd as epl
+from tests.utils import run_multiprocess
+
+
+def _build():
+    torch.manual_seed(11)
+    with epl.replicate(1):
+        m = nn.Sequential(nn.Linear(8, 32), nn.Tanh(), nn.Linear(32, 4))
+    return m
+
+
+def test_global_grad_norm_matches_torch():
+    epl.init()
+    engine = epl.Engine(_build(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(12)
+    x, y = torch.randn(16, 8), torch.randn(16, 4)
+    engine.zero_grad()
+    loss = engine.loss_fn(engine.model(x), y)
+    loss.backward()
+    engine.finish_grad_sync()
+    want = torch.sqrt(sum((p.grad.float() ** 2).sum()
+                          for p in engine.model.parameters()))
+    got = engine._global_grad_norm()
+    assert abs(got - float(want)) < 1e-5, (got, float(want))
+
+
+def test_clip_changes_trajectory_only_when_binding():
+    def run(max_norm):
+        from easyparallellibrary_amd.env import Env
+        from easyparallellibrary_amd.parallel import hooks
+        hooks.remove_hooks()
+        Env._instance = None
+        epl.init(epl.Config({"optimizer.max_grad_norm": max_norm}))
+        engine = epl.Engine(_build(), loss_fn=nn.MSELoss(),
+                            optimizer="adamw", lr=1e-2)
+        torch.manual_seed(12)
+        x, y = torch.randn(16, 8), torch.randn(16, 4)
+        return [float(engine.train_step(x, y)) for _ in range(3)]
+
+    base = run(0.0)
+    loose = run(1e9)     # never binds -> identical
+    tight = run(1e-3)    # always binds -> different
+    assert base == loose, (base, loose)
+    assert base != tight
+
+
+def _norm_worker(rank, world, pp):
+    import easyparallellibrary_amd as epl
+    del pp
+    epl.init()
+    torch.manual_seed(11)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(8, 32), nn.Tanh(),
+                              nn.Linear(32, 4))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(12)
+    x, y = torch.randn(16, 8), torch.randn(16, 4)
+    engine.zero_grad()
+    loss = engine.loss_fn(engine.model(x), y)
+    loss.backward()
+    engine.finish_grad_sync()
+    return engine._global_grad_norm()
+
+
+def test_global_norm_dedup_dp2():
+    """DP2 (grads averaged identically on both ranks) reports the same
+    global norm as the single-rank run."""
+    single = _norm_worker(0, 1, False)
+    # reset global state left by the in-process run
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    hooks.remove_hooks()
+    Env._instance = None
+    dp = run_multiprocess(_norm_worker, world=2, args=(False,))
+    assert abs(dp[0] - dp[1]) < 1e-6
+    assert abs(dp[0] - single) < 1e-4, (dp[0], single)
