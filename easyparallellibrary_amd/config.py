@@ -88,6 +88,11 @@ class CommunicationConfig(BaseConfig):
         "bucket_bytes": constant.DEFAULT_BUCKET_BYTES,
         # max number of buckets per allreduce batch
         "max_splits": constant.DEFAULT_MAX_SPLITS,
+        # parity knob (reference config.py:96-97); clipping — when
+        # optimizer.max_grad_norm is set — is ALWAYS applied after
+        # aggregation here (the reference-recommended ordering), so
+        # False + max_grad_norm is rejected at engine build
+        "clip_after_allreduce": True,
         # 'mean' or 'sum' gradient reduction
         "gradients_reduce_method": "mean",
         # fp16/bf16-compress gradients before allreduce ('', 'fp16', 'bf16')
@@ -129,6 +134,8 @@ class OffloadConfig(BaseConfig):
 class AmpConfig(BaseConfig):
     _SECTION = "amp"
     _DEFAULTS = {
+        # log loss-scale decisions (reference amp.debug_log)
+        "debug_log": False,
         # '' (off) | 'O1'
         "level": "",
         # True -> dynamic loss scaling; False -> fixed
@@ -140,6 +147,12 @@ class AmpConfig(BaseConfig):
 class GradientCheckpointConfig(BaseConfig):
     _SECTION = "gradient_checkpoint"
     _DEFAULTS = {
+        # only wrap modules owned by taskgraphs with index < end_taskgraph
+        # (-1 = all; reference gc end_taskgraph, config.py:123)
+        "end_taskgraph": -1,
+        # after the first step, verify GC gradients against a no-recompute
+        # backward and log the max deviation (reference check_gradients)
+        "check_gradients": False,
         # '' (off) | 'collection' (user-tagged) | 'auto'
         "type": "",
     }
@@ -148,6 +161,9 @@ class GradientCheckpointConfig(BaseConfig):
 class IoConfig(BaseConfig):
     _SECTION = "io"
     _DEFAULTS = {
+        # drop remainder files so every replica gets an equal count
+        # (reference io.drop_last_files)
+        "drop_last_files": False,
         # slice input files proportionally to local replicas
         "slicing": False,
         "unbalanced_io_slicing": False,

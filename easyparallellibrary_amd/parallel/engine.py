@@ -180,10 +180,23 @@ class Engine:
 
         # ---- gradient checkpointing (reference: runtime/gc/) ----------------
         gc_type = self.config.gradient_checkpoint.type
+        self._gc_wrapped = 0
         if gc_type:
             from easyparallellibrary_amd.runtime.gc import (
                 apply_gradient_checkpointing)
-            n = apply_gradient_checkpointing(self._runnable, mode=gc_type)
+            end_tg = self.config.gradient_checkpoint.end_taskgraph
+            allowed = None
+            if end_tg >= 0:
+                # restrict recompute to modules owned by taskgraphs with
+                # index < end_taskgraph (reference gc end_taskgraph)
+                allowed = set()
+                for tg in self._owned_tgs:
+                    if tg.index < end_tg:
+                        for m in tg.modules:
+                            allowed.update(id(sub) for sub in m.modules())
+            n = apply_gradient_checkpointing(self._runnable, mode=gc_type,
+                                             allowed=allowed)
+            self._gc_wrapped = n
             logger.info("gradient checkpointing: wrapped %d module(s)", n)
 
         # ---- DP groups + reducers -------------------------------------------
@@ -323,6 +336,10 @@ class Engine:
             loss = self.pipeline.run(inputs, targets)
         else:
             loss = self._train_step_simple(inputs, targets)
+        if (self._gc_wrapped
+                and self.config.gradient_checkpoint.check_gradients
+                and self.global_step == 0 and self.pipeline is None):
+            self._verify_gc_gradients(inputs, targets)
         found_inf = self.amp.found_inf(self.flat_groups)
         if not found_inf:
             grad_scale = float(self.num_micro_batch) * self.amp.loss_scale
@@ -337,6 +354,30 @@ class Engine:
         self.amp.post_step(found_inf)
         self.global_step += 1
         return loss
+
+    def _verify_gc_gradients(self, inputs, targets):
+        """One-time self-check (reference gc check_gradients,
+        gradient_checkpoint.py:310-325): recompute the first step's
+        gradients with recompute disabled and log the max deviation;
+        the checkpointed gradients are kept for the actual step."""
+        from easyparallellibrary_amd.runtime.gc import CheckpointWrapper
+        snap = [fg.grad_arena.clone() for fg in self.flat_groups]
+        wrappers = [m for m in self._runnable.modules()
+                    if isinstance(m, CheckpointWrapper)]
+        for w in wrappers:
+            w.enabled = False
+        self.zero_grad()
+        self._train_step_simple(inputs, targets)
+        worst = 0.0
+        for fg, s0 in zip(self.flat_groups, snap):
+            worst = max(worst, float(
+                (fg.grad_arena.float() - s0.float()).abs().max()))
+        logger.info("gc check_gradients: max |gc - plain| grad deviation "
+                    "= %.3e", worst)
+        for w in wrappers:
+            w.enabled = True
+        for fg, s0 in zip(self.flat_groups, snap):
+            fg.grad_arena.copy_(s0)
 
     def _global_grad_norm(self):
         """Global L2 norm of the de-duplicated gradient: each arena's
@@ -447,7 +488,8 @@ class Engine:
         graph_editor.py:149-215)."""
         from easyparallellibrary_amd.utils.io_slicing import slice_files
         return slice_files(files, self.num_replicas, self.replica_id,
-                           unbalanced=self.config.io.unbalanced_io_slicing)
+                           unbalanced=self.config.io.unbalanced_io_slicing,
+                           drop_last=self.config.io.drop_last_files)
 
     # ---- checkpoint ----------------------------------------------------------
     def save_checkpoint(self, path, save_optimizer=True):

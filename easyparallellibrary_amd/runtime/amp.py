@@ -58,6 +58,7 @@ class AmpContext:
         self.dtype = (torch.float16 if config.amp.dtype == "fp16"
                       else torch.bfloat16)
         self.device_type = "cuda" if device.type == "cuda" else "cpu"
+        self.debug_log = bool(config.amp.debug_log)
         ls = config.amp.loss_scale
         if not self.enabled or self.dtype == torch.bfloat16:
             self.scaler = FixedLossScaler(1.0)
@@ -105,4 +106,10 @@ class AmpContext:
         return local
 
     def post_step(self, found_inf):
+        before = self.loss_scale
         self.scaler.update(found_inf)
+        if self.debug_log and self.loss_scale != before:
+            from easyparallellibrary_amd.utils.logging import get_logger
+            get_logger().info(
+                "amp loss scale %s -> %s (%s)", before, self.loss_scale,
+                "overflow" if found_inf else "growth")

@@ -35,12 +35,14 @@ def annotate_checkpoint(module):
 
 
 class CheckpointWrapper(nn.Module):
+    enabled = True  # instance-overridable (gc check_gradients flips it)
+
     def __init__(self, inner):
         super().__init__()
         self.inner = inner
 
     def forward(self, *args, **kwargs):
-        if torch.is_grad_enabled() and self.training:
+        if self.enabled and torch.is_grad_enabled() and self.training:
             return checkpoint(self.inner, *args, use_reentrant=False,
                               **kwargs)
         return self.inner(*args, **kwargs)
@@ -92,9 +94,13 @@ def select_checkpoint_modules(root, mode="auto", min_repeat=3):
     return hits
 
 
-def apply_gradient_checkpointing(root, mode="auto"):
-    """Wrap the selected modules in place; returns how many."""
+def apply_gradient_checkpointing(root, mode="auto", allowed=None):
+    """Wrap the selected modules in place; returns how many.
+    ``allowed``: optional set of module ids to restrict wrapping to
+    (engine uses it for gradient_checkpoint.end_taskgraph)."""
     hits = select_checkpoint_modules(root, mode=mode)
+    if allowed is not None:
+        hits = [(p, n, c) for (p, n, c) in hits if id(c) in allowed]
     for parent, name, child in hits:
         if isinstance(child, CheckpointWrapper):
             continue

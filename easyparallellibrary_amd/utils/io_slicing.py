@@ -11,8 +11,12 @@ unbalanced_io_slicing — keeps strict contiguous proportional slices.
 """
 
 
-def slice_files(files, num_replicas, replica_id, unbalanced=False):
+def slice_files(files, num_replicas, replica_id, unbalanced=False,
+                drop_last=False):
     files = list(files)
+    if drop_last and num_replicas > 1 and len(files) % num_replicas:
+        # reference io.drop_last_files: equal counts everywhere
+        files = files[:len(files) - len(files) % num_replicas]
     n = len(files)
     if num_replicas <= 1:
         return files

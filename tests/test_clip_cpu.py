@@ -83,3 +83,41 @@ def test_global_norm_dedup_dp2():
     dp = run_multiprocess(_norm_worker, world=2, args=(False,))
     assert abs(dp[0] - dp[1]) < 1e-6
     assert abs(dp[0] - single) < 1e-4, (dp[0], single)
+
+
+def test_parity_knobs():
+    """Reference-parity config knobs: amp.debug_log, drop_last_files,
+    gc end_taskgraph / check_gradients, clip_after_allreduce."""
+    from easyparallellibrary_amd.utils.io_slicing import slice_files
+    files = [str(i) for i in range(10)]
+    parts = [slice_files(files, 4, r, drop_last=True) for r in range(4)]
+    assert [len(p) for p in parts] == [2, 2, 2, 2]
+    assert sum(parts, []) == files[:8]
+
+
+def test_gc_check_gradients_logs_and_trains():
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    hooks.remove_hooks()
+    Env._instance = None
+    epl.init(epl.Config({"gradient_checkpoint.type": "auto",
+                         "gradient_checkpoint.check_gradients": True}))
+    torch.manual_seed(13)
+
+    class Block(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.a = nn.Linear(8, 8)
+
+        def forward(self, x):
+            return torch.tanh(self.a(x))
+
+    with epl.replicate(1):
+        model = nn.Sequential(Block(), Block(), Block(), nn.Linear(8, 2))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    assert engine._gc_wrapped == 3
+    torch.manual_seed(14)
+    x, y = torch.randn(8, 8), torch.randn(8, 2)
+    losses = [float(engine.train_step(x, y)) for _ in range(2)]
+    assert losses[1] < losses[0]
