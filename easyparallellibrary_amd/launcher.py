@@ -31,6 +31,11 @@ def main(argv=None):
     parser = argparse.ArgumentParser(prog="epl-launch")
     parser.add_argument("--num_workers", type=int, required=True)
     parser.add_argument("--gpu_per_worker", type=int, default=1)
+    parser.add_argument("--visible_devices", default=os.environ.get(
+        "EPL_CLUSTER_RUN_VISIBLE_DEVICES", ""),
+        help="comma-separated physical GPU ids the job may use "
+             "(reference cluster.run_visible_devices); workers are "
+             "assigned from this list in rank order")
     parser.add_argument("--master_addr", default="127.0.0.1")
     parser.add_argument("--master_port", type=int, default=0)
     parser.add_argument("--log-dir", default="")
@@ -53,9 +58,19 @@ def main(argv=None):
                 "MASTER_PORT": str(port),
             })
             if args.gpu_per_worker > 0:
+                pool = ([int(d) for d in args.visible_devices.split(",")]
+                        if args.visible_devices else None)
                 first = rank * args.gpu_per_worker
-                env["HIP_VISIBLE_DEVICES"] = ",".join(
-                    str(first + i) for i in range(args.gpu_per_worker))
+                ids = range(first, first + args.gpu_per_worker)
+                if pool is not None:
+                    if first + args.gpu_per_worker > len(pool):
+                        raise SystemExit(
+                            "visible_devices lists {} GPUs; rank {} needs "
+                            "ids {}..{}".format(len(pool), rank, first,
+                                                first + args.gpu_per_worker
+                                                - 1))
+                    ids = [pool[i] for i in ids]
+                env["HIP_VISIBLE_DEVICES"] = ",".join(str(i) for i in ids)
                 env["LOCAL_RANK"] = "0"
             stdout = None
             if args.log_dir:
