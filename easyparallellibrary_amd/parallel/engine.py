@@ -71,6 +71,14 @@ class Engine:
                 "(pure-bf16 training needs no AMP)")
 
         # ---- plan -----------------------------------------------------------
+        if (self.config.auto.auto_parallel
+                and not self.env.strategy_context.strategies):
+            # reference hooks.py:130-134: auto mode tags the whole model
+            # Replicate(1); auto-stage below then splits it if
+            # pipeline.num_stages > 1
+            from easyparallellibrary_amd.strategies.replicate import (
+                Replicate)
+            self.env.strategy_context.set_default_strategy(Replicate(1))
         self.plan = Plan.build(model, self.env.strategy_context)
         # auto-stage: partition an un-staged sequential model into
         # pipeline.num_stages stages (reference: parallel/planner.py

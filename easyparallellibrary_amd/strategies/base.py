@@ -45,6 +45,10 @@ class ParallelStrategy:
 
     def __enter__(self):
         from easyparallellibrary_amd.env import Env
+        if Env.get().config.auto.auto_parallel:
+            # reference: strategies/parallel_strategy.py:61-63
+            raise ValueError("auto.auto_parallel is enabled; do not use "
+                             "explicit strategy scopes at the same time")
         Env.get().strategy_context.add_context(self)
         return self
 

@@ -81,3 +81,21 @@ def test_split_plan_device_counts():
     plan = Plan.build(M(), Env.get().strategy_context)
     assert plan.effective_device_counts(False) == [8, 8]
     assert plan.effective_device_counts(True) == [8, 0]
+
+
+def test_auto_parallel_mode():
+    """auto.auto_parallel: untagged models train (whole model replicate);
+    explicit scopes are rejected (reference parallel_strategy.py:61)."""
+    import torch
+    import pytest
+    epl.init(epl.Config({"auto.auto_parallel": True}))
+    with pytest.raises(ValueError):
+        with epl.replicate(1):
+            pass
+    torch.manual_seed(2)
+    model = nn.Sequential(nn.Linear(8, 8), nn.Tanh(), nn.Linear(8, 2))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    x, y = torch.randn(4, 8), torch.randn(4, 2)
+    losses = [float(engine.train_step(x, y)) for _ in range(3)]
+    assert losses[-1] < losses[0]
