@@ -397,14 +397,22 @@ class Engine:
         from easyparallellibrary_amd.ops.dispatch import (native_ext,
                                                           use_native)
         total = torch.zeros(1, dtype=torch.float32, device=self.device)
-        for info in self._group_infos:
+        for info, red in zip(self._group_infos, self.reducers):
             g = info["fg"].grad_arena
+            n_copies = max(1, len(info["ranks"]))
+            if red.shard_owners and n_copies > 1:
+                # ZeRO v1: only my shard's slice holds fully reduced
+                # grads (the rest is unreduced partials) — count it once
+                shard = info["fg"].total // n_copies
+                r = info["bcomm"].rank
+                g = g[r * shard:(r + 1) * shard]
+                n_copies = 1
             if use_native(g):
                 sq = torch.zeros(1, dtype=torch.float32, device=g.device)
                 native_ext().sqnorm(g, sq)
             else:
                 sq = g.float().pow(2).sum().reshape(1)
-            total += sq / max(1, len(info["ranks"]))
+            total += sq / n_copies
         if dist.is_initialized():
             use_dev = (dist.get_backend() == "nccl"
                        and self.device.type == "cuda")

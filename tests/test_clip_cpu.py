@@ -121,3 +121,31 @@ def test_gc_check_gradients_logs_and_trains():
     x, y = torch.randn(8, 8), torch.randn(8, 2)
     losses = [float(engine.train_step(x, y)) for _ in range(2)]
     assert losses[1] < losses[0]
+
+
+def _zero_norm_worker(rank, world, zero_level):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"zero.level": zero_level}))
+    torch.manual_seed(11)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(8, 32), nn.Tanh(),
+                              nn.Linear(32, 4))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(12)
+    x, y = torch.randn(16, 8), torch.randn(16, 4)
+    engine.zero_grad()
+    loss = engine.loss_fn(engine.model(x), y)
+    loss.backward()
+    engine.finish_grad_sync()
+    return engine._global_grad_norm()
+
+
+def test_global_norm_zero_v1_dedup():
+    """Under ZeRO v1 only the owned shard is counted; the norm still
+    equals the plain-DP value."""
+    plain = run_multiprocess(_zero_norm_worker, world=2, args=("",))
+    z1 = run_multiprocess(_zero_norm_worker, world=2, args=("v1",))
+    assert abs(plain[0] - plain[1]) < 1e-6
+    assert abs(z1[0] - z1[1]) < 1e-6
+    assert abs(plain[0] - z1[0]) < 1e-5, (plain[0], z1[0])
