@@ -387,6 +387,19 @@ class Engine:
         for fg, s0 in zip(self.flat_groups, snap):
             fg.grad_arena.copy_(s0)
 
+    def set_lr(self, lr):
+        """Update the learning rate for subsequent steps (schedules are
+        the caller's loop; the fused optimizers read ``lr`` each step)."""
+        opt = self.optimizer
+        inner = getattr(opt, "inner", None)
+        (inner if inner is not None else opt).lr = float(lr)
+
+    @property
+    def lr(self):
+        opt = self.optimizer
+        inner = getattr(opt, "inner", None)
+        return (inner if inner is not None else opt).lr
+
     def _global_grad_norm(self):
         """Global L2 norm of the de-duplicated gradient: each arena's
         ||g||^2 is divided by its DP-group size (the number of ranks

@@ -111,3 +111,22 @@ def test_grad_compression():
     res = run_multiprocess(_compress_worker, world=2)
     assert res[0] == res[1]
     assert res[0][-1] < res[0][0]
+
+
+def test_set_lr():
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(1)
+    with epl.replicate(1):
+        model = nn.Linear(4, 2)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    assert engine.lr == 1e-2
+    x, y = torch.randn(4, 4), torch.randn(4, 2)
+    engine.train_step(x, y)
+    w0 = model.weight.detach().clone()
+    engine.set_lr(0.0)
+    engine.train_step(x, y)
+    # lr=0 scales the whole update (incl. decoupled weight decay) to
+    # zero -> parameters must be bit-identical
+    assert torch.equal(model.weight.detach(), w0)
