@@ -66,6 +66,11 @@ def get_collection(name):
     return Env.get().get_collection(name)
 
 
+def get_all_collections():
+    """reference: epl.get_all_collections"""
+    return dict(Env.get()._collections)
+
+
 def __getattr__(name):
     # lazy to avoid import cycles (engine imports config/env/strategies)
     if name == "Engine":
@@ -77,4 +82,5 @@ __all__ = [
     "init", "set_default_strategy", "replicate", "split", "Replicate",
     "Split", "Config", "Cluster", "VirtualDevice", "Env", "Engine",
     "GraphKeys", "add_to_collection", "get_collection",
+    "get_all_collections",
 ]
