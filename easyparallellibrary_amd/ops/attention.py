@@ -64,7 +64,6 @@ class _FlashAttention(torch.autograd.Function):
         dk = grad_like(orig_q)
         dv = grad_like(orig_q)
         delta = torch.empty_like(lse)
-        import os
         # split dK/dV kernels measure ~4% faster than the combined one
         # (3 vs 2 waves/SIMD; A/B in profiles/r01_attention_ab.txt)
         split = os.environ.get("EPL_ATTN_BWD_SPLIT", "1") == "1"
