@@ -215,6 +215,144 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // ============================================================================
 // backward: D_i = rowsum(dO * O) precompute
 // ============================================================================
+// causal-only forward variant (EPL_ATTN_CAUSAL_V2): 64 q-rows per block
+// (2 waves, 128 threads).  Halves the per-block kv range so waves idle
+// through at most ~0.5 masked tiles (vs ~1.5 in the 128-row kernel) and
+// doubles the grid for a shorter dispatch tail; the cost is staging the
+// same K/V tiles into LDS twice as often.  UNMEASURED as of round 1 —
+// flag-gated off until a GPU A/B; numerics covered by
+// tests/test_attention_gpu.py when the flag is on.
+// ============================================================================
+__global__ __launch_bounds__(128) void attn_fwd_kernel_c64(
+    const short* __restrict__ q, const short* __restrict__ k,
+    const short* __restrict__ v, short* __restrict__ out,
+    float* __restrict__ lse, int64_t seq, float scale, int causal,
+    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
+    int64_t o_sb, int64_t o_sh, int64_t o_ss) {
+  __shared__ short ldsV[64][72];       // V^T: [d][kv]
+  __shared__ short ldsK[64][72];       // K: [kv][d]
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;   // 0..1
+  const int hi = lane >> 5;
+  const int lq = lane & 31;
+  const int64_t bh = blockIdx.y;
+  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
+  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
+                            : (int64_t)blockIdx.x;
+  const int64_t q0_blk = qb * 64;
+  const int64_t q0 = q0_blk + wave * 32;
+  const bool active = q0 < seq;
+  const short* qp = q + boff;
+  const short* kp = k + boff;
+  const short* vp = v + boff;
+
+  const int64_t myq = q0 + lq;
+  const int64_t qrow = myq < seq ? myq : seq - 1;
+
+  bf16x8 qfrag[4];
+#pragma unroll
+  for (int c = 0; c < 4; ++c)
+    qfrag[c] = *reinterpret_cast<const bf16x8*>(
+        qp + qrow * in_ss + hi * 8 + 16 * c);
+
+  f32x16 ot0 = {}, ot1 = {};
+  float m = -1e30f, l = 0.f;
+
+  const int64_t blk_kv_end =
+      causal ? (q0_blk + 64 < seq ? q0_blk + 64 : seq) : seq;
+  const int stage_kv = threadIdx.x & 63;
+  const int stage_d0 = (threadIdx.x >> 6) * 8;   // 0 or 8
+
+  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 64) {
+    __syncthreads();
+    {
+      int64_t vrow = kv0 + stage_kv;
+      if (vrow >= seq) vrow = seq - 1;
+#pragma unroll
+      for (int h2 = 0; h2 < 4; ++h2) {     // 2 waves x 4 strips = 64 cols
+        const int sd = stage_d0 + h2 * 16;
+        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+            vp + vrow * in_ss + sd);
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ldsV[sd + j][stage_kv] = vv[j];
+        *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
+            *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
+      }
+    }
+    __syncthreads();
+    if (active) {
+      const int64_t wave_kv_end = causal
+          ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
+#pragma unroll
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t kvs = kv0 + sub;
+        if (kvs >= wave_kv_end) break;
+        f32x16 st = {};
+#pragma unroll
+        for (int c = 0; c < 4; ++c) {
+          bf16x8 kfrag = *reinterpret_cast<const bf16x8*>(
+              &ldsK[sub + lq][hi * 8 + 16 * c]);
+          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c],
+                                                       st, 0, 0, 0);
+        }
+        float s[16];
+        float tile_max = -1e30f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          float sv = st[r] * scale;
+          const int64_t kvg = kvs + crow(r, hi);
+          if (kvg >= seq || (causal && kvg > myq)) sv = -1e30f;
+          s[r] = sv;
+          tile_max = fmaxf(tile_max, sv);
+        }
+        tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+        const float m_new = fmaxf(m, tile_max);
+        const float alpha = __expf(m - m_new);
+        float rowsum = 0.f;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          s[r] = __expf(s[r] - m_new);
+          rowsum += s[r];
+        }
+        rowsum += __shfl_xor(rowsum, 32, 64);
+        l = l * alpha + rowsum;
+        m = m_new;
+#pragma unroll
+        for (int r = 0; r < 16; ++r) {
+          ot0[r] *= alpha;
+          ot1[r] *= alpha;
+        }
+        bf16x8 pf0 = assemble_pfrag(&s[0]);
+        bf16x8 pf1 = assemble_pfrag(&s[8]);
+#pragma unroll
+        for (int kc = 0; kc < 2; ++kc) {
+          bf16x8 pf = kc == 0 ? pf0 : pf1;
+          bf16x8 vt0 = *reinterpret_cast<const bf16x8*>(
+              &ldsV[lq][sub + kc * 16 + hi * 8]);
+          bf16x8 vt1 = *reinterpret_cast<const bf16x8*>(
+              &ldsV[32 + lq][sub + kc * 16 + hi * 8]);
+          ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0,
+                                                        0, 0);
+          ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0,
+                                                        0, 0);
+        }
+      }
+    }
+  }
+
+  if (!active || myq >= seq) return;
+  const float inv_l = 1.f / l;
+  short* op = out + (bh / heads) * o_sb + (bh % heads) * o_sh + myq * o_ss;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    op[crow(r, hi)] = (short)f2bf(ot0[r] * inv_l);
+    op[32 + crow(r, hi)] = (short)f2bf(ot1[r] * inv_l);
+  }
+  if (hi == 0) lse[bh * seq + myq] = m + __logf(l);
+}
+
+// ============================================================================
 __global__ void attn_bwd_prep_kernel(const short* __restrict__ dout,
                                      const short* __restrict__ out,
                                      float* __restrict__ delta,
@@ -698,6 +836,15 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   }
 }
 
+static bool attn_fwd_c64_enabled() {
+  static int v = -1;
+  if (v < 0) {
+    const char* e = getenv("EPL_ATTN_CAUSAL_V2");
+    v = (e && e[0] == '1') ? 1 : 0;
+  }
+  return v == 1;
+}
+
 }  // namespace
 
 extern "C" {
@@ -706,6 +853,17 @@ void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
                   float* lse, int64_t batch_heads, int64_t seq, float scale,
                   bool causal, int64_t heads, const int64_t* in_strides,
                   const int64_t* o_strides, hipStream_t stream) {
+  if (causal && attn_fwd_c64_enabled()) {
+    dim3 cgrid((unsigned)((seq + 63) / 64), (unsigned)batch_heads);
+    hipLaunchKernelGGL(attn_fwd_kernel_c64, cgrid, dim3(128), 0, stream,
+                       reinterpret_cast<const short*>(q),
+                       reinterpret_cast<const short*>(k),
+                       reinterpret_cast<const short*>(v),
+                       reinterpret_cast<short*>(out), lse, seq, scale, 1,
+                       heads, in_strides[0], in_strides[1], in_strides[2],
+                       o_strides[0], o_strides[1], o_strides[2]);
+    return;
+  }
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
