@@ -130,3 +130,18 @@ def test_set_lr():
     # lr=0 scales the whole update (incl. decoupled weight decay) to
     # zero -> parameters must be bit-identical
     assert torch.equal(model.weight.detach(), w0)
+
+
+def test_kernel_fused_off_still_trains():
+    """kernel.fused=False forces the eager op paths everywhere."""
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"kernel.fused": False}))
+    torch.manual_seed(3)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(8, 16), nn.Tanh(),
+                              nn.Linear(16, 2))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    x, y = torch.randn(8, 8), torch.randn(8, 2)
+    losses = [float(engine.train_step(x, y)) for _ in range(3)]
+    assert losses[-1] < losses[0]
