@@ -79,3 +79,44 @@ def test_dynamic_scaler_growth():
     for _ in range(3):
         s.update(False)
     assert s.scale == 16
+
+
+def _amp_pp_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    from tests.utils import run_multiprocess  # noqa: F401 (import guard)
+    epl.init(epl.Config({
+        "amp.level": "O1", "amp.loss_scale": 128,
+        "pipeline.num_micro_batch": 2,
+    }))
+    torch.manual_seed(21)
+    with epl.replicate(1, name="stage_0"):
+        s0 = nn.Sequential(nn.Linear(8, 16), nn.Tanh())
+    with epl.replicate(1, name="stage_1"):
+        s1 = nn.Linear(16, 4)
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.s0, self.s1 = s0, s1
+
+        def forward(self, x):
+            return self.s1(self.s0(x))
+
+    engine = epl.Engine(M(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(22)
+    x, y = torch.randn(8, 8), torch.randn(8, 4)
+    out = []
+    for _ in range(3):
+        loss = engine.train_step(x, y)
+        out.append(None if loss is None else float(loss))
+    return out
+
+
+def test_amp_with_pipeline():
+    """AMP loss scaling under PP2: the overflow verdict is allreduced so
+    stages step in lockstep; training converges."""
+    from tests.utils import run_multiprocess
+    res = run_multiprocess(_amp_pp_worker, world=2)
+    assert res[0][0] is None
+    assert res[1][-1] < res[1][0], res[1]
