@@ -308,6 +308,7 @@ class Engine:
             self.pipeline = PipelineRuntime(self)
 
         self.global_step = 0
+        self._accum_count = 0
         logger.info(
             "Engine ready: world=%d stages=%d replicas=%d per_replica=%d "
             "dtype=%s groups=%d", self.world_size, self.num_stages,
@@ -335,11 +336,30 @@ class Engine:
         return self._runnable(x)
 
     # ---- the training step ---------------------------------------------------
-    def train_step(self, inputs, targets):
+    def train_step(self, inputs, targets, accumulate=False):
         """One optimizer step: micro-batch loop (pipeline or GA), overlapped
         DP gradient reduction, fused optimizer.  Returns the local mean loss
-        tensor (on the last stage; None elsewhere for pipelines)."""
-        self.zero_grad()
+        tensor (on the last stage; None elsewhere for pipelines).
+
+        ``accumulate=True`` runs forward+backward only (no reduction, no
+        optimizer) so callers can drive their own accumulation loop; the
+        next ``accumulate=False`` call reduces everything accumulated and
+        steps once, dividing by the number of accumulated calls."""
+        if accumulate:
+            if self.pipeline is not None:
+                raise ValueError(
+                    "manual accumulation is not supported with pipeline "
+                    "parallelism (use pipeline.num_micro_batch)")
+            if self._accum_count == 0:
+                self.zero_grad()
+            self._set_reducers_enabled(False)
+            with self.amp.autocast():
+                loss = self.loss_fn(self._runnable(inputs), targets)
+            self.amp.scale_loss(loss).backward()
+            self._accum_count += 1
+            return loss.detach()
+        if self._accum_count == 0:
+            self.zero_grad()
         if self.pipeline is not None:
             loss = self.pipeline.run(inputs, targets)
         else:
@@ -349,8 +369,11 @@ class Engine:
                 and self.global_step == 0 and self.pipeline is None):
             self._verify_gc_gradients(inputs, targets)
         found_inf = self.amp.found_inf(self.flat_groups)
+        n_accum = self._accum_count + 1
+        self._accum_count = 0
         if not found_inf:
-            grad_scale = float(self.num_micro_batch) * self.amp.loss_scale
+            grad_scale = (float(self.num_micro_batch) * n_accum
+                          * self.amp.loss_scale)
             max_norm = self.config.optimizer.max_grad_norm
             if max_norm:
                 gnorm = self._global_grad_norm() / grad_scale

@@ -145,3 +145,35 @@ def test_kernel_fused_off_still_trains():
     x, y = torch.randn(8, 8), torch.randn(8, 2)
     losses = [float(engine.train_step(x, y)) for _ in range(3)]
     assert losses[-1] < losses[0]
+
+
+def test_manual_accumulation_matches_big_batch():
+    import easyparallellibrary_amd as epl
+
+    def build_engine():
+        from easyparallellibrary_amd.env import Env
+        from easyparallellibrary_amd.parallel import hooks
+        hooks.remove_hooks()
+        Env._instance = None
+        epl.init()
+        torch.manual_seed(5)
+        with epl.replicate(1):
+            m = nn.Sequential(nn.Linear(8, 16), nn.Tanh(),
+                              nn.Linear(16, 2))
+        return epl.Engine(m, loss_fn=nn.MSELoss(), optimizer="adamw",
+                          lr=1e-2)
+
+    torch.manual_seed(6)
+    x = torch.randn(8, 8)
+    y = torch.randn(8, 2)
+
+    e1 = build_engine()
+    big = [float(e1.train_step(x, y)) for _ in range(3)]
+
+    e2 = build_engine()
+    man = []
+    for _ in range(3):
+        l0 = e2.train_step(x[:4], y[:4], accumulate=True)
+        l1 = e2.train_step(x[4:], y[4:])
+        man.append((float(l0) + float(l1)) / 2)
+    assert all(abs(a - b) < 1e-6 for a, b in zip(big, man)), (big, man)
