@@ -88,3 +88,37 @@ def test_dp_matches_serial():
     dp = run_multiprocess(_serial_worker, world=2)
     assert all(abs(a - b) < 1e-5 for a, b in zip(serial, dp[0]))
     assert all(abs(a - b) < 1e-5 for a, b in zip(serial, dp[1]))
+
+
+def _manual_accum_worker(rank, world, manual):
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(5)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(8, 16), nn.Tanh(),
+                              nn.Linear(16, 2))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(50 + rank)
+    x = torch.randn(8, 8)
+    y = torch.randn(8, 2)
+    losses = []
+    for _ in range(3):
+        if manual:
+            engine.train_step(x[:4], y[:4], accumulate=True)
+            engine.train_step(x[4:], y[4:])
+        else:
+            engine.train_step(x, y)
+        losses.append(float(engine.all_reduce_metric(
+            engine.loss_fn(engine.model(x), y))))
+    return losses
+
+
+def test_manual_accumulation_dp2():
+    """accumulate=True under DP2: the deferred allreduce still fires on
+    the closing step and matches plain big-batch DP."""
+    base = run_multiprocess(_manual_accum_worker, world=2, args=(False,))
+    man = run_multiprocess(_manual_accum_worker, world=2, args=(True,))
+    assert base[0] == base[1] and man[0] == man[1]
+    assert all(abs(a - b) < 1e-6 for a, b in zip(base[0], man[0])), (
+        base[0], man[0])
