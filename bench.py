@@ -5,7 +5,7 @@ Headline metric: samples/sec (whole node) for BERT-Large DP(+PP) on
 synthetic MLM data, bf16, random-init weights (BASELINE.json configs).
 
   python bench.py --gpus N --steps K --warmup W
-      [--config bert_dp|bert_pp|bert_zero|resnet_tp|gpt2_xl|moe]
+      [--config bert_dp|bert_pp|bert_zero|resnet_tp|gpt2_xl|moe|moe_pp]
       [--pp S] [--micro-batch M] [--batch B] [--zero v0|v1]
 
 For N>1 the driver launches this under torch.distributed.run with one rank
@@ -28,7 +28,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--config", default="bert_dp",
                    choices=["bert_dp", "bert_pp", "bert_zero", "resnet_tp",
-                            "gpt2_xl", "moe"])
+                            "gpt2_xl", "moe", "moe_pp"])
     p.add_argument("--model", default=None)
     p.add_argument("--batch", type=int, default=0,
                    help="per-GPU samples per step (0 = config default)")
@@ -170,6 +170,41 @@ def build_moe_bench(args, epl, world, on_gpu, dtype):
     return engine, (ids, tgt), batch * world, meta
 
 
+def build_moe_pp_bench(args, epl, world, on_gpu, dtype):
+    """PP2 x (DP+EP) MoE hybrid (multi-rank pipeline stages).  ep=2 when
+    world divides by 4, else 1; world==1 degrades to the plain MoE
+    config."""
+    if world < 2:
+        return build_moe_bench(args, epl, world, on_gpu, dtype)
+    from easyparallellibrary_amd.models.moe_transformer import (
+        build_moe_pipeline)
+    from easyparallellibrary_amd.models import gpt2
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    batch = args.batch or 8
+    seq = args.seq_len if args.seq_len != 512 else 1024
+    ep = 2 if world % 4 == 0 else 1
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "pipeline.num_micro_batch": min(4, batch),
+    }))
+    vocab = 32000
+    model = build_moe_pipeline(stages=2, ep=ep, layers=12, hidden=1024,
+                               heads=16, ffn=4096,
+                               num_experts=max(8, 2 * ep),
+                               vocab_size=vocab, max_pos=seq)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=1e-4, dtype=dtype)
+    rank = int(os.environ.get("RANK", "0"))
+    ids, tgt = gpt2.synthetic_lm_batch(batch, seq, vocab,
+                                       device=engine.device, seed=55 + rank)
+    streams = world // 2  # per-stage positions x replicas
+    meta = {"model": "moe-transformer-12L", "global_batch": batch * streams,
+            "seq_len": seq,
+            "parallelism": "pp2_ep{}_dp{}".format(ep, world // (2 * ep))}
+    return engine, (ids, tgt), batch * streams, meta
+
+
 BUILDERS = {
     "bert_dp": build_bert_bench,
     "bert_pp": build_bert_bench,
@@ -177,6 +212,7 @@ BUILDERS = {
     "gpt2_xl": build_gpt2_bench,
     "resnet_tp": build_resnet_bench,
     "moe": build_moe_bench,
+    "moe_pp": build_moe_pp_bench,
 }
 
 

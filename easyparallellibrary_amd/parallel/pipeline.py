@@ -160,6 +160,11 @@ class PipelineRuntime:
                         if self.s == 0 else [None] * M)
         target_chunks = (torch.chunk(targets, M, dim=0)
                          if self.s == self.S - 1 else [None] * M)
+        if self.s == 0 and len(input_chunks) < M:
+            raise ValueError(
+                "batch dim {} is smaller than pipeline.num_micro_batch={}"
+                " — torch.chunk would silently produce fewer micro-"
+                "batches".format(inputs.shape[0], M))
         self._shape_cache_out = {}
         self.engine._set_reducers_enabled(False)
         if self.schedule == constant.SCHEDULER_PREFER_FORWARD:
