@@ -177,3 +177,36 @@ def test_manual_accumulation_matches_big_batch():
         l1 = e2.train_step(x[4:], y[4:])
         man.append((float(l0) + float(l1)) / 2)
     assert all(abs(a - b) < 1e-6 for a, b in zip(big, man)), (big, man)
+
+
+def test_random_architectures_train():
+    """Robustness sweep: a handful of randomized architectures step
+    twice through the engine without shape/edge assumptions breaking."""
+    import random
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    rng = random.Random(1234)
+    for trial in range(5):
+        hooks.remove_hooks()
+        Env._instance = None
+        epl.init()
+        torch.manual_seed(trial)
+        depth = rng.randint(1, 4)
+        dims = [rng.choice([3, 8, 17, 32])]
+        layers = []
+        for _ in range(depth):
+            nxt = rng.choice([5, 16, 31])
+            layers += [nn.Linear(dims[-1], nxt), nn.Tanh()]
+            dims.append(nxt)
+        layers.append(nn.Linear(dims[-1], 4))
+        with epl.replicate(1):
+            model = nn.Sequential(*layers)
+        engine = epl.Engine(model, loss_fn=nn.MSELoss(),
+                            optimizer=rng.choice(["adamw", "lamb"]),
+                            lr=1e-3)
+        b = rng.choice([1, 3, 8])
+        x, y = torch.randn(b, dims[0]), torch.randn(b, 4)
+        for _ in range(2):
+            loss = engine.train_step(x, y)
+        assert torch.isfinite(loss), (trial, loss)
