@@ -59,3 +59,13 @@ def test_freeze():
     c.freeze()
     with pytest.raises(AttributeError):
         c.pipeline.num_micro_batch = 5
+
+
+def test_env_override_new_knobs(monkeypatch):
+    monkeypatch.setenv("EPL_OPTIMIZER_MAX_GRAD_NORM", "2.5")
+    monkeypatch.setenv("EPL_IO_DROP_LAST_FILES", "true")
+    monkeypatch.setenv("EPL_GRADIENT_CHECKPOINT_END_TASKGRAPH", "3")
+    c = Config()
+    assert c.optimizer.max_grad_norm == 2.5
+    assert c.io.drop_last_files is True
+    assert c.gradient_checkpoint.end_taskgraph == 3
