@@ -480,6 +480,9 @@ class Engine:
 
     # ---- eval ----------------------------------------------------------------
     def eval_step(self, inputs):
+        """Forward without grad.  Under pipeline parallelism this runs
+        ONLY this rank's stage (feed it the previous stage's activations
+        or run evaluation data-parallel on a non-pipelined engine)."""
         with torch.no_grad():
             return self._runnable(inputs)
 
