@@ -17,7 +17,11 @@ def replica_to_split(x, comm):
 
 def split_to_replica(x, comm):
     """Take this replica's slice of a full-batch tensor produced by a split
-    scope (inverse bridge; backward allgathers)."""
+    scope (inverse bridge).  Backward is the plain slice adjoint (zeros
+    outside this replica's rows); the other replicas' contributions to
+    the split scope's parameter gradients arrive through the scope's own
+    cross-rank gradient reduction, mirroring the reference's
+    Split2Replica."""
     if comm.size == 1:
         return x
     n = x.shape[0] // comm.size
