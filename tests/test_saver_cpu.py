@@ -182,3 +182,52 @@ def test_pp2_save_resume_exact():
     assert all(l is not None for l in cont)
     assert all(abs(a - b) < 1e-6 for a, b in zip(cont, resumed)), (
         cont, resumed)
+
+
+def test_assign_map_renamed_restore(tmp_path):
+    """assign_map restores a checkpoint into a model whose module path
+    was renamed (reference ShardingLoader name remap)."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+
+    epl.init()
+    torch.manual_seed(31)
+
+    class Old(nn.Module):
+        def __init__(self):
+            super().__init__()
+            with epl.replicate(1):
+                self.enc = nn.Linear(8, 4)
+
+        def forward(self, x):
+            return self.enc(x)
+
+    old = Old()
+    engine = epl.Engine(old, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    x, y = torch.randn(4, 8), torch.randn(4, 4)
+    engine.train_step(x, y)
+    engine.save_checkpoint(str(tmp_path))
+    want = old.enc.weight.detach().clone()
+
+    hooks.remove_hooks()
+    Env._instance = None
+    epl.init()
+    torch.manual_seed(99)
+
+    class New(nn.Module):
+        def __init__(self):
+            super().__init__()
+            with epl.replicate(1):
+                self.backbone = nn.Linear(8, 4)
+
+        def forward(self, x):
+            return self.backbone(x)
+
+    new = New()
+    engine2 = epl.Engine(new, loss_fn=nn.MSELoss(), optimizer="adamw",
+                         lr=1e-2)
+    engine2.load_checkpoint(str(tmp_path), load_optimizer=False,
+                            assign_map={"backbone": "enc"})
+    assert torch.allclose(new.backbone.weight.detach(), want)
