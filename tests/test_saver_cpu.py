@@ -231,3 +231,34 @@ def test_assign_map_renamed_restore(tmp_path):
     engine2.load_checkpoint(str(tmp_path), load_optimizer=False,
                             assign_map={"backbone": "enc"})
     assert torch.allclose(new.backbone.weight.detach(), want)
+
+
+def test_non_strict_partial_restore(tmp_path):
+    """strict=False tolerates missing checkpoint tensors (new heads)."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+
+    epl.init()
+    torch.manual_seed(41)
+    with epl.replicate(1):
+        small = nn.Sequential(nn.Linear(8, 4))
+    e1 = epl.Engine(small, loss_fn=nn.MSELoss(), optimizer="adamw",
+                    lr=1e-2)
+    e1.save_checkpoint(str(tmp_path))
+    want = small[0].weight.detach().clone()
+
+    hooks.remove_hooks()
+    Env._instance = None
+    epl.init()
+    torch.manual_seed(42)
+    with epl.replicate(1):
+        bigger = nn.Sequential(nn.Linear(8, 4), nn.Linear(4, 2))
+    e2 = epl.Engine(bigger, loss_fn=nn.MSELoss(), optimizer="adamw",
+                    lr=1e-2)
+    import pytest
+    with pytest.raises(KeyError):
+        e2.load_checkpoint(str(tmp_path), load_optimizer=False,
+                           strict=True)
+    e2.load_checkpoint(str(tmp_path), load_optimizer=False, strict=False)
+    assert torch.allclose(bigger[0].weight.detach(), want)
