@@ -558,3 +558,6 @@ class Engine:
                               assign_map=assign_map, strict=strict)
         meta = saver.ShardingLoader(path).meta
         self.global_step = meta.get("global_step", 0)
+        scale = meta.get("amp_loss_scale")
+        if scale is not None and self.amp.enabled:
+            self.amp.scaler.scale = float(scale)

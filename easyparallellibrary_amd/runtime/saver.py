@@ -43,6 +43,7 @@ def save_checkpoint(engine, path, save_optimizer=True):
             "per_replica": engine.per_replica,
             "num_stages": engine.num_stages,
             "global_step": engine.global_step,
+            "amp_loss_scale": engine.amp.loss_scale,
             "taskgraphs": [
                 {"index": tg.index, "type": tg.strategy_type,
                  "device_count": tg.device_count,
