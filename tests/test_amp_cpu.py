@@ -120,3 +120,31 @@ def test_amp_with_pipeline():
     res = run_multiprocess(_amp_pp_worker, world=2)
     assert res[0][0] is None
     assert res[1][-1] < res[1][0], res[1]
+
+
+def test_loss_scale_persists_across_checkpoint(tmp_path):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+
+    def build(seed):
+        hooks.remove_hooks()
+        Env._instance = None
+        epl.init(epl.Config({"amp.level": "O1", "amp.dtype": "fp16",
+                             "amp.loss_scale": "dynamic"}))
+        torch.manual_seed(seed)
+        with epl.replicate(1):
+            m = nn.Sequential(nn.Linear(8, 8), nn.Tanh(), nn.Linear(8, 2))
+        return epl.Engine(m, loss_fn=nn.MSELoss(), optimizer="adamw",
+                          lr=1e-3)
+
+    e1 = build(1)
+    e1.amp.scaler.scale = 4096.0  # pretend overflows shrank it
+    x, y = torch.randn(4, 8), torch.randn(4, 2)
+    e1.train_step(x, y)
+    scale_after = e1.amp.loss_scale
+    e1.save_checkpoint(str(tmp_path))
+    e2 = build(2)
+    assert e2.amp.loss_scale != scale_after  # fresh default
+    e2.load_checkpoint(str(tmp_path))
+    assert e2.amp.loss_scale == scale_after
