@@ -138,3 +138,26 @@ def test_tp_embedding_sharded():
     res = run_multiprocess(_tp_embedding_worker, world=2)
     assert res[0] == res[1]
     assert res[0][-1] < res[0][0]
+
+
+def _argmax_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.comm.backend import create_communicator
+    from easyparallellibrary_amd.ops.distributed_ops import (
+        distributed_argmax, distributed_equal)
+    epl.init()
+    comm = create_communicator("am", list(range(world)))
+    torch.manual_seed(70)
+    full = torch.randn(6, 10)          # same on every rank
+    shard = 10 // world
+    lo = rank * shard
+    local = full[:, lo:lo + shard]
+    pred = distributed_argmax(local, comm, vocab_begin=lo)
+    eq = distributed_equal(pred, full.argmax(dim=-1))
+    return pred, bool(eq.all())
+
+
+def test_distributed_argmax_matches_full():
+    res = run_multiprocess(_argmax_worker, world=2)
+    assert res[0][1] and res[1][1]
+    assert torch.equal(res[0][0], res[1][0])
