@@ -143,9 +143,11 @@ def test_tp_embedding_sharded():
 def _argmax_worker(rank, world):
     import easyparallellibrary_amd as epl
     from easyparallellibrary_amd.comm.backend import create_communicator
+    from easyparallellibrary_amd.env import Env
     from easyparallellibrary_amd.ops.distributed_ops import (
         distributed_argmax, distributed_equal)
     epl.init()
+    Env.get().get_or_create_process_group()
     comm = create_communicator("am", list(range(world)))
     torch.manual_seed(70)
     full = torch.randn(6, 10)          # same on every rank
