@@ -1,5 +1,7 @@
 """BERT model family (the flagship benchmark model: BERT-Large MLM on
-synthetic data — BASELINE.json configs 2 and 3).
+synthetic data — BASELINE.json configs 2 and 3; the reference's model
+zoo is external, see /root/reference/docs/en/tutorials/pipe.md for the
+staged-BERT usage this mirrors).
 
 Built with the epl annotation API: ``build_bert(num_stages=S)`` wraps the
 layer ranges in ``epl.replicate(name='stage_i')`` scopes so the engine

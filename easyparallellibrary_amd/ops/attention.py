@@ -1,5 +1,8 @@
 """Fused flash attention (hand-written CDNA4 kernels, head_dim 64, bf16).
 
+No reference counterpart (the reference's attention is plain TF ops);
+this is MI355X-native hot-path work — design notes in docs/kernels.md.
+
 Replaces torch SDPA (AOTriton) on the transformer hot path.  The HIP
 kernels (csrc/kernels/attention.hip) keep the whole online softmax in
 registers — swapped QK^T so each lane owns one query row, O accumulated

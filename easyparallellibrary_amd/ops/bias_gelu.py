@@ -1,5 +1,8 @@
 """Fused bias + GeLU (tanh approximation), CDNA4 kernel.
 
+No reference counterpart (plain TF ops there); MI355X-native fusion for
+the HBM-bound FFN epilogue.
+
 The FFN's first Linear runs without bias; the bias-add and activation fuse
 into one streaming kernel pass (saves two full activation read/writes per
 FFN on the 8 TB/s HBM bound).  CPU fallback in torch.
