@@ -118,12 +118,21 @@ def broadcast(inp, comm, root=0):
 
 class _AllToAll(torch.autograd.Function):
     """Equal-split all-to-all on dim 0; backward is all-to-all
-    (reference: nccl_ops.py:99-124)."""
+    (reference: nccl_ops.py:99-124).  ``compress`` ('fp16'/'bf16')
+    halves the wire bytes for fp32 payloads in BOTH directions
+    (reference: parallel/ops.py:485-495 alltoall fp16 option)."""
 
     @staticmethod
-    def forward(ctx, inp, comm):
+    def forward(ctx, inp, comm, compress):
         ctx.comm = comm
+        ctx.compress = compress
         inp = inp.contiguous()
+        wire_dt = _wire_dtype(inp, compress)
+        if wire_dt is not None:
+            w = inp.to(wire_dt)
+            o = w.new_empty(w.shape)
+            comm.all_to_all_single(o, w)
+            return o.to(inp.dtype)
         out = inp.new_empty(inp.shape)
         comm.all_to_all_single(out, inp)
         return out
@@ -131,12 +140,24 @@ class _AllToAll(torch.autograd.Function):
     @staticmethod
     def backward(ctx, grad):
         grad = grad.contiguous()
+        wire_dt = _wire_dtype(grad, ctx.compress)
+        if wire_dt is not None:
+            w = grad.to(wire_dt)
+            o = w.new_empty(w.shape)
+            ctx.comm.all_to_all_single(o, w)
+            return o.to(grad.dtype), None, None
         out = grad.new_empty(grad.shape)
         ctx.comm.all_to_all_single(out, grad)
-        return out, None
+        return out, None, None
 
 
-def all_to_all(inp, comm):
+def _wire_dtype(t, compress):
+    if not compress or t.dtype != torch.float32:
+        return None
+    return torch.float16 if compress == "fp16" else torch.bfloat16
+
+
+def all_to_all(inp, comm, compress=""):
     if comm.size == 1:
         return inp
-    return _AllToAll.apply(inp, comm)
+    return _AllToAll.apply(inp, comm, compress)

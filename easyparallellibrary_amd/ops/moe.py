@@ -69,6 +69,12 @@ class ExpertParallelMLP(nn.Module):
         self.w1._epl_shard_dim = 0
         self.w2._epl_shard_dim = 0
 
+    def _wire_compression(self):
+        # reference parallel/ops.py:485-495: optional fp16 all-to-all;
+        # only meaningful for fp32 activations (bf16 ships bf16 anyway)
+        from easyparallellibrary_amd.env import Env
+        return Env.get().config.communication.compression
+
     def forward(self, x):
         orig_shape = x.shape
         x = x.reshape(-1, self.hidden)
@@ -105,7 +111,8 @@ class ExpertParallelMLP(nn.Module):
         d = dispatched.reshape(self.world,
                                self.local_experts * capacity, self.hidden)
         if self.comm is not None and self.world > 1:
-            d = functional.all_to_all(d.contiguous(), self.comm)
+            d = functional.all_to_all(d.contiguous(), self.comm,
+                                      compress=self._wire_compression())
         # now d[w] = tokens sent by rank w for MY local experts
         d = d.reshape(self.world, self.local_experts, capacity, self.hidden)
         d = d.transpose(0, 1).reshape(self.local_experts,
@@ -123,7 +130,8 @@ class ExpertParallelMLP(nn.Module):
         h = h.transpose(0, 1).reshape(
             self.world, self.local_experts * capacity, self.hidden)
         if self.comm is not None and self.world > 1:
-            h = functional.all_to_all(h.contiguous(), self.comm)
+            h = functional.all_to_all(h.contiguous(), self.comm,
+                                      compress=self._wire_compression())
         h = h.reshape(self.num_experts, capacity, self.hidden)
 
         out = x.new_zeros(n_tokens, self.hidden)

@@ -160,3 +160,31 @@ def test_pp2_ep2_dp2_full_3d():
     assert last[0] == last[1] == last[2] == last[3]
     for a, b in zip(base[1], last[0]):
         assert abs(a - b) < 1e-4, (base[1], last[0])
+
+
+def _moe_comp_worker(rank, world, comp):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models.moe_transformer import (
+        build_moe_transformer)
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True,
+                         "communication.compression": comp}))
+    torch.manual_seed(30)
+    model = build_moe_transformer(world=world, layers=1, hidden=32,
+                                  heads=4, ffn=64, num_experts=4,
+                                  vocab_size=128, max_pos=32)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=1e-3)
+    torch.manual_seed(31)
+    ids = torch.randint(0, 128, (4, 16))
+    tgt = torch.randint(0, 128, (4, 16))
+    return [float(engine.train_step(ids, tgt)) for _ in range(3)]
+
+
+def test_moe_a2a_wire_compression_tracks_fp32():
+    base = run_multiprocess(_moe_comp_worker, world=2, args=("",))
+    comp = run_multiprocess(_moe_comp_worker, world=2, args=("bf16",))
+    assert base[0] == base[1] and comp[0] == comp[1]
+    assert all(abs(a - b) < 5e-2 for a, b in zip(base[0], comp[0])), (
+        base[0], comp[0])
