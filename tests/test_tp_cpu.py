@@ -163,3 +163,17 @@ def test_distributed_argmax_matches_full():
     res = run_multiprocess(_argmax_worker, world=2)
     assert res[0][1] and res[1][1]
     assert torch.equal(res[0][0], res[1][0])
+
+
+def test_distributed_glorot_full_fan_variance():
+    from easyparallellibrary_amd.ops.initializers import (
+        distributed_glorot_uniform_)
+    import math
+    g = torch.Generator().manual_seed(0)
+    shard = torch.empty(64, 256)           # half of a [128, 256] layer
+    distributed_glorot_uniform_(shard, full_fan_in=256, full_fan_out=128,
+                                generator=g)
+    limit = math.sqrt(6.0 / (256 + 128))
+    assert shard.abs().max() <= limit
+    # variance of U(-l, l) is l^2/3; sampled variance within 10%
+    assert abs(shard.var().item() - limit ** 2 / 3) < 0.1 * limit ** 2 / 3
