@@ -333,6 +333,8 @@ class Engine:
             fg.zero_grad()
 
     def forward(self, x):
+        """Plain forward (this rank's stage under pipeline parallelism —
+        see ``eval_step``)."""
         return self._runnable(x)
 
     # ---- the training step ---------------------------------------------------
