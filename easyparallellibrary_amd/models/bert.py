@@ -17,6 +17,7 @@ from easyparallellibrary_amd.models.transformer import (Block, Embeddings,
                                                         LMHead, init_weights)
 
 BERT_CONFIGS = {
+    "bert-tiny": dict(layers=2, hidden=128, heads=2, ffn=512),
     "bert-base": dict(layers=12, hidden=768, heads=12, ffn=3072),
     "bert-large": dict(layers=24, hidden=1024, heads=16, ffn=4096),
 }
