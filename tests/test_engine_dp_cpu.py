@@ -122,3 +122,44 @@ def test_manual_accumulation_dp2():
     assert base[0] == base[1] and man[0] == man[1]
     assert all(abs(a - b) < 1e-6 for a, b in zip(base[0], man[0])), (
         base[0], man[0])
+
+
+def test_bucket_layout_invariants_randomized():
+    """GradReducer bucket invariants over random param shapes:
+    buckets tile the arena exactly, split at ZeRO shard edges, and every
+    param gates every bucket its extent overlaps."""
+    import random
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.comm.backend import create_communicator
+    from easyparallellibrary_amd.comm.pool import CommunicationPool
+    from easyparallellibrary_amd.parallel.dp import (FlatParamGroup,
+                                                     GradReducer, _aligned)
+    rng = random.Random(99)
+    epl.init()
+    for trial in range(25):
+        n_params = rng.randint(1, 12)
+        params = [nn.Parameter(torch.randn(rng.randint(1, 5000)))
+                  for _ in range(n_params)]
+        shard_owners = rng.random() < 0.5
+        world = rng.choice([2, 4]) if shard_owners else 1
+        fg = FlatParamGroup(params, torch.device("cpu"),
+                            pad_to_multiple=world if shard_owners else 1)
+        pool = CommunicationPool("bl{}".format(trial), [0], 1)
+        red = GradReducer(fg, pool, bucket_bytes=rng.choice(
+            [1 << 10, 1 << 14, 1 << 22]), overlap=False,
+            shard_owners=False)
+        # tiling: contiguous, ordered, covers [0, total)
+        pos = 0
+        for (start, end, ps, owner) in red.buckets:
+            assert start == pos and end > start
+            pos = end
+        assert pos == fg.total
+        # gating: every param's extent is covered by its gated buckets
+        for p, off in zip(fg.ordered, fg.offsets):
+            lo, hi = off, off + _aligned(p.numel())
+            gated = [b for b in red.buckets
+                     if any(q is p for q in b[2])]
+            assert gated, "param gates no bucket"
+            assert min(b[0] for b in gated) <= lo
+            assert max(b[1] for b in gated) >= hi
+        red.remove_hooks()
