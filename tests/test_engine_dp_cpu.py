@@ -163,3 +163,43 @@ def test_bucket_layout_invariants_randomized():
             assert min(b[0] for b in gated) <= lo
             assert max(b[1] for b in gated) >= hi
         red.remove_hooks()
+
+
+class _FakePool:
+    """Pool stub exposing size>1 comms so shard_owners layout activates
+    without multi-process setup (layout logic only; no comm calls)."""
+
+    class _C:
+        def __init__(self, n):
+            self.size = n
+            self.rank = 0
+
+    def __init__(self, world):
+        self.comms = [self._C(world)]
+
+
+def test_zero_bucket_shard_edges_randomized():
+    import random
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.parallel.dp import (FlatParamGroup,
+                                                     GradReducer)
+    rng = random.Random(7)
+    epl.init()
+    for trial in range(25):
+        world = rng.choice([2, 4, 8])
+        params = [nn.Parameter(torch.randn(rng.randint(1, 4000)))
+                  for _ in range(rng.randint(1, 10))]
+        fg = FlatParamGroup(params, torch.device("cpu"),
+                            pad_to_multiple=world)
+        red = GradReducer(fg, _FakePool(world), bucket_bytes=1 << 12,
+                          overlap=False, shard_owners=True)
+        shard = fg.total // world
+        pos = 0
+        for (start, end, ps, owner) in red.buckets:
+            assert start == pos
+            pos = end
+            # bucket never crosses a shard edge; owner matches its shard
+            assert start // shard == (end - 1) // shard
+            assert owner == min(start // shard, world - 1)
+        assert pos == fg.total
+        red.remove_hooks()
