@@ -177,3 +177,21 @@ def test_distributed_glorot_full_fan_variance():
     assert shard.abs().max() <= limit
     # variance of U(-l, l) is l^2/3; sampled variance within 10%
     assert abs(shard.var().item() - limit ** 2 / 3) < 0.1 * limit ** 2 / 3
+
+
+def test_shard_roundtrip_randomized():
+    """shard_offset/shard_size tile [0, n) exactly for random (n, w)."""
+    import random
+    from easyparallellibrary_amd.ops.distributed_dense import (
+        shard_offset, shard_size)
+    rng = random.Random(3)
+    for _ in range(200):
+        w = rng.randint(1, 9)
+        n = rng.randint(w, 10000)
+        pos = 0
+        for s in range(w):
+            assert shard_offset(n, w, s) == pos
+            sz = shard_size(n, w, s)
+            assert sz >= n // w
+            pos += sz
+        assert pos == n
