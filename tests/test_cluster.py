@@ -37,3 +37,23 @@ def test_specific_layout():
     vds = sl.slices()
     assert vds[0].local_ranks(1) == [1]
     assert vds[1].replica_of_rank(3) == 1
+
+
+def test_layout_tiling_randomized():
+    """Layout.slices() covers every rank exactly once per replica set."""
+    import random
+    rng = random.Random(5)
+    for _ in range(100):
+        counts = [rng.randint(1, 4) for _ in range(rng.randint(1, 5))]
+        per = sum(counts)
+        replicas = rng.randint(1, 4)
+        world = per * replicas
+        lay = Layout(world, counts)
+        assert lay.num_replicas == replicas
+        seen = set()
+        for vd in lay.slices():
+            for rep in range(replicas):
+                for r in vd.local_ranks(rep):
+                    assert r not in seen
+                    seen.add(r)
+        assert seen == set(range(world))
