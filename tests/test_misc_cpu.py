@@ -154,3 +154,22 @@ def test_examples_tiny_cpu(script):
         env=env, capture_output=True, text=True, timeout=240)
     assert r.returncode == 0, r.stderr[-2000:]
     assert "step 9" in r.stdout, r.stdout[-500:]
+
+
+def test_io_slicing_tiling_randomized():
+    import random
+    from easyparallellibrary_amd.utils.io_slicing import slice_files
+    rng = random.Random(8)
+    for _ in range(100):
+        w = rng.randint(1, 8)
+        n = rng.randint(w, 500)
+        files = [str(i) for i in range(n)]
+        for unbalanced in (False, True):
+            parts = [slice_files(files, w, r, unbalanced=unbalanced)
+                     for r in range(w)]
+            assert sum(parts, []) == files
+        dropped = [slice_files(files, w, r, drop_last=True)
+                   for r in range(w)]
+        sizes = {len(p) for p in dropped}
+        assert len(sizes) == 1
+        assert sum(dropped, []) == files[:n - n % w]
