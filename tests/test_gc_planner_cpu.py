@@ -130,3 +130,26 @@ def test_cost_model_and_profiler():
     mp = MemoryProfiler()
     mp.after_step(1)
     assert len(mp.records) == 1
+
+
+def test_partition_balance_randomized():
+    """Contiguity/coverage + optimality bound: DP max-chunk cost is never
+    above the trivial upper bound (total) and never below total/k."""
+    import random
+    from easyparallellibrary_amd.parallel.partitioner import (
+        partition_balance)
+    rng = random.Random(12)
+    for _ in range(100):
+        n = rng.randint(1, 40)
+        k = rng.randint(1, 8)
+        w = [rng.uniform(0.1, 10.0) for _ in range(n)]
+        parts = partition_balance(w, k)
+        assert len(parts) == min(k, n)
+        flat = sum(parts, [])
+        assert flat == list(range(n))          # contiguous, in order
+        assert all(parts)                       # no empty chunk
+        costs = [sum(w[i] for i in p) for p in parts]
+        assert max(costs) >= sum(w) / len(parts) - 1e-9
+        # DP optimality sanity: no single element exceeds... the max chunk
+        # must be at least the largest single weight
+        assert max(costs) >= max(w) - 1e-9
