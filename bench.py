@@ -38,6 +38,10 @@ def parse_args():
     p.add_argument("--zero", default=None, choices=[None, "", "v0", "v1"])
     p.add_argument("--gc", default=None, choices=[None, "", "auto"])
     p.add_argument("--offload", default=None, choices=[None, "", "v0"])
+    p.add_argument("--layers", type=int, default=0,
+                   help="override model depth (moe/moe_pp smoke runs)")
+    p.add_argument("--hidden", type=int, default=0,
+                   help="override hidden size (moe/moe_pp smoke runs)")
     p.add_argument("--device", default=None)
     return p.parse_args()
 
@@ -156,8 +160,10 @@ def build_moe_bench(args, epl, world, on_gpu, dtype):
     seq = args.seq_len if args.seq_len != 512 else 1024
     epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
     vocab = 32000
-    model = build_moe_transformer(world=world, layers=12, hidden=1024,
-                                  heads=16, ffn=4096,
+    layers = args.layers or 12
+    hidden = args.hidden or 1024
+    model = build_moe_transformer(world=world, layers=layers, hidden=hidden,
+                                  heads=16, ffn=4 * hidden,
                                   num_experts=max(8, world),
                                   vocab_size=vocab, max_pos=seq)
     engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
@@ -165,7 +171,8 @@ def build_moe_bench(args, epl, world, on_gpu, dtype):
     rank = int(os.environ.get("RANK", "0"))
     ids, tgt = gpt2.synthetic_lm_batch(batch, seq, vocab,
                                        device=engine.device, seed=55 + rank)
-    meta = {"model": "moe-transformer-12L", "global_batch": batch * world,
+    meta = {"model": "moe-transformer-{}L".format(layers),
+            "global_batch": batch * world,
             "seq_len": seq, "parallelism": "dp{}_ep{}".format(world, world)}
     return engine, (ids, tgt), batch * world, meta
 
@@ -189,8 +196,10 @@ def build_moe_pp_bench(args, epl, world, on_gpu, dtype):
         "pipeline.num_micro_batch": min(4, batch),
     }))
     vocab = 32000
-    model = build_moe_pipeline(stages=2, ep=ep, layers=12, hidden=1024,
-                               heads=16, ffn=4096,
+    layers = args.layers or 12
+    hidden = args.hidden or 1024
+    model = build_moe_pipeline(stages=2, ep=ep, layers=layers, hidden=hidden,
+                               heads=16, ffn=4 * hidden,
                                num_experts=max(8, 2 * ep),
                                vocab_size=vocab, max_pos=seq)
     engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
@@ -199,7 +208,8 @@ def build_moe_pp_bench(args, epl, world, on_gpu, dtype):
     ids, tgt = gpt2.synthetic_lm_batch(batch, seq, vocab,
                                        device=engine.device, seed=55 + rank)
     streams = world // 2  # per-stage positions x replicas
-    meta = {"model": "moe-transformer-12L", "global_batch": batch * streams,
+    meta = {"model": "moe-transformer-{}L".format(layers),
+            "global_batch": batch * streams,
             "seq_len": seq,
             "parallelism": "pp2_ep{}_dp{}".format(ep, world // (2 * ep))}
     return engine, (ids, tgt), batch * streams, meta
