@@ -64,6 +64,15 @@ def test_ring_attention_matches_serial():
             assert dq_err < 1e-4, (causal, dq_err)
 
 
+def test_ring_attention_world4_causal():
+    """More hops (3 rotations), causal masking across blocks."""
+    res = run_multiprocess(_ring_worker, world=4, args=(True,),
+                           timeout=240)
+    for fwd_err, dq_err in res:
+        assert fwd_err < 1e-5, fwd_err
+        assert dq_err < 1e-4, dq_err
+
+
 def test_ring_module_single_rank_matches_plain():
     """world=1 degenerate: RingSelfAttention == plain attention."""
     import easyparallellibrary_amd as epl
