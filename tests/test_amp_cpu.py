@@ -5,7 +5,7 @@ import torch
 import torch.nn as nn
 
 import easyparallellibrary_amd as epl
-from easyparallellibrary_amd.runtime.amp import AmpContext, DynamicLossScaler
+from easyparallellibrary_amd.runtime.amp import DynamicLossScaler
 
 
 def test_amp_o1_bf16_trains():

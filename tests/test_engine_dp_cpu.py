@@ -130,7 +130,6 @@ def test_bucket_layout_invariants_randomized():
     param gates every bucket its extent overlaps."""
     import random
     import easyparallellibrary_amd as epl
-    from easyparallellibrary_amd.comm.backend import create_communicator
     from easyparallellibrary_amd.comm.pool import CommunicationPool
     from easyparallellibrary_amd.parallel.dp import (FlatParamGroup,
                                                      GradReducer, _aligned)
