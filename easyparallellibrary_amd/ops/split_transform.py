@@ -65,6 +65,8 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
     its parent by dotted name."""
     from easyparallellibrary_amd.ops.moe import ExpertParallelMLP
     replaced = 0
+    done = set()  # module ids already transformed (named_modules visits
+    #               a nested module both as child and as parent)
     for root_i, root in enumerate(tg.modules):
         if isinstance(root, (nn.Linear, nn.Embedding)):
             new = _make_sharded(root, comm, gather_input)
@@ -84,14 +86,20 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
             continue
         for parent_name, parent in root.named_modules():
             if isinstance(parent, ExpertParallelMLP):
-                parent.set_comm(comm)
-                replaced += 1
+                if id(parent) not in done:
+                    done.add(id(parent))
+                    parent.set_comm(comm)
+                    replaced += 1
                 continue
             for child_name, child in list(parent.named_children()):
+                if id(child) in done:
+                    continue
                 if isinstance(child, ExpertParallelMLP):
+                    done.add(id(child))
                     child.set_comm(comm)
                     replaced += 1
                 elif isinstance(child, (nn.Linear, nn.Embedding)):
+                    done.add(id(child))
                     _replace_module(parent, child_name,
                                     _make_sharded(child, comm, gather_input))
                     replaced += 1

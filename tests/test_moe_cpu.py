@@ -188,3 +188,42 @@ def test_moe_a2a_wire_compression_tracks_fp32():
     assert base[0] == base[1] and comp[0] == comp[1]
     assert all(abs(a - b) < 5e-2 for a, b in zip(base[0], comp[0])), (
         base[0], comp[0])
+
+
+def _moe_container_worker(rank, world):
+    """ExpertParallelMLP nested under a CONTAINER split root must be
+    sharded exactly once (regression: named_modules visits it both as
+    child and as parent)."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.ops.moe import ExpertParallelMLP
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    torch.manual_seed(30)
+    with epl.replicate(world, name="trunk"):
+        emb = nn.Embedding(64, 32)
+    with epl.split(world, name="experts"):
+        box = nn.Sequential()
+        box.add_module("moe", ExpertParallelMLP(32, 64, 4, top_k=2))
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb, self.box = emb, box
+
+        def forward(self, ids):
+            return self.box(self.emb(ids))
+
+    engine = epl.Engine(M(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-3)
+    torch.manual_seed(31)
+    ids = torch.randint(0, 64, (4, 8))
+    y = torch.randn(4, 8, 32)
+    losses = [float(engine.train_step(ids, y)) for _ in range(2)]
+    return losses, box.moe.local_experts
+
+
+def test_moe_nested_in_container_split_root():
+    res = run_multiprocess(_moe_container_worker, world=2)
+    assert res[0][1] == 2 and res[1][1] == 2   # sharded once: 4/2
+    assert res[0][0] == res[1][0]
