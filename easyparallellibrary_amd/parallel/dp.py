@@ -235,7 +235,12 @@ class GradReducer:
             return
         self._launched[bi] = True
         start, end, _, owner = self.buckets[bi]
-        comm = self.pool.next_comm()
+        # DETERMINISTIC bucket->communicator mapping: every rank must
+        # issue the same bucket on the same comm, but buckets become
+        # ready in backward-completion order which is only guaranteed to
+        # match across ranks for identical graphs — index-based mapping
+        # stays correct even if launch order ever diverges.
+        comm = self.pool.comms[bi % self.pool.size]
         buf = self.group.grad_arena[start:end]
         wire = buf
         if self.compression and comm.size > 1:
