@@ -49,13 +49,21 @@ class CheckpointWrapper(nn.Module):
 
 
 def _excluded(mod):
+    """Modules whose forward issues COLLECTIVES never recompute
+    (reference avoids recomputing a2a ops, constant.py:97): replaying a
+    collective inside a checkpointed backward risks cross-rank ordering
+    divergence."""
     from easyparallellibrary_amd.ops.distributed_dense import (
         ColumnParallelLinear, RowParallelLinear)
     from easyparallellibrary_amd.ops.moe import ExpertParallelMLP
+    from easyparallellibrary_amd.ops.ring_attention import (
+        RingSelfAttention)
     from easyparallellibrary_amd.ops.split_transform import (
         VocabParallelEmbedding)
+    from easyparallellibrary_amd.ops.ulysses import UlyssesSelfAttention
     return isinstance(mod, (ColumnParallelLinear, RowParallelLinear,
-                            ExpertParallelMLP, VocabParallelEmbedding))
+                            ExpertParallelMLP, VocabParallelEmbedding,
+                            UlyssesSelfAttention, RingSelfAttention))
 
 
 def select_checkpoint_modules(root, mode="auto", min_repeat=3):
