@@ -30,5 +30,6 @@ for step in range(10):
     ids, tgt = bert.synthetic_mlm_batch(BATCH, SEQ, device=engine.device,
                                         seed=step)
     loss = engine.train_step(ids, tgt)
+    merged = engine.all_reduce_metric(loss)   # collective: every rank
     if engine.rank == 0:
-        print("step", step, "loss", float(engine.all_reduce_metric(loss)))
+        print("step", step, "loss", float(merged))

@@ -143,7 +143,8 @@ print("worker ok", engine.rank)
 
 @pytest.mark.parametrize("script", ["train_bert_dp.py",
                                     "train_bert_pipeline.py",
-                                    "train_moe.py"])
+                                    "train_moe.py",
+                                    "train_long_context.py"])
 def test_examples_tiny_cpu(script):
     """Every example runs end-to-end in tiny mode on CPU."""
     import subprocess
