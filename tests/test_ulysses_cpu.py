@@ -55,3 +55,62 @@ def test_ulysses_sp2_matches_serial():
     assert torch.allclose(total_qkv, attn.qkv.weight.grad, atol=1e-5)
     total_proj = results[0][3] + results[1][3]
     assert torch.allclose(total_proj, attn.proj.weight.grad, atol=1e-5)
+
+
+def _ulysses_engine_worker(rank, world):
+    """Engine-integrated Ulysses SP (mirror of the ring-attention
+    engine test): sequence-sharded data, replicated params, DP-averaged
+    grads across the SP group."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.comm.backend import create_communicator
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.ops.ulysses import UlyssesSelfAttention
+
+    epl.init()
+    Env.get().get_or_create_process_group()
+    sp = (create_communicator("sp_ul_eng", list(range(world)))
+          if world > 1 else None)
+    torch.manual_seed(90)
+
+    class LM(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb = nn.Embedding(64, H)
+            self.attn = UlyssesSelfAttention(H, HEADS, comm=sp,
+                                             causal=True)
+            self.head = nn.Linear(H, 64)
+
+        def forward(self, ids):
+            return self.head(self.attn(self.emb(ids)))
+
+    def lm_loss(logits, targets):
+        return nn.functional.cross_entropy(logits.reshape(-1, 64),
+                                           targets)
+
+    with epl.replicate(1):
+        model = LM()
+    engine = epl.Engine(model, loss_fn=lm_loss, optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(91)
+    ids = torch.randint(0, 64, (2, S))
+    tgt = torch.randint(0, 64, (2, S))
+    sl = S // world
+    lo = rank * sl
+    out = []
+    for _ in range(3):
+        loss = engine.train_step(ids[:, lo:lo + sl],
+                                 tgt[:, lo:lo + sl].reshape(-1))
+        out.append(float(engine.all_reduce_metric(loss)))
+    return out
+
+
+def test_ulysses_sp_engine_matches_serial():
+    serial = _ulysses_engine_worker(0, 1)
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    hooks.remove_hooks()
+    Env._instance = None
+    sp2 = run_multiprocess(_ulysses_engine_worker, world=2)
+    assert sp2[0] == sp2[1]
+    for a, b in zip(serial, sp2[0]):
+        assert abs(a - b) < 1e-5, (serial, sp2[0])
