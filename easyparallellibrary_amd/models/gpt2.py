@@ -10,6 +10,7 @@ from easyparallellibrary_amd.models.transformer import (Block, Embeddings,
                                                         LMHead, init_weights)
 
 GPT2_CONFIGS = {
+    "gpt2-tiny": dict(layers=2, hidden=128, heads=2, ffn=512),
     "gpt2-small": dict(layers=12, hidden=768, heads=12, ffn=3072),
     "gpt2-medium": dict(layers=24, hidden=1024, heads=16, ffn=4096),
     "gpt2-large": dict(layers=36, hidden=1280, heads=20, ffn=5120),

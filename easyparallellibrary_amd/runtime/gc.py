@@ -81,11 +81,14 @@ def select_checkpoint_modules(root, mode="auto", min_repeat=3):
     for m in root.modules():
         if any(True for _ in m.parameters(recurse=True)) and not _excluded(m):
             counts[type(m)] += 1
+    from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
+    from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
     candidates = [
         (cls, n) for cls, n in counts.items()
         if n >= min_repeat and cls not in (nn.Linear, nn.Embedding,
                                            nn.LayerNorm, nn.Conv2d,
-                                           nn.BatchNorm2d)]
+                                           nn.BatchNorm2d, FusedLayerNorm,
+                                           FusedBiasGelu)]
     if not candidates:
         return hits
     # the repeated class with the most parameters per instance wins

@@ -153,3 +153,20 @@ def test_partition_balance_randomized():
         # DP optimality sanity: no single element exceeds... the max chunk
         # must be at least the largest single weight
         assert max(costs) >= max(w) - 1e-9
+
+
+def test_gc_auto_never_wraps_fused_elementwise():
+    """Regression: with few repeated blocks, auto-GC must not fall back
+    to wrapping FusedLayerNorm/FusedBiasGelu (their duck-typed methods —
+    forward_with_sum — break under a wrapper)."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models import gpt2
+    from easyparallellibrary_amd.runtime.gc import (
+        select_checkpoint_modules)
+    epl.init()
+    model = gpt2.build_gpt2("gpt2-tiny", vocab_size=128, max_pos=32)
+    hits = select_checkpoint_modules(model, mode="auto")
+    from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
+    from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
+    for _, _, child in hits:
+        assert not isinstance(child, (FusedLayerNorm, FusedBiasGelu))
