@@ -482,11 +482,18 @@ class Engine:
 
     # ---- eval ----------------------------------------------------------------
     def eval_step(self, inputs):
-        """Forward without grad.  Under pipeline parallelism this runs
-        ONLY this rank's stage (feed it the previous stage's activations
-        or run evaluation data-parallel on a non-pipelined engine)."""
-        with torch.no_grad():
-            return self._runnable(inputs)
+        """Forward in eval mode without grad (dropout off, BatchNorm uses
+        running stats and does not update them).  Under pipeline
+        parallelism this runs ONLY this rank's stage (feed it the
+        previous stage's activations or run evaluation data-parallel on
+        a non-pipelined engine)."""
+        was_training = self._runnable.training
+        self._runnable.eval()
+        try:
+            with torch.no_grad():
+                return self._runnable(inputs)
+        finally:
+            self._runnable.train(was_training)
 
     # ---- merged outputs (reference: parallel/parallel.py:233-353) ------------
     def all_reduce_metric(self, value, op="mean"):

@@ -210,3 +210,24 @@ def test_random_architectures_train():
         for _ in range(2):
             loss = engine.train_step(x, y)
         assert torch.isfinite(loss), (trial, loss)
+
+
+def test_eval_step_uses_eval_mode():
+    """eval_step must not update BatchNorm running stats or apply
+    dropout, and must restore training mode afterwards."""
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(4)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(8, 8), nn.BatchNorm1d(8),
+                              nn.Dropout(0.5), nn.Linear(8, 2))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-3)
+    x = torch.randn(16, 8)
+    engine.train_step(x, torch.randn(16, 2))
+    rm = model[1].running_mean.clone()
+    y1 = engine.eval_step(x)
+    y2 = engine.eval_step(x)
+    assert torch.equal(y1, y2)                     # dropout off
+    assert torch.equal(model[1].running_mean, rm)  # stats frozen
+    assert model.training                          # mode restored
