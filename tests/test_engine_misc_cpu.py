@@ -231,3 +231,20 @@ def test_eval_step_uses_eval_mode():
     assert torch.equal(y1, y2)                     # dropout off
     assert torch.equal(model[1].running_mean, rm)  # stats frozen
     assert model.training                          # mode restored
+
+
+def test_bf16_training_on_cpu():
+    """Pure-bf16 engine on CPU (eager fallbacks + fp32 master)."""
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(0)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(8, 16), nn.Tanh(),
+                              nn.Linear(16, 2))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2, dtype=torch.bfloat16)
+    x = torch.randn(8, 8).bfloat16()
+    y = torch.randn(8, 2).bfloat16()
+    losses = [float(engine.train_step(x, y)) for _ in range(3)]
+    assert losses[-1] < losses[0]
+    assert model[0].weight.dtype == torch.bfloat16
