@@ -161,3 +161,47 @@ def all_to_all(inp, comm, compress=""):
     if comm.size == 1:
         return inp
     return _AllToAll.apply(inp, comm, compress)
+
+
+class _CopyToGroup(torch.autograd.Function):
+    """Megatron 'f' operator: identity forward, all-reduce backward —
+    marks the point where a replicated activation enters sharded
+    compute (each shard's input-gradient is a partial sum)."""
+
+    @staticmethod
+    def forward(ctx, inp, comm):
+        ctx.comm = comm
+        return inp
+
+    @staticmethod
+    def backward(ctx, grad):
+        g = grad.contiguous().clone()
+        ctx.comm.all_reduce(g, op="sum")
+        return g, None
+
+
+def copy_to_group(inp, comm):
+    if comm is None or comm.size == 1:
+        return inp
+    return _CopyToGroup.apply(inp, comm)
+
+
+class _ReduceFromGroup(torch.autograd.Function):
+    """Megatron 'g' operator: all-reduce forward, identity backward —
+    for losses computed identically on every rank of the group."""
+
+    @staticmethod
+    def forward(ctx, inp, comm):
+        out = inp.contiguous().clone()
+        comm.all_reduce(out, op="sum")
+        return out
+
+    @staticmethod
+    def backward(ctx, grad):
+        return grad, None
+
+
+def reduce_from_group(inp, comm):
+    if comm is None or comm.size == 1:
+        return inp
+    return _ReduceFromGroup.apply(inp, comm)

@@ -84,10 +84,14 @@ class RowParallelLinear(nn.Module):
     partial outputs (the TP all-reduce over xGMI of BASELINE config 4)."""
 
     def __init__(self, in_features, out_features, comm, bias=True,
-                 source=None):
+                 source=None, pre_sharded=False):
         super().__init__()
         from easyparallellibrary_amd.comm import functional
         self._fn = functional
+        # pre_sharded: the input already IS this rank's in-feature shard
+        # (e.g. it came from a paired ColumnParallelLinear) — skip the
+        # local slice
+        self.pre_sharded = pre_sharded
         self.comm = comm
         self.nshards = comm.size
         self.shard = max(comm.rank, 0)
@@ -109,9 +113,12 @@ class RowParallelLinear(nn.Module):
                                         out_features)
 
     def forward(self, x):
-        xs = x[..., self.in_offset:self.in_offset + self.local_in]
+        xs = (x if self.pre_sharded
+              else x[..., self.in_offset:self.in_offset + self.local_in])
         out = F.linear(xs, self.weight)
-        out = self._fn.all_reduce(out, self.comm)
+        # Megatron 'g': sum the partials forward, pass the (replicated)
+        # output-gradient straight through backward
+        out = self._fn.reduce_from_group(out, self.comm)
         if self.bias is not None:
             out = out + self.bias
         return out

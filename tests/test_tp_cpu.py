@@ -195,3 +195,37 @@ def test_shard_roundtrip_randomized():
             assert sz >= n // w
             pos += sz
         assert pos == n
+
+
+def _tp_mlp_worker(rank, world):
+    """Megatron-style column->row MLP pair: full output on every rank
+    after ONE all-reduce; matches the serial MLP exactly."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.comm.backend import create_communicator
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.ops.tp_mlp import TensorParallelMLP
+    epl.init()
+    Env.get().get_or_create_process_group()
+    comm = create_communicator("tpmlp", list(range(world)))
+    torch.manual_seed(21)
+    fc1 = nn.Linear(16, 32)
+    fc2 = nn.Linear(32, 16)
+    tp = TensorParallelMLP(16, 32, comm, act=nn.GELU(),
+                           source_fc1=fc1, source_fc2=fc2)
+    torch.manual_seed(22)
+    x = torch.randn(4, 16, requires_grad=True)
+    out = tp(x)
+    out.square().sum().backward()
+
+    xr = x.detach().clone().requires_grad_(True)
+    ref = fc2(nn.functional.gelu(fc1(xr)))
+    ref.square().sum().backward()
+    return ((out - ref).abs().max().item(),
+            (x.grad - xr.grad).abs().max().item())
+
+
+def test_tensor_parallel_mlp_pair():
+    res = run_multiprocess(_tp_mlp_worker, world=2)
+    for fwd_err, dx_err in res:
+        assert fwd_err < 1e-5, fwd_err
+        assert dx_err < 1e-4, dx_err
