@@ -1,4 +1,4 @@
-"""Megatron-style tensor-parallel MLP pair.
+"""Megatron-style tensor-parallel transformer blocks (MLP + attention).
 
 Beyond the reference's single-sharded-layer pattern: the FIRST linear is
 column-parallel (output features sharded, NO gather), the activation runs
@@ -34,3 +34,47 @@ class TensorParallelMLP(nn.Module):
         # input-gradients backward
         x = functional.copy_to_group(x, self.fc1.comm)
         return self.fc2(self.act(self.fc1(x)))
+
+
+class TensorParallelSelfAttention(nn.Module):
+    """Head-sharded self-attention: q/k/v projections column-parallel
+    (each rank owns heads/W full heads), attention runs on the local
+    heads over the full sequence, the output projection is row-parallel
+    — full tensor out after ONE all-reduce (the Megatron attention
+    block).  Requires heads % comm.size == 0."""
+
+    def __init__(self, hidden, num_heads, comm, causal=False,
+                 source=None):
+        super().__init__()
+        assert hidden % num_heads == 0
+        w = comm.size if comm is not None else 1
+        assert num_heads % max(1, w) == 0, "heads must divide TP degree"
+        self.comm = comm
+        self.hidden = hidden
+        self.head_dim = hidden // num_heads
+        self.local_heads = num_heads // max(1, w)
+        self.causal = causal
+
+        def col(src):
+            return ColumnParallelLinear(hidden, hidden, comm, bias=True,
+                                        gather_input=False, source=src)
+
+        sq = sk = sv = sp = None
+        if source is not None:   # an ops-style module with q/k/v/proj
+            sq, sk, sv, sp = (source.q, source.k, source.v, source.proj)
+        self.q = col(sq)
+        self.k = col(sk)
+        self.v = col(sv)
+        self.proj = RowParallelLinear(hidden, hidden, comm, bias=True,
+                                      source=sp, pre_sharded=True)
+
+    def forward(self, x):
+        from easyparallellibrary_amd.ops.attention import flash_attention
+        x = functional.copy_to_group(x, self.comm)
+        b, s, _ = x.shape
+        shp = (b, s, self.local_heads, self.head_dim)
+        q = self.q(x).reshape(shp).transpose(1, 2)
+        k = self.k(x).reshape(shp).transpose(1, 2)
+        v = self.v(x).reshape(shp).transpose(1, 2)
+        o = flash_attention(q, k, v, causal=self.causal)
+        return self.proj(o.transpose(1, 2).reshape(b, s, -1))

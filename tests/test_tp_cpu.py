@@ -229,3 +229,60 @@ def test_tensor_parallel_mlp_pair():
     for fwd_err, dx_err in res:
         assert fwd_err < 1e-5, fwd_err
         assert dx_err < 1e-4, dx_err
+
+
+class _PlainQKVAttention(nn.Module):
+    """Serial reference with separate q/k/v projections (the layout
+    TensorParallelSelfAttention shards)."""
+
+    def __init__(self, hidden, heads, causal):
+        super().__init__()
+        self.q = nn.Linear(hidden, hidden)
+        self.k = nn.Linear(hidden, hidden)
+        self.v = nn.Linear(hidden, hidden)
+        self.proj = nn.Linear(hidden, hidden)
+        self.heads = heads
+        self.d = hidden // heads
+        self.causal = causal
+
+    def forward(self, x):
+        b, s, _ = x.shape
+        shp = (b, s, self.heads, self.d)
+        q = self.q(x).reshape(shp).transpose(1, 2)
+        k = self.k(x).reshape(shp).transpose(1, 2)
+        v = self.v(x).reshape(shp).transpose(1, 2)
+        o = torch.nn.functional.scaled_dot_product_attention(
+            q, k, v, is_causal=self.causal)
+        return self.proj(o.transpose(1, 2).reshape(b, s, -1))
+
+
+def _tp_attn_worker(rank, world, causal):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.comm.backend import create_communicator
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.ops.tp_mlp import (
+        TensorParallelSelfAttention)
+    epl.init()
+    Env.get().get_or_create_process_group()
+    comm = create_communicator("tpattn", list(range(world)))
+    torch.manual_seed(41)
+    ref = _PlainQKVAttention(32, 4, causal)
+    tp = TensorParallelSelfAttention(32, 4, comm, causal=causal,
+                                     source=ref)
+    torch.manual_seed(42)
+    x = torch.randn(2, 8, 32, requires_grad=True)
+    out = tp(x)
+    out.square().sum().backward()
+    xr = x.detach().clone().requires_grad_(True)
+    r = ref(xr)
+    r.square().sum().backward()
+    return ((out - r).abs().max().item(),
+            (x.grad - xr.grad).abs().max().item())
+
+
+def test_tensor_parallel_attention():
+    for causal in (False, True):
+        res = run_multiprocess(_tp_attn_worker, world=2, args=(causal,))
+        for fwd_err, dx_err in res:
+            assert fwd_err < 1e-5, (causal, fwd_err)
+            assert dx_err < 1e-4, (causal, dx_err)
