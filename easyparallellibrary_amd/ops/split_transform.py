@@ -64,6 +64,10 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
     module) lets a taskgraph root that is itself a Linear be rewired into
     its parent by dotted name."""
     from easyparallellibrary_amd.ops.moe import ExpertParallelMLP
+    from easyparallellibrary_amd.ops.tp_mlp import (
+        TensorParallelMLP, TensorParallelSelfAttention)
+    deferred = (ExpertParallelMLP, TensorParallelMLP,
+                TensorParallelSelfAttention)
     replaced = 0
     done = set()  # module ids already transformed (named_modules visits
     #               a nested module both as child and as parent)
@@ -80,12 +84,13 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
                 setattr(parent, parts[-1], new)
             replaced += 1
             continue
-        if isinstance(root, ExpertParallelMLP):
+        if isinstance(root, deferred):
             root.set_comm(comm)
             replaced += 1
             continue
-        for parent_name, parent in root.named_modules():
-            if isinstance(parent, ExpertParallelMLP):
+        # materialize: set_comm / child replacement mutates the tree
+        for parent_name, parent in list(root.named_modules()):
+            if isinstance(parent, deferred):
                 if id(parent) not in done:
                     done.add(id(parent))
                     parent.set_comm(comm)
@@ -94,7 +99,7 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
             for child_name, child in list(parent.named_children()):
                 if id(child) in done:
                     continue
-                if isinstance(child, ExpertParallelMLP):
+                if isinstance(child, deferred):
                     done.add(id(child))
                     child.set_comm(comm)
                     replaced += 1
