@@ -3,8 +3,9 @@ families (the reference's model zoo is external; these are the in-repo
 equivalents the benchmarks run).
 
 Hot ops ride the hand-written CDNA4 kernels: FusedLayerNorm,
-FusedBiasGelu (FFN first Linear runs bias-free); attention uses torch SDPA
-(hipBLASLt GEMMs).
+FusedBiasGelu (FFN first Linear runs bias-free), and the fused-qkv flash
+attention (default for non-causal; causal routes to torch SDPA until the
+staged 64-row variants are measured — see _NATIVE_ATTN below).
 """
 
 import os
