@@ -248,3 +248,29 @@ def test_bf16_training_on_cpu():
     losses = [float(engine.train_step(x, y)) for _ in range(3)]
     assert losses[-1] < losses[0]
     assert model[0].weight.dtype == torch.bfloat16
+
+
+def test_no_memory_growth_over_steps():
+    """200 steps leave RSS flat (hooks/pending-tensor bookkeeping must
+    not accumulate)."""
+    import gc
+    import psutil
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(0)
+    with epl.replicate(1):
+        model = nn.Sequential(nn.Linear(64, 256), nn.Tanh(),
+                              nn.Linear(256, 64))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-3)
+    x, y = torch.randn(32, 64), torch.randn(32, 64)
+    proc = psutil.Process()
+    for _ in range(20):
+        engine.train_step(x, y)
+    gc.collect()
+    rss0 = proc.memory_info().rss
+    for _ in range(200):
+        engine.train_step(x, y)
+    gc.collect()
+    growth = proc.memory_info().rss - rss0
+    assert growth < 64 * 1024 * 1024, "RSS grew {} bytes".format(growth)
