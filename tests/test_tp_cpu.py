@@ -422,3 +422,42 @@ def test_pp2_dense_tp2_hybrid():
     assert hyb[2] == hyb[3]
     for a, b in zip(base[1], hyb[2]):
         assert abs(a - b) < 1e-5, (base[1], hyb[2])
+
+
+def _tp_pipeline_builder_worker(rank, world, tp):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models.tp_transformer import (
+        build_tp_pipeline)
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "pipeline.num_micro_batch": 2,
+    }))
+    torch.manual_seed(71)
+    model = build_tp_pipeline(stages=2, tp=tp, layers=2, hidden=32,
+                              heads=4, ffn=64, vocab_size=128,
+                              max_pos=32)
+
+    def lm_loss(logits, targets):
+        return nn.functional.cross_entropy(logits.reshape(-1, 128),
+                                           targets)
+
+    engine = epl.Engine(model, loss_fn=lm_loss, optimizer="adamw",
+                        lr=1e-3)
+    torch.manual_seed(72)
+    ids = torch.randint(0, 128, (4, 16))
+    tgt = torch.randint(0, 128, (4 * 16,))
+    out = []
+    for _ in range(3):
+        loss = engine.train_step(ids, tgt)
+        out.append(None if loss is None else float(loss))
+    return out
+
+
+def test_build_tp_pipeline_matches_plain_pp():
+    base = run_multiprocess(_tp_pipeline_builder_worker, world=2,
+                            args=(1,))
+    hyb = run_multiprocess(_tp_pipeline_builder_worker, world=4,
+                           args=(2,), timeout=300)
+    assert hyb[2] == hyb[3]
+    for a, b in zip(base[1], hyb[2]):
+        assert abs(a - b) < 1e-5, (base[1], hyb[2])
