@@ -28,7 +28,7 @@ def parse_args():
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--config", default="bert_dp",
                    choices=["bert_dp", "bert_pp", "bert_zero", "resnet_tp",
-                            "gpt2_xl", "moe", "moe_pp"])
+                            "gpt2_xl", "moe", "moe_pp", "tp_pp"])
     p.add_argument("--model", default=None)
     p.add_argument("--batch", type=int, default=0,
                    help="per-GPU samples per step (0 = config default)")
@@ -215,6 +215,47 @@ def build_moe_pp_bench(args, epl, world, on_gpu, dtype):
     return engine, (ids, tgt), batch * streams, meta
 
 
+def build_tp_pp_bench(args, epl, world, on_gpu, dtype):
+    """PP2 x dense Megatron-TP hybrid (build_tp_pipeline).  tp=2 when
+    world divides by 4, else 1; world==1 falls back to bert_dp."""
+    if world < 2:
+        return build_bert_bench(args, epl, world, on_gpu, dtype)
+    from easyparallellibrary_amd.models import gpt2
+    from easyparallellibrary_amd.models.tp_transformer import (
+        build_tp_pipeline)
+    batch = args.batch or 8
+    seq = args.seq_len if args.seq_len != 512 else 1024
+    tp = 2 if world % 4 == 0 else 1
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "pipeline.num_micro_batch": min(4, batch),
+    }))
+    vocab = 32000
+    layers = args.layers or 24
+    hidden = args.hidden or 1024
+    model = build_tp_pipeline(stages=2, tp=tp, layers=layers,
+                              hidden=hidden, heads=16, ffn=4 * hidden,
+                              vocab_size=vocab, max_pos=seq)
+
+    def lm_loss(logits, targets):
+        import torch.nn.functional as F2
+        return F2.cross_entropy(logits.reshape(-1, vocab), targets)
+
+    engine = epl.Engine(model, loss_fn=lm_loss, optimizer="adamw",
+                        lr=1e-4, dtype=dtype)
+    # TP positions shard ONE data stream: seed by replica so every rank
+    # of a replica feeds identical tokens (unlike moe_pp, where
+    # positions are independent DP+EP streams)
+    ids, tgt = gpt2.synthetic_lm_batch(batch, seq, vocab,
+                                       device=engine.device,
+                                       seed=77 + engine.replica_id)
+    streams = world // (2 * tp)
+    meta = {"model": "tp-transformer-{}L".format(layers),
+            "global_batch": batch * max(1, streams), "seq_len": seq,
+            "parallelism": "pp2_tp{}_dp{}".format(tp, max(1, streams))}
+    return engine, (ids, tgt.reshape(-1)), batch * max(1, streams), meta
+
+
 BUILDERS = {
     "bert_dp": build_bert_bench,
     "bert_pp": build_bert_bench,
@@ -223,6 +264,7 @@ BUILDERS = {
     "resnet_tp": build_resnet_bench,
     "moe": build_moe_bench,
     "moe_pp": build_moe_pp_bench,
+    "tp_pp": build_tp_pp_bench,
 }
 
 
