@@ -41,6 +41,13 @@ class TaskGraph:
         return self.strategy_type == constant.SPLIT
 
     @property
+    def replicated_io(self):
+        """True when this stage's boundary activations are identical on
+        every position (dense-TP stages) — required at mixed-width
+        pipeline boundaries (parallel/pipeline.py)."""
+        return getattr(self.strategy, "replicated_io", False)
+
+    @property
     def device_count(self):
         return self.strategy.device_count
 

@@ -36,10 +36,10 @@ class PipelineRuntime:
         rank_lists = [tg.virtual_device.local_ranks(replica)
                       for tg in engine.stage_tgs]
         widths = {len(r) for r in rank_lists}
-        if len(widths) != 1:
-            raise NotImplementedError(
-                "pipeline stages must share one device count per replica; "
-                "got widths {}".format(sorted(widths)))
+        self.mixed = len(widths) != 1
+        if self.mixed:
+            self._init_mixed(engine, rank_lists)
+            return
         self.width = widths.pop()
         self.pos = rank_lists[self.s].index(engine.rank)
         self.stage_ranks = [r[self.pos] for r in rank_lists]
@@ -70,6 +70,108 @@ class PipelineRuntime:
         self._shape_cache = {}  # microbatch idx -> (shape, dtype)
         self._shapes_known = False
 
+    # ---- mixed stage widths ---------------------------------------------------
+    def _init_mixed(self, engine, rank_lists):
+        """Stages with different device counts (e.g. a width-1 embedding
+        stage feeding a dense-TP-2 stage).
+
+        Only legal when every unequal boundary is 1<->k and the wide side
+        is a ``replicate(..., replicated_io=True)`` stage: dense-TP blocks
+        (ops/tp_mlp.py) keep boundary activations IDENTICAL across the
+        stage's k positions (copy_to_group all-reduces input grads, the
+        row-parallel all-reduce replicates outputs), so the boundary
+        adapters are pure fan-out/fan-in of one tensor:
+
+        * 1->k forward: the narrow rank sends the activation to all k
+          positions (the Megatron 'copy' fan-out across the pipe).
+        * 1->k backward: every position already holds the FULL summed
+          input-grad after copy_to_group's backward all-reduce — the
+          narrow rank receives from position 0 only (receiving k copies
+          and summing would multiply the gradient by k).
+        * k->1 forward: outputs are replicated — position 0 sends.
+        * k->1 backward: the 'reduce' op's backward is identity per rank,
+          so every position needs the same grad_out — the narrow rank
+          sends it to all k positions.
+
+        Expert-parallel stages carry per-position DATA STREAMS, not
+        replicas, so they cannot change width; the replicated_io marker
+        is how a stage declares itself safe (never guessed from module
+        types).  Runs the PreferForward (GPipe) schedule: the 1F1B fused
+        bidirectional exchanges assume uniform position chains.
+        """
+        self.widths = [len(r) for r in rank_lists]
+        S = self.S
+        for s in range(S - 1):
+            a, b = self.widths[s], self.widths[s + 1]
+            if a != b:
+                if 1 not in (a, b):
+                    raise NotImplementedError(
+                        "pipeline boundary stage_{} ({} ranks) -> stage_{}"
+                        " ({} ranks): only 1<->k width changes are "
+                        "supported".format(s, a, s + 1, b))
+                wide = engine.stage_tgs[s if a > b else s + 1]
+                if not wide.replicated_io:
+                    raise ValueError(
+                        "mixed-width pipeline boundary stage_{}({}) -> "
+                        "stage_{}({}): the wide stage must be built with "
+                        "epl.replicate(..., replicated_io=True) — only "
+                        "stages whose boundary activations are identical "
+                        "on every position (dense-TP blocks, ops/tp_mlp) "
+                        "can change width; expert-parallel stages carry "
+                        "per-position data streams and cannot".format(
+                            s, a, s + 1, b))
+        self.pos = rank_lists[self.s].index(engine.rank)
+        w = self.widths
+        self._next_is_narrow = self.s < S - 1 and w[self.s + 1] < w[self.s]
+        self._prev_is_narrow = self.s > 0 and w[self.s - 1] < w[self.s]
+        self._prev_is_wide = self.s > 0 and w[self.s - 1] > w[self.s]
+        # per-boundary link communicators: (up_rank, down_rank) pairs.
+        # gloo group creation is collective over ALL ranks — every rank
+        # walks every (replica, boundary, link) in the same order.
+        self._links_prev = []  # (comm, peer_group_rank, peer_global_rank)
+        self._links_next = []
+        for rep in range(engine.num_replicas):
+            lists = [tg.virtual_device.local_ranks(rep)
+                     for tg in engine.stage_tgs]
+            for s in range(S - 1):
+                up, dn = lists[s], lists[s + 1]
+                if len(up) == len(dn):
+                    pairs = list(zip(up, dn))
+                elif len(up) == 1:
+                    pairs = [(up[0], d) for d in dn]
+                else:
+                    pairs = [(u, dn[0]) for u in up]
+                for li, (u, v) in enumerate(pairs):
+                    mine = engine.rank in (u, v)
+                    if not mine and (torch.cuda.is_available()
+                                     or not dist.is_initialized()):
+                        continue
+                    comm = create_communicator(
+                        "{}_pipeb_rep{}_s{}_l{}".format(
+                            engine._ns, rep, s, li), [u, v])
+                    if mine and rep == engine.replica_id:
+                        if engine.rank == u:
+                            self._links_next.append(
+                                (comm, comm.ranks.index(v), v))
+                        else:
+                            self._links_prev.append(
+                                (comm, comm.ranks.index(u), u))
+        self.comm = None
+        self.prev_rank = self._links_prev[0][2] if self._links_prev else None
+        self.next_rank = self._links_next[0][2] if self._links_next else None
+        self.schedule = constant.SCHEDULER_PREFER_FORWARD
+        if engine.config.pipeline.strategy != \
+                constant.SCHEDULER_PREFER_FORWARD and engine.rank == 0:
+            import logging
+            logging.getLogger("epl").warning(
+                "mixed-width pipeline stages run the PreferForward "
+                "(GPipe) schedule; %s requested",
+                engine.config.pipeline.strategy)
+        self.dtype = engine.dtype
+        self.device = engine.device
+        self._shape_cache = {}
+        self._shapes_known = False
+
     # ---- shape handshake (first step only) -----------------------------------
     def _send_shape(self, t, peer):
         if self._shapes_known:
@@ -95,22 +197,55 @@ class PipelineRuntime:
         return self.group_rank[global_rank]
 
     def recv_forward(self, i):
+        if self.mixed:
+            # narrow-after-wide receives from position 0 only; every other
+            # case has exactly one upstream link
+            comm, pg, peer = self._links_prev[0]
+            shape, dtype = self._recv_shape(i, peer)
+            t = torch.empty(shape, dtype=dtype, device=self.device)
+            comm.recv(t, pg)
+            return t
         shape, dtype = self._recv_shape(i, self.prev_rank)
         t = torch.empty(shape, dtype=dtype, device=self.device)
         self.comm.recv(t, self._g(self.prev_rank))
         return t
 
     def send_forward(self, t, i):
+        if self.mixed:
+            if self._next_is_narrow and self.pos != 0:
+                return  # replicated output: position 0 alone sends
+            t = t.contiguous()
+            for comm, pg, peer in self._links_next:  # 1->k fans out
+                self._send_shape(t, peer)
+                comm.send(t, pg)
+            return
         self._send_shape(t, self.next_rank)
         self.comm.send(t.contiguous(), self._g(self.next_rank))
 
     def recv_backward(self, i):
         shape, dtype = self._shape_cache_out[i]
         t = torch.empty(shape, dtype=dtype, device=self.device)
+        if self.mixed:
+            # narrow-before-wide takes the already-all-reduced grad from
+            # position 0; wide positions each get their copy from narrow
+            comm, pg, _ = self._links_next[0]
+            comm.recv(t, pg)
+            return t
         self.comm.recv(t, self._g(self.next_rank))
         return t
 
     def send_backward(self, g, i):
+        if self.mixed:
+            if self._prev_is_wide:  # k->1 backward: same grad to all k
+                g = g.contiguous()
+                for comm, pg, _ in self._links_prev:
+                    comm.send(g, pg)
+                return
+            if self._prev_is_narrow and self.pos != 0:
+                return  # grads already summed by copy_to_group's backward
+            comm, pg, _ = self._links_prev[0]
+            comm.send(g.contiguous(), pg)
+            return
         self.comm.send(g.contiguous(), self._g(self.prev_rank))
 
     def send_forward_recv_backward(self, t, i_send, i_recv):

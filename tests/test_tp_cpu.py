@@ -461,3 +461,43 @@ def test_build_tp_pipeline_matches_plain_pp():
     assert hyb[2] == hyb[3]
     for a, b in zip(base[1], hyb[2]):
         assert abs(a - b) < 1e-5, (base[1], hyb[2])
+
+
+def test_mixed_width_tp_pipeline_matches_plain_pp():
+    """Stage widths [1, 2]: a plain width-1 stage feeding a dense-TP-2
+    stage (pipeline runtime fans the activation out to both positions
+    and takes the already-all-reduced grad from position 0).  Losses
+    must match the all-width-1 pipeline exactly."""
+    base = run_multiprocess(_tp_pipeline_builder_worker, world=2,
+                            args=(1,))
+    mixed = run_multiprocess(_tp_pipeline_builder_worker, world=3,
+                             args=([1, 2],), timeout=300)
+    # last stage spans ranks 1 and 2 — identical replicated losses
+    assert mixed[1] == mixed[2]
+    for a, b in zip(base[1], mixed[1]):
+        assert abs(a - b) < 1e-5, (base[1], mixed[1])
+
+
+def test_mixed_width_tp_pipeline_wide_first():
+    """Stage widths [2, 1]: the k->1 direction — position 0 sends the
+    replicated stage output; the narrow rank fans the grad back to both
+    positions (the 'reduce' op's backward is identity per rank)."""
+    base = run_multiprocess(_tp_pipeline_builder_worker, world=2,
+                            args=(1,))
+    mixed = run_multiprocess(_tp_pipeline_builder_worker, world=3,
+                             args=([2, 1],), timeout=300)
+    for a, b in zip(base[1], mixed[2]):
+        assert abs(a - b) < 1e-5, (base[1], mixed[2])
+
+
+def test_mixed_width_requires_replicated_io():
+    """An unmarked wide stage next to a narrow one must be refused —
+    EP-style stages carry per-position data streams."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.ir.plan import TaskGraph
+    from easyparallellibrary_amd.strategies.replicate import Replicate
+    tg_n = TaskGraph(0, Replicate(1, name="s0"))
+    tg_w = TaskGraph(1, Replicate(2, name="s1"))
+    assert not tg_w.replicated_io
+    tg_ok = TaskGraph(2, Replicate(2, name="s2", replicated_io=True))
+    assert tg_ok.replicated_io
