@@ -196,6 +196,11 @@ class AutoConfig(BaseConfig):
     _SECTION = "auto"
     _DEFAULTS = {
         "auto_parallel": False,
+        # pair consecutive Linears inside nn.Sequential containers under
+        # a split scope into Megatron column->row blocks (one all-reduce
+        # instead of two all-gathers).  Safe only where execution order
+        # is known from structure, hence Sequential-only and opt-in.
+        "auto_pair_sequential": False,
     }
 
 

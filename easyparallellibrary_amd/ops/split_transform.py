@@ -58,6 +58,76 @@ def _replace_module(parent, name, new):
     setattr(parent, name, new)
 
 
+# Between a paired column and row linear the features are SHARDED, so only
+# feature-independent elementwise modules may sit in between (LayerNorm and
+# friends normalise over the full feature dim and would be wrong).
+_ELEMENTWISE_SAFE = (nn.ReLU, nn.GELU, nn.SiLU, nn.Tanh, nn.Sigmoid,
+                     nn.LeakyReLU, nn.ELU, nn.Hardswish, nn.Mish,
+                     nn.Dropout, nn.Identity)
+
+
+class PairedColumnLinear(nn.Module):
+    """Entry half of an auto-paired Megatron block: the 'f' copy operator
+    (identity forward, all-reduce of input grads backward) in front of a
+    column-parallel linear, so the pair's backward matches ops/tp_mlp."""
+
+    def __init__(self, col, comm):
+        super().__init__()
+        self.col = col
+        self.comm = comm
+
+    def forward(self, x):
+        return self.col(functional.copy_to_group(x, self.comm))
+
+
+def _pair_sequential(seq, comm):
+    """Greedy left-to-right pairing of Linear..Linear runs inside an
+    ``nn.Sequential`` whose in-between modules are all elementwise-safe:
+    first Linear -> column-parallel (sharded out, no gather), second ->
+    row-parallel over the pre-sharded features (ONE all-reduce instead of
+    two all-gathers).  Sequential-only because there the execution order
+    is known from structure — pairing arbitrary sibling Linears would be
+    guessing at the user's forward.  Returns ids of consumed modules."""
+    from easyparallellibrary_amd.ops.distributed_dense import (
+        RowParallelLinear)
+    children = list(seq.named_children())
+    consumed = set()
+    i = 0
+    while i < len(children):
+        name_i, mod_i = children[i]
+        if not isinstance(mod_i, nn.Linear) or id(mod_i) in consumed:
+            i += 1
+            continue
+        j = i + 1
+        mod_j = None
+        while j < len(children):
+            cand = children[j][1]
+            if isinstance(cand, nn.Linear):
+                mod_j = cand
+                break
+            if not isinstance(cand, _ELEMENTWISE_SAFE):
+                break
+            j += 1
+        if mod_j is None or mod_j.in_features != mod_i.out_features:
+            i += 1
+            continue
+        name_j = children[j][0]
+        col = ColumnParallelLinear(mod_i.in_features, mod_i.out_features,
+                                   comm, bias=mod_i.bias is not None,
+                                   gather_input=False, source=mod_i)
+        row = RowParallelLinear(mod_j.in_features, mod_j.out_features,
+                                comm, bias=mod_j.bias is not None,
+                                source=mod_j, pre_sharded=True)
+        setattr(seq, name_i, PairedColumnLinear(col, comm))
+        setattr(seq, name_j, row)
+        consumed.add(id(mod_i))
+        consumed.add(id(mod_j))
+        logger.info("auto-paired Sequential linears [%s] -> column / "
+                    "[%s] -> row (%d-way)", name_i, name_j, comm.size)
+        i = j + 1
+    return consumed
+
+
 def transform_taskgraph(tg, comm, gather_input=True, model=None):
     """Rewrite every Linear/Embedding under the taskgraph's module roots
     into its sharded equivalent over ``comm``.  ``model`` (the user's root
@@ -68,6 +138,9 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
         TensorParallelMLP, TensorParallelSelfAttention)
     deferred = (ExpertParallelMLP, TensorParallelMLP,
                 TensorParallelSelfAttention)
+    from easyparallellibrary_amd.env import Env
+    auto_pair = (Env.get().config.auto.auto_pair_sequential
+                 and comm.size > 1)
     replaced = 0
     done = set()  # module ids already transformed (named_modules visits
     #               a nested module both as child and as parent)
@@ -89,7 +162,16 @@ def transform_taskgraph(tg, comm, gather_input=True, model=None):
             replaced += 1
             continue
         # materialize: set_comm / child replacement mutates the tree
+        if auto_pair and isinstance(root, nn.Sequential):
+            paired = _pair_sequential(root, comm)
+            done |= paired
+            replaced += len(paired)
         for parent_name, parent in list(root.named_modules()):
+            if auto_pair and isinstance(parent, nn.Sequential) \
+                    and parent is not root:
+                paired = _pair_sequential(parent, comm)
+                done |= paired
+                replaced += len(paired)
             if isinstance(parent, deferred):
                 if id(parent) not in done:
                     done.add(id(parent))

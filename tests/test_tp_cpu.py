@@ -490,6 +490,56 @@ def test_mixed_width_tp_pipeline_wide_first():
         assert abs(a - b) < 1e-5, (base[1], mixed[2])
 
 
+def _auto_pair_worker(rank, world):
+    """auto.auto_pair_sequential: plain Linears inside an nn.Sequential
+    under a split scope become a Megatron column->row pair (one
+    all-reduce), numerically identical to serial training."""
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True,
+                         "auto.auto_pair_sequential": True}))
+    torch.manual_seed(61)
+    with epl.replicate(world, name="trunk"):
+        emb = nn.Linear(8, 16)
+    with epl.split(world, name="tp"):
+        mlp = nn.Sequential(nn.Linear(16, 32), nn.GELU(),
+                            nn.Linear(32, 16))
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb, self.mlp = emb, mlp
+            self.head = nn.Linear(16, 4)
+
+        def forward(self, x):
+            return self.head(self.mlp(self.emb(x)))
+
+    engine = epl.Engine(M(), loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    if world > 1:
+        from easyparallellibrary_amd.ops.split_transform import (
+            PairedColumnLinear)
+        from easyparallellibrary_amd.ops.distributed_dense import (
+            RowParallelLinear)
+        assert isinstance(mlp[0], PairedColumnLinear), type(mlp[0])
+        assert isinstance(mlp[2], RowParallelLinear), type(mlp[2])
+    torch.manual_seed(62)
+    x = torch.randn(2, 8, 8)
+    y = torch.randn(2, 8, 4)
+    return [float(engine.train_step(x, y)) for _ in range(3)]
+
+
+def test_auto_pair_sequential_matches_serial():
+    serial = _auto_pair_worker(0, 1)
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+    hooks.remove_hooks()
+    Env._instance = None
+    tp2 = run_multiprocess(_auto_pair_worker, world=2)
+    assert tp2[0] == tp2[1]
+    for a, b in zip(serial, tp2[0]):
+        assert abs(a - b) < 1e-5, (serial, tp2[0])
+
+
 def test_mixed_width_requires_replicated_io():
     """An unmarked wide stage next to a narrow one must be refused —
     EP-style stages carry per-position data streams."""
