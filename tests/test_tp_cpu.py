@@ -478,6 +478,21 @@ def test_mixed_width_tp_pipeline_matches_plain_pp():
         assert abs(a - b) < 1e-5, (base[1], mixed[1])
 
 
+def test_mixed_width_tp_pipeline_dp2():
+    """DP x mixed-width hybrid: 2 replicas of a [1, 2]-width pipeline on
+    6 ranks (exercises the multi-replica boundary-link creation in
+    _init_mixed).  Same seed => both replicas see identical data, so the
+    DP allreduce is an identity and losses must still match plain PP2."""
+    base = run_multiprocess(_tp_pipeline_builder_worker, world=2,
+                            args=(1,))
+    dp2 = run_multiprocess(_tp_pipeline_builder_worker, world=6,
+                           args=([1, 2],), timeout=300)
+    # last stage of replica 0 = ranks 1,2; of replica 1 = ranks 4,5
+    assert dp2[1] == dp2[2] == dp2[4] == dp2[5]
+    for a, b in zip(base[1], dp2[1]):
+        assert abs(a - b) < 1e-5, (base[1], dp2[1])
+
+
 def test_mixed_width_tp_pipeline_wide_first():
     """Stage widths [2, 1]: the k->1 direction — position 0 sends the
     replicated stage output; the narrow rank fans the grad back to both
