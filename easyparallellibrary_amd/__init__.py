@@ -54,6 +54,10 @@ class GraphKeys:
     LOCAL_MEAN_OBJECTS = "local_mean_objects"
     LOCAL_SUM_OBJECTS = "local_sum_objects"
     LOCAL_CONCAT_OBJECTS = "local_concat_objects"
+    # (name, value-or-callable) pairs registered ONCE; evaluated and
+    # replica-merged by Engine.write_summaries (reference: summary
+    # rewiring, parallel/parallel.py:355-413)
+    SUMMARIES = "epl_summaries"
 
 
 def add_to_collection(value, name):

@@ -547,6 +547,35 @@ class Engine:
                 out[name] = vals
         return out
 
+    def write_summaries(self, writer, step, scalars=None):
+        """Merge scalar summaries across every rank (mean) and write them
+        on rank 0 — the reference rewires ``tf.summary`` inputs to the
+        replica-merged tensors so TensorBoard shows whole-job values
+        (parallel/parallel.py:355-413, §5 observability); here the engine
+        merges at write time instead of editing a graph.
+
+        COLLECTIVE when world>1: every rank must call it each step.
+        ``scalars`` is a per-step dict name->value; additionally every
+        ``(name, value_or_callable)`` registered once via
+        ``epl.add_to_collection((name, fn), GraphKeys.SUMMARIES)`` is
+        evaluated (callables play the role of the reference's live graph
+        tensors).  ``writer`` is duck-typed — a
+        ``torch.utils.tensorboard.SummaryWriter`` or anything with an
+        ``add_scalar(name, value, step)`` method; it may be None on
+        ranks that do not write.  Returns the merged {name: float}."""
+        import easyparallellibrary_amd as epl
+        items = list((scalars or {}).items())
+        for entry in self.env.get_collection(epl.GraphKeys.SUMMARIES):
+            name, v = entry
+            items.append((name, v() if callable(v) else v))
+        merged = {}
+        for name, v in items:
+            merged[name] = float(self.all_reduce_metric(v, op="mean"))
+        if self.rank == 0 and writer is not None:
+            for name, v in merged.items():
+                writer.add_scalar(name, v, step)
+        return merged
+
     def slice_input_files(self, files):
         """Per-replica IO slicing (config io.slicing; reference
         graph_editor.py:149-215)."""

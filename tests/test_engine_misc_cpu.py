@@ -274,3 +274,41 @@ def test_no_memory_growth_over_steps():
     gc.collect()
     growth = proc.memory_info().rss - rss0
     assert growth < 64 * 1024 * 1024, "RSS grew {} bytes".format(growth)
+
+
+def _summaries_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(31)
+    with epl.replicate(1):
+        m = nn.Linear(4, 2)
+    engine = epl.Engine(m, loss_fn=nn.MSELoss())
+    # once-registered callable (plays the reference's live summary tensor)
+    epl.add_to_collection(("lr", lambda: engine.lr),
+                          epl.GraphKeys.SUMMARIES)
+
+    class FakeWriter:
+        def __init__(self):
+            self.rows = []
+
+        def add_scalar(self, name, value, step):
+            self.rows.append((name, value, step))
+
+    w = FakeWriter()
+    # per-rank value 10*rank: mean across 2 ranks = 5.0
+    merged = engine.write_summaries(w, step=3,
+                                    scalars={"loss": 10.0 * rank})
+    return merged, w.rows
+
+
+def test_write_summaries_merges_across_ranks():
+    out = run_multiprocess(_summaries_worker, world=2)
+    for rank in (0, 1):
+        merged, rows = out[rank]
+        assert abs(merged["loss"] - 5.0) < 1e-6, merged
+        assert abs(merged["lr"] - 1e-3) < 1e-8
+    # only rank 0 writes
+    names = [(n, s) for n, _, s in out[0][1]]
+    assert names == [("loss", 3), ("lr", 3)]
+    assert abs(out[0][1][0][1] - 5.0) < 1e-6
+    assert out[1][1] == []
