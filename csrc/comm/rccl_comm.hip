@@ -326,6 +326,33 @@ void all_to_all_v(const std::string& name, at::Tensor out, at::Tensor in,
   end_op(c, async, {out, in});
 }
 
+// Varying-size all-gather: rank r's `in` lands in every rank's `outs[r]`
+// (reference: AllGatherv = grouped per-rank ncclBroadcast loop,
+// tensorflow_nccl.h:169-183 — same shape here; sizes may differ per rank,
+// so the caller exchanges counts first, e.g. over the gloo control group).
+void all_gather_v(const std::string& name, std::vector<at::Tensor> outs,
+                  at::Tensor in, bool async) {
+  check_gpu_contig(in);
+  Comm& c = get_comm(name);
+  TORCH_CHECK((int)outs.size() == c.size,
+              "all_gather_v needs one output tensor per rank");
+  TORCH_CHECK(outs[c.rank].numel() == in.numel(),
+              "all_gather_v: outs[rank] must match input size");
+  hipStream_t s = begin_op(c, async);
+  RCCL_CHECK(ncclGroupStart());
+  for (int r = 0; r < c.size; ++r) {
+    check_gpu_contig(outs[r]);
+    if (outs[r].numel() == 0) continue;
+    const void* src = (r == c.rank) ? in.data_ptr() : outs[r].data_ptr();
+    RCCL_CHECK(ncclBroadcast(src, outs[r].data_ptr(), outs[r].numel(),
+                             to_rccl_dtype(outs[r]), r, c.comm, s));
+  }
+  RCCL_CHECK(ncclGroupEnd());
+  std::vector<at::Tensor> keep(outs.begin(), outs.end());
+  keep.push_back(in);
+  end_op(c, async, keep);
+}
+
 void send(const std::string& name, at::Tensor t, int64_t peer, bool async) {
   check_gpu_contig(t);
   Comm& c = get_comm(name);
@@ -386,6 +413,7 @@ void register_comm(py::module& m) {
   m.def("reduce_scatter", &reduce_scatter);
   m.def("all_to_all_single", &all_to_all_single);
   m.def("all_to_all_v", &all_to_all_v);
+  m.def("all_gather_v", &all_gather_v);
   m.def("send", &send);
   m.def("recv", &recv);
   m.def("batch_p2p", &batch_p2p);

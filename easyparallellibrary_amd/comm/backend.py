@@ -58,6 +58,12 @@ class Communicator:
     def all_to_all_v(self, out, inp, out_counts, in_counts, async_op=False):
         raise NotImplementedError
 
+    def all_gather_v(self, outs, inp, async_op=False):
+        """Varying-size all-gather: rank r's ``inp`` lands in ``outs[r]``
+        on every rank (reference AllGatherv, tensorflow_nccl.h:169-183:
+        a grouped per-rank broadcast — sizes may differ per rank)."""
+        raise NotImplementedError
+
     def send(self, t, peer):
         raise NotImplementedError
 
@@ -107,6 +113,10 @@ class LocalCommunicator(Communicator):
     def all_to_all_v(self, out, inp, out_counts, in_counts, async_op=False):
         out.copy_(inp.reshape(out.shape))
         return out
+
+    def all_gather_v(self, outs, inp, async_op=False):
+        outs[0].copy_(inp.reshape(outs[0].shape))
+        return outs
 
 
 class RcclCommunicator(Communicator):
@@ -169,6 +179,10 @@ class RcclCommunicator(Communicator):
         _ext().all_to_all_v(self.name, out, inp, list(out_counts),
                             list(in_counts), async_op)
         return out
+
+    def all_gather_v(self, outs, inp, async_op=False):
+        _ext().all_gather_v(self.name, list(outs), inp, async_op)
+        return outs
 
     def send(self, t, peer):
         _ext().send(self.name, t, self._peer(peer), False)
@@ -286,6 +300,13 @@ class GlooCommunicator(Communicator):
         for rq in reqs:
             rq.wait()
         return out
+
+    def all_gather_v(self, outs, inp, async_op=False):
+        for r in range(self.size):
+            if r == self.rank:
+                outs[r].reshape(-1).copy_(inp.reshape(-1))
+            dist.broadcast(outs[r], self.ranks[r], group=self.group)
+        return outs
 
     def send(self, t, peer):
         dist.send(t, self.ranks[peer], group=self.group)

@@ -5,6 +5,8 @@ Capability parity: /root/reference/epl/communicators/collective_communicator.py
 :151-157, reduce :159-177, bucket-size estimation :183-204).
 """
 
+import torch
+
 from easyparallellibrary_amd.comm import functional
 from easyparallellibrary_amd.comm.backend import create_communicator
 from easyparallellibrary_amd.comm.pool import CommunicationPool
@@ -67,6 +69,13 @@ class CollectiveCommunicator:
 
     def all_to_all(self, t):
         return functional.all_to_all(t, self.inline)
+
+    def all_gather_v(self, t, sizes):
+        """Varying-size all-gather (reference AllGatherv): ``sizes[r]`` is
+        rank r's dim-0 length; returns the list of per-rank tensors."""
+        outs = [torch.empty((int(n),) + tuple(t.shape[1:]), dtype=t.dtype,
+                            device=t.device) for n in sizes]
+        return self.inline.all_gather_v(outs, t.contiguous())
 
     def broadcast(self, t, root=0):
         return functional.broadcast(t, self.inline, root)

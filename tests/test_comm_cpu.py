@@ -48,6 +48,12 @@ def _collectives_worker(rank, world):
     o = torch.empty(world)
     comm.all_to_all_v(o, inp, [1] * world, [1] * world)
     out["a2av"] = o.clone()
+
+    # all_gather_v: rank r contributes r+1 elements of value r
+    mine = torch.full((rank + 1,), float(rank))
+    outs = [torch.empty(r + 1) for r in range(world)]
+    comm.all_gather_v(outs, mine)
+    out["agv"] = torch.cat(outs)
     return out
 
 
@@ -65,6 +71,9 @@ def test_collectives_numeric():
     assert torch.equal(r[1]["a2a"], torch.tensor([2., 3., 102., 103.]))
     assert torch.equal(r[0]["a2av"], torch.tensor([0., 1.]))
     assert torch.equal(r[1]["a2av"], torch.tensor([0., 1.]))
+    # all_gather_v: [rank0's 1 elem, rank1's 2 elems] on every rank
+    assert torch.equal(r[0]["agv"], torch.tensor([0., 1., 1.]))
+    assert torch.equal(r[1]["agv"], torch.tensor([0., 1., 1.]))
 
 
 def _autograd_worker(rank, world):
