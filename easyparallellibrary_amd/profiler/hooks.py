@@ -79,6 +79,52 @@ class MemoryProfiler:
             w.writeheader()
             w.writerows(self.records)
 
+    def dump_png(self, path):
+        """Memory timeline plot (reference: MemoryProfilerHook's
+        matplotlib PNG with per-phase coloring, memory_profiler_hook.py
+        :207-271 — notes recorded per step play the phase role here).
+        No-op when matplotlib is unavailable."""
+        if not self.records:
+            return False
+        try:
+            import matplotlib
+            matplotlib.use("Agg")
+            import matplotlib.pyplot as plt
+        except ImportError:
+            return False
+        steps = [r["step"] for r in self.records]
+        gib = 1 << 30
+        fig, ax = plt.subplots(figsize=(8, 4))
+        ax.plot(steps, [r["allocated_bytes"] / gib for r in self.records],
+                label="allocated")
+        ax.plot(steps, [r["peak_bytes"] / gib for r in self.records],
+                label="peak", linestyle="--")
+        ax.plot(steps, [r["reserved_bytes"] / gib for r in self.records],
+                label="reserved", linestyle=":")
+        # color the background by note runs (the phase coloring analogue)
+        notes = [r["note"] for r in self.records]
+        colors = {}
+        palette = ["#fde2cf", "#d4e7fa", "#d9f2d9", "#f5d9f0", "#f7f6cf"]
+        start = 0
+        for i in range(1, len(notes) + 1):
+            if i == len(notes) or notes[i] != notes[start]:
+                note = notes[start]
+                if note:
+                    c = colors.setdefault(
+                        note, palette[len(colors) % len(palette)])
+                    ax.axvspan(steps[start], steps[i - 1], color=c,
+                               alpha=0.5,
+                               label=note if note not in ax.get_legend_handles_labels()[1] else None)
+                start = i
+        ax.set_xlabel("step")
+        ax.set_ylabel("GiB")
+        ax.set_title("device memory timeline")
+        ax.legend(loc="upper left", fontsize=8)
+        fig.tight_layout()
+        fig.savefig(path)
+        plt.close(fig)
+        return True
+
     @property
     def peak_gb(self):
         if not self.records:

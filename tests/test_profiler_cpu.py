@@ -47,6 +47,19 @@ def test_memory_profiler_csv(tmp_path):
     assert prof.peak_gb >= 0.0
 
 
+def test_memory_profiler_png(tmp_path):
+    prof = MemoryProfiler(every_n_steps=1)
+    engine = _engine()
+    x, y = torch.randn(4, 8), torch.randn(4, 4)
+    for s in range(4):
+        engine.train_step(x, y)
+        prof.after_step(step=s, note="fwd" if s < 2 else "bwd")
+    png = tmp_path / "mem.png"
+    ok = prof.dump_png(str(png))
+    if ok:  # matplotlib present in this image
+        assert png.exists() and png.stat().st_size > 0
+
+
 def test_flops_profiler_bert_tiny():
     from easyparallellibrary_amd.models import bert
     epl.init()
