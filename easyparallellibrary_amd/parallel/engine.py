@@ -547,6 +547,17 @@ class Engine:
                 out[name] = vals
         return out
 
+    def broadcast_signal(self, value=0.0, root=0):
+        """Broadcast a scalar from ``root`` to every rank — the
+        reference's eval-barrier sync signal (hooks.py:915-933: workers
+        block on a broadcast until the chief finishes evaluation).
+        COLLECTIVE when world>1: acts as a barrier; returns the root's
+        value as a float on every rank."""
+        t = torch.tensor(float(value))
+        if dist.is_initialized() and self.world_size > 1:
+            dist.broadcast(t, root, group=self._control_group)
+        return float(t)
+
     def write_summaries(self, writer, step, scalars=None):
         """Merge scalar summaries across every rank (mean) and write them
         on rank 0 — the reference rewires ``tf.summary`` inputs to the

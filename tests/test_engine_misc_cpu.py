@@ -312,3 +312,21 @@ def test_write_summaries_merges_across_ranks():
     assert names == [("loss", 3), ("lr", 3)]
     assert abs(out[0][1][0][1] - 5.0) < 1e-6
     assert out[1][1] == []
+
+
+def _signal_worker(rank, world):
+    import time
+    import easyparallellibrary_amd as epl
+    epl.init()
+    with epl.replicate(1):
+        m = nn.Linear(4, 2)
+    engine = epl.Engine(m, loss_fn=nn.MSELoss())
+    if rank == 0:
+        time.sleep(0.3)  # "chief evaluates"; others block on the signal
+        return engine.broadcast_signal(7.5)
+    return engine.broadcast_signal(-1.0)  # non-root value is ignored
+
+
+def test_broadcast_signal_eval_barrier():
+    out = run_multiprocess(_signal_worker, world=2)
+    assert out == [7.5, 7.5]
