@@ -65,7 +65,9 @@ def test_env_override_new_knobs(monkeypatch):
     monkeypatch.setenv("EPL_OPTIMIZER_MAX_GRAD_NORM", "2.5")
     monkeypatch.setenv("EPL_IO_DROP_LAST_FILES", "true")
     monkeypatch.setenv("EPL_GRADIENT_CHECKPOINT_END_TASKGRAPH", "3")
+    monkeypatch.setenv("EPL_AUTO_AUTO_PAIR_SEQUENTIAL", "1")
     c = Config()
     assert c.optimizer.max_grad_norm == 2.5
     assert c.io.drop_last_files is True
     assert c.gradient_checkpoint.end_taskgraph == 3
+    assert c.auto.auto_pair_sequential is True
