@@ -159,14 +159,7 @@ class PipelineRuntime:
         self.comm = None
         self.prev_rank = self._links_prev[0][2] if self._links_prev else None
         self.next_rank = self._links_next[0][2] if self._links_next else None
-        self.schedule = constant.SCHEDULER_PREFER_FORWARD
-        if engine.config.pipeline.strategy != \
-                constant.SCHEDULER_PREFER_FORWARD and engine.rank == 0:
-            import logging
-            logging.getLogger("epl").warning(
-                "mixed-width pipeline stages run the PreferForward "
-                "(GPipe) schedule; %s requested",
-                engine.config.pipeline.strategy)
+        self.schedule = engine.config.pipeline.strategy
         self.dtype = engine.dtype
         self.device = engine.device
         self._shape_cache = {}
@@ -249,9 +242,27 @@ class PipelineRuntime:
         self.comm.send(g.contiguous(), self._g(self.prev_rank))
 
     def send_forward_recv_backward(self, t, i_send, i_recv):
-        self._send_shape(t, self.next_rank)
         shape, dtype = self._shape_cache_out[i_recv]
         g = torch.empty(shape, dtype=dtype, device=self.device)
+        if self.mixed:
+            # every pairwise link exchange is either a matching batched
+            # send+recv (both ends batch on the same 2-rank comm) or a
+            # matching plain send/recv — per-link pairing keeps the
+            # bidirectional steady state deadlock-free at 1<->k
+            # boundaries too
+            if self._next_is_narrow and self.pos != 0:
+                comm, pg, _ = self._links_next[0]
+                comm.recv(g, pg)  # pure grad recv: position 0 sent fwd
+                return g
+            t = t.contiguous()
+            comm0, pg0, peer0 = self._links_next[0]
+            self._send_shape(t, peer0)
+            comm0.batch_p2p([(True, t, pg0), (False, g, pg0)])
+            for comm, pg, peer in self._links_next[1:]:  # 1->k fan-out
+                self._send_shape(t, peer)
+                comm.send(t, pg)
+            return g
+        self._send_shape(t, self.next_rank)
         self.comm.batch_p2p([
             (True, t.contiguous(), self._g(self.next_rank)),
             (False, g, self._g(self.next_rank)),
@@ -259,6 +270,20 @@ class PipelineRuntime:
         return g
 
     def send_backward_recv_forward(self, g, i_recv):
+        if self.mixed:
+            if self._prev_is_narrow and self.pos != 0:
+                # grads were summed by copy_to_group's backward; position
+                # 0 alone sends — this position only receives forward
+                return self.recv_forward(i_recv)
+            g = g.contiguous()
+            comm0, pg0, peer0 = self._links_prev[0]
+            shape, dtype = self._recv_shape(i_recv, peer0)
+            t = torch.empty(shape, dtype=dtype, device=self.device)
+            comm0.batch_p2p([(True, g, pg0), (False, t, pg0)])
+            if self._prev_is_wide:  # k->1: same grad to the other k-1
+                for comm, pg, _ in self._links_prev[1:]:
+                    comm.send(g, pg)
+            return t
         shape, dtype = self._recv_shape(i_recv, self.prev_rank)
         t = torch.empty(shape, dtype=dtype, device=self.device)
         self.comm.batch_p2p([

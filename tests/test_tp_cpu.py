@@ -424,14 +424,17 @@ def test_pp2_dense_tp2_hybrid():
         assert abs(a - b) < 1e-5, (base[1], hyb[2])
 
 
-def _tp_pipeline_builder_worker(rank, world, tp):
+def _tp_pipeline_builder_worker(rank, world, tp, strategy=None):
     import easyparallellibrary_amd as epl
     from easyparallellibrary_amd.models.tp_transformer import (
         build_tp_pipeline)
-    epl.init(epl.Config({
+    cfg = {
         "cluster.colocate_split_and_replicate": True,
         "pipeline.num_micro_batch": 2,
-    }))
+    }
+    if strategy:
+        cfg["pipeline.strategy"] = strategy
+    epl.init(epl.Config(cfg))
     torch.manual_seed(71)
     model = build_tp_pipeline(stages=2, tp=tp, layers=2, hidden=32,
                               heads=4, ffn=64, vocab_size=128,
@@ -553,6 +556,19 @@ def test_auto_pair_sequential_matches_serial():
     assert tp2[0] == tp2[1]
     for a, b in zip(serial, tp2[0]):
         assert abs(a - b) < 1e-5, (serial, tp2[0])
+
+
+def test_mixed_width_gpipe_schedule():
+    """Mixed widths under the PreferForward (GPipe) schedule (the
+    default-schedule tests above exercise the per-link 1F1B path)."""
+    base = run_multiprocess(_tp_pipeline_builder_worker, world=2,
+                            args=(1,))
+    mixed = run_multiprocess(_tp_pipeline_builder_worker, world=3,
+                             args=([1, 2], "prefer_forward"),
+                             timeout=300)
+    assert mixed[1] == mixed[2]
+    for a, b in zip(base[1], mixed[1]):
+        assert abs(a - b) < 1e-5, (base[1], mixed[1])
 
 
 def test_mixed_width_requires_replicated_io():
