@@ -430,7 +430,9 @@ def _tp_pipeline_builder_worker(rank, world, tp, strategy=None):
         build_tp_pipeline)
     cfg = {
         "cluster.colocate_split_and_replicate": True,
-        "pipeline.num_micro_batch": 2,
+        # 4 micro-batches so 1F1B's steady-state loop (fused per-link
+        # exchanges) runs repeatedly, not just warmup/cooldown
+        "pipeline.num_micro_batch": 4,
     }
     if strategy:
         cfg["pipeline.strategy"] = strategy
