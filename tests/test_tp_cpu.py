@@ -560,6 +560,45 @@ def test_auto_pair_sequential_matches_serial():
         assert abs(a - b) < 1e-5, (serial, tp2[0])
 
 
+def _tp3_worker(rank, world, tp):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models.tp_transformer import (
+        build_tp_pipeline)
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "pipeline.num_micro_batch": 4,
+    }))
+    torch.manual_seed(73)
+    model = build_tp_pipeline(stages=3, tp=tp, layers=3, hidden=32,
+                              heads=4, ffn=64, vocab_size=128,
+                              max_pos=32)
+
+    def lm_loss(logits, targets):
+        return nn.functional.cross_entropy(logits.reshape(-1, 128),
+                                           targets)
+
+    engine = epl.Engine(model, loss_fn=lm_loss, optimizer="adamw",
+                        lr=1e-3)
+    torch.manual_seed(74)
+    ids = torch.randint(0, 128, (4, 16))
+    tgt = torch.randint(0, 128, (4 * 16,))
+    return [float(l) if (l := engine.train_step(ids, tgt)) is not None
+            else None for _ in range(3)]
+
+
+def test_mixed_width_equal_wide_boundary():
+    """Widths [1, 2, 2]: covers the wide<->wide equal boundary inside a
+    mixed pipeline (replicated tensors on both per-position links) on
+    top of the 1->k entry boundary."""
+    base = run_multiprocess(_tp3_worker, world=3, args=([1, 1, 1],),
+                            timeout=300)
+    mixed = run_multiprocess(_tp3_worker, world=5, args=([1, 2, 2],),
+                             timeout=300)
+    assert mixed[3] == mixed[4]
+    for a, b in zip(base[2], mixed[3]):
+        assert abs(a - b) < 1e-5, (base[2], mixed[3])
+
+
 def test_mixed_width_gpipe_schedule():
     """Mixed widths under the PreferForward (GPipe) schedule (the
     default-schedule tests above exercise the per-link 1F1B path)."""
