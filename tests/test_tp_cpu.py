@@ -612,14 +612,41 @@ def test_mixed_width_gpipe_schedule():
         assert abs(a - b) < 1e-5, (base[1], mixed[1])
 
 
+def _unmarked_mixed_worker(rank, world):
+    """Mixed widths where the wide stage is NOT replicated_io: the
+    pipeline runtime must refuse (EP-style stages carry per-position
+    data streams and cannot change width)."""
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"pipeline.num_micro_batch": 2}))
+    torch.manual_seed(81)
+    with epl.replicate(1, name="stage_0"):
+        s0 = nn.Linear(8, 16)
+    with epl.replicate(2, name="stage_1"):  # wide but unmarked
+        s1 = nn.Linear(16, 4)
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.s0, self.s1 = s0, s1
+
+        def forward(self, x):
+            return self.s1(self.s0(x))
+
+    try:
+        epl.Engine(M(), loss_fn=nn.MSELoss())
+    except ValueError as e:
+        return "replicated_io" in str(e)
+    return False
+
+
 def test_mixed_width_requires_replicated_io():
     """An unmarked wide stage next to a narrow one must be refused —
-    EP-style stages carry per-position data streams."""
-    import easyparallellibrary_amd as epl
+    both at the TaskGraph level and when the engine builds the runtime."""
     from easyparallellibrary_amd.ir.plan import TaskGraph
     from easyparallellibrary_amd.strategies.replicate import Replicate
-    tg_n = TaskGraph(0, Replicate(1, name="s0"))
     tg_w = TaskGraph(1, Replicate(2, name="s1"))
     assert not tg_w.replicated_io
     tg_ok = TaskGraph(2, Replicate(2, name="s2", replicated_io=True))
     assert tg_ok.replicated_io
+    out = run_multiprocess(_unmarked_mixed_worker, world=3, timeout=300)
+    assert all(out), out
