@@ -342,6 +342,9 @@ void all_gather_v(const std::string& name, std::vector<at::Tensor> outs,
   RCCL_CHECK(ncclGroupStart());
   for (int r = 0; r < c.size; ++r) {
     check_gpu_contig(outs[r]);
+    TORCH_CHECK(outs[r].scalar_type() == in.scalar_type(),
+                "all_gather_v: outs[", r, "] dtype ", outs[r].scalar_type(),
+                " != input dtype ", in.scalar_type());
     if (outs[r].numel() == 0) continue;
     const void* src = (r == c.rank) ? in.data_ptr() : outs[r].data_ptr();
     RCCL_CHECK(ncclBroadcast(src, outs[r].data_ptr(), outs[r].numel(),

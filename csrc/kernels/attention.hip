@@ -74,12 +74,81 @@ DEVI bf16x8 assemble_pfrag(const float* p) {
 }
 
 // ============================================================================
-// forward (v2.1): KVBLK = 64; K (row copies) and V^T (transposed, 144-
+// forward (v3): KVBLK = 64; K (row copies) and V^T (transposed, 144-
 // byte row pad -> conflict-free ds_read_b128 fragments) staged in LDS
 // once per block per tile.  Measured: explicit double-buffering and
 // register prefetch both REGRESS (hipcc schedules the simple form best
 // at 3 waves/SIMD) - keep this structure.
+//
+// v3: the kv loop is SPLIT into a branch-free bulk phase (tiles that are
+// provably full: no seq-bound or causal masking, no per-wave break — the
+// compiler pipelines it like the non-causal fast path) and a masked
+// phase covering the causal diagonal (<= 2 tiles) and the seq tail.
+// Round-1 measurement showed the single masked loop ran EVERY causal
+// tile 33-46% slower per tile than the non-causal loop ran the same
+// machine code — the masking ALU + dynamic break cost lands on all
+// tiles, not just diagonal ones.
 // ============================================================================
+
+// one 32-key sub-tile of the fwd online-softmax loop; MASKED adds the
+// seq-bound + causal-diagonal masking (bulk tiles skip all of it)
+template <bool MASKED>
+DEVI void fwd_tile(const short (&ldsK)[64][72], const short (&ldsV)[64][72],
+                   int sub, int64_t kvs, int64_t seq, int64_t myq,
+                   float scale, int causal, const bf16x8 (&qfrag)[4],
+                   f32x16& ot0, f32x16& ot1, float& m, float& l, int lq,
+                   int hi) {
+  f32x16 st = {};
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    bf16x8 kfrag =
+        *reinterpret_cast<const bf16x8*>(&ldsK[sub + lq][hi * 8 + 16 * c]);
+    st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c], st, 0, 0,
+                                                 0);
+  }
+  float s[16];
+  float tile_max = -1e30f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    float sv = st[r] * scale;
+    if (MASKED) {
+      const int64_t kvg = kvs + crow(r, hi);
+      if (kvg >= seq || (causal && kvg > myq)) sv = -1e30f;
+    }
+    s[r] = sv;
+    tile_max = fmaxf(tile_max, sv);
+  }
+  tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
+  const float m_new = fmaxf(m, tile_max);
+  const float alpha = __expf(m - m_new);
+  float rowsum = 0.f;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    s[r] = __expf(s[r] - m_new);
+    rowsum += s[r];
+  }
+  rowsum += __shfl_xor(rowsum, 32, 64);
+  l = l * alpha + rowsum;
+  m = m_new;
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    ot0[r] *= alpha;
+    ot1[r] *= alpha;
+  }
+  bf16x8 pf0 = assemble_pfrag(&s[0]);
+  bf16x8 pf1 = assemble_pfrag(&s[8]);
+#pragma unroll
+  for (int kc = 0; kc < 2; ++kc) {
+    bf16x8 pf = kc == 0 ? pf0 : pf1;
+    bf16x8 vt0 =
+        *reinterpret_cast<const bf16x8*>(&ldsV[lq][sub + kc * 16 + hi * 8]);
+    bf16x8 vt1 = *reinterpret_cast<const bf16x8*>(
+        &ldsV[32 + lq][sub + kc * 16 + hi * 8]);
+    ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0, 0, 0);
+    ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0, 0, 0);
+  }
+}
+
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
@@ -117,30 +186,42 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   f32x16 ot0 = {}, ot1 = {};
   float m = -1e30f, l = 0.f;
 
+  // bulk tiles are provably full for EVERY lane of the block: below the
+  // causal diagonal (kv < q0_blk <= myq) and inside the seq bound
+  const int64_t bulk_end = causal ? q0_blk : (seq & ~(int64_t)63);
   const int64_t blk_kv_end =
       causal ? (q0_blk + 128 < seq ? q0_blk + 128 : seq) : seq;
   const int stage_kv = threadIdx.x & 63;
   const int stage_d0 = (threadIdx.x >> 6) * 8;
 
-  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 64) {
-    // ---- stage K (rows) and V^T (transposed) cooperatively ---------------
+  auto stage_tile = [&](int64_t kv0) {
     __syncthreads();
-    {
-      int64_t vrow = kv0 + stage_kv;
-      if (vrow >= seq) vrow = seq - 1;   // masked columns never contribute
+    int64_t vrow = kv0 + stage_kv;
+    if (vrow >= seq) vrow = seq - 1;   // masked columns never contribute
 #pragma unroll
-      for (int h2 = 0; h2 < 2; ++h2) {
-        const int sd = stage_d0 + h2 * 32;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-            vp + vrow * in_ss + sd);
+    for (int h2 = 0; h2 < 2; ++h2) {
+      const int sd = stage_d0 + h2 * 32;
+      bf16x8 vv = *reinterpret_cast<const bf16x8*>(
+          vp + vrow * in_ss + sd);
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          ldsV[sd + j][stage_kv] = vv[j];
-        *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
-            *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
-      }
+      for (int j = 0; j < 8; ++j)
+        ldsV[sd + j][stage_kv] = vv[j];
+      *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
+          *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
     }
     __syncthreads();
+  };
+
+  int64_t kv0 = 0;
+  for (; kv0 < bulk_end; kv0 += 64) {
+    stage_tile(kv0);
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32)
+      fwd_tile<false>(ldsK, ldsV, sub, kv0 + sub, seq, myq, scale, causal,
+                      qfrag, ot0, ot1, m, l, lq, hi);
+  }
+  for (; kv0 < blk_kv_end; kv0 += 64) {
+    stage_tile(kv0);
     if (active) {
       const int64_t wave_kv_end = causal
           ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
@@ -148,55 +229,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int sub = 0; sub < 64; sub += 32) {
         const int64_t kvs = kv0 + sub;
         if (kvs >= wave_kv_end) break;
-        f32x16 st = {};
-#pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          bf16x8 kfrag = *reinterpret_cast<const bf16x8*>(
-              &ldsK[sub + lq][hi * 8 + 16 * c]);
-          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c],
-                                                       st, 0, 0, 0);
-        }
-        float s[16];
-        float tile_max = -1e30f;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          float sv = st[r] * scale;
-          const int64_t kvg = kvs + crow(r, hi);
-          if (kvg >= seq || (causal && kvg > myq)) sv = -1e30f;
-          s[r] = sv;
-          tile_max = fmaxf(tile_max, sv);
-        }
-        tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
-        const float m_new = fmaxf(m, tile_max);
-        const float alpha = __expf(m - m_new);
-        float rowsum = 0.f;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          s[r] = __expf(s[r] - m_new);
-          rowsum += s[r];
-        }
-        rowsum += __shfl_xor(rowsum, 32, 64);
-        l = l * alpha + rowsum;
-        m = m_new;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          ot0[r] *= alpha;
-          ot1[r] *= alpha;
-        }
-        bf16x8 pf0 = assemble_pfrag(&s[0]);
-        bf16x8 pf1 = assemble_pfrag(&s[8]);
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
-          bf16x8 pf = kc == 0 ? pf0 : pf1;
-          bf16x8 vt0 = *reinterpret_cast<const bf16x8*>(
-              &ldsV[lq][sub + kc * 16 + hi * 8]);
-          bf16x8 vt1 = *reinterpret_cast<const bf16x8*>(
-              &ldsV[32 + lq][sub + kc * 16 + hi * 8]);
-          ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0,
-                                                        0, 0);
-          ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0,
-                                                        0, 0);
-        }
+        fwd_tile<true>(ldsK, ldsV, sub, kvs, seq, myq, scale, causal,
+                       qfrag, ot0, ot1, m, l, lq, hi);
       }
     }
   }
@@ -512,8 +546,103 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 // ============================================================================
 // split backward A kernels: dV-only and dK-only.  Same structure as the
 // combined kernel but half the accumulators each -> 3 waves/SIMD instead
-// of 2 (A/B-selectable via epl_attn_bwd mode).
+// of 2 (A/B-selectable via epl_attn_bwd mode).  The q loop runs in three
+// phases: masked causal-diagonal tiles, branch-free bulk tiles, masked
+// seq tail (same rationale as the forward v3 split).
 // ============================================================================
+
+// one 32-row q tile of the dV accumulation
+template <bool MASKED>
+DEVI void dv_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
+                  int64_t q0, int64_t seq, int64_t mykv, float scale,
+                  int causal, const float* lsep, const bf16x8 (&kfrag)[4],
+                  f32x16& dvt0, f32x16& dvt1, int lkv, int hi) {
+  f32x16 st = {};
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+        &ldsQ[lkv][hi * 8 + 16 * c]);
+    st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0, 0);
+  }
+  float p[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int64_t qg = q0 + crow(r, hi);
+    if (MASKED) {
+      const int64_t qgc = qg < seq ? qg : seq - 1;
+      bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
+      p[r] = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
+    } else {
+      p[r] = __expf(st[r] * scale - lsep[qg]);
+    }
+  }
+  bf16x8 pb0 = assemble_pfrag(&p[0]);
+  bf16x8 pb1 = assemble_pfrag(&p[8]);
+#pragma unroll
+  for (int qc = 0; qc < 2; ++qc) {
+    bf16x8 dot0, dot1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int qr = qc * 16 + hi * 8 + j;
+      dot0[j] = ldsDO[qr][lkv];
+      dot1[j] = ldsDO[qr][32 + lkv];
+    }
+    bf16x8 pb = qc == 0 ? pb0 : pb1;
+    dvt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot0, pb, dvt0, 0, 0, 0);
+    dvt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot1, pb, dvt1, 0, 0, 0);
+  }
+}
+
+// one 32-row q tile of the dK accumulation
+template <bool MASKED>
+DEVI void dk_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
+                  int64_t q0, int64_t seq, int64_t mykv, float scale,
+                  int causal, const float* lsep, const float* dltp,
+                  const bf16x8 (&kfrag)[4], const bf16x8 (&vfrag)[4],
+                  f32x16& dkt0, f32x16& dkt1, int lkv, int hi) {
+  f32x16 st = {};
+  f32x16 dpt = {};
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+        &ldsQ[lkv][hi * 8 + 16 * c]);
+    bf16x8 dof = *reinterpret_cast<const bf16x8*>(
+        &ldsDO[lkv][hi * 8 + 16 * c]);
+    st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0, 0);
+    dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfrag[c], dpt, 0, 0,
+                                                  0);
+  }
+  float ds[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    const int64_t qg = q0 + crow(r, hi);
+    if (MASKED) {
+      const int64_t qgc = qg < seq ? qg : seq - 1;
+      bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
+      float pv = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
+      ds[r] = masked ? 0.f : pv * (dpt[r] - dltp[qgc]) * scale;
+    } else {
+      float pv = __expf(st[r] * scale - lsep[qg]);
+      ds[r] = pv * (dpt[r] - dltp[qg]) * scale;
+    }
+  }
+  bf16x8 db0 = assemble_pfrag(&ds[0]);
+  bf16x8 db1 = assemble_pfrag(&ds[8]);
+#pragma unroll
+  for (int qc = 0; qc < 2; ++qc) {
+    bf16x8 qt0, qt1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int qr = qc * 16 + hi * 8 + j;
+      qt0[j] = ldsQ[qr][lkv];
+      qt1[j] = ldsQ[qr][32 + lkv];
+    }
+    bf16x8 db = qc == 0 ? db0 : db1;
+    dkt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt0, db, dkt0, 0, 0, 0);
+    dkt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt1, db, dkt1, 0, 0, 0);
+  }
+}
+
 __global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ dout, const float* __restrict__ lse,
@@ -548,52 +677,39 @@ __global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
   f32x16 dvt0 = {}, dvt1 = {};
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
-  const int64_t q_start = causal ? (kv0_blk / 32) * 32 : 0;
-  for (int64_t q0 = q_start; q0 < seq; q0 += 32) {
+
+  auto stage_q_tile = [&](int64_t q0) {
     __syncthreads();
-    {
-      int64_t qr = q0 + stage_row;
-      if (qr >= seq) qr = seq - 1;
-      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
-      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
-    }
+    int64_t qr = q0 + stage_row;
+    if (qr >= seq) qr = seq - 1;
+    *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
+        *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
+    *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
+        *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
     __syncthreads();
+  };
+
+  // phases: masked causal diagonal -> branch-free bulk -> masked tail
+  const int64_t diag_end =
+      causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
+  const int64_t bulk_end = seq & ~(int64_t)31;
+  int64_t q0 = causal ? kv0_blk : 0;
+  for (; q0 < diag_end; q0 += 32) {
+    stage_q_tile(q0);
     if (!active || (causal && q0 + 31 < kv0)) continue;
-    f32x16 st = {};
-#pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 qf = *reinterpret_cast<const bf16x8*>(
-          &ldsQ[lkv][hi * 8 + 16 * c]);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0,
-                                                   0);
-    }
-    float p[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int64_t qg = q0 + crow(r, hi);
-      const int64_t qgc = qg < seq ? qg : seq - 1;
-      bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
-      p[r] = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
-    }
-    bf16x8 pb0 = assemble_pfrag(&p[0]);
-    bf16x8 pb1 = assemble_pfrag(&p[8]);
-#pragma unroll
-    for (int qc = 0; qc < 2; ++qc) {
-      bf16x8 dot0, dot1;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int qr = qc * 16 + hi * 8 + j;
-        dot0[j] = ldsDO[qr][lkv];
-        dot1[j] = ldsDO[qr][32 + lkv];
-      }
-      bf16x8 pb = qc == 0 ? pb0 : pb1;
-      dvt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot0, pb, dvt0, 0, 0,
-                                                     0);
-      dvt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot1, pb, dvt1, 0, 0,
-                                                     0);
-    }
+    dv_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, kfrag,
+                  dvt0, dvt1, lkv, hi);
+  }
+  for (; q0 < bulk_end; q0 += 32) {
+    stage_q_tile(q0);
+    dv_tile<false>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, kfrag,
+                   dvt0, dvt1, lkv, hi);
+  }
+  for (; q0 < seq; q0 += 32) {
+    stage_q_tile(q0);
+    if (!active) continue;
+    dv_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, kfrag,
+                  dvt0, dvt1, lkv, hi);
   }
   if (!active || mykv >= seq) return;
   short* dvp = dv + (bh / heads) * g_sb + (bh % heads) * g_sh +
@@ -645,58 +761,38 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
   f32x16 dkt0 = {}, dkt1 = {};
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
-  const int64_t q_start = causal ? (kv0_blk / 32) * 32 : 0;
-  for (int64_t q0 = q_start; q0 < seq; q0 += 32) {
+
+  auto stage_q_tile = [&](int64_t q0) {
     __syncthreads();
-    {
-      int64_t qr = q0 + stage_row;
-      if (qr >= seq) qr = seq - 1;
-      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
-      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
-    }
+    int64_t qr = q0 + stage_row;
+    if (qr >= seq) qr = seq - 1;
+    *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
+        *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
+    *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
+        *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
     __syncthreads();
+  };
+
+  const int64_t diag_end =
+      causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
+  const int64_t bulk_end = seq & ~(int64_t)31;
+  int64_t q0 = causal ? kv0_blk : 0;
+  for (; q0 < diag_end; q0 += 32) {
+    stage_q_tile(q0);
     if (!active || (causal && q0 + 31 < kv0)) continue;
-    f32x16 st = {};
-    f32x16 dpt = {};
-#pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 qf = *reinterpret_cast<const bf16x8*>(
-          &ldsQ[lkv][hi * 8 + 16 * c]);
-      bf16x8 dof = *reinterpret_cast<const bf16x8*>(
-          &ldsDO[lkv][hi * 8 + 16 * c]);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0,
-                                                   0);
-      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfrag[c], dpt, 0,
-                                                    0, 0);
-    }
-    float ds[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int64_t qg = q0 + crow(r, hi);
-      const int64_t qgc = qg < seq ? qg : seq - 1;
-      bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
-      float pv = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
-      ds[r] = masked ? 0.f : pv * (dpt[r] - dltp[qgc]) * scale;
-    }
-    bf16x8 db0 = assemble_pfrag(&ds[0]);
-    bf16x8 db1 = assemble_pfrag(&ds[8]);
-#pragma unroll
-    for (int qc = 0; qc < 2; ++qc) {
-      bf16x8 qt0, qt1;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int qr = qc * 16 + hi * 8 + j;
-        qt0[j] = ldsQ[qr][lkv];
-        qt1[j] = ldsQ[qr][32 + lkv];
-      }
-      bf16x8 db = qc == 0 ? db0 : db1;
-      dkt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt0, db, dkt0, 0, 0,
-                                                     0);
-      dkt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt1, db, dkt1, 0, 0,
-                                                     0);
-    }
+    dk_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, dltp,
+                  kfrag, vfrag, dkt0, dkt1, lkv, hi);
+  }
+  for (; q0 < bulk_end; q0 += 32) {
+    stage_q_tile(q0);
+    dk_tile<false>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, dltp,
+                   kfrag, vfrag, dkt0, dkt1, lkv, hi);
+  }
+  for (; q0 < seq; q0 += 32) {
+    stage_q_tile(q0);
+    if (!active) continue;
+    dk_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, dltp,
+                  kfrag, vfrag, dkt0, dkt1, lkv, hi);
   }
   if (!active || mykv >= seq) return;
   short* dkp = dk + (bh / heads) * g_sb + (bh % heads) * g_sh +
@@ -711,8 +807,57 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
 // ============================================================================
 // backward kernel B (q-parallel): dQ.  K and V tiles staged in LDS as
 // row copies per block; row fragments via ds_read_b128, K-transposed
-// fragments via 2-byte LDS reads.
+// fragments via 2-byte LDS reads.  Like the forward, the kv loop is
+// split into a branch-free bulk phase (below the causal diagonal /
+// inside the seq bound) and a masked diagonal+tail phase.
 // ============================================================================
+
+// one 32-key tile of the dQ loop
+template <bool MASKED>
+DEVI void dq_tile(const short (&ldsK)[32][72], const short (&ldsVr)[32][72],
+                  int64_t kv0, int64_t seq, int64_t myq, float scale,
+                  int causal, float mylse, float mydelta,
+                  const bf16x8 (&qfrag)[4], const bf16x8 (&dofrag)[4],
+                  f32x16& dqt0, f32x16& dqt1, int lq, int hi) {
+  f32x16 st = {};
+  f32x16 dpt = {};
+#pragma unroll
+  for (int c = 0; c < 4; ++c) {
+    bf16x8 kf = *reinterpret_cast<const bf16x8*>(
+        &ldsK[lq][hi * 8 + 16 * c]);
+    bf16x8 vf = *reinterpret_cast<const bf16x8*>(
+        &ldsVr[lq][hi * 8 + 16 * c]);
+    st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[c], st, 0, 0, 0);
+    dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[c], dpt, 0, 0,
+                                                  0);
+  }
+  float ds[16];
+#pragma unroll
+  for (int r = 0; r < 16; ++r) {
+    float pv = __expf(st[r] * scale - mylse);
+    if (MASKED) {
+      const int64_t kvg = kv0 + crow(r, hi);
+      if (myq >= seq || kvg >= seq || (causal && kvg > myq)) pv = 0.f;
+    }
+    ds[r] = pv * (dpt[r] - mydelta) * scale;
+  }
+  bf16x8 db0 = assemble_pfrag(&ds[0]);
+  bf16x8 db1 = assemble_pfrag(&ds[8]);
+#pragma unroll
+  for (int kc = 0; kc < 2; ++kc) {
+    bf16x8 kt0, kt1;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      const int kr = kc * 16 + hi * 8 + j;
+      kt0[j] = ldsK[kr][lq];
+      kt1[j] = ldsK[kr][32 + lq];
+    }
+    bf16x8 db = kc == 0 ? db0 : db1;
+    dqt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt0, db, dqt0, 0, 0, 0);
+    dqt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt1, db, dqt1, 0, 0, 0);
+  }
+}
+
 __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -771,60 +916,34 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
+  const int64_t bulk_end = causal ? q0_blk : (seq & ~(int64_t)31);
   const int64_t blk_kv_end =
       causal ? (q0_blk + 128 < seq ? q0_blk + 128 : seq) : seq;
-  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
+
+  auto stage_kv_tile = [&](int64_t kv0) {
     __syncthreads();
-    {
-      int64_t kr = kv0 + stage_row;
-      if (kr >= seq) kr = seq - 1;   // dS there is 0
-      *reinterpret_cast<bf16x8*>(&ldsK[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
-      *reinterpret_cast<bf16x8*>(&ldsVr[stage_row][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
-    }
+    int64_t kr = kv0 + stage_row;
+    if (kr >= seq) kr = seq - 1;   // dS there is 0
+    *reinterpret_cast<bf16x8*>(&ldsK[stage_row][stage_seg]) =
+        *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
+    *reinterpret_cast<bf16x8*>(&ldsVr[stage_row][stage_seg]) =
+        *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
     __syncthreads();
+  };
+
+  int64_t kv0 = 0;
+  for (; kv0 < bulk_end; kv0 += 32) {
+    stage_kv_tile(kv0);
+    dq_tile<false>(ldsK, ldsVr, kv0, seq, myq, scale, causal, mylse,
+                   mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
+  }
+  for (; kv0 < blk_kv_end; kv0 += 32) {
+    stage_kv_tile(kv0);
     const int64_t wave_kv_end = causal
         ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
     if (!active || kv0 >= wave_kv_end) continue;
-    f32x16 st = {};
-    f32x16 dpt = {};
-#pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-          &ldsK[lq][hi * 8 + 16 * c]);
-      bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-          &ldsVr[lq][hi * 8 + 16 * c]);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[c], st, 0, 0,
-                                                   0);
-      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[c], dpt, 0,
-                                                    0, 0);
-    }
-    float ds[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int64_t kvg = kv0 + crow(r, hi);
-      bool masked = myq >= seq || kvg >= seq || (causal && kvg > myq);
-      float pv = masked ? 0.f : __expf(st[r] * scale - mylse);
-      ds[r] = masked ? 0.f : pv * (dpt[r] - mydelta) * scale;
-    }
-    bf16x8 db0 = assemble_pfrag(&ds[0]);
-    bf16x8 db1 = assemble_pfrag(&ds[8]);
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf16x8 kt0, kt1;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int kr = kc * 16 + hi * 8 + j;
-        kt0[j] = ldsK[kr][lq];
-        kt1[j] = ldsK[kr][32 + lq];
-      }
-      bf16x8 db = kc == 0 ? db0 : db1;
-      dqt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt0, db, dqt0, 0, 0,
-                                                     0);
-      dqt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt1, db, dqt1, 0, 0,
-                                                     0);
-    }
+    dq_tile<true>(ldsK, ldsVr, kv0, seq, myq, scale, causal, mylse,
+                  mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
   }
 
   if (!active || myq >= seq) return;

@@ -260,14 +260,11 @@ class GlooCommunicator(Communicator):
         return out
 
     def all_to_all_single(self, out, inp, async_op=False):
-        full = torch.empty(self.size * inp.numel(), dtype=inp.dtype,
-                           device=inp.device)
-        self.all_gather(full, inp)
-        # full[r] = rank r's input; my row of each rank's input
+        # pairwise p2p (O(W) traffic per rank), not allgather (O(W^2)):
+        # keeps CPU plumbing tests representative of the RCCL a2a cost
         chunk = inp.numel() // self.size
-        full = full.reshape(self.size, self.size, chunk)
-        out.copy_(full[:, self.rank, :].reshape(out.shape))
-        return out
+        counts = [chunk] * self.size
+        return self.all_to_all_v(out, inp, counts, counts, async_op)
 
     def all_to_all_v(self, out, inp, out_counts, in_counts, async_op=False):
         reqs = []
@@ -304,7 +301,10 @@ class GlooCommunicator(Communicator):
     def all_gather_v(self, outs, inp, async_op=False):
         for r in range(self.size):
             if r == self.rank:
-                outs[r].reshape(-1).copy_(inp.reshape(-1))
+                # copy_ on the tensor itself: reshape(-1) on a
+                # non-contiguous out would return a copy and drop the
+                # write (advisor finding r1)
+                outs[r].copy_(inp.reshape(outs[r].shape))
             dist.broadcast(outs[r], self.ranks[r], group=self.group)
         return outs
 

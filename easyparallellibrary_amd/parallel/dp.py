@@ -130,10 +130,17 @@ class FlatParamGroup:
 
     def adopt_grad(self, p):
         """Called from the post-accumulate hook: fold autograd's fresh
-        grad into the arena view (first touch this step) and re-point."""
+        grad into the arena view (first touch this step) and re-point.
+        Sparse COO grads (nn.Embedding(sparse=True) under
+        communication.sparse_as_dense=True) densify into the view; later
+        micro-batches accumulate sparse-into-dense in place."""
         view = self._grad_views[id(p)]
         if p.grad is not view:
-            view.copy_(p.grad)
+            if p.grad.is_sparse:
+                view.zero_()
+                view.add_(p.grad.to(view.dtype))
+            else:
+                view.copy_(p.grad)
             p.grad = view
 
     def sync_master_to_params(self):

@@ -220,6 +220,24 @@ class Engine:
         overlap = self.config.communication.overlap_grad_reduce
         zero_level = self.config.zero.level
         self._bcast_jobs = []
+        # ---- sparse-grad params (nn.Embedding(sparse=True)) -----------------
+        # reference: rewriters/sparse_allreduce.py:39-173 + the
+        # communication.sparse_as_dense knob (config.py:81).  Dense mode
+        # keeps them in the arena (adopt_grad densifies); gather mode
+        # excludes them and wires a SparseGradHandler per DP group.
+        from easyparallellibrary_amd.parallel.sparse import (
+            find_sparse_grad_params)
+        self._sparse_ids = {
+            id(p) for p in find_sparse_grad_params([self.model])}
+        self._sparse_dense = self.config.communication.sparse_as_dense
+        self.sparse_handlers = []
+        _pending_sparse = []
+        if self._sparse_ids and not self._sparse_dense and (
+                zero_level or self.config.offload.level):
+            raise ValueError(
+                "sparse-grad embeddings with zero/offload need "
+                "communication.sparse_as_dense=True (the gather-based "
+                "sparse path keeps params outside the shard arenas)")
         for tg in self.plan.taskgraphs:
             vd = tg.virtual_device
             k = len(vd.local_ranks(0))
@@ -257,6 +275,12 @@ class Engine:
                 elif kind == "replicated":
                     params = [p for p in params
                               if getattr(p, "_epl_shard_dim", None) is None]
+                if self._sparse_ids and not self._sparse_dense:
+                    sp = [p for p in params if id(p) in self._sparse_ids]
+                    params = [p for p in params
+                              if id(p) not in self._sparse_ids]
+                    if sp:
+                        _pending_sparse.append((sp, bcomm, granks))
                 if not params:
                     continue
                 fg = FlatParamGroup(
@@ -282,6 +306,10 @@ class Engine:
             if bcomm is not None and bcomm.size > 1:
                 bcomm.broadcast(fg.param_arena, root=0)
                 fg.refresh_master()
+        for sp, bcomm, _ in _pending_sparse:
+            if bcomm is not None and bcomm.size > 1:
+                for p in sp:
+                    bcomm.broadcast(p.data, root=0)
 
         # ---- optimizer -------------------------------------------------------
         okw = dict(optimizer_kwargs or {})
@@ -299,6 +327,20 @@ class Engine:
             self.optimizer = CPUOffloadAdamW(self.flat_groups, **okw)
         else:
             self.optimizer = OPTIMIZERS[optimizer](self.flat_groups, **okw)
+        for sp, bcomm, granks in _pending_sparse:
+            from easyparallellibrary_amd.parallel.sparse import (
+                SparseGradHandler)
+            h = SparseGradHandler(
+                sp, bcomm, reduce_method=reduce_method,
+                lr=okw.get("lr", lr), betas=okw.get("betas", (0.9, 0.999)),
+                eps=okw.get("eps", 1e-8),
+                weight_decay=okw.get("weight_decay", 0.01))
+            h.n_copies = max(1, len(granks))
+            self.sparse_handlers.append(h)
+        if self.sparse_handlers:
+            logger.info("sparse-grad path: %d param(s) on the gather wire "
+                        "(communication.sparse_as_dense=False)",
+                        sum(len(h.params) for h in self.sparse_handlers))
 
         # ---- pipeline runtime ------------------------------------------------
         self.pipeline = None
@@ -327,10 +369,14 @@ class Engine:
     def finish_grad_sync(self):
         for r in self.reducers:
             r.finish()
+        for h in self.sparse_handlers:
+            h.reduce()
 
     def zero_grad(self):
         for fg in self.flat_groups:
             fg.zero_grad()
+        for h in self.sparse_handlers:
+            h.zero_grad()
 
     def forward(self, x):
         """Plain forward (this rank's stage under pipeline parallelism —
@@ -384,6 +430,8 @@ class Engine:
                     # unscale — no extra pass over the arenas
                     grad_scale *= gnorm / max_norm
             self.optimizer.step(grad_scale=grad_scale)
+            for h in self.sparse_handlers:
+                h.step(grad_scale=grad_scale)
         self.amp.post_step(found_inf)
         self.global_step += 1
         return loss
@@ -418,6 +466,8 @@ class Engine:
         opt = self.optimizer
         inner = getattr(opt, "inner", None)
         (inner if inner is not None else opt).lr = float(lr)
+        for h in self.sparse_handlers:
+            h.lr = float(lr)
 
     @property
     def lr(self):
@@ -451,6 +501,8 @@ class Engine:
             else:
                 sq = g.float().pow(2).sum().reshape(1)
             total += sq / n_copies
+        for h in self.sparse_handlers:
+            total += h.sqnorm().to(total.device) / h.n_copies
         if dist.is_initialized():
             use_dev = (dist.get_backend() == "nccl"
                        and self.device.type == "cuda")

@@ -96,8 +96,10 @@ class PipelineRuntime:
         Expert-parallel stages carry per-position DATA STREAMS, not
         replicas, so they cannot change width; the replicated_io marker
         is how a stage declares itself safe (never guessed from module
-        types).  Runs the PreferForward (GPipe) schedule: the 1F1B fused
-        bidirectional exchanges assume uniform position chains.
+        types).  Both schedules run: 1F1B's fused bidirectional
+        exchanges decompose into per-link pairwise batches (the narrow
+        rank batches with position 0; plain send/recv covers the other
+        k-1 links), exercised by the mixed-width exact-match tests.
         """
         self.widths = [len(r) for r in rank_lists]
         S = self.S

@@ -71,6 +71,10 @@ def save_checkpoint(engine, path, save_optimizer=True):
         sd = engine.optimizer.state_dict()
         sd_cpu = _to_cpu(sd)
         torch.save(sd_cpu, os.path.join(path, "opt_rank{}.pt".format(rank)))
+        if getattr(engine, "sparse_handlers", None):
+            torch.save(
+                [_to_cpu(h.state_dict()) for h in engine.sparse_handlers],
+                os.path.join(path, "opt_sparse_rank{}.pt".format(rank)))
     _barrier(engine)
 
 
@@ -129,6 +133,13 @@ class ShardingLoader:
             if os.path.exists(opt_path):
                 engine.optimizer.load_state_dict(
                     torch.load(opt_path, weights_only=False))
+            sp_path = os.path.join(
+                self.path, "opt_sparse_rank{}.pt".format(engine.rank))
+            if os.path.exists(sp_path) and getattr(
+                    engine, "sparse_handlers", None):
+                blobs = torch.load(sp_path, weights_only=False)
+                for h, b in zip(engine.sparse_handlers, blobs):
+                    h.load_state_dict(b)
         _barrier(engine)
 
     def _load_replicated_tg(self, tg, strict):
