@@ -551,17 +551,19 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 // seq tail (same rationale as the forward v3 split).
 // ============================================================================
 
-// one 32-row q tile of the dV accumulation
+// one 32-row q sub-tile (of the 64-row staged chunk) of the dV
+// accumulation; `sub` selects the LDS half
 template <bool MASKED>
-DEVI void dv_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
-                  int64_t q0, int64_t seq, int64_t mykv, float scale,
-                  int causal, const float* lsep, const bf16x8 (&kfrag)[4],
-                  f32x16& dvt0, f32x16& dvt1, int lkv, int hi) {
+DEVI void dv_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
+                  int sub, int64_t q0, int64_t seq, int64_t mykv,
+                  float scale, int causal, const float* lsep,
+                  const bf16x8 (&kfrag)[4], f32x16& dvt0, f32x16& dvt1,
+                  int lkv, int hi) {
   f32x16 st = {};
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     bf16x8 qf = *reinterpret_cast<const bf16x8*>(
-        &ldsQ[lkv][hi * 8 + 16 * c]);
+        &ldsQ[sub + lkv][hi * 8 + 16 * c]);
     st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0, 0);
   }
   float p[16];
@@ -583,7 +585,7 @@ DEVI void dv_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
     bf16x8 dot0, dot1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int qr = qc * 16 + hi * 8 + j;
+      const int qr = sub + qc * 16 + hi * 8 + j;
       dot0[j] = ldsDO[qr][lkv];
       dot1[j] = ldsDO[qr][32 + lkv];
     }
@@ -593,21 +595,22 @@ DEVI void dv_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
   }
 }
 
-// one 32-row q tile of the dK accumulation
+// one 32-row q sub-tile of the dK accumulation
 template <bool MASKED>
-DEVI void dk_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
-                  int64_t q0, int64_t seq, int64_t mykv, float scale,
-                  int causal, const float* lsep, const float* dltp,
-                  const bf16x8 (&kfrag)[4], const bf16x8 (&vfrag)[4],
-                  f32x16& dkt0, f32x16& dkt1, int lkv, int hi) {
+DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
+                  int sub, int64_t q0, int64_t seq, int64_t mykv,
+                  float scale, int causal, const float* lsep,
+                  const float* dltp, const bf16x8 (&kfrag)[4],
+                  const bf16x8 (&vfrag)[4], f32x16& dkt0, f32x16& dkt1,
+                  int lkv, int hi) {
   f32x16 st = {};
   f32x16 dpt = {};
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     bf16x8 qf = *reinterpret_cast<const bf16x8*>(
-        &ldsQ[lkv][hi * 8 + 16 * c]);
+        &ldsQ[sub + lkv][hi * 8 + 16 * c]);
     bf16x8 dof = *reinterpret_cast<const bf16x8*>(
-        &ldsDO[lkv][hi * 8 + 16 * c]);
+        &ldsDO[sub + lkv][hi * 8 + 16 * c]);
     st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0, 0);
     dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dof, vfrag[c], dpt, 0, 0,
                                                   0);
@@ -633,7 +636,7 @@ DEVI void dk_tile(const short (&ldsQ)[32][72], const short (&ldsDO)[32][72],
     bf16x8 qt0, qt1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int qr = qc * 16 + hi * 8 + j;
+      const int qr = sub + qc * 16 + hi * 8 + j;
       qt0[j] = ldsQ[qr][lkv];
       qt1[j] = ldsQ[qr][32 + lkv];
     }
@@ -650,8 +653,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
     int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
     int64_t do_sb, int64_t do_sh, int64_t do_ss, int64_t g_sb,
     int64_t g_sh, int64_t g_ss) {
-  __shared__ short ldsQ[32][72];
-  __shared__ short ldsDO[32][72];
+  __shared__ short ldsQ[64][72];
+  __shared__ short ldsDO[64][72];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -678,38 +681,57 @@ __global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
 
-  auto stage_q_tile = [&](int64_t q0) {
+  // stage 64 q rows (two 32-row halves) per barrier pair
+  auto stage_q64 = [&](int64_t q0) {
     __syncthreads();
-    int64_t qr = q0 + stage_row;
-    if (qr >= seq) qr = seq - 1;
-    *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
-        *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
-    *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
-        *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int64_t qr = q0 + stage_row + half * 32;
+      if (qr >= seq) qr = seq - 1;
+      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row + half * 32][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row + half * 32][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+    }
     __syncthreads();
   };
 
-  // phases: masked causal diagonal -> branch-free bulk -> masked tail
+  // phases (64-row chunks): masked causal diagonal -> branch-free bulk
+  // -> masked tail
   const int64_t diag_end =
       causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
-  const int64_t bulk_end = seq & ~(int64_t)31;
+  const int64_t bulk_end = seq & ~(int64_t)63;
   int64_t q0 = causal ? kv0_blk : 0;
-  for (; q0 < diag_end; q0 += 32) {
-    stage_q_tile(q0);
-    if (!active || (causal && q0 + 31 < kv0)) continue;
-    dv_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, kfrag,
-                  dvt0, dvt1, lkv, hi);
-  }
-  for (; q0 < bulk_end; q0 += 32) {
-    stage_q_tile(q0);
-    dv_tile<false>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, kfrag,
-                   dvt0, dvt1, lkv, hi);
-  }
-  for (; q0 < seq; q0 += 32) {
-    stage_q_tile(q0);
+  for (; q0 < diag_end; q0 += 64) {
+    stage_q64(q0);
     if (!active) continue;
-    dv_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, kfrag,
-                  dvt0, dvt1, lkv, hi);
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32) {
+      const int64_t q0s = q0 + sub;
+      if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
+      dv_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
+                    kfrag, dvt0, dvt1, lkv, hi);
+    }
+  }
+  q0 = diag_end > q0 ? diag_end : q0;  // 64-aligned when < seq (diag is
+                                       // 2x64 or clipped by seq)
+  for (; q0 + 63 < bulk_end; q0 += 64) {
+    stage_q64(q0);
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32)
+      dv_tile<false>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale, causal,
+                     lsep, kfrag, dvt0, dvt1, lkv, hi);
+  }
+  for (; q0 < seq; q0 += 64) {
+    stage_q64(q0);
+    if (!active) continue;
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32) {
+      const int64_t q0s = q0 + sub;
+      if (q0s >= seq) break;
+      dv_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
+                    kfrag, dvt0, dvt1, lkv, hi);
+    }
   }
   if (!active || mykv >= seq) return;
   short* dvp = dv + (bh / heads) * g_sb + (bh % heads) * g_sh +
@@ -729,8 +751,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
     int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
     int64_t do_sb, int64_t do_sh, int64_t do_ss, int64_t g_sb,
     int64_t g_sh, int64_t g_ss) {
-  __shared__ short ldsQ[32][72];
-  __shared__ short ldsDO[32][72];
+  __shared__ short ldsQ[64][72];
+  __shared__ short ldsDO[64][72];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -762,37 +784,53 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
 
-  auto stage_q_tile = [&](int64_t q0) {
+  auto stage_q64 = [&](int64_t q0) {
     __syncthreads();
-    int64_t qr = q0 + stage_row;
-    if (qr >= seq) qr = seq - 1;
-    *reinterpret_cast<bf16x8*>(&ldsQ[stage_row][stage_seg]) =
-        *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
-    *reinterpret_cast<bf16x8*>(&ldsDO[stage_row][stage_seg]) =
-        *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int64_t qr = q0 + stage_row + half * 32;
+      if (qr >= seq) qr = seq - 1;
+      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row + half * 32][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row + half * 32][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+    }
     __syncthreads();
   };
 
   const int64_t diag_end =
       causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
-  const int64_t bulk_end = seq & ~(int64_t)31;
+  const int64_t bulk_end = seq & ~(int64_t)63;
   int64_t q0 = causal ? kv0_blk : 0;
-  for (; q0 < diag_end; q0 += 32) {
-    stage_q_tile(q0);
-    if (!active || (causal && q0 + 31 < kv0)) continue;
-    dk_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, dltp,
-                  kfrag, vfrag, dkt0, dkt1, lkv, hi);
-  }
-  for (; q0 < bulk_end; q0 += 32) {
-    stage_q_tile(q0);
-    dk_tile<false>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, dltp,
-                   kfrag, vfrag, dkt0, dkt1, lkv, hi);
-  }
-  for (; q0 < seq; q0 += 32) {
-    stage_q_tile(q0);
+  for (; q0 < diag_end; q0 += 64) {
+    stage_q64(q0);
     if (!active) continue;
-    dk_tile<true>(ldsQ, ldsDO, q0, seq, mykv, scale, causal, lsep, dltp,
-                  kfrag, vfrag, dkt0, dkt1, lkv, hi);
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32) {
+      const int64_t q0s = q0 + sub;
+      if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
+      dk_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
+                    dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi);
+    }
+  }
+  q0 = diag_end > q0 ? diag_end : q0;
+  for (; q0 + 63 < bulk_end; q0 += 64) {
+    stage_q64(q0);
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32)
+      dk_tile<false>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale, causal,
+                     lsep, dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi);
+  }
+  for (; q0 < seq; q0 += 64) {
+    stage_q64(q0);
+    if (!active) continue;
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32) {
+      const int64_t q0s = q0 + sub;
+      if (q0s >= seq) break;
+      dk_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
+                    dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi);
+    }
   }
   if (!active || mykv >= seq) return;
   short* dkp = dk + (bh / heads) * g_sb + (bh % heads) * g_sh +
@@ -812,11 +850,11 @@ __global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
 // inside the seq bound) and a masked diagonal+tail phase.
 // ============================================================================
 
-// one 32-key tile of the dQ loop
+// one 32-key sub-tile (of the 64-row staged chunk) of the dQ loop
 template <bool MASKED>
-DEVI void dq_tile(const short (&ldsK)[32][72], const short (&ldsVr)[32][72],
-                  int64_t kv0, int64_t seq, int64_t myq, float scale,
-                  int causal, float mylse, float mydelta,
+DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
+                  int sub, int64_t kv0, int64_t seq, int64_t myq,
+                  float scale, int causal, float mylse, float mydelta,
                   const bf16x8 (&qfrag)[4], const bf16x8 (&dofrag)[4],
                   f32x16& dqt0, f32x16& dqt1, int lq, int hi) {
   f32x16 st = {};
@@ -824,9 +862,9 @@ DEVI void dq_tile(const short (&ldsK)[32][72], const short (&ldsVr)[32][72],
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
     bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-        &ldsK[lq][hi * 8 + 16 * c]);
+        &ldsK[sub + lq][hi * 8 + 16 * c]);
     bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-        &ldsVr[lq][hi * 8 + 16 * c]);
+        &ldsVr[sub + lq][hi * 8 + 16 * c]);
     st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[c], st, 0, 0, 0);
     dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[c], dpt, 0, 0,
                                                   0);
@@ -848,7 +886,7 @@ DEVI void dq_tile(const short (&ldsK)[32][72], const short (&ldsVr)[32][72],
     bf16x8 kt0, kt1;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
-      const int kr = kc * 16 + hi * 8 + j;
+      const int kr = sub + kc * 16 + hi * 8 + j;
       kt0[j] = ldsK[kr][lq];
       kt1[j] = ldsK[kr][32 + lq];
     }
@@ -867,8 +905,8 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
     int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss,
     int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
     int64_t o_ss) {
-  __shared__ short ldsK[32][72];
-  __shared__ short ldsVr[32][72];
+  __shared__ short ldsK[64][72];
+  __shared__ short ldsVr[64][72];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -916,34 +954,46 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
 
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
-  const int64_t bulk_end = causal ? q0_blk : (seq & ~(int64_t)31);
+  // bulk chunks (64 kv rows) are full for every lane; q0_blk is a
+  // multiple of 128 so the causal bulk region is 64-aligned
+  const int64_t bulk_end = causal ? q0_blk : (seq & ~(int64_t)63);
   const int64_t blk_kv_end =
       causal ? (q0_blk + 128 < seq ? q0_blk + 128 : seq) : seq;
 
-  auto stage_kv_tile = [&](int64_t kv0) {
+  auto stage_kv64 = [&](int64_t kv0) {
     __syncthreads();
-    int64_t kr = kv0 + stage_row;
-    if (kr >= seq) kr = seq - 1;   // dS there is 0
-    *reinterpret_cast<bf16x8*>(&ldsK[stage_row][stage_seg]) =
-        *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
-    *reinterpret_cast<bf16x8*>(&ldsVr[stage_row][stage_seg]) =
-        *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int64_t kr = kv0 + stage_row + half * 32;
+      if (kr >= seq) kr = seq - 1;   // dS there is 0
+      *reinterpret_cast<bf16x8*>(&ldsK[stage_row + half * 32][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
+      *reinterpret_cast<bf16x8*>(&ldsVr[stage_row + half * 32][stage_seg]) =
+          *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
+    }
     __syncthreads();
   };
 
   int64_t kv0 = 0;
-  for (; kv0 < bulk_end; kv0 += 32) {
-    stage_kv_tile(kv0);
-    dq_tile<false>(ldsK, ldsVr, kv0, seq, myq, scale, causal, mylse,
-                   mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
+  for (; kv0 < bulk_end; kv0 += 64) {
+    stage_kv64(kv0);
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32)
+      dq_tile<false>(ldsK, ldsVr, sub, kv0 + sub, seq, myq, scale, causal,
+                     mylse, mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
   }
-  for (; kv0 < blk_kv_end; kv0 += 32) {
-    stage_kv_tile(kv0);
+  for (; kv0 < blk_kv_end; kv0 += 64) {
+    stage_kv64(kv0);
     const int64_t wave_kv_end = causal
         ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
-    if (!active || kv0 >= wave_kv_end) continue;
-    dq_tile<true>(ldsK, ldsVr, kv0, seq, myq, scale, causal, mylse,
-                  mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
+    if (!active) continue;
+#pragma unroll
+    for (int sub = 0; sub < 64; sub += 32) {
+      const int64_t kvs = kv0 + sub;
+      if (kvs >= wave_kv_end) break;
+      dq_tile<true>(ldsK, ldsVr, sub, kvs, seq, myq, scale, causal, mylse,
+                    mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
+    }
   }
 
   if (!active || myq >= seq) return;

@@ -167,6 +167,13 @@ class IoConfig(BaseConfig):
         # slice input files proportionally to local replicas
         "slicing": False,
         "unbalanced_io_slicing": False,
+        # checkpoint writes stream in buckets of this size instead of
+        # materializing a full host copy (reference MemoryEfficientBuilder
+        # 50 MB buckets, /root/reference/epl/runtime/saver.py:145-207)
+        "checkpoint_bucket_mb": 50,
+        # serialize checkpoint writes across ranks (one writer at a time,
+        # reference saver's serialized shard writes)
+        "serial_checkpoint_writes": False,
     }
 
 

@@ -34,6 +34,37 @@ def _tg_params(tg):
     return out
 
 
+def _save_tg_bucketed(tg, pos, nshards, path, bucket_bytes):
+    """Stream one taskgraph's params to disk in ~bucket-sized part files:
+    never more than one bucket of host copies alive at a time (reference
+    MemoryEfficientBuilder's 50 MB bucketed serialized writes,
+    /root/reference/epl/runtime/saver.py:145-207)."""
+    params = _tg_params(tg)
+    parts = 0
+    cur, cur_bytes = {}, 0
+
+    def flush():
+        nonlocal parts, cur, cur_bytes
+        if not cur:
+            return
+        torch.save(cur, os.path.join(
+            path, "tg{}_pos{}.part{}.pt".format(tg.index, pos, parts)))
+        parts += 1
+        cur, cur_bytes = {}, 0
+
+    for k, t in params.items():
+        key = "{}.shard{}".format(k, pos) if tg.is_split else k
+        cur[key] = t.detach().to("cpu")
+        cur_bytes += t.numel() * t.element_size()
+        if cur_bytes >= bucket_bytes:
+            flush()
+    flush()
+    torch.save(
+        {"taskgraph": tg.index, "position": pos, "is_split": tg.is_split,
+         "nshards": nshards, "nparts": parts},
+        os.path.join(path, "tg{}_pos{}.pt".format(tg.index, pos)))
+
+
 def save_checkpoint(engine, path, save_optimizer=True):
     os.makedirs(path, exist_ok=True)
     rank = engine.rank
@@ -50,23 +81,22 @@ def save_checkpoint(engine, path, save_optimizer=True):
                  "module_names": tg.module_names}
                 for tg in engine.plan.taskgraphs],
         }, os.path.join(path, "meta.pt"))
-    if engine.replica_id == 0:
-        for tg in engine._owned_tgs:
-            ranks = tg.virtual_device.local_ranks(0)
-            pos = ranks.index(rank)
-            if not tg.is_split and pos != 0:
-                continue  # replicated: first position saves
-            params = _tg_params(tg)
-            blob = {}
-            for k, t in params.items():
-                key = "{}.shard{}".format(k, pos) if tg.is_split else k
-                blob[key] = t.detach().to("cpu")
-            torch.save(
-                {"taskgraph": tg.index, "position": pos,
-                 "is_split": tg.is_split,
-                 "nshards": len(ranks) if tg.is_split else 1,
-                 "params": blob},
-                os.path.join(path, "tg{}_pos{}.pt".format(tg.index, pos)))
+    bucket_bytes = int(engine.config.io.checkpoint_bucket_mb) * 1024 * 1024
+    serial = engine.config.io.serial_checkpoint_writes
+    turns = range(engine.world_size) if serial else [None]
+    for turn in turns:
+        if turn is None or turn == rank:
+            if engine.replica_id == 0:
+                for tg in engine._owned_tgs:
+                    ranks = tg.virtual_device.local_ranks(0)
+                    pos = ranks.index(rank)
+                    if not tg.is_split and pos != 0:
+                        continue  # replicated: first position saves
+                    _save_tg_bucketed(tg, pos,
+                                      len(ranks) if tg.is_split else 1,
+                                      path, bucket_bytes)
+        if turn is not None:
+            _barrier(engine)
     if save_optimizer:
         sd = engine.optimizer.state_dict()
         sd_cpu = _to_cpu(sd)
@@ -142,23 +172,61 @@ class ShardingLoader:
                     h.load_state_dict(b)
         _barrier(engine)
 
+    def _head_and_blobs(self, tgindex, pos):
+        """Yield the param dicts of tg/pos — one per bucketed part file,
+        or the single legacy blob.  Returns (head, iterator)."""
+        f = os.path.join(self.path, "tg{}_pos{}.pt".format(tgindex, pos))
+        head = torch.load(f, weights_only=False)
+
+        def gen():
+            if "params" in head:   # legacy single-blob format
+                yield head["params"]
+                return
+            for i in range(head.get("nparts", 0)):
+                yield torch.load(os.path.join(
+                    self.path, "tg{}_pos{}.part{}.pt".format(
+                        tgindex, pos, i)), weights_only=False)
+        return head, gen()
+
+    def _merged_params(self, tgindex, pos):
+        head, blobs = self._head_and_blobs(tgindex, pos)
+        out = {}
+        for b in blobs:
+            out.update(b)
+        head = dict(head)
+        head["params"] = out
+        return head
+
     def _load_replicated_tg(self, tg, strict):
-        f = os.path.join(self.path, "tg{}_pos0.pt".format(tg.index))
-        blob = torch.load(f, weights_only=False)["params"]
         params = _tg_params(tg)
-        for k, t in params.items():
-            src = self._remap(k)
-            if src in blob:
-                t.data.copy_(blob[src].to(t.device, t.dtype))
-            elif strict:
-                raise KeyError("missing checkpoint tensor {}".format(src))
+        want = {self._remap(k): (k, t) for k, t in params.items()}
+        found = set()
+        _, blobs = self._head_and_blobs(tg.index, 0)
+        for blob in blobs:   # one bucket of host tensors at a time
+            for src, t in blob.items():
+                if src in want:
+                    want[src][1].data.copy_(
+                        t.to(want[src][1].device, want[src][1].dtype))
+                    found.add(src)
+        if strict:
+            missing = set(want) - found
+            if missing:
+                raise KeyError(
+                    "missing checkpoint tensor(s) {}".format(sorted(missing)))
 
     def _load_split_tg(self, tg, pos, nshards, strict):
-        # gather available shard files
+        # gather available shard files (reshard path merges each
+        # position's parts; the same-count fast path below could stream,
+        # but resharding needs all shards of a tensor at once anyway)
         import glob
         files = sorted(glob.glob(os.path.join(
             self.path, "tg{}_pos*.pt".format(tg.index))))
-        shards = [torch.load(f, weights_only=False) for f in files]
+        files = [f for f in files if ".part" not in os.path.basename(f)]
+        shards = [
+            self._merged_params(
+                tg.index,
+                torch.load(f, weights_only=False)["position"])
+            for f in files]
         saved_n = shards[0]["nshards"] if shards else 0
         params = _tg_params(tg)
         for k, t in params.items():
