@@ -249,143 +249,6 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
 // ============================================================================
 // backward: D_i = rowsum(dO * O) precompute
 // ============================================================================
-// causal-only forward variant (EPL_ATTN_CAUSAL_V2): 64 q-rows per block
-// (2 waves, 128 threads).  Halves the per-block kv range so waves idle
-// through at most ~0.5 masked tiles (vs ~1.5 in the 128-row kernel) and
-// doubles the grid for a shorter dispatch tail; the cost is staging the
-// same K/V tiles into LDS twice as often.  UNMEASURED as of round 1 —
-// flag-gated off until a GPU A/B; numerics covered by
-// tests/test_attention_gpu.py when the flag is on.
-// ============================================================================
-__global__ __launch_bounds__(128) void attn_fwd_kernel_c64(
-    const short* __restrict__ q, const short* __restrict__ k,
-    const short* __restrict__ v, short* __restrict__ out,
-    float* __restrict__ lse, int64_t seq, float scale, int causal,
-    int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
-    int64_t o_sb, int64_t o_sh, int64_t o_ss) {
-  __shared__ short ldsV[64][72];       // V^T: [d][kv]
-  __shared__ short ldsK[64][72];       // K: [kv][d]
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;   // 0..1
-  const int hi = lane >> 5;
-  const int lq = lane & 31;
-  const int64_t bh = blockIdx.y;
-  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
-  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
-                            : (int64_t)blockIdx.x;
-  const int64_t q0_blk = qb * 64;
-  const int64_t q0 = q0_blk + wave * 32;
-  const bool active = q0 < seq;
-  const short* qp = q + boff;
-  const short* kp = k + boff;
-  const short* vp = v + boff;
-
-  const int64_t myq = q0 + lq;
-  const int64_t qrow = myq < seq ? myq : seq - 1;
-
-  bf16x8 qfrag[4];
-#pragma unroll
-  for (int c = 0; c < 4; ++c)
-    qfrag[c] = *reinterpret_cast<const bf16x8*>(
-        qp + qrow * in_ss + hi * 8 + 16 * c);
-
-  f32x16 ot0 = {}, ot1 = {};
-  float m = -1e30f, l = 0.f;
-
-  const int64_t blk_kv_end =
-      causal ? (q0_blk + 64 < seq ? q0_blk + 64 : seq) : seq;
-  const int stage_kv = threadIdx.x & 63;
-  const int stage_d0 = (threadIdx.x >> 6) * 8;   // 0 or 8
-
-  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 64) {
-    __syncthreads();
-    {
-      int64_t vrow = kv0 + stage_kv;
-      if (vrow >= seq) vrow = seq - 1;
-#pragma unroll
-      for (int h2 = 0; h2 < 4; ++h2) {     // 2 waves x 4 strips = 64 cols
-        const int sd = stage_d0 + h2 * 16;
-        bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-            vp + vrow * in_ss + sd);
-#pragma unroll
-        for (int j = 0; j < 8; ++j)
-          ldsV[sd + j][stage_kv] = vv[j];
-        *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
-            *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
-      }
-    }
-    __syncthreads();
-    if (active) {
-      const int64_t wave_kv_end = causal
-          ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
-#pragma unroll
-      for (int sub = 0; sub < 64; sub += 32) {
-        const int64_t kvs = kv0 + sub;
-        if (kvs >= wave_kv_end) break;
-        f32x16 st = {};
-#pragma unroll
-        for (int c = 0; c < 4; ++c) {
-          bf16x8 kfrag = *reinterpret_cast<const bf16x8*>(
-              &ldsK[sub + lq][hi * 8 + 16 * c]);
-          st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c],
-                                                       st, 0, 0, 0);
-        }
-        float s[16];
-        float tile_max = -1e30f;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          float sv = st[r] * scale;
-          const int64_t kvg = kvs + crow(r, hi);
-          if (kvg >= seq || (causal && kvg > myq)) sv = -1e30f;
-          s[r] = sv;
-          tile_max = fmaxf(tile_max, sv);
-        }
-        tile_max = fmaxf(tile_max, __shfl_xor(tile_max, 32, 64));
-        const float m_new = fmaxf(m, tile_max);
-        const float alpha = __expf(m - m_new);
-        float rowsum = 0.f;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          s[r] = __expf(s[r] - m_new);
-          rowsum += s[r];
-        }
-        rowsum += __shfl_xor(rowsum, 32, 64);
-        l = l * alpha + rowsum;
-        m = m_new;
-#pragma unroll
-        for (int r = 0; r < 16; ++r) {
-          ot0[r] *= alpha;
-          ot1[r] *= alpha;
-        }
-        bf16x8 pf0 = assemble_pfrag(&s[0]);
-        bf16x8 pf1 = assemble_pfrag(&s[8]);
-#pragma unroll
-        for (int kc = 0; kc < 2; ++kc) {
-          bf16x8 pf = kc == 0 ? pf0 : pf1;
-          bf16x8 vt0 = *reinterpret_cast<const bf16x8*>(
-              &ldsV[lq][sub + kc * 16 + hi * 8]);
-          bf16x8 vt1 = *reinterpret_cast<const bf16x8*>(
-              &ldsV[32 + lq][sub + kc * 16 + hi * 8]);
-          ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0,
-                                                        0, 0);
-          ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0,
-                                                        0, 0);
-        }
-      }
-    }
-  }
-
-  if (!active || myq >= seq) return;
-  const float inv_l = 1.f / l;
-  short* op = out + (bh / heads) * o_sb + (bh % heads) * o_sh + myq * o_ss;
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    op[crow(r, hi)] = (short)f2bf(ot0[r] * inv_l);
-    op[32 + crow(r, hi)] = (short)f2bf(ot1[r] * inv_l);
-  }
-  if (hi == 0) lse[bh * seq + myq] = m + __logf(l);
-}
-
 // ============================================================================
 __global__ void attn_bwd_prep_kernel(const short* __restrict__ dout,
                                      const short* __restrict__ out,
@@ -646,7 +509,7 @@ DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
   }
 }
 
-__global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
+__global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ dout, const float* __restrict__ lse,
     short* __restrict__ dv, int64_t seq, float scale, int causal,
@@ -743,7 +606,7 @@ __global__ __launch_bounds__(256) void attn_bwd_dv_kernel(
   }
 }
 
-__global__ __launch_bounds__(256) void attn_bwd_dk_kernel(
+__global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -896,7 +759,7 @@ DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
   }
 }
 
-__global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
+__global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -1005,148 +868,6 @@ __global__ __launch_bounds__(256) void attn_bwd_dq_kernel(
   }
 }
 
-// ============================================================================
-// causal-only dQ variant (EPL_ATTN_CAUSAL_V2): 64 q-rows / 2 waves per
-// block — same rationale as attn_fwd_kernel_c64.  UNMEASURED; flag-
-// gated off.  Staging covers 32 kv rows with 128 threads (2 rows per
-// thread).
-// ============================================================================
-__global__ __launch_bounds__(128) void attn_bwd_dq_kernel_c64(
-    const short* __restrict__ q, const short* __restrict__ k,
-    const short* __restrict__ v, const short* __restrict__ dout,
-    const float* __restrict__ lse, const float* __restrict__ delta,
-    short* __restrict__ dq, const short* __restrict__ out, int64_t seq,
-    float scale, int causal, int64_t heads, int64_t in_sb, int64_t in_sh,
-    int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss,
-    int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
-    int64_t o_ss) {
-  __shared__ short ldsK[32][72];
-  __shared__ short ldsVr[32][72];
-  const int lane = threadIdx.x & 63;
-  const int wave = threadIdx.x >> 6;
-  const int hi = lane >> 5;
-  const int lq = lane & 31;
-  const int64_t bh = blockIdx.y;
-  // causal: longest q-blocks first (see attn_fwd_kernel)
-  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
-                            : (int64_t)blockIdx.x;
-  const int64_t q0_blk = qb * 64;
-  const int64_t q0 = q0_blk + wave * 32;
-  const bool active = q0 < seq;
-  const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
-  const short* qp = q + boff;
-  const short* kp = k + boff;
-  const short* vp = v + boff;
-  const short* dop = dout + (bh / heads) * do_sb + (bh % heads) * do_sh;
-
-  const int64_t myq = q0 + lq;
-  const int64_t qrow = myq < seq ? myq : seq - 1;
-  const float mylse = lse[bh * seq + qrow];
-
-  const short* op_ = out + (bh / heads) * o_sb + (bh % heads) * o_sh;
-  bf16x8 qfrag[4], dofrag[4];
-  float dsum = 0.f;
-#pragma unroll
-  for (int c = 0; c < 4; ++c) {
-    qfrag[c] = *reinterpret_cast<const bf16x8*>(
-        qp + qrow * in_ss + hi * 8 + 16 * c);
-    dofrag[c] = *reinterpret_cast<const bf16x8*>(
-        dop + qrow * do_ss + hi * 8 + 16 * c);
-    bf16x8 of = *reinterpret_cast<const bf16x8*>(
-        op_ + qrow * o_ss + hi * 8 + 16 * c);
-#pragma unroll
-    for (int j = 0; j < 8; ++j)
-      dsum += bf2f(dofrag[c][j]) * bf2f(of[j]);
-  }
-  // D_i = rowsum(dO * O): this lane covers 32 of the 64 d-elements, the
-  // partner lane (^32) the rest
-  const float mydelta = dsum + __shfl_xor(dsum, 32, 64);
-  // publish delta for the dK kernel that follows on the same stream
-  if (myq < seq && hi == 0)
-    const_cast<float*>(delta)[bh * seq + myq] = mydelta;
-
-  f32x16 dqt0 = {}, dqt1 = {};
-
-  const int stage_row = threadIdx.x >> 3;
-  const int stage_seg = (threadIdx.x & 7) * 8;
-  const int64_t blk_kv_end =
-      causal ? (q0_blk + 64 < seq ? q0_blk + 64 : seq) : seq;
-  for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 32) {
-    __syncthreads();
-    {
-#pragma unroll
-      for (int r2 = 0; r2 < 2; ++r2) {   // 128 threads -> 2 rows each
-        const int srow = stage_row + 16 * r2;
-        int64_t kr = kv0 + srow;
-        if (kr >= seq) kr = seq - 1;   // dS there is 0
-        *reinterpret_cast<bf16x8*>(&ldsK[srow][stage_seg]) =
-            *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
-        *reinterpret_cast<bf16x8*>(&ldsVr[srow][stage_seg]) =
-            *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
-      }
-    }
-    __syncthreads();
-    const int64_t wave_kv_end = causal
-        ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
-    if (!active || kv0 >= wave_kv_end) continue;
-    f32x16 st = {};
-    f32x16 dpt = {};
-#pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      bf16x8 kf = *reinterpret_cast<const bf16x8*>(
-          &ldsK[lq][hi * 8 + 16 * c]);
-      bf16x8 vf = *reinterpret_cast<const bf16x8*>(
-          &ldsVr[lq][hi * 8 + 16 * c]);
-      st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kf, qfrag[c], st, 0, 0,
-                                                   0);
-      dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[c], dpt, 0,
-                                                    0, 0);
-    }
-    float ds[16];
-#pragma unroll
-    for (int r = 0; r < 16; ++r) {
-      const int64_t kvg = kv0 + crow(r, hi);
-      bool masked = myq >= seq || kvg >= seq || (causal && kvg > myq);
-      float pv = masked ? 0.f : __expf(st[r] * scale - mylse);
-      ds[r] = masked ? 0.f : pv * (dpt[r] - mydelta) * scale;
-    }
-    bf16x8 db0 = assemble_pfrag(&ds[0]);
-    bf16x8 db1 = assemble_pfrag(&ds[8]);
-#pragma unroll
-    for (int kc = 0; kc < 2; ++kc) {
-      bf16x8 kt0, kt1;
-#pragma unroll
-      for (int j = 0; j < 8; ++j) {
-        const int kr = kc * 16 + hi * 8 + j;
-        kt0[j] = ldsK[kr][lq];
-        kt1[j] = ldsK[kr][32 + lq];
-      }
-      bf16x8 db = kc == 0 ? db0 : db1;
-      dqt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt0, db, dqt0, 0, 0,
-                                                     0);
-      dqt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt1, db, dqt1, 0, 0,
-                                                     0);
-    }
-  }
-
-  if (!active || myq >= seq) return;
-  short* dqp = dq + (bh / heads) * g_sb + (bh % heads) * g_sh + myq * g_ss;
-#pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    dqp[crow(r, hi)] = (short)f2bf(dqt0[r]);
-    dqp[32 + crow(r, hi)] = (short)f2bf(dqt1[r]);
-  }
-}
-
-static bool attn_fwd_c64_enabled() {
-  static int v = -1;
-  if (v < 0) {
-    const char* e = getenv("EPL_ATTN_CAUSAL_V2");
-    v = (e && e[0] == '1') ? 1 : 0;
-  }
-  return v == 1;
-}
-
 }  // namespace
 
 extern "C" {
@@ -1155,17 +876,6 @@ void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
                   float* lse, int64_t batch_heads, int64_t seq, float scale,
                   bool causal, int64_t heads, const int64_t* in_strides,
                   const int64_t* o_strides, hipStream_t stream) {
-  if (causal && attn_fwd_c64_enabled()) {
-    dim3 cgrid((unsigned)((seq + 63) / 64), (unsigned)batch_heads);
-    hipLaunchKernelGGL(attn_fwd_kernel_c64, cgrid, dim3(128), 0, stream,
-                       reinterpret_cast<const short*>(q),
-                       reinterpret_cast<const short*>(k),
-                       reinterpret_cast<const short*>(v),
-                       reinterpret_cast<short*>(out), lse, seq, scale, 1,
-                       heads, in_strides[0], in_strides[1], in_strides[2],
-                       o_strides[0], o_strides[1], o_strides[2]);
-    return;
-  }
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
   hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
@@ -1200,21 +910,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                        in_strides[2], do_strides[0], do_strides[1],
                        do_strides[2], g_strides[0], g_strides[1],
                        g_strides[2]);
-    if (causal && attn_fwd_c64_enabled()) {
-      dim3 cgrid((unsigned)((seq + 63) / 64), (unsigned)batch_heads);
-      hipLaunchKernelGGL(attn_bwd_dq_kernel_c64, cgrid, dim3(128), 0,
-                         stream, reinterpret_cast<const short*>(q),
-                         reinterpret_cast<const short*>(k),
-                         reinterpret_cast<const short*>(v),
-                         reinterpret_cast<const short*>(dout), lse,
-                         delta_ws, reinterpret_cast<short*>(dq),
-                         reinterpret_cast<const short*>(out), seq, scale,
-                         1, heads, in_strides[0], in_strides[1],
-                         in_strides[2], do_strides[0], do_strides[1],
-                         do_strides[2], g_strides[0], g_strides[1],
-                         g_strides[2], o_strides[0], o_strides[1],
-                         o_strides[2]);
-    } else {
+    {
       hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
                          reinterpret_cast<const short*>(q),
                          reinterpret_cast<const short*>(k),

@@ -14,15 +14,13 @@ import torch
 import torch.nn as nn
 import torch.nn.functional as F
 
-# Hand-written attention kernels (tests/test_attention_gpu.py).  The
-# fully fused qkv path (kernels read the qkv buffer views, write d_qkv
-# slices directly) beats SDPA+assembly per-op on non-causal BERT shapes
-# (2225 vs 2263 us f+b; profiles/r01_fused_qkv_ab.txt) and ties it
-# end-to-end, so it is the DEFAULT for non-causal attention.  Causal
-# shapes stay on SDPA (AOTriton's causal kernel is still ahead, 1468 vs
-# 1751 us) until the triangle-skipping rework (NOTES.md item 1b).
-# EPL_NATIVE_ATTENTION: "auto" (default) = native for non-causal only;
-# "1" = native everywhere; "0" = SDPA everywhere.
+# Hand-written attention kernels (tests/test_attention_gpu.py) are the
+# DEFAULT for causal AND non-causal shapes: the round-2 split-phase
+# (branch-free bulk + masked diagonal) + 64-row bwd staging + occupancy
+# tuning beats AOTriton SDPA on both hot shapes — BERT b128 h16 s512
+# f+b 2049 vs 2392 us, GPT-2 b16 h25 s1024 causal f+b 1364 vs 1463 us
+# (profiles/r02_attention_ab.txt).
+# EPL_NATIVE_ATTENTION: "auto"/"1" (default) = native; "0" = SDPA.
 _NATIVE_ATTN = os.environ.get("EPL_NATIVE_ATTENTION", "auto")
 
 from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
@@ -84,8 +82,7 @@ class SelfAttention(nn.Module):
             flash_attention, qkv_flash_attention, qkv_native_ok)
         b, s, h = x.shape
         qkv = self.qkv(x).reshape(b, s, 3, self.num_heads, self.head_dim)
-        native_here = (_NATIVE_ATTN == "1"
-                       or (_NATIVE_ATTN == "auto" and not self.causal))
+        native_here = _NATIVE_ATTN in ("1", "auto")
         if (native_here and qkv_native_ok(qkv)
                 and not (self.dropout and self.training)):
             # fully fused: kernels read the qkv views and write d_qkv
@@ -101,11 +98,15 @@ class SelfAttention(nn.Module):
             k = k.transpose(1, 2)
             v = v.transpose(1, 2)
         if self.dropout and self.training:
+            from easyparallellibrary_amd.ops.attention import _warn_fallback
+            if native_here and q.is_cuda:
+                _warn_fallback("dropout_p={} (native kernels are "
+                               "dropout-free)".format(self.dropout))
             o = F.scaled_dot_product_attention(
                 q, k, v, is_causal=self.causal, dropout_p=self.dropout)
         else:
             o = flash_attention(q, k, v, causal=self.causal,
-                                allow_native=(_NATIVE_ATTN == "1"))
+                                allow_native=native_here)
         o = o.transpose(1, 2).reshape(b, s, h)
         return self.proj(o)
 

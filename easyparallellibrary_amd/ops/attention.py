@@ -14,6 +14,21 @@ import torch
 import torch.nn.functional as F
 
 from easyparallellibrary_amd.ops.dispatch import native_ext, use_native
+from easyparallellibrary_amd.utils.logging import get_logger
+
+logger = get_logger()
+_warned = set()
+
+
+def _warn_fallback(reason):
+    """LOUD (once per reason) SDPA fallback: silent multi-backend
+    dispatch on the hot path is exactly what the project brief
+    disallows — if a model leaves the native kernels, say so."""
+    if reason not in _warned:
+        _warned.add(reason)
+        logger.warning(
+            "native attention unavailable (%s): falling back to torch "
+            "SDPA for these shapes", reason)
 
 
 def _kernel_ok(t):
@@ -138,8 +153,9 @@ def flash_attention(q, k, v, causal=False, scale=None, allow_native=True):
     SDPA otherwise."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
-    if (allow_native and use_native(q) and q.dtype == torch.bfloat16
-            and q.shape[-1] == 64):
-        return _FlashAttention.apply(q, k, v, causal, scale)
+    if allow_native and use_native(q):
+        if q.dtype == torch.bfloat16 and q.shape[-1] == 64:
+            return _FlashAttention.apply(q, k, v, causal, scale)
+        _warn_fallback("dtype={} head_dim={}".format(q.dtype, q.shape[-1]))
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
                                           scale=scale)

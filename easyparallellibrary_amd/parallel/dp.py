@@ -225,6 +225,20 @@ class GradReducer:
         self._launched = [False] * len(self.buckets)
         self._decompress = []
         self.enabled = True          # pipeline sets False until last ubatch
+        # Optional per-bucket completion callback: called right after a
+        # bucket's collective is ENQUEUED (grads in [start,end) are final
+        # in the arena once that comm's stream reaches the op's end).
+        # Consumers (offload D2H overlap, PreferBackwardOptimizer eager
+        # apply) fence their own stream behind the comm stream via
+        # comm.join() inside the callback.  NOT fired when the bucket
+        # rode a compressed wire buffer (grads land in the arena only at
+        # finish()'s decompress) — callers must check
+        # `supports_bucket_callbacks`.
+        self.on_bucket_reduced = None
+
+    @property
+    def supports_bucket_callbacks(self):
+        return not self.compression
         self._world = (self.pool.comms[0].size if world_scale is None
                        else world_scale)
         self._hook_handles = []
@@ -262,6 +276,8 @@ class GradReducer:
             # decompress on the comm's completion order: since async ops on
             # one comm serialize on its stream, enqueue the copy after join
             self._decompress.append((bi, wire, buf))
+        elif self.on_bucket_reduced is not None:
+            self.on_bucket_reduced(bi, start, end, comm, owner)
 
     def _on_grad_ready(self, p):
         if self.group.grad_copy_first:

@@ -58,6 +58,49 @@ class CPUOffloadAdamW:
             # free the device-side fp32 master: CPU owns the truth now
             if g.master_arena is not g.param_arena:
                 g.master_arena = g.state["master_cpu"]
+        # per-bucket D2H overlap state (attach_reducers)
+        self._events = {}       # group idx -> [(lo, hi, event)]
+        self._n_buckets = {}    # group idx -> bucket count
+
+    def attach_reducers(self, reducers, groups_of_reducers):
+        """Overlap grad D2H with the tail of backward: as each bucket's
+        allreduce is enqueued (grads final in the arena on the comm
+        stream), copy that span to pinned host memory on the d2h stream
+        (NOTES.md round-2 design; closes the ~0.56 s D2H bound of the
+        post-backward copy).  Gated to DP reducers without ZeRO shard
+        owners or wire compression."""
+        if self._d2h is None:
+            return 0
+        attached = 0
+        gi_of = {id(g): i for i, g in enumerate(self.groups)}
+        for red, fg in zip(reducers, groups_of_reducers):
+            gi = gi_of.get(id(fg))
+            if gi is None or red.shard_owners \
+                    or not red.supports_bucket_callbacks \
+                    or not fg.grad_arena.is_cuda:
+                continue
+            self._n_buckets[gi] = len(red.buckets)
+            red.on_bucket_reduced = self._make_cb(gi)
+            attached += 1
+        return attached
+
+    def _make_cb(self, gi):
+        def cb(bi, lo, hi, comm, owner):
+            g = self.groups[gi]
+            gc = g.state["grad_cpu"]
+            # fence the d2h stream behind BOTH the producing compute
+            # stream (grads written during backward; covers the size-1
+            # LocalCommunicator whose join is a no-op) and the comm
+            # stream (reduced values)
+            self._d2h.wait_stream(torch.cuda.current_stream())
+            with torch.cuda.stream(self._d2h):
+                comm.join()
+                gc[lo:hi].copy_(g.grad_arena[lo:hi].to(torch.float32),
+                                non_blocking=True)
+                ev = torch.cuda.Event()
+                ev.record(self._d2h)
+            self._events.setdefault(gi, []).append((lo, hi, ev))
+        return cb
 
     def _chunk_update(self, w, gc, m, v, grad_scale):
         """One AdamW update on a CPU arena slice."""

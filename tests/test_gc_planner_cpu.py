@@ -170,3 +170,39 @@ def test_gc_auto_never_wraps_fused_elementwise():
     from easyparallellibrary_amd.ops.bias_gelu import FusedBiasGelu
     for _, _, child in hits:
         assert not isinstance(child, (FusedLayerNorm, FusedBiasGelu))
+
+
+def test_auto_stage_repeated_block_policy():
+    """Auto-stage must cut at repeated-block boundaries by STRUCTURE
+    (reference planner.py:66-112 policy order), keeping the embedding
+    prefix in stage 0 and the head in the last stage."""
+    import torch.nn as nn
+    from easyparallellibrary_amd.parallel.planner import AutoStageGenerator
+
+    class Block(nn.Module):
+        def __init__(self, h):
+            super().__init__()
+            self.fc = nn.Linear(h, h)
+
+        def forward(self, x):
+            return self.fc(x)
+
+    h = 8
+    spine = nn.Sequential(
+        nn.Embedding(1000, h),            # heavy prefix
+        *[Block(h) for _ in range(6)],
+        nn.Linear(h, 1000),               # heavy head
+    )
+    stages = AutoStageGenerator(spine, 2).search()
+    assert stages is not None and len(stages) == 2
+    # every cut lands at a Block boundary: stage 0 starts with the
+    # embedding, stage 1 starts with a Block (never mid-prefix/suffix)
+    assert isinstance(stages[0][0], nn.Embedding)
+    assert type(stages[1][0]).__name__ == "Block"
+    assert isinstance(stages[1][-1], nn.Linear)
+    assert sum(len(s) for s in stages) == len(spine)
+    # the heavy embedding/head did NOT drag blocks unevenly: both stages
+    # hold at least one Block
+    n_blocks = [sum(1 for m in s if type(m).__name__ == "Block")
+                for s in stages]
+    assert all(n >= 1 for n in n_blocks), n_blocks
