@@ -235,10 +235,6 @@ class GradReducer:
         # finish()'s decompress) — callers must check
         # `supports_bucket_callbacks`.
         self.on_bucket_reduced = None
-
-    @property
-    def supports_bucket_callbacks(self):
-        return not self.compression
         self._world = (self.pool.comms[0].size if world_scale is None
                        else world_scale)
         self._hook_handles = []
@@ -246,6 +242,10 @@ class GradReducer:
             for p in group.ordered:
                 h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
                 self._hook_handles.append(h)
+
+    @property
+    def supports_bucket_callbacks(self):
+        return not self.compression
 
     @property
     def op(self):

@@ -325,6 +325,11 @@ class Engine:
             if optimizer != "adamw":
                 raise NotImplementedError("offload supports adamw")
             self.optimizer = CPUOffloadAdamW(self.flat_groups, **okw)
+            n = self.optimizer.attach_reducers(
+                self.reducers, [info["fg"] for info in self._group_infos])
+            if n:
+                logger.info("offload: per-bucket D2H overlap on %d "
+                            "reducer(s)", n)
         else:
             self.optimizer = OPTIMIZERS[optimizer](self.flat_groups, **okw)
         for sp, bcomm, granks in _pending_sparse:

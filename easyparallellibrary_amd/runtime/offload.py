@@ -127,7 +127,7 @@ class CPUOffloadAdamW:
 
     def step(self, grad_scale=1.0):
         self.step_count += 1
-        for g in self.groups:
+        for gi, g in enumerate(self.groups):
             gc = g.state["grad_cpu"]
             m, v = g.state["exp_avg"], g.state["exp_avg_sq"]
             w = g.state["master_cpu"]
@@ -141,20 +141,32 @@ class CPUOffloadAdamW:
                     g.param_arena[lo:hi].copy_(
                         w[lo:hi].to(g.param_arena.dtype))
                 continue
-            # pre-issue every D2H grad chunk on the copy stream; fence
-            # each with an event so the CPU starts as soon as ITS chunk
-            # lands while later chunks are still in flight
-            self._d2h.wait_stream(torch.cuda.current_stream())
-            events = []
-            with torch.cuda.stream(self._d2h):
-                for lo, hi in spans:
-                    gc[lo:hi].copy_(g.grad_arena[lo:hi].to(torch.float32),
-                                    non_blocking=True)
-                    ev = torch.cuda.Event()
-                    ev.record(self._d2h)
-                    events.append(ev)
-            for (lo, hi), ev in zip(spans, events):
-                ev.synchronize()
+            # D2H: bucket callbacks already copied every span DURING
+            # backward (one event per bucket, arena-order) — reuse them;
+            # otherwise pre-issue every grad chunk on the copy stream
+            # now.  Either way: one event per copied span so the CPU
+            # starts as soon as ITS chunk lands.
+            bucket_evs = self._events.pop(gi, [])
+            if len(bucket_evs) == self._n_buckets.get(gi, -1):
+                copies = sorted(bucket_evs)  # (lo, hi, event) by lo
+            else:
+                self._d2h.wait_stream(torch.cuda.current_stream())
+                copies = []
+                with torch.cuda.stream(self._d2h):
+                    for lo, hi in spans:
+                        gc[lo:hi].copy_(
+                            g.grad_arena[lo:hi].to(torch.float32),
+                            non_blocking=True)
+                        ev = torch.cuda.Event()
+                        ev.record(self._d2h)
+                        copies.append((lo, hi, ev))
+            ci = 0
+            done = 0   # arena prefix whose events are synchronized
+            for lo, hi in spans:
+                while done < hi and ci < len(copies):
+                    copies[ci][2].synchronize()
+                    done = copies[ci][1]
+                    ci += 1
                 self._chunk_update(w[lo:hi], gc[lo:hi], m[lo:hi],
                                    v[lo:hi], grad_scale)
                 # direct pinned-fp32 -> device-bf16 copy: the transfer
@@ -166,6 +178,7 @@ class CPUOffloadAdamW:
             torch.cuda.current_stream().wait_stream(self._h2d)
 
     def zero_grad(self):
+        self._events.clear()   # drop stale bucket copies (skipped step)
         for g in self.groups:
             g.zero_grad()
 
