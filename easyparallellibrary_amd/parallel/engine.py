@@ -354,6 +354,33 @@ class Engine:
                 PipelineRuntime)
             self.pipeline = PipelineRuntime(self)
 
+        # ---- PreferBackwardOptimizer: bucket-wise eager apply ---------------
+        # (reference scheduler.py:87-116).  Gated to the plain fused-
+        # AdamW path: eager applies cannot precede a found_inf check or
+        # global grad-norm clip, and sharded owners/compression change
+        # when a bucket's grads are final in the arena.
+        from easyparallellibrary_amd.runtime.optim import FusedAdamW
+        self._pbo_eager = False
+        if (self.pipeline is not None
+                and self.config.pipeline.strategy
+                == constant.SCHEDULER_PREFER_BACKWARD_OPT
+                and type(self.optimizer) is FusedAdamW
+                and not self.amp.enabled
+                and not self.config.optimizer.max_grad_norm
+                and not zero_level and not self.config.offload.level
+                and not self.sparse_handlers
+                and all(r.supports_bucket_callbacks and not r.shard_owners
+                        for r in self.reducers)):
+            def _mk(fg):
+                def cb(bi, lo, hi, comm, owner):
+                    self.optimizer.eager_apply(fg, lo, hi, comm)
+                return cb
+            for red, info in zip(self.reducers, self._group_infos):
+                red.on_bucket_reduced = _mk(info["fg"])
+            self._pbo_eager = True
+            logger.info("prefer_backward_optimizer: eager bucket apply "
+                        "on %d reducer(s)", len(self.reducers))
+
         self.global_step = 0
         self._accum_count = 0
         logger.info(
@@ -413,6 +440,10 @@ class Engine:
             return loss.detach()
         if self._accum_count == 0:
             self.zero_grad()
+        if self._pbo_eager:
+            # grad scale is known before backward (amp off, no manual
+            # accumulation under pipeline): buckets apply as they reduce
+            self.optimizer.begin_eager(float(self.num_micro_batch))
         if self.pipeline is not None:
             loss = self.pipeline.run(inputs, targets)
         else:

@@ -30,11 +30,70 @@ class FusedAdamW:
         self.eps = eps
         self.weight_decay = weight_decay
         self.step_count = 0
+        self._eager = False
+        self._eager_scale = 1.0
+        self._apply_stream = None
         for g in self.groups:
             g.state["exp_avg"] = torch.zeros_like(g.master_arena)
             g.state["exp_avg_sq"] = torch.zeros_like(g.master_arena)
 
+    # ---- PreferBackwardOptimizer eager apply --------------------------------
+    # (reference: strategies/scheduler.py:87-116 — the apply of a
+    # parameter group starts as soon as its gradients are final, instead
+    # of after the whole backward; here "group" = an arena bucket, and
+    # the apply runs on a side stream fenced behind that bucket's comm.)
+    def begin_eager(self, grad_scale):
+        self.step_count += 1
+        self._eager_scale = float(grad_scale)
+        self._eager = True
+        if self._apply_stream is None and torch.cuda.is_available():
+            self._apply_stream = torch.cuda.Stream()
+
+    def eager_apply(self, g, lo, hi, comm):
+        """Apply AdamW to the arena slice [lo, hi) whose grads just
+        became final (bucket reduced).  Called from the reducer's
+        bucket callback during the final backward."""
+        inv = 1.0 / self._eager_scale
+        if use_native(g.master_arena):
+            with torch.cuda.stream(self._apply_stream):
+                comm.join()   # fence apply stream behind the comm stream
+                native_ext().fused_adamw(
+                    g.master_arena[lo:hi],
+                    g.param_arena[lo:hi]
+                    if g.param_arena.dtype == torch.bfloat16 else None,
+                    g.grad_arena[lo:hi], g.state["exp_avg"][lo:hi],
+                    g.state["exp_avg_sq"][lo:hi],
+                    self.lr, self.beta1, self.beta2, self.eps,
+                    self.weight_decay, self.step_count, inv)
+        else:
+            self._step_torch_slice(g, lo, hi, inv)
+
+    def _step_torch_slice(self, g, lo, hi, inv_scale):
+        grad = g.grad_arena[lo:hi].to(torch.float32)
+        if inv_scale != 1.0:
+            grad = grad * inv_scale
+        m = g.state["exp_avg"][lo:hi]
+        v = g.state["exp_avg_sq"][lo:hi]
+        m.mul_(self.beta1).add_(grad, alpha=1 - self.beta1)
+        v.mul_(self.beta2).addcmul_(grad, grad, value=1 - self.beta2)
+        bc1 = 1 - self.beta1 ** self.step_count
+        bc2 = 1 - self.beta2 ** self.step_count
+        denom = (v / bc2).sqrt_().add_(self.eps)
+        update = (m / bc1) / denom
+        update.add_(g.master_arena[lo:hi], alpha=self.weight_decay)
+        g.master_arena[lo:hi].add_(update, alpha=-self.lr)
+        if g.master_arena is not g.param_arena:
+            g.param_arena[lo:hi].copy_(
+                g.master_arena[lo:hi].to(g.param_arena.dtype))
+
     def step(self, grad_scale=1.0):
+        if self._eager:
+            # buckets were applied during backward; just fence the next
+            # forward behind the apply stream
+            self._eager = False
+            if self._apply_stream is not None:
+                torch.cuda.current_stream().wait_stream(self._apply_stream)
+            return
         self.step_count += 1
         inv_scale = 1.0 / grad_scale
         from easyparallellibrary_amd.env import Env

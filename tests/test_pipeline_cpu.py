@@ -138,3 +138,36 @@ def test_pp2_with_gradient_checkpoint():
     gc = run_multiprocess(_pipeline_gc_worker, world=2, args=("auto",))
     assert all(abs(a - b) < 1e-5 for a, b in zip(plain[1], gc[1])), (
         plain[1], gc[1])
+
+
+def test_pp2_prefer_backward_optimizer_matches_serial():
+    """PBO is a REAL schedule now: buckets apply eagerly during the
+    final backward via reducer callbacks (reference scheduler.py:87-116)
+    and the trajectory must still match serial exactly."""
+    serial = run_multiprocess(_serial_worker, world=1, args=(4,))[0]
+    pp = run_multiprocess(_pipeline_worker, world=2,
+                          args=("prefer_backward_optimizer", 4))
+    assert pp[0][0] is None
+    assert all(abs(a - b) < 1e-5 for a, b in zip(serial, pp[1])), (
+        serial, pp[1])
+
+
+def _pbo_eager_active_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({
+        "pipeline.num_micro_batch": 4,
+        "pipeline.strategy": "prefer_backward_optimizer",
+    }))
+    model = _build(world_stages=world)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(33)
+    engine.train_step(torch.randn(8, 8), torch.randn(8, 4))
+    return engine._pbo_eager, engine.optimizer.step_count
+
+
+def test_pbo_eager_path_is_active():
+    res = run_multiprocess(_pbo_eager_active_worker, world=2)
+    for eager, steps in res:
+        assert eager is True
+        assert steps == 1
