@@ -351,3 +351,17 @@ def destroy_all_communicators():
         if isinstance(comm, RcclCommunicator):
             comm.destroy()
         del _REGISTRY[name]
+
+
+def destroy_namespace(prefix):
+    """Destroy every communicator whose name starts with ``prefix`` —
+    Engine.close() uses this to reclaim RCCL comms + HIP streams (the
+    registry otherwise grows for the life of the process)."""
+    n = 0
+    for name, comm in list(_REGISTRY.items()):
+        if name.startswith(prefix):
+            if isinstance(comm, RcclCommunicator):
+                comm.destroy()
+            del _REGISTRY[name]
+            n += 1
+    return n

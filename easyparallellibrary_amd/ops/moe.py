@@ -19,6 +19,11 @@ import torch.nn.functional as F
 from easyparallellibrary_amd.comm import functional
 
 
+def _moe_bmm():
+    import os
+    return os.environ.get("EPL_MOE_BMM", "1") == "1"
+
+
 class ExpertParallelMLP(nn.Module):
     """num_experts split across the comm group; each rank holds
     num_experts // world experts (reference tests use the same layout,
@@ -117,15 +122,19 @@ class ExpertParallelMLP(nn.Module):
         d = d.reshape(self.world, self.local_experts, capacity, self.hidden)
         d = d.transpose(0, 1).reshape(self.local_experts,
                                       self.world * capacity, self.hidden)
-        # per-expert 2D GEMMs instead of torch.bmm: the bf16 batched-GEMM
-        # BACKWARD ([E,c,4h]x[E,4h,h]) memory-faults in hipBLASLt on
-        # ROCm 7.0 / gfx950 (isolated in tests/moe_bisect_gpu.py bmm2);
-        # 2D mms hit the well-tested non-batched path.
-        outs = []
-        for e in range(self.local_experts):
-            he = F.gelu(d[e] @ self.w1[e])
-            outs.append(he @ self.w2[e])
-        h = torch.stack(outs)
+        # expert FFN: batched GEMM (one hipBLASLt grouped call per
+        # matmul) by default; EPL_MOE_BMM=0 falls back to the per-expert
+        # 2D-GEMM loop that worked around a bf16 batched-GEMM BACKWARD
+        # memory fault on ROCm 7.0/gfx950 (tests/moe_bisect_gpu.py bmm2;
+        # re-tested fixed on the round-2 ROCm 7.0.2 stack)
+        if _moe_bmm():
+            h = torch.bmm(F.gelu(torch.bmm(d, self.w1)), self.w2)
+        else:
+            outs = []
+            for e in range(self.local_experts):
+                he = F.gelu(d[e] @ self.w1[e])
+                outs.append(he @ self.w2[e])
+            h = torch.stack(outs)
         h = h.reshape(self.local_experts, self.world, capacity, self.hidden)
         h = h.transpose(0, 1).reshape(
             self.world, self.local_experts * capacity, self.hidden)

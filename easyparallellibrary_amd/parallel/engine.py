@@ -572,9 +572,12 @@ class Engine:
     def eval_step(self, inputs):
         """Forward in eval mode without grad (dropout off, BatchNorm uses
         running stats and does not update them).  Under pipeline
-        parallelism this runs ONLY this rank's stage (feed it the
-        previous stage's activations or run evaluation data-parallel on
-        a non-pipelined engine)."""
+        parallelism this runs the PIPELINED forward chain — collective
+        across the replica's stages; stage 0 consumes ``inputs``, the
+        last stage returns the outputs, other stages return None
+        (PipelineRuntime.run_eval)."""
+        if self.pipeline is not None:
+            return self.pipeline.run_eval(inputs)
         was_training = self._runnable.training
         self._runnable.eval()
         try:
@@ -682,6 +685,19 @@ class Engine:
         return slice_files(files, self.num_replicas, self.replica_id,
                            unbalanced=self.config.io.unbalanced_io_slicing,
                            drop_last=self.config.io.drop_last_files)
+
+    # ---- teardown ------------------------------------------------------------
+    def close(self):
+        """Release this engine's resources: autograd hooks and every
+        communicator created under its namespace (RCCL comms + HIP
+        streams are destroyed; the global registry shrinks).  The engine
+        must not be used afterwards.  COLLECTIVE-free."""
+        for r in self.reducers:
+            r.remove_hooks()
+        from easyparallellibrary_amd.comm.backend import destroy_namespace
+        n = destroy_namespace(self._ns + "_")
+        logger.info("engine %s closed: %d communicator(s) destroyed",
+                    self._ns, n)
 
     # ---- checkpoint ----------------------------------------------------------
     def save_checkpoint(self, path, save_optimizer=True):

@@ -294,6 +294,77 @@ class PipelineRuntime:
         ])
         return t
 
+    # ---- pipelined evaluation -------------------------------------------------
+    def _send_eval(self, t):
+        """Eval-time forward send with a per-call shape handshake (eval
+        batches may differ from training shapes; eval is not hot)."""
+        t = t.contiguous()
+        targets = (self._links_next if self.mixed
+                   else [(self.comm, self._g(self.next_rank),
+                          self.next_rank)])
+        if self.mixed and self._next_is_narrow and self.pos != 0:
+            return  # replicated output: position 0 alone sends
+        for comm, pg, peer in targets:
+            meta = [tuple(t.shape), str(t.dtype)]
+            dist.send_object_list(meta, dst=peer,
+                                  group=self.engine._control_group)
+            comm.send(t, pg)
+
+    def _recv_eval(self):
+        if self.mixed:
+            comm, pg, peer = self._links_prev[0]
+        else:
+            comm, pg, peer = (self.comm, self._g(self.prev_rank),
+                              self.prev_rank)
+        meta = [None, None]
+        dist.recv_object_list(meta, src=peer,
+                              group=self.engine._control_group)
+        shape = tuple(meta[0])
+        dtype = getattr(torch, meta[1].replace("torch.", ""))
+        t = torch.empty(shape, dtype=dtype, device=self.device)
+        comm.recv(t, pg)
+        return t
+
+    def run_eval(self, inputs, num_micro_batch=None):
+        """Forward-only pipelined evaluation (closes the reference's
+        eval story under PP — hooks.py:915-933 has the chief evaluate
+        while workers barrier; here the whole chain evaluates).
+
+        COLLECTIVE across the replica's pipeline chain: every rank must
+        call it with the same ``num_micro_batch`` (defaults to the
+        training value).  ``inputs`` is consumed on stage 0; the LAST
+        stage returns the concatenated outputs, every other stage
+        returns None.  Runs in eval mode (dropout off, BatchNorm frozen)
+        without grad.
+        """
+        M = num_micro_batch or self.engine.num_micro_batch
+        mod = self.engine.stage_module
+        was_training = mod.training
+        mod.eval()
+        outs = []
+        try:
+            with torch.no_grad():
+                chunks = (torch.chunk(inputs, M, dim=0)
+                          if self.s == 0 else [None] * M)
+                if self.s == 0 and len(chunks) < M:
+                    raise ValueError(
+                        "eval batch dim {} < num_micro_batch {}".format(
+                            inputs.shape[0], M))
+                for i in range(M):
+                    if self.s == 0:
+                        x = chunks[i].to(self.device)
+                    else:
+                        x = self._recv_eval()
+                    with self.engine.amp.autocast():
+                        out = mod(x)
+                    if self.s < self.S - 1:
+                        self._send_eval(out)
+                    else:
+                        outs.append(out)
+        finally:
+            mod.train(was_training)
+        return torch.cat(outs, dim=0) if outs else None
+
     # ---- forward/backward wrappers -------------------------------------------
     def _forward(self, inp, target):
         with self.engine.amp.autocast():

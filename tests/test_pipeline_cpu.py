@@ -171,3 +171,59 @@ def test_pbo_eager_path_is_active():
     for eager, steps in res:
         assert eager is True
         assert steps == 1
+
+
+def _eval_pp_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"pipeline.num_micro_batch": 4}))
+    model = _build(world_stages=world)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(33)
+    x = torch.randn(8, 8)
+    out = engine.eval_step(x)
+    # a second eval with a DIFFERENT batch size (shape handshake is
+    # per-call, not cached from training)
+    out2 = engine.eval_step(torch.randn(4, 8))
+    engine.close()
+    return (None if out is None else out.clone(),
+            None if out2 is None else out2.shape)
+
+
+def _eval_serial_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    epl.init()
+    model = _build(world_stages=1)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss())
+    torch.manual_seed(33)
+    x = torch.randn(8, 8)
+    return engine.eval_step(x).clone()
+
+
+def test_pipelined_eval_matches_serial():
+    """run_eval: forward-only pipelined evaluation — last stage output
+    must equal the serial model's eval output; works with eval batch
+    shapes different from training (fresh shape handshake)."""
+    serial = run_multiprocess(_eval_serial_worker, world=1)[0]
+    pp = run_multiprocess(_eval_pp_worker, world=2)
+    assert pp[0][0] is None             # stage 0 yields no output
+    assert torch.allclose(pp[1][0], serial, atol=1e-6)
+    assert tuple(pp[1][1]) == (4, 4)    # different eval batch worked
+
+
+def _close_worker(rank, world):
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.comm import backend
+    epl.init()
+    with epl.replicate(device_count=1):
+        model = nn.Linear(4, 2)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss())
+    before = len(backend._REGISTRY)
+    engine.close()
+    after = len(backend._REGISTRY)
+    return before, after
+
+
+def test_engine_close_shrinks_registry():
+    before, after = run_multiprocess(_close_worker, world=1)[0]
+    assert before > 0 and after == 0, (before, after)
