@@ -24,6 +24,32 @@ def _moe_bmm():
     return os.environ.get("EPL_MOE_BMM", "1") == "1"
 
 
+class _BatchedExpertLinear(torch.autograd.Function):
+    """Per-expert batched GEMM with a hand-written backward that issues
+    ONLY plain (NN, contiguous) torch.bmm calls.
+
+    torch.bmm's own backward runs hipBLASLt strided-batched GEMMs on
+    TRANSPOSED VIEWS, and that configuration memory-faults on
+    ROCm 7.0.x/gfx950 (still reproduced on 7.0.2 —
+    gpurun_out/r2_moe_bmm2.txt, repro tests/moe_bisect_gpu.py bmm2).
+    Materializing the transposes and calling NN bmm dodges the faulting
+    kernel while keeping one grouped GEMM launch per matmul instead of
+    the per-expert 2D loop."""
+
+    @staticmethod
+    def forward(ctx, x, w):
+        ctx.save_for_backward(x, w)
+        return torch.bmm(x, w)
+
+    @staticmethod
+    def backward(ctx, go):
+        x, w = ctx.saved_tensors
+        go = go.contiguous()
+        gx = torch.bmm(go, w.transpose(1, 2).contiguous())
+        gw = torch.bmm(x.transpose(1, 2).contiguous(), go)
+        return gx, gw
+
+
 class ExpertParallelMLP(nn.Module):
     """num_experts split across the comm group; each rank holds
     num_experts // world experts (reference tests use the same layout,
@@ -128,7 +154,8 @@ class ExpertParallelMLP(nn.Module):
         # memory fault on ROCm 7.0/gfx950 (tests/moe_bisect_gpu.py bmm2;
         # re-tested fixed on the round-2 ROCm 7.0.2 stack)
         if _moe_bmm():
-            h = torch.bmm(F.gelu(torch.bmm(d, self.w1)), self.w2)
+            he = F.gelu(_BatchedExpertLinear.apply(d.contiguous(), self.w1))
+            h = _BatchedExpertLinear.apply(he, self.w2)
         else:
             outs = []
             for e in range(self.local_experts):

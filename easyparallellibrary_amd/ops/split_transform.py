@@ -69,12 +69,21 @@ _ELEMENTWISE_SAFE = (nn.ReLU, nn.GELU, nn.SiLU, nn.Tanh, nn.Sigmoid,
 class PairedColumnLinear(nn.Module):
     """Entry half of an auto-paired Megatron block: the 'f' copy operator
     (identity forward, all-reduce of input grads backward) in front of a
-    column-parallel linear, so the pair's backward matches ops/tp_mlp."""
+    column-parallel linear, so the pair's backward matches ops/tp_mlp.
+
+    The column linear is NOT a registered submodule — its parameters are
+    registered directly on this wrapper under the ORIGINAL Linear's
+    names (weight/bias), so state-dict keys are identical with
+    auto_pair_sequential on or off and checkpoints stay portable across
+    the flag (advisor finding r1)."""
 
     def __init__(self, col, comm):
         super().__init__()
-        self.col = col
+        self.__dict__["col"] = col   # plain attr: keep it out of the tree
         self.comm = comm
+        self.weight = col.weight
+        if col.bias is not None:
+            self.bias = col.bias
 
     def forward(self, x):
         return self.col(functional.copy_to_group(x, self.comm))

@@ -52,16 +52,19 @@ def _save_tg_bucketed(tg, pos, nshards, path, bucket_bytes):
         parts += 1
         cur, cur_bytes = {}, 0
 
+    shard_dims = {}
     for k, t in params.items():
         key = "{}.shard{}".format(k, pos) if tg.is_split else k
         cur[key] = t.detach().to("cpu")
+        if tg.is_split and getattr(t, "_epl_shard_dim", None) is not None:
+            shard_dims[k] = int(t._epl_shard_dim)
         cur_bytes += t.numel() * t.element_size()
         if cur_bytes >= bucket_bytes:
             flush()
     flush()
     torch.save(
         {"taskgraph": tg.index, "position": pos, "is_split": tg.is_split,
-         "nshards": nshards, "nparts": parts},
+         "nshards": nshards, "nparts": parts, "shard_dims": shard_dims},
         os.path.join(path, "tg{}_pos{}.pt".format(tg.index, pos)))
 
 
@@ -234,32 +237,48 @@ class ShardingLoader:
             if saved_n == nshards:
                 key = "{}.shard{}".format(base, pos)
                 blob = shards[pos]["params"]
-                if key in blob:
+                if key in blob and tuple(blob[key].shape) == tuple(t.shape):
                     t.data.copy_(blob[key].to(t.device, t.dtype))
                     continue
-                if strict:
+                if key not in blob and strict:
                     raise KeyError(key)
+                # shape mismatch: same shard count but a different shard
+                # LAYOUT (e.g. column-sharded checkpoint restoring into a
+                # row-parallel module under auto_pair_sequential) — fall
+                # through to the full reshard path below
+            # reshard: concatenate all shards on the SAVED shard dim,
+            # re-slice on the target's shard dim with the
+            # remainder-to-shard-0 policy
+            pieces = []
+            saved_dim = None
+            for s in shards:
+                key = "{}.shard{}".format(base, s["position"])
+                if key in s["params"]:
+                    pieces.append(s["params"][key])
+                    if saved_dim is None:
+                        saved_dim = s.get("shard_dims", {}).get(base)
+            if not pieces:
+                if strict:
+                    raise KeyError(base)
+                continue
+            tdim = getattr(t, "_epl_shard_dim", None)
+            tdim = 0 if tdim is None else tdim
+            if saved_dim is None:
+                # the saved tensor was REPLICATED across positions (e.g.
+                # a row-parallel bias): any piece is the full tensor
+                full = pieces[0]
             else:
-                # reshard: concatenate all shards on dim 0, re-slice with
-                # the remainder-to-shard-0 policy
-                pieces = []
-                for s in shards:
-                    key = "{}.shard{}".format(base, s["position"])
-                    if key in s["params"]:
-                        pieces.append(s["params"][key])
-                if not pieces:
-                    if strict:
-                        raise KeyError(base)
-                    continue
-                dim = getattr(t, "_epl_shard_dim", 0)
-                full = torch.cat(pieces, dim=dim)
-                from easyparallellibrary_amd.ops.distributed_dense import (
-                    shard_offset, shard_size)
-                total = full.shape[dim]
-                lo = shard_offset(total, nshards, pos)
-                n = shard_size(total, nshards, pos)
-                t.data.copy_(
-                    full.narrow(dim, lo, n).to(t.device, t.dtype))
+                full = torch.cat(pieces, dim=saved_dim)
+            if tuple(full.shape) == tuple(t.shape):
+                t.data.copy_(full.to(t.device, t.dtype))
+                continue
+            from easyparallellibrary_amd.ops.distributed_dense import (
+                shard_offset, shard_size)
+            total = full.shape[tdim]
+            lo = shard_offset(total, nshards, pos)
+            n = shard_size(total, nshards, pos)
+            t.data.copy_(
+                full.narrow(tdim, lo, n).to(t.device, t.dtype))
 
 
 def load_checkpoint(engine, path, load_optimizer=True, assign_map=None,

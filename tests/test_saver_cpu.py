@@ -368,3 +368,105 @@ def test_bucketed_serial_checkpoint_roundtrip():
     for cont, resumed, nparts in results:
         assert nparts == 4, nparts  # 2 Linears x (weight + bias)
         assert all(abs(a - b) < 1e-6 for a, b in zip(cont, resumed))
+
+
+def _auto_pair_model(world):
+    import easyparallellibrary_amd as epl
+    torch.manual_seed(61)
+    with epl.replicate(world, name="trunk"):
+        emb = nn.Linear(8, 16)
+    with epl.split(world, name="tp"):
+        mlp = nn.Sequential(nn.Linear(16, 32), nn.GELU(),
+                            nn.Linear(32, 16))
+
+    class M(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.emb, self.mlp = emb, mlp
+            self.head = nn.Linear(16, 4)
+
+        def forward(self, x):
+            return self.head(self.mlp(self.emb(x)))
+
+    return M()
+
+
+def _auto_pair_save_worker(rank, world, path):
+    """auto_pair_sequential checkpoints (advisor finding r1):
+    PairedColumnLinear registers its params under the original Linear
+    names, shard dims are recorded at save, and the loader re-shards on
+    mismatch — so a pairing-ON checkpoint both roundtrips and restores
+    into a serial (1-way) engine."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.env import Env
+    from easyparallellibrary_amd.parallel import hooks
+
+    def build():
+        epl.init(epl.Config({
+            "cluster.colocate_split_and_replicate": True,
+            "auto.auto_pair_sequential": True}))
+        return epl.Engine(_auto_pair_model(world), loss_fn=nn.MSELoss(),
+                          optimizer="adamw", lr=1e-2)
+
+    engine = build()
+    torch.manual_seed(62)
+    x = torch.randn(2, 8, 8)
+    y = torch.randn(2, 8, 4)
+    engine.train_step(x, y)
+    engine.save_checkpoint(path, save_optimizer=False)
+    cont = float(engine.train_step(x, y))
+    # key names: identical to the unsharded model (no '.col.' nesting)
+    keys = set(engine.model.mlp.state_dict().keys())
+    assert keys == {"0.weight", "0.bias", "2.weight", "2.bias"}, keys
+
+    hooks.remove_hooks()
+    Env._instance = None
+    engine2 = build()
+    engine2.load_checkpoint(path, load_optimizer=False)
+    resumed = float(engine2.train_step(x, y))
+    return cont, resumed
+
+
+def test_auto_pair_checkpoint_roundtrip():
+    res = run_multiprocess(_auto_pair_save_worker, world=2,
+                           args=(CKPT + "_pair",))
+    for cont, resumed in res:
+        assert abs(cont - resumed) < 5e-5, (cont, resumed)
+
+
+def _auto_pair_serial_restore_worker(rank, world, path):
+    """Restore the pairing-ON 2-way checkpoint into a SERIAL engine:
+    the loader concatenates shards on their SAVED dims (column=0 for the
+    entry linear, row=1 for the exit linear) and installs full tensors."""
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    engine = epl.Engine(_auto_pair_model(1), loss_fn=nn.MSELoss(),
+                        optimizer="adamw", lr=1e-2)
+    engine.load_checkpoint(path, load_optimizer=False)
+    torch.manual_seed(63)
+    x = torch.randn(2, 8, 8)
+    return engine.eval_step(x).clone()
+
+
+def _auto_pair_eval_worker(rank, world, path):
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "auto.auto_pair_sequential": True}))
+    engine = epl.Engine(_auto_pair_model(world), loss_fn=nn.MSELoss(),
+                        optimizer="adamw", lr=1e-2)
+    engine.load_checkpoint(path, load_optimizer=False)
+    torch.manual_seed(63)
+    x = torch.randn(2, 8, 8)
+    return engine.eval_step(x).clone()
+
+
+def test_auto_pair_checkpoint_restores_into_serial():
+    run_multiprocess(_auto_pair_save_worker, world=2,
+                     args=(CKPT + "_pair2",))
+    paired = run_multiprocess(_auto_pair_eval_worker, world=2,
+                              args=(CKPT + "_pair2",))
+    serial = run_multiprocess(_auto_pair_serial_restore_worker, world=1,
+                              args=(CKPT + "_pair2",))
+    assert torch.allclose(paired[0], serial[0], atol=1e-5), (
+        (paired[0] - serial[0]).abs().max())
