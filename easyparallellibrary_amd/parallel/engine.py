@@ -450,7 +450,7 @@ class Engine:
             loss = self._train_step_simple(inputs, targets)
         if (self._gc_wrapped
                 and self.config.gradient_checkpoint.check_gradients
-                and self.global_step == 0 and self.pipeline is None):
+                and self.global_step == 0):
             self._verify_gc_gradients(inputs, targets)
         found_inf = self.amp.found_inf(self.flat_groups)
         n_accum = self._accum_count + 1
@@ -483,8 +483,18 @@ class Engine:
                     if isinstance(m, CheckpointWrapper)]
         for w in wrappers:
             w.enabled = False
+        # suppress bucket callbacks for the verification rerun (eager
+        # PBO applies / offload D2H must not fire on throwaway grads)
+        saved_cbs = [r.on_bucket_reduced for r in self.reducers]
+        for r in self.reducers:
+            r.on_bucket_reduced = None
         self.zero_grad()
-        self._train_step_simple(inputs, targets)
+        if self.pipeline is not None:
+            # collective: every pipeline rank reruns the schedule with
+            # recompute off and compares its local stage's gradients
+            self.pipeline.run(inputs, targets)
+        else:
+            self._train_step_simple(inputs, targets)
         worst = 0.0
         for fg, s0 in zip(self.flat_groups, snap):
             worst = max(worst, float(
@@ -493,6 +503,8 @@ class Engine:
                     "= %.3e", worst)
         for w in wrappers:
             w.enabled = True
+        for r, cb in zip(self.reducers, saved_cbs):
+            r.on_bucket_reduced = cb
         for fg, s0 in zip(self.flat_groups, snap):
             fg.grad_arena.copy_(s0)
 
@@ -685,6 +697,45 @@ class Engine:
         return slice_files(files, self.num_replicas, self.replica_id,
                            unbalanced=self.config.io.unbalanced_io_slicing,
                            drop_last=self.config.io.drop_last_files)
+
+    # ---- replica consistency -------------------------------------------------
+    def check_param_consistency(self):
+        """Detect silent parameter drift between ranks that hold
+        replicated copies (nondeterministic GPU kernels can desync e.g.
+        the positions of a wide pipeline stage over many steps —
+        advisor finding r1).  COLLECTIVE.  Returns the worst absolute
+        checksum deviation across all replicated groups."""
+        worst = 0.0
+        for info in self._group_infos:
+            if len(info["ranks"]) <= 1:
+                continue
+            fg, bcomm = info["fg"], info["bcomm"]
+            local = fg.param_arena.to(torch.float64).sum().reshape(1)
+            local = local.to(torch.float32)
+            if fg.param_arena.is_cuda:
+                out = torch.empty(bcomm.size, dtype=torch.float32,
+                                  device=fg.param_arena.device)
+                bcomm.all_gather(out, local.to(fg.param_arena.device))
+            else:
+                out = torch.empty(bcomm.size, dtype=torch.float32)
+                bcomm.all_gather(out, local)
+            dev = float((out - out[0]).abs().max())
+            worst = max(worst, dev)
+        if worst > 0:
+            logger.warning(
+                "replicated parameters drifted across ranks (max checksum "
+                "delta %.3e) — call engine.resync_params()", worst)
+        return worst
+
+    def resync_params(self):
+        """Force bitwise consistency of replicated parameters: broadcast
+        every group's arena from its rank 0 and refresh fp32 masters.
+        COLLECTIVE; use at checkpoint/eval points if
+        check_param_consistency reported drift."""
+        for fg, bcomm in self._bcast_jobs:
+            if bcomm is not None and bcomm.size > 1:
+                bcomm.broadcast(fg.param_arena, root=0)
+                fg.refresh_master()
 
     # ---- teardown ------------------------------------------------------------
     def close(self):

@@ -330,3 +330,30 @@ def _signal_worker(rank, world):
 def test_broadcast_signal_eval_barrier():
     out = run_multiprocess(_signal_worker, world=2)
     assert out == [7.5, 7.5]
+
+
+def _drift_worker(rank, world):
+    import torch
+    import torch.nn as nn
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(12)
+    with epl.replicate(device_count=1):
+        model = nn.Linear(4, 2)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss())
+    clean = engine.check_param_consistency()
+    if rank == 1:   # inject drift on one replica
+        engine.flat_groups[0].param_arena[0] += 0.5
+    drifted = engine.check_param_consistency()
+    engine.resync_params()
+    fixed = engine.check_param_consistency()
+    return clean, drifted, fixed
+
+
+def test_param_consistency_check_and_resync():
+    from tests.utils import run_multiprocess
+    res = run_multiprocess(_drift_worker, world=2)
+    for clean, drifted, fixed in res:
+        assert clean == 0.0
+        assert drifted > 0.1
+        assert fixed == 0.0

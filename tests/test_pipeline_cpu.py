@@ -227,3 +227,27 @@ def _close_worker(rank, world):
 def test_engine_close_shrinks_registry():
     before, after = run_multiprocess(_close_worker, world=1)[0]
     assert before > 0 and after == 0, (before, after)
+
+
+def _gc_verify_pp_worker(rank, world):
+    """check_gradients now runs under pipeline too: the schedule reruns
+    with recompute off and the local stage's gradients must match."""
+    import easyparallellibrary_amd as epl
+    epl.init(epl.Config({
+        "pipeline.num_micro_batch": 4,
+        "gradient_checkpoint.type": "auto",
+        "gradient_checkpoint.check_gradients": True,
+    }))
+    model = _build(world_stages=world)
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(33)
+    x = torch.randn(8, 8)
+    y = torch.randn(8, 4)
+    losses = [engine.train_step(x, y) for _ in range(2)]
+    return [None if l is None else float(l) for l in losses]
+
+
+def test_gc_check_gradients_under_pipeline():
+    res = run_multiprocess(_gc_verify_pp_worker, world=2)
+    assert res[1][0] is not None and res[1][1] is not None
