@@ -47,12 +47,13 @@ void run_mfma_probe(const unsigned short*, const unsigned short*, float*,
                     hipStream_t);
 void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
                   int64_t, int64_t, float, bool, int64_t, const int64_t*,
-                  const int64_t*, hipStream_t);
+                  const int64_t*, unsigned int*, int64_t,
+                  unsigned long long, int, float, hipStream_t);
 void epl_attn_bwd(const void*, const void*, const void*, const void*,
                   const void*, const float*, float*, void*, void*, void*,
                   int64_t, int64_t, float, bool, int64_t, const int64_t*,
                   const int64_t*, const int64_t*, const int64_t*, int,
-                  hipStream_t);
+                  const unsigned int*, int64_t, float, hipStream_t);
 }
 
 namespace {
@@ -277,8 +278,13 @@ static void attn_strides(const at::Tensor& t, int64_t* out3,
   out3[2] = t.stride(2);
 }
 
+// drop_mask: optional int32 [bh*seq*ceil(seq/32)] keep-bit tensor the
+// kernel fills when dropout is on (drop_thresh in (0,256); the actual
+// drop probability is drop_thresh/256).
 void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
-              at::Tensor lse, double scale, bool causal) {
+              at::Tensor lse, double scale, bool causal,
+              c10::optional<at::Tensor> drop_mask, int64_t seed,
+              int64_t drop_thresh, double inv_keep) {
   TORCH_CHECK(q.is_cuda() && q.scalar_type() == at::kBFloat16);
   int64_t in_s[3], o_s[3], tmp[3];
   attn_strides(q, in_s, "q");
@@ -293,15 +299,26 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
   const int64_t heads = q.size(1);
   const int64_t seq = q.size(2);
   const int64_t bh = q.size(0) * heads;
+  unsigned int* mptr = nullptr;
+  int64_t mask_w = (seq + 31) / 32;
+  if (drop_mask.has_value()) {
+    check(*drop_mask, at::kInt, "drop_mask");
+    TORCH_CHECK(drop_mask->numel() == bh * seq * mask_w,
+                "drop_mask size mismatch");
+    TORCH_CHECK(drop_thresh > 0 && drop_thresh < 256, "drop_thresh range");
+    mptr = reinterpret_cast<unsigned int*>(drop_mask->data_ptr());
+  }
   epl_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
                lse.data_ptr<float>(), bh, seq, (float)scale, causal, heads,
-               in_s, o_s, cur_stream());
+               in_s, o_s, mptr, mask_w, (unsigned long long)seed,
+               (int)drop_thresh, (float)inv_keep, cur_stream());
 }
 
 void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
               at::Tensor dout, at::Tensor lse, at::Tensor delta_ws,
               at::Tensor dq, at::Tensor dk, at::Tensor dv, double scale,
-              bool causal, bool split_dkdv) {
+              bool causal, bool split_dkdv,
+              c10::optional<at::Tensor> drop_mask, double inv_keep) {
   int64_t in_s[3], o_s[3], do_s[3], g_s[3], tmp[3];
   attn_strides(q, in_s, "q");
   attn_strides(out, o_s, "out");
@@ -317,11 +334,18 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
   const int64_t seq = q.size(2);
   const int64_t bh = q.size(0) * heads;
   check(delta_ws, at::kFloat, "delta_ws");
+  const unsigned int* mptr = nullptr;
+  int64_t mask_w = (seq + 31) / 32;
+  if (drop_mask.has_value()) {
+    check(*drop_mask, at::kInt, "drop_mask");
+    mptr = reinterpret_cast<const unsigned int*>(drop_mask->data_ptr());
+  }
   epl_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
                dout.data_ptr(), lse.data_ptr<float>(),
                delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                dv.data_ptr(), bh, seq, (float)scale, causal, heads, in_s,
-               o_s, do_s, g_s, split_dkdv ? 1 : 0, cur_stream());
+               o_s, do_s, g_s, split_dkdv ? 1 : 0, mptr, mask_w,
+               (float)inv_keep, cur_stream());
 }
 
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {

@@ -73,6 +73,39 @@ DEVI bf16x8 assemble_pfrag(const float* p) {
   return out.v;
 }
 
+// ----------------------------------------------------------------------------
+// philox4x32-10 counter-based RNG for in-kernel attention dropout.
+// One call yields 16 bytes = the 16 Bernoulli draws of one lane's
+// 32-key sub-tile (byte u >= thresh keeps; p is quantized to 1/256 —
+// documented in ops/attention.py).  The keep-mask is packed into a
+// [bh, seq, ceil(seq/32)] uint32 tensor during the FORWARD; the three
+// backward kernels re-read bits instead of re-running philox.
+// ----------------------------------------------------------------------------
+DEVI void philox_round(unsigned int& c0, unsigned int& c1,
+                       unsigned int& c2, unsigned int& c3,
+                       unsigned int k0, unsigned int k1) {
+  const unsigned int hi0 = __umulhi(0xD2511F53u, c0);
+  const unsigned int lo0 = 0xD2511F53u * c0;
+  const unsigned int hi1 = __umulhi(0xCD9E8D57u, c2);
+  const unsigned int lo1 = 0xCD9E8D57u * c2;
+  c0 = hi1 ^ c1 ^ k0;
+  c1 = lo1;
+  c2 = hi0 ^ c3 ^ k1;
+  c3 = lo0;
+}
+
+DEVI void philox4x32(unsigned int c0, unsigned int c1, unsigned int c2,
+                     unsigned int c3, unsigned int k0, unsigned int k1,
+                     unsigned int out[4]) {
+#pragma unroll
+  for (int i = 0; i < 10; ++i) {
+    philox_round(c0, c1, c2, c3, k0, k1);
+    k0 += 0x9E3779B9u;
+    k1 += 0xBB67AE85u;
+  }
+  out[0] = c0; out[1] = c1; out[2] = c2; out[3] = c3;
+}
+
 // ============================================================================
 // forward (v3): KVBLK = 64; K (row copies) and V^T (transposed, 144-
 // byte row pad -> conflict-free ds_read_b128 fragments) staged in LDS
@@ -91,13 +124,18 @@ DEVI bf16x8 assemble_pfrag(const float* p) {
 // ============================================================================
 
 // one 32-key sub-tile of the fwd online-softmax loop; MASKED adds the
-// seq-bound + causal-diagonal masking (bulk tiles skip all of it)
-template <bool MASKED>
+// seq-bound + causal-diagonal masking (bulk tiles skip all of it);
+// DROP generates + applies + publishes the dropout keep-mask (the
+// normalizer l accumulates UNDROPPED exp values, so lse and the
+// backward's P are dropout-free; O accumulates P*keep/(1-p)*V)
+template <bool MASKED, bool DROP>
 DEVI void fwd_tile(const short (&ldsK)[64][72], const short (&ldsV)[64][72],
                    int sub, int64_t kvs, int64_t seq, int64_t myq,
                    float scale, int causal, const bf16x8 (&qfrag)[4],
                    f32x16& ot0, f32x16& ot1, float& m, float& l, int lq,
-                   int hi) {
+                   int hi, unsigned int* maskrow, int64_t mask_w,
+                   unsigned int seed0, unsigned int seed1,
+                   unsigned int bh32, int drop_thresh, float inv_keep) {
   f32x16 st = {};
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
@@ -130,6 +168,23 @@ DEVI void fwd_tile(const short (&ldsK)[64][72], const short (&ldsV)[64][72],
   rowsum += __shfl_xor(rowsum, 32, 64);
   l = l * alpha + rowsum;
   m = m_new;
+  if (DROP) {
+    unsigned int rnd[4];
+    philox4x32((unsigned int)myq,
+               (unsigned int)(kvs >> 5) * 2u + (unsigned int)hi,
+               bh32, 0x2E1B2137u, seed0, seed1, rnd);
+    unsigned int mybits = 0;
+#pragma unroll
+    for (int r = 0; r < 16; ++r) {
+      const unsigned int byte = (rnd[r >> 2] >> ((r & 3) * 8)) & 0xFFu;
+      const bool keep = (int)byte >= drop_thresh;
+      s[r] = keep ? s[r] * inv_keep : 0.f;
+      mybits |= (unsigned int)keep << crow(r, hi);
+    }
+    const unsigned int word = mybits | __shfl_xor(mybits, 32, 64);
+    if (hi == 0 && myq < seq)
+      maskrow[myq * mask_w + (kvs >> 5)] = word;
+  }
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     ot0[r] *= alpha;
@@ -149,12 +204,15 @@ DEVI void fwd_tile(const short (&ldsK)[64][72], const short (&ldsV)[64][72],
   }
 }
 
+template <bool DROP>
 __global__ __launch_bounds__(256) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
     float* __restrict__ lse, int64_t seq, float scale, int causal,
     int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
-    int64_t o_sb, int64_t o_sh, int64_t o_ss) {
+    int64_t o_sb, int64_t o_sh, int64_t o_ss,
+    unsigned int* __restrict__ mask, int64_t mask_w,
+    unsigned long long seed, int drop_thresh, float inv_keep) {
   __shared__ short ldsV[64][72];       // V^T: [d][kv]
   __shared__ short ldsK[64][72];       // K: [kv][d]
   const int lane = threadIdx.x & 63;
@@ -212,13 +270,21 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     __syncthreads();
   };
 
+  unsigned int* maskrow =
+      DROP ? mask + bh * seq * mask_w : (unsigned int*)nullptr;
+  const unsigned int seed0 = (unsigned int)seed;
+  const unsigned int seed1 = (unsigned int)(seed >> 32);
+  const unsigned int bh32 = (unsigned int)bh;
+
   int64_t kv0 = 0;
   for (; kv0 < bulk_end; kv0 += 64) {
     stage_tile(kv0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      fwd_tile<false>(ldsK, ldsV, sub, kv0 + sub, seq, myq, scale, causal,
-                      qfrag, ot0, ot1, m, l, lq, hi);
+      fwd_tile<false, DROP>(ldsK, ldsV, sub, kv0 + sub, seq, myq, scale,
+                            causal, qfrag, ot0, ot1, m, l, lq, hi,
+                            maskrow, mask_w, seed0, seed1, bh32,
+                            drop_thresh, inv_keep);
   }
   for (; kv0 < blk_kv_end; kv0 += 64) {
     stage_tile(kv0);
@@ -229,8 +295,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int sub = 0; sub < 64; sub += 32) {
         const int64_t kvs = kv0 + sub;
         if (kvs >= wave_kv_end) break;
-        fwd_tile<true>(ldsK, ldsV, sub, kvs, seq, myq, scale, causal,
-                       qfrag, ot0, ot1, m, l, lq, hi);
+        fwd_tile<true, DROP>(ldsK, ldsV, sub, kvs, seq, myq, scale,
+                             causal, qfrag, ot0, ot1, m, l, lq, hi,
+                             maskrow, mask_w, seed0, seed1, bh32,
+                             drop_thresh, inv_keep);
       }
     }
   }
@@ -415,13 +483,16 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 // ============================================================================
 
 // one 32-row q sub-tile (of the 64-row staged chunk) of the dV
-// accumulation; `sub` selects the LDS half
-template <bool MASKED>
+// accumulation; `sub` selects the LDS half.  DROP: dV = P_d^T dO, so P
+// is masked+rescaled with the forward's keep bits (bit mykv&31 of the
+// mask word of row qg — one broadcast word load per row).
+template <bool MASKED, bool DROP>
 DEVI void dv_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
                   int sub, int64_t q0, int64_t seq, int64_t mykv,
                   float scale, int causal, const float* lsep,
                   const bf16x8 (&kfrag)[4], f32x16& dvt0, f32x16& dvt1,
-                  int lkv, int hi) {
+                  int lkv, int hi, const unsigned int* maskrow,
+                  int64_t mask_w, float inv_keep) {
   f32x16 st = {};
 #pragma unroll
   for (int c = 0; c < 4; ++c) {
@@ -433,12 +504,17 @@ DEVI void dv_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int64_t qg = q0 + crow(r, hi);
+    const int64_t qgc = qg < seq ? qg : seq - 1;
     if (MASKED) {
-      const int64_t qgc = qg < seq ? qg : seq - 1;
       bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
       p[r] = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
     } else {
       p[r] = __expf(st[r] * scale - lsep[qg]);
+    }
+    if (DROP) {
+      const int64_t kvc = mykv < seq ? mykv : seq - 1;
+      const unsigned int w = maskrow[qgc * mask_w + (kvc >> 5)];
+      p[r] *= ((w >> (kvc & 31)) & 1u) ? inv_keep : 0.f;
     }
   }
   bf16x8 pb0 = assemble_pfrag(&p[0]);
@@ -458,14 +534,16 @@ DEVI void dv_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
   }
 }
 
-// one 32-row q sub-tile of the dK accumulation
-template <bool MASKED>
+// one 32-row q sub-tile of the dK accumulation.  DROP mirrors dq_tile:
+// P undropped, the dO.V term masked+rescaled.
+template <bool MASKED, bool DROP>
 DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
                   int sub, int64_t q0, int64_t seq, int64_t mykv,
                   float scale, int causal, const float* lsep,
                   const float* dltp, const bf16x8 (&kfrag)[4],
                   const bf16x8 (&vfrag)[4], f32x16& dkt0, f32x16& dkt1,
-                  int lkv, int hi) {
+                  int lkv, int hi, const unsigned int* maskrow,
+                  int64_t mask_w, float inv_keep) {
   f32x16 st = {};
   f32x16 dpt = {};
 #pragma unroll
@@ -482,14 +560,20 @@ DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
     const int64_t qg = q0 + crow(r, hi);
+    const int64_t qgc = qg < seq ? qg : seq - 1;
+    float dp = dpt[r];
+    if (DROP) {
+      const int64_t kvc = mykv < seq ? mykv : seq - 1;
+      const unsigned int w = maskrow[qgc * mask_w + (kvc >> 5)];
+      dp *= ((w >> (kvc & 31)) & 1u) ? inv_keep : 0.f;
+    }
     if (MASKED) {
-      const int64_t qgc = qg < seq ? qg : seq - 1;
       bool masked = qg >= seq || (causal && mykv > qg) || mykv >= seq;
       float pv = masked ? 0.f : __expf(st[r] * scale - lsep[qgc]);
-      ds[r] = masked ? 0.f : pv * (dpt[r] - dltp[qgc]) * scale;
+      ds[r] = masked ? 0.f : pv * (dp - dltp[qgc]) * scale;
     } else {
       float pv = __expf(st[r] * scale - lsep[qg]);
-      ds[r] = pv * (dpt[r] - dltp[qg]) * scale;
+      ds[r] = pv * (dp - dltp[qg]) * scale;
     }
   }
   bf16x8 db0 = assemble_pfrag(&ds[0]);
@@ -509,13 +593,16 @@ DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
   }
 }
 
+template <bool DROP>
 __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ dout, const float* __restrict__ lse,
     short* __restrict__ dv, int64_t seq, float scale, int causal,
     int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
     int64_t do_sb, int64_t do_sh, int64_t do_ss, int64_t g_sb,
-    int64_t g_sh, int64_t g_ss) {
+    int64_t g_sh, int64_t g_ss,
+    const unsigned int* __restrict__ mask, int64_t mask_w,
+    float inv_keep) {
   __shared__ short ldsQ[64][72];
   __shared__ short ldsDO[64][72];
   const int lane = threadIdx.x & 63;
@@ -561,6 +648,8 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 
   // phases (64-row chunks): masked causal diagonal -> branch-free bulk
   // -> masked tail
+  const unsigned int* maskrow =
+      DROP ? mask + bh * seq * mask_w : (const unsigned int*)nullptr;
   const int64_t diag_end =
       causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
   const int64_t bulk_end = seq & ~(int64_t)63;
@@ -572,8 +661,9 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
-      dv_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
-                    kfrag, dvt0, dvt1, lkv, hi);
+      dv_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
+                          lsep, kfrag, dvt0, dvt1, lkv, hi, maskrow,
+                          mask_w, inv_keep);
     }
   }
   q0 = diag_end > q0 ? diag_end : q0;  // 64-aligned when < seq (diag is
@@ -582,8 +672,9 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     stage_q64(q0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      dv_tile<false>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale, causal,
-                     lsep, kfrag, dvt0, dvt1, lkv, hi);
+      dv_tile<false, DROP>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale,
+                           causal, lsep, kfrag, dvt0, dvt1, lkv, hi,
+                           maskrow, mask_w, inv_keep);
   }
   for (; q0 < seq; q0 += 64) {
     stage_q64(q0);
@@ -592,8 +683,9 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= seq) break;
-      dv_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
-                    kfrag, dvt0, dvt1, lkv, hi);
+      dv_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
+                          lsep, kfrag, dvt0, dvt1, lkv, hi, maskrow,
+                          mask_w, inv_keep);
     }
   }
   if (!active || mykv >= seq) return;
@@ -606,6 +698,7 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
   }
 }
 
+template <bool DROP>
 __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -613,7 +706,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     short* __restrict__ dk, int64_t seq, float scale, int causal,
     int64_t heads, int64_t in_sb, int64_t in_sh, int64_t in_ss,
     int64_t do_sb, int64_t do_sh, int64_t do_ss, int64_t g_sb,
-    int64_t g_sh, int64_t g_ss) {
+    int64_t g_sh, int64_t g_ss,
+    const unsigned int* __restrict__ mask, int64_t mask_w,
+    float inv_keep) {
   __shared__ short ldsQ[64][72];
   __shared__ short ldsDO[64][72];
   const int lane = threadIdx.x & 63;
@@ -661,6 +756,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     __syncthreads();
   };
 
+  const unsigned int* maskrow =
+      DROP ? mask + bh * seq * mask_w : (const unsigned int*)nullptr;
   const int64_t diag_end =
       causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
   const int64_t bulk_end = seq & ~(int64_t)63;
@@ -672,8 +769,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
-      dk_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
-                    dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi);
+      dk_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
+                          lsep, dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi,
+                          maskrow, mask_w, inv_keep);
     }
   }
   q0 = diag_end > q0 ? diag_end : q0;
@@ -681,8 +779,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     stage_q64(q0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      dk_tile<false>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale, causal,
-                     lsep, dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi);
+      dk_tile<false, DROP>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale,
+                           causal, lsep, dltp, kfrag, vfrag, dkt0, dkt1,
+                           lkv, hi, maskrow, mask_w, inv_keep);
   }
   for (; q0 < seq; q0 += 64) {
     stage_q64(q0);
@@ -691,8 +790,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= seq) break;
-      dk_tile<true>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal, lsep,
-                    dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi);
+      dk_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
+                          lsep, dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi,
+                          maskrow, mask_w, inv_keep);
     }
   }
   if (!active || mykv >= seq) return;
@@ -713,13 +813,17 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
 // inside the seq bound) and a masked diagonal+tail phase.
 // ============================================================================
 
-// one 32-key sub-tile (of the 64-row staged chunk) of the dQ loop
-template <bool MASKED>
+// one 32-key sub-tile (of the 64-row staged chunk) of the dQ loop.
+// DROP: dS = P o (dP_d o M/(1-p) - D) — P stays undropped, the V^T dO
+// term is masked+rescaled with the forward's published keep bits.
+template <bool MASKED, bool DROP>
 DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
                   int sub, int64_t kv0, int64_t seq, int64_t myq,
                   float scale, int causal, float mylse, float mydelta,
                   const bf16x8 (&qfrag)[4], const bf16x8 (&dofrag)[4],
-                  f32x16& dqt0, f32x16& dqt1, int lq, int hi) {
+                  f32x16& dqt0, f32x16& dqt1, int lq, int hi,
+                  const unsigned int* maskrow, int64_t mask_w,
+                  float inv_keep) {
   f32x16 st = {};
   f32x16 dpt = {};
 #pragma unroll
@@ -732,6 +836,11 @@ DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
     dpt = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vf, dofrag[c], dpt, 0, 0,
                                                   0);
   }
+  unsigned int mword = 0;
+  if (DROP) {
+    const int64_t qc = myq < seq ? myq : seq - 1;
+    mword = maskrow[qc * mask_w + (kv0 >> 5)];
+  }
   float ds[16];
 #pragma unroll
   for (int r = 0; r < 16; ++r) {
@@ -740,7 +849,10 @@ DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
       const int64_t kvg = kv0 + crow(r, hi);
       if (myq >= seq || kvg >= seq || (causal && kvg > myq)) pv = 0.f;
     }
-    ds[r] = pv * (dpt[r] - mydelta) * scale;
+    float dp = dpt[r];
+    if (DROP)
+      dp *= ((mword >> crow(r, hi)) & 1u) ? inv_keep : 0.f;
+    ds[r] = pv * (dp - mydelta) * scale;
   }
   bf16x8 db0 = assemble_pfrag(&ds[0]);
   bf16x8 db1 = assemble_pfrag(&ds[8]);
@@ -759,6 +871,7 @@ DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
   }
 }
 
+template <bool DROP>
 __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -767,7 +880,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     float scale, int causal, int64_t heads, int64_t in_sb, int64_t in_sh,
     int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss,
     int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
-    int64_t o_ss) {
+    int64_t o_ss, const unsigned int* __restrict__ mask, int64_t mask_w,
+    float inv_keep) {
   __shared__ short ldsK[64][72];
   __shared__ short ldsVr[64][72];
   const int lane = threadIdx.x & 63;
@@ -837,13 +951,16 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     __syncthreads();
   };
 
+  const unsigned int* maskrow =
+      DROP ? mask + bh * seq * mask_w : (const unsigned int*)nullptr;
   int64_t kv0 = 0;
   for (; kv0 < bulk_end; kv0 += 64) {
     stage_kv64(kv0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      dq_tile<false>(ldsK, ldsVr, sub, kv0 + sub, seq, myq, scale, causal,
-                     mylse, mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
+      dq_tile<false, DROP>(ldsK, ldsVr, sub, kv0 + sub, seq, myq, scale,
+                           causal, mylse, mydelta, qfrag, dofrag, dqt0,
+                           dqt1, lq, hi, maskrow, mask_w, inv_keep);
   }
   for (; kv0 < blk_kv_end; kv0 += 64) {
     stage_kv64(kv0);
@@ -854,8 +971,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t kvs = kv0 + sub;
       if (kvs >= wave_kv_end) break;
-      dq_tile<true>(ldsK, ldsVr, sub, kvs, seq, myq, scale, causal, mylse,
-                    mydelta, qfrag, dofrag, dqt0, dqt1, lq, hi);
+      dq_tile<true, DROP>(ldsK, ldsVr, sub, kvs, seq, myq, scale, causal,
+                          mylse, mydelta, qfrag, dofrag, dqt0, dqt1, lq,
+                          hi, maskrow, mask_w, inv_keep);
     }
   }
 
@@ -875,16 +993,23 @@ extern "C" {
 void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
                   float* lse, int64_t batch_heads, int64_t seq, float scale,
                   bool causal, int64_t heads, const int64_t* in_strides,
-                  const int64_t* o_strides, hipStream_t stream) {
+                  const int64_t* o_strides, unsigned int* drop_mask,
+                  int64_t mask_w, unsigned long long seed, int drop_thresh,
+                  float inv_keep, hipStream_t stream) {
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
-  hipLaunchKernelGGL(attn_fwd_kernel, grid, dim3(256), 0, stream,
-                     reinterpret_cast<const short*>(q),
-                     reinterpret_cast<const short*>(k),
-                     reinterpret_cast<const short*>(v),
-                     reinterpret_cast<short*>(out), lse, seq, scale,
-                     causal ? 1 : 0, heads, in_strides[0], in_strides[1],
-                     in_strides[2], o_strides[0], o_strides[1],
-                     o_strides[2]);
+#define FWD_ARGS                                                         \
+  reinterpret_cast<const short*>(q), reinterpret_cast<const short*>(k),  \
+      reinterpret_cast<const short*>(v), reinterpret_cast<short*>(out),  \
+      lse, seq, scale, causal ? 1 : 0, heads, in_strides[0],             \
+      in_strides[1], in_strides[2], o_strides[0], o_strides[1],          \
+      o_strides[2], drop_mask, mask_w, seed, drop_thresh, inv_keep
+  if (drop_mask != nullptr)
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<true>), grid,
+                       dim3(256), 0, stream, FWD_ARGS);
+  else
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<false>), grid,
+                       dim3(256), 0, stream, FWD_ARGS);
+#undef FWD_ARGS
 }
 
 void epl_attn_bwd(const void* q, const void* k, const void* v,
@@ -894,46 +1019,57 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   bool causal, int64_t heads, const int64_t* in_strides,
                   const int64_t* o_strides, const int64_t* do_strides,
                   const int64_t* g_strides, int split_dkdv,
-                  hipStream_t stream) {
+                  const unsigned int* drop_mask, int64_t mask_w,
+                  float inv_keep, hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
-  if (split_dkdv) {
+  if (split_dkdv || drop_mask != nullptr) {
     // order: dV (needs no delta) -> dQ (computes + publishes delta from
     // the dO/O rows it already loads) -> dK (consumes delta).  The prep
-    // pass disappears.
-    hipLaunchKernelGGL(attn_bwd_dv_kernel, grid, dim3(256), 0, stream,
-                       reinterpret_cast<const short*>(q),
-                       reinterpret_cast<const short*>(k),
-                       reinterpret_cast<const short*>(dout), lse,
-                       reinterpret_cast<short*>(dv), seq, scale,
-                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
-                       in_strides[2], do_strides[0], do_strides[1],
-                       do_strides[2], g_strides[0], g_strides[1],
-                       g_strides[2]);
-    {
-      hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
-                         reinterpret_cast<const short*>(q),
-                         reinterpret_cast<const short*>(k),
-                         reinterpret_cast<const short*>(v),
-                         reinterpret_cast<const short*>(dout), lse,
-                         delta_ws, reinterpret_cast<short*>(dq),
-                         reinterpret_cast<const short*>(out), seq, scale,
-                         causal ? 1 : 0, heads, in_strides[0],
-                         in_strides[1], in_strides[2], do_strides[0],
-                         do_strides[1], do_strides[2], g_strides[0],
-                         g_strides[1], g_strides[2], o_strides[0],
-                         o_strides[1], o_strides[2]);
+    // pass disappears.  Dropout always runs the split kernels (the
+    // combined dkdv kernel has no mask path).
+#define DV_ARGS                                                          \
+  reinterpret_cast<const short*>(q), reinterpret_cast<const short*>(k),  \
+      reinterpret_cast<const short*>(dout), lse,                         \
+      reinterpret_cast<short*>(dv), seq, scale, causal ? 1 : 0, heads,   \
+      in_strides[0], in_strides[1], in_strides[2], do_strides[0],        \
+      do_strides[1], do_strides[2], g_strides[0], g_strides[1],          \
+      g_strides[2], drop_mask, mask_w, inv_keep
+#define DQ_ARGS                                                          \
+  reinterpret_cast<const short*>(q), reinterpret_cast<const short*>(k),  \
+      reinterpret_cast<const short*>(v),                                 \
+      reinterpret_cast<const short*>(dout), lse, delta_ws,               \
+      reinterpret_cast<short*>(dq), reinterpret_cast<const short*>(out), \
+      seq, scale, causal ? 1 : 0, heads, in_strides[0], in_strides[1],   \
+      in_strides[2], do_strides[0], do_strides[1], do_strides[2],        \
+      g_strides[0], g_strides[1], g_strides[2], o_strides[0],            \
+      o_strides[1], o_strides[2], drop_mask, mask_w, inv_keep
+#define DK_ARGS                                                          \
+  reinterpret_cast<const short*>(q), reinterpret_cast<const short*>(k),  \
+      reinterpret_cast<const short*>(v),                                 \
+      reinterpret_cast<const short*>(dout), lse, delta_ws,               \
+      reinterpret_cast<short*>(dk), seq, scale, causal ? 1 : 0, heads,   \
+      in_strides[0], in_strides[1], in_strides[2], do_strides[0],        \
+      do_strides[1], do_strides[2], g_strides[0], g_strides[1],          \
+      g_strides[2], drop_mask, mask_w, inv_keep
+    if (drop_mask != nullptr) {
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<true>), grid,
+                         dim3(256), 0, stream, DV_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true>), grid,
+                         dim3(256), 0, stream, DQ_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<true>), grid,
+                         dim3(256), 0, stream, DK_ARGS);
+    } else {
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<false>), grid,
+                         dim3(256), 0, stream, DV_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false>), grid,
+                         dim3(256), 0, stream, DQ_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<false>), grid,
+                         dim3(256), 0, stream, DK_ARGS);
     }
-    hipLaunchKernelGGL(attn_bwd_dk_kernel, grid, dim3(256), 0, stream,
-                       reinterpret_cast<const short*>(q),
-                       reinterpret_cast<const short*>(k),
-                       reinterpret_cast<const short*>(v),
-                       reinterpret_cast<const short*>(dout), lse, delta_ws,
-                       reinterpret_cast<short*>(dk), seq, scale,
-                       causal ? 1 : 0, heads, in_strides[0], in_strides[1],
-                       in_strides[2], do_strides[0], do_strides[1],
-                       do_strides[2], g_strides[0], g_strides[1],
-                       g_strides[2]);
+#undef DV_ARGS
+#undef DQ_ARGS
+#undef DK_ARGS
     return;
   }
   {
@@ -958,7 +1094,8 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      in_strides[2], do_strides[0], do_strides[1],
                      do_strides[2], g_strides[0], g_strides[1],
                      g_strides[2]);
-  hipLaunchKernelGGL(attn_bwd_dq_kernel, grid, dim3(256), 0, stream,
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false>), grid,
+                     dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),
                      reinterpret_cast<const short*>(v),
@@ -969,7 +1106,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      in_strides[2], do_strides[0], do_strides[1],
                      do_strides[2], g_strides[0], g_strides[1],
                      g_strides[2], o_strides[0], o_strides[1],
-                     o_strides[2]);
+                     o_strides[2], drop_mask, mask_w, inv_keep);
 }
 
 }  // extern "C"

@@ -83,11 +83,12 @@ class SelfAttention(nn.Module):
         b, s, h = x.shape
         qkv = self.qkv(x).reshape(b, s, 3, self.num_heads, self.head_dim)
         native_here = _NATIVE_ATTN in ("1", "auto")
-        if (native_here and qkv_native_ok(qkv)
-                and not (self.dropout and self.training)):
+        p = self.dropout if self.training else 0.0
+        if native_here and qkv_native_ok(qkv):
             # fully fused: kernels read the qkv views and write d_qkv
-            # slices directly — no unbind/stack copies at all
-            o = qkv_flash_attention(qkv, causal=self.causal)
+            # slices directly — no unbind/stack copies at all; dropout
+            # runs in-kernel (philox keep-mask, p quantized to 1/256)
+            o = qkv_flash_attention(qkv, causal=self.causal, dropout_p=p)
             o = o.transpose(1, 2).reshape(b, s, h)
             return self.proj(o)
         if _QKV_SPLIT:
@@ -97,16 +98,8 @@ class SelfAttention(nn.Module):
             q = q.transpose(1, 2)
             k = k.transpose(1, 2)
             v = v.transpose(1, 2)
-        if self.dropout and self.training:
-            from easyparallellibrary_amd.ops.attention import _warn_fallback
-            if native_here and q.is_cuda:
-                _warn_fallback("dropout_p={} (native kernels are "
-                               "dropout-free)".format(self.dropout))
-            o = F.scaled_dot_product_attention(
-                q, k, v, is_causal=self.causal, dropout_p=self.dropout)
-        else:
-            o = flash_attention(q, k, v, causal=self.causal,
-                                allow_native=native_here)
+        o = flash_attention(q, k, v, causal=self.causal,
+                            allow_native=native_here, dropout_p=p)
         o = o.transpose(1, 2).reshape(b, s, h)
         return self.proj(o)
 

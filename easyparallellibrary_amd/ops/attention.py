@@ -36,13 +36,33 @@ def _kernel_ok(t):
             and t.stride(1) % 8 == 0 and t.stride(2) % 8 == 0)
 
 
+def _drop_params(dropout_p):
+    """Quantize the drop probability to 1/256 (the kernel compares
+    philox BYTES against an 8-bit threshold): returns (thresh, inv_keep,
+    seed).  thresh 0 disables."""
+    if not dropout_p:
+        return 0, 1.0, 0
+    thresh = min(255, max(1, int(round(float(dropout_p) * 256.0))))
+    inv_keep = 256.0 / (256 - thresh)
+    seed = int(torch.randint(0, 2 ** 62, (1,)).item())
+    return thresh, inv_keep, seed
+
+
+def _mask_tensor(b, h, seq, device):
+    mask_w = (seq + 31) // 32
+    return torch.empty(b * h * seq * mask_w, dtype=torch.int32,
+                       device=device)
+
+
 class _FlashAttention(torch.autograd.Function):
     """Strided-input flash attention: the kernels consume the qkv-unbind
     views directly (no contiguous() copies); the output is written in
-    [B,S,H,D] memory order so the caller's transpose+reshape is free."""
+    [B,S,H,D] memory order so the caller's transpose+reshape is free.
+    dropout_p > 0 generates the philox keep-mask in-kernel (bit tensor
+    republished to the backward kernels; p quantized to 1/256)."""
 
     @staticmethod
-    def forward(ctx, q, k, v, causal, scale):
+    def forward(ctx, q, k, v, causal, scale, dropout_p=0.0):
         if not (_kernel_ok(q) and _kernel_ok(k) and _kernel_ok(v)
                 and q.stride() == k.stride() == v.stride()):
             q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
@@ -51,16 +71,24 @@ class _FlashAttention(torch.autograd.Function):
                           device=q.device).permute(0, 2, 1, 3)
         lse = torch.empty(b * h * seq, dtype=torch.float32,
                           device=q.device)
-        native_ext().attn_fwd(q, k, v, out, lse, scale, causal)
-        ctx.save_for_backward(q, k, v, out, lse)
+        thresh, inv_keep, seed = _drop_params(dropout_p)
+        mask = _mask_tensor(b, h, seq, q.device) if thresh else None
+        native_ext().attn_fwd(q, k, v, out, lse, scale, causal, mask,
+                              seed, thresh, inv_keep)
+        if mask is None:
+            ctx.save_for_backward(q, k, v, out, lse)
+        else:
+            ctx.save_for_backward(q, k, v, out, lse, mask)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.inv_keep = inv_keep
         return out
 
     @staticmethod
     def backward(ctx, dout):
         import os
-        q, k, v, out, lse = ctx.saved_tensors
+        q, k, v, out, lse = ctx.saved_tensors[:5]
+        mask = ctx.saved_tensors[5] if len(ctx.saved_tensors) > 5 else None
         # the bwd kernels re-stage q/k/v/dout tiles every 32-row iteration;
         # strided rows (6 KB apart in the qkv views) measured +405us/call
         # vs 4 contiguization copies at ~43us each — copy for backward only
@@ -86,8 +114,9 @@ class _FlashAttention(torch.autograd.Function):
         # (3 vs 2 waves/SIMD; A/B in profiles/r01_attention_ab.txt)
         split = os.environ.get("EPL_ATTN_BWD_SPLIT", "1") == "1"
         native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
-                              ctx.scale, ctx.causal, split)
-        return dq, dk, dv, None, None
+                              ctx.scale, ctx.causal, split, mask,
+                              ctx.inv_keep)
+        return dq, dk, dv, None, None, None
 
 
 class _QKVFlashAttention(torch.autograd.Function):
@@ -99,7 +128,7 @@ class _QKVFlashAttention(torch.autograd.Function):
     3x-oversized ``empty_strided`` transients."""
 
     @staticmethod
-    def forward(ctx, qkv, causal, scale):
+    def forward(ctx, qkv, causal, scale, dropout_p=0.0):
         b, s, three, h, d = qkv.shape
         q = qkv[:, :, 0].transpose(1, 2)
         k = qkv[:, :, 1].transpose(1, 2)
@@ -107,15 +136,23 @@ class _QKVFlashAttention(torch.autograd.Function):
         out = torch.empty(b, s, h, d, dtype=qkv.dtype,
                           device=qkv.device).permute(0, 2, 1, 3)
         lse = torch.empty(b * h * s, dtype=torch.float32, device=qkv.device)
-        native_ext().attn_fwd(q, k, v, out, lse, scale, causal)
-        ctx.save_for_backward(qkv, out, lse)
+        thresh, inv_keep, seed = _drop_params(dropout_p)
+        mask = _mask_tensor(b, h, s, qkv.device) if thresh else None
+        native_ext().attn_fwd(q, k, v, out, lse, scale, causal, mask,
+                              seed, thresh, inv_keep)
+        if mask is None:
+            ctx.save_for_backward(qkv, out, lse)
+        else:
+            ctx.save_for_backward(qkv, out, lse, mask)
         ctx.causal = causal
         ctx.scale = scale
+        ctx.inv_keep = inv_keep
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        qkv, out, lse = ctx.saved_tensors
+        qkv, out, lse = ctx.saved_tensors[:3]
+        mask = ctx.saved_tensors[3] if len(ctx.saved_tensors) > 3 else None
         q = qkv[:, :, 0].transpose(1, 2)
         k = qkv[:, :, 1].transpose(1, 2)
         v = qkv[:, :, 2].transpose(1, 2)
@@ -127,19 +164,21 @@ class _QKVFlashAttention(torch.autograd.Function):
         dv = dqkv[:, :, 2].transpose(1, 2)
         delta = torch.empty_like(lse)
         native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
-                              ctx.scale, ctx.causal, True)
-        return dqkv, None, None
+                              ctx.scale, ctx.causal, True, mask,
+                              ctx.inv_keep)
+        return dqkv, None, None, None
 
 
-def qkv_flash_attention(qkv, causal=False, scale=None):
+def qkv_flash_attention(qkv, causal=False, scale=None, dropout_p=0.0):
     """qkv: [B, S, 3, H, D] (the fused projection output).  Returns
     attention output as a [B, H, S, D] view whose memory order is
     [B, S, H, D] (the caller's transpose+reshape is free).  Requires the
     native kernels (bf16, D=64); callers fall back to the split-views +
-    SDPA path otherwise."""
+    SDPA path otherwise.  dropout_p is applied in-kernel (philox mask,
+    p quantized to 1/256) — pass it only in training mode."""
     if scale is None:
         scale = qkv.shape[-1] ** -0.5
-    return _QKVFlashAttention.apply(qkv, causal, scale)
+    return _QKVFlashAttention.apply(qkv, causal, scale, dropout_p)
 
 
 def qkv_native_ok(qkv):
@@ -148,14 +187,16 @@ def qkv_native_ok(qkv):
             and qkv.is_contiguous())
 
 
-def flash_attention(q, k, v, causal=False, scale=None, allow_native=True):
+def flash_attention(q, k, v, causal=False, scale=None, allow_native=True,
+                    dropout_p=0.0):
     """q,k,v: [B, H, S, D].  Native kernel when bf16/D=64 on GPU; torch
-    SDPA otherwise."""
+    SDPA otherwise.  dropout_p runs in-kernel on the native path
+    (training-mode callers only)."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if allow_native and use_native(q):
         if q.dtype == torch.bfloat16 and q.shape[-1] == 64:
-            return _FlashAttention.apply(q, k, v, causal, scale)
+            return _FlashAttention.apply(q, k, v, causal, scale, dropout_p)
         _warn_fallback("dtype={} head_dim={}".format(q.dtype, q.shape[-1]))
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,
-                                          scale=scale)
+                                          scale=scale, dropout_p=dropout_p)

@@ -148,3 +148,91 @@ def test_qkv_flash_attention_fused(causal, s):
     gerr = (qkv.grad.float() - ref_in.grad).abs().max().item()
     grel = gerr / ref_in.grad.abs().max().clamp_min(1e-6).item()
     assert gerr < 0.1 or grel < 5e-2, (gerr, grel)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("p", [0.1, 0.5])
+def test_flash_attention_dropout_exact(causal, p):
+    """In-kernel philox dropout: unpack the mask the forward published
+    and verify fwd AND bwd EXACTLY against a torch reference using the
+    same mask (O = (P*M/(1-p)) V; normalizer undropped)."""
+    from easyparallellibrary_amd.ops.attention import _FlashAttention
+    torch.manual_seed(5)
+    b, h, s, d = 2, 3, 256, 64
+    scale = d ** -0.5
+    q = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    dout = torch.randn_like(q)
+    out = _FlashAttention.apply(q, k, v, causal, scale, p)
+    saved = out.grad_fn.saved_tensors
+    assert len(saved) == 6, "mask not saved"
+    mask_words = saved[5].reshape(b * h, s, (s + 31) // 32)
+    out.backward(dout)
+
+    # unpack keep bits -> [bh, s, s]
+    bits = torch.zeros(b * h, s, s, device="cuda")
+    for w in range((s + 31) // 32):
+        word = mask_words[:, :, w].unsqueeze(-1)  # int32
+        shifts = torch.arange(32, device="cuda")
+        cols = w * 32 + shifts
+        keep = (word >> shifts) & 1
+        bits[:, :, cols[cols < s]] = keep.float()[:, :, (cols < s)]
+    thresh = min(255, max(1, int(round(p * 256))))
+    inv_keep = 256.0 / (256 - thresh)
+    # drop-rate sanity (valid region only)
+    valid = torch.ones(s, s, device="cuda")
+    if causal:
+        valid = torch.tril(valid)
+    rate = (bits * valid).sum() / (b * h * valid.sum())
+    assert abs(rate.item() - (1 - thresh / 256.0)) < 0.02, rate.item()
+
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    sc = torch.bmm(qf.reshape(b * h, s, d),
+                   kf.reshape(b * h, s, d).transpose(1, 2)) * scale
+    if causal:
+        sc = sc.masked_fill(
+            torch.triu(torch.ones(s, s, device="cuda", dtype=torch.bool),
+                       1), float("-inf"))
+    P = sc.softmax(-1)
+    Pd = P * bits * inv_keep
+    ref = torch.bmm(Pd, vf.reshape(b * h, s, d)).reshape(b, h, s, d)
+    assert torch.allclose(out.float(), ref, atol=3e-2), (
+        (out.float() - ref).abs().max())
+    ref.backward(dout.float().reshape(b, h, s, d))
+    for g, rg, name in ((q.grad, qf.grad, "dq"), (k.grad, kf.grad, "dk"),
+                        (v.grad, vf.grad, "dv")):
+        rg = rg.reshape(b, h, s, d)
+        assert torch.allclose(g.float(), rg, atol=6e-2), (
+            name, (g.float() - rg).abs().max())
+
+
+@pytest.mark.gpu
+def test_selfattention_dropout_stays_native():
+    """SelfAttention with dropout>0 must stay on the native kernels in
+    training mode (dropout silently left the native path)."""
+    from easyparallellibrary_amd.models.transformer import SelfAttention
+    m = SelfAttention(256, 4, causal=True, dropout=0.1).to(
+        "cuda", torch.bfloat16)
+    m.train()
+    x = torch.randn(2, 128, 256, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    y = m(x)
+    # native path: the fused qkv Function appears in the autograd graph
+    seen, stack, found = set(), [y.grad_fn], False
+    while stack:
+        node = stack.pop()
+        if node is None or id(node) in seen:
+            continue
+        seen.add(id(node))
+        if "QKVFlashAttention" in type(node).__name__:
+            found = True
+            break
+        stack.extend(n for n, _ in getattr(node, "next_functions", ()))
+    assert found, "native fused-qkv node not in graph"
+    y.sum().backward()
+    assert torch.isfinite(x.grad.float()).all()
