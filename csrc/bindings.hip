@@ -46,14 +46,16 @@ void epl_colsum(void*, const void*, float*, int64_t, int64_t, int64_t,
 void run_mfma_probe(const unsigned short*, const unsigned short*, float*,
                     hipStream_t);
 void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
-                  int64_t, int64_t, float, bool, int64_t, const int64_t*,
+                  int64_t, int64_t, float, bool, int64_t, int64_t,
+                  const int64_t*,
                   const int64_t*, unsigned int*, int64_t,
                   unsigned long long, int, float, hipStream_t);
 void epl_attn_bwd(const void*, const void*, const void*, const void*,
                   const void*, const float*, float*, void*, void*, void*,
                   int64_t, int64_t, float, bool, int64_t, const int64_t*,
                   const int64_t*, const int64_t*, const int64_t*, int,
-                  const unsigned int*, int64_t, float, hipStream_t);
+                  const unsigned int*, int64_t, float, int64_t,
+                  hipStream_t);
 }
 
 namespace {
@@ -268,8 +270,8 @@ void bf16_to_f32(at::Tensor dst, at::Tensor src) {
 
 static void attn_strides(const at::Tensor& t, int64_t* out3,
                          const char* name) {
-  TORCH_CHECK(t.dim() == 4 && t.size(3) == 64, name,
-              " must be [B,H,S,64]");
+  TORCH_CHECK(t.dim() == 4 && (t.size(3) == 64 || t.size(3) == 128), name,
+              " must be [B,H,S,64|128]");
   TORCH_CHECK(t.stride(3) == 1, name, " last dim must be contiguous");
   TORCH_CHECK(t.stride(2) % 8 == 0 && t.stride(1) % 8 == 0,
               name, " row strides must be 16-byte aligned");
@@ -298,6 +300,7 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
   check(lse, at::kFloat, "lse");
   const int64_t heads = q.size(1);
   const int64_t seq = q.size(2);
+  const int64_t head_dim = q.size(3);
   const int64_t bh = q.size(0) * heads;
   unsigned int* mptr = nullptr;
   int64_t mask_w = (seq + 31) / 32;
@@ -310,8 +313,9 @@ void attn_fwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
   }
   epl_attn_fwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
                lse.data_ptr<float>(), bh, seq, (float)scale, causal, heads,
-               in_s, o_s, mptr, mask_w, (unsigned long long)seed,
-               (int)drop_thresh, (float)inv_keep, cur_stream());
+               head_dim, in_s, o_s, mptr, mask_w,
+               (unsigned long long)seed, (int)drop_thresh,
+               (float)inv_keep, cur_stream());
 }
 
 void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
@@ -345,7 +349,7 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
                delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                dv.data_ptr(), bh, seq, (float)scale, causal, heads, in_s,
                o_s, do_s, g_s, split_dkdv ? 1 : 0, mptr, mask_w,
-               (float)inv_keep, cur_stream());
+               (float)inv_keep, q.size(3), cur_stream());
 }
 
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {

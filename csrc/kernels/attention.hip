@@ -128,17 +128,18 @@ DEVI void philox4x32(unsigned int c0, unsigned int c1, unsigned int c2,
 // DROP generates + applies + publishes the dropout keep-mask (the
 // normalizer l accumulates UNDROPPED exp values, so lse and the
 // backward's P are dropout-free; O accumulates P*keep/(1-p)*V)
-template <bool MASKED, bool DROP>
-DEVI void fwd_tile(const short (&ldsK)[64][72], const short (&ldsV)[64][72],
+template <bool MASKED, bool DROP, int D>
+DEVI void fwd_tile(const short (&ldsK)[64][D + 8],
+                   const short (&ldsV)[D][72],
                    int sub, int64_t kvs, int64_t seq, int64_t myq,
-                   float scale, int causal, const bf16x8 (&qfrag)[4],
-                   f32x16& ot0, f32x16& ot1, float& m, float& l, int lq,
+                   float scale, int causal, const bf16x8 (&qfrag)[D / 16],
+                   f32x16 (&ot)[D / 32], float& m, float& l, int lq,
                    int hi, unsigned int* maskrow, int64_t mask_w,
                    unsigned int seed0, unsigned int seed1,
                    unsigned int bh32, int drop_thresh, float inv_keep) {
   f32x16 st = {};
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < D / 16; ++c) {
     bf16x8 kfrag =
         *reinterpret_cast<const bf16x8*>(&ldsK[sub + lq][hi * 8 + 16 * c]);
     st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kfrag, qfrag[c], st, 0, 0,
@@ -186,26 +187,27 @@ DEVI void fwd_tile(const short (&ldsK)[64][72], const short (&ldsV)[64][72],
       maskrow[myq * mask_w + (kvs >> 5)] = word;
   }
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    ot0[r] *= alpha;
-    ot1[r] *= alpha;
-  }
+  for (int j = 0; j < D / 32; ++j)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      ot[j][r] *= alpha;
   bf16x8 pf0 = assemble_pfrag(&s[0]);
   bf16x8 pf1 = assemble_pfrag(&s[8]);
 #pragma unroll
   for (int kc = 0; kc < 2; ++kc) {
     bf16x8 pf = kc == 0 ? pf0 : pf1;
-    bf16x8 vt0 =
-        *reinterpret_cast<const bf16x8*>(&ldsV[lq][sub + kc * 16 + hi * 8]);
-    bf16x8 vt1 = *reinterpret_cast<const bf16x8*>(
-        &ldsV[32 + lq][sub + kc * 16 + hi * 8]);
-    ot0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt0, pf, ot0, 0, 0, 0);
-    ot1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt1, pf, ot1, 0, 0, 0);
+#pragma unroll
+    for (int j = 0; j < D / 32; ++j) {
+      bf16x8 vt = *reinterpret_cast<const bf16x8*>(
+          &ldsV[j * 32 + lq][sub + kc * 16 + hi * 8]);
+      ot[j] = __builtin_amdgcn_mfma_f32_32x32x16_bf16(vt, pf, ot[j], 0, 0,
+                                                      0);
+    }
   }
 }
 
-template <bool DROP>
-__global__ __launch_bounds__(256) void attn_fwd_kernel(
+template <bool DROP, int D>
+__global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_fwd_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, short* __restrict__ out,
     float* __restrict__ lse, int64_t seq, float scale, int causal,
@@ -213,8 +215,8 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     int64_t o_sb, int64_t o_sh, int64_t o_ss,
     unsigned int* __restrict__ mask, int64_t mask_w,
     unsigned long long seed, int drop_thresh, float inv_keep) {
-  __shared__ short ldsV[64][72];       // V^T: [d][kv]
-  __shared__ short ldsK[64][72];       // K: [kv][d]
+  __shared__ short ldsV[D][72];        // V^T: [d][kv]
+  __shared__ short ldsK[64][D + 8];    // K: [kv][d]
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -235,13 +237,13 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const int64_t myq = q0 + lq;
   const int64_t qrow = myq < seq ? myq : seq - 1;
 
-  bf16x8 qfrag[4];
+  bf16x8 qfrag[D / 16];
 #pragma unroll
-  for (int c = 0; c < 4; ++c)
+  for (int c = 0; c < D / 16; ++c)
     qfrag[c] = *reinterpret_cast<const bf16x8*>(
         qp + qrow * in_ss + hi * 8 + 16 * c);
 
-  f32x16 ot0 = {}, ot1 = {};
+  f32x16 ot[D / 32] = {};
   float m = -1e30f, l = 0.f;
 
   // bulk tiles are provably full for EVERY lane of the block: below the
@@ -257,7 +259,7 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     int64_t vrow = kv0 + stage_kv;
     if (vrow >= seq) vrow = seq - 1;   // masked columns never contribute
 #pragma unroll
-    for (int h2 = 0; h2 < 2; ++h2) {
+    for (int h2 = 0; h2 < D / 32; ++h2) {
       const int sd = stage_d0 + h2 * 32;
       bf16x8 vv = *reinterpret_cast<const bf16x8*>(
           vp + vrow * in_ss + sd);
@@ -281,10 +283,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
     stage_tile(kv0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      fwd_tile<false, DROP>(ldsK, ldsV, sub, kv0 + sub, seq, myq, scale,
-                            causal, qfrag, ot0, ot1, m, l, lq, hi,
-                            maskrow, mask_w, seed0, seed1, bh32,
-                            drop_thresh, inv_keep);
+      fwd_tile<false, DROP, D>(ldsK, ldsV, sub, kv0 + sub, seq, myq,
+                               scale, causal, qfrag, ot, m, l, lq, hi,
+                               maskrow, mask_w, seed0, seed1, bh32,
+                               drop_thresh, inv_keep);
   }
   for (; kv0 < blk_kv_end; kv0 += 64) {
     stage_tile(kv0);
@@ -295,10 +297,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
       for (int sub = 0; sub < 64; sub += 32) {
         const int64_t kvs = kv0 + sub;
         if (kvs >= wave_kv_end) break;
-        fwd_tile<true, DROP>(ldsK, ldsV, sub, kvs, seq, myq, scale,
-                             causal, qfrag, ot0, ot1, m, l, lq, hi,
-                             maskrow, mask_w, seed0, seed1, bh32,
-                             drop_thresh, inv_keep);
+        fwd_tile<true, DROP, D>(ldsK, ldsV, sub, kvs, seq, myq, scale,
+                                causal, qfrag, ot, m, l, lq, hi,
+                                maskrow, mask_w, seed0, seed1, bh32,
+                                drop_thresh, inv_keep);
       }
     }
   }
@@ -307,10 +309,10 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   const float inv_l = 1.f / l;
   short* op = out + (bh / heads) * o_sb + (bh % heads) * o_sh + myq * o_ss;
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    op[crow(r, hi)] = (short)f2bf(ot0[r] * inv_l);
-    op[32 + crow(r, hi)] = (short)f2bf(ot1[r] * inv_l);
-  }
+  for (int j = 0; j < D / 32; ++j)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      op[j * 32 + crow(r, hi)] = (short)f2bf(ot[j][r] * inv_l);
   if (hi == 0) lse[bh * seq + myq] = m + __logf(l);
 }
 
@@ -486,16 +488,17 @@ __global__ __launch_bounds__(256) void attn_bwd_dkdv_kernel(
 // accumulation; `sub` selects the LDS half.  DROP: dV = P_d^T dO, so P
 // is masked+rescaled with the forward's keep bits (bit mykv&31 of the
 // mask word of row qg — one broadcast word load per row).
-template <bool MASKED, bool DROP>
-DEVI void dv_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
+template <bool MASKED, bool DROP, int D>
+DEVI void dv_tile(const short (&ldsQ)[64][D + 8],
+                  const short (&ldsDO)[64][D + 8],
                   int sub, int64_t q0, int64_t seq, int64_t mykv,
                   float scale, int causal, const float* lsep,
-                  const bf16x8 (&kfrag)[4], f32x16& dvt0, f32x16& dvt1,
+                  const bf16x8 (&kfrag)[D / 16], f32x16 (&dvt)[D / 32],
                   int lkv, int hi, const unsigned int* maskrow,
                   int64_t mask_w, float inv_keep) {
   f32x16 st = {};
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < D / 16; ++c) {
     bf16x8 qf = *reinterpret_cast<const bf16x8*>(
         &ldsQ[sub + lkv][hi * 8 + 16 * c]);
     st = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qf, kfrag[c], st, 0, 0, 0);
@@ -521,33 +524,36 @@ DEVI void dv_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
   bf16x8 pb1 = assemble_pfrag(&p[8]);
 #pragma unroll
   for (int qc = 0; qc < 2; ++qc) {
-    bf16x8 dot0, dot1;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int qr = sub + qc * 16 + hi * 8 + j;
-      dot0[j] = ldsDO[qr][lkv];
-      dot1[j] = ldsDO[qr][32 + lkv];
-    }
     bf16x8 pb = qc == 0 ? pb0 : pb1;
-    dvt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot0, pb, dvt0, 0, 0, 0);
-    dvt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot1, pb, dvt1, 0, 0, 0);
+#pragma unroll
+    for (int dj = 0; dj < D / 32; ++dj) {
+      bf16x8 dot;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int qr = sub + qc * 16 + hi * 8 + j;
+        dot[j] = ldsDO[qr][dj * 32 + lkv];
+      }
+      dvt[dj] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(dot, pb, dvt[dj], 0, 0, 0);
+    }
   }
 }
 
 // one 32-row q sub-tile of the dK accumulation.  DROP mirrors dq_tile:
 // P undropped, the dO.V term masked+rescaled.
-template <bool MASKED, bool DROP>
-DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
+template <bool MASKED, bool DROP, int D>
+DEVI void dk_tile(const short (&ldsQ)[64][D + 8],
+                  const short (&ldsDO)[64][D + 8],
                   int sub, int64_t q0, int64_t seq, int64_t mykv,
                   float scale, int causal, const float* lsep,
-                  const float* dltp, const bf16x8 (&kfrag)[4],
-                  const bf16x8 (&vfrag)[4], f32x16& dkt0, f32x16& dkt1,
+                  const float* dltp, const bf16x8 (&kfrag)[D / 16],
+                  const bf16x8 (&vfrag)[D / 16], f32x16 (&dkt)[D / 32],
                   int lkv, int hi, const unsigned int* maskrow,
                   int64_t mask_w, float inv_keep) {
   f32x16 st = {};
   f32x16 dpt = {};
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < D / 16; ++c) {
     bf16x8 qf = *reinterpret_cast<const bf16x8*>(
         &ldsQ[sub + lkv][hi * 8 + 16 * c]);
     bf16x8 dof = *reinterpret_cast<const bf16x8*>(
@@ -580,21 +586,23 @@ DEVI void dk_tile(const short (&ldsQ)[64][72], const short (&ldsDO)[64][72],
   bf16x8 db1 = assemble_pfrag(&ds[8]);
 #pragma unroll
   for (int qc = 0; qc < 2; ++qc) {
-    bf16x8 qt0, qt1;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int qr = sub + qc * 16 + hi * 8 + j;
-      qt0[j] = ldsQ[qr][lkv];
-      qt1[j] = ldsQ[qr][32 + lkv];
-    }
     bf16x8 db = qc == 0 ? db0 : db1;
-    dkt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt0, db, dkt0, 0, 0, 0);
-    dkt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt1, db, dkt1, 0, 0, 0);
+#pragma unroll
+    for (int dj = 0; dj < D / 32; ++dj) {
+      bf16x8 qt;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int qr = sub + qc * 16 + hi * 8 + j;
+        qt[j] = ldsQ[qr][dj * 32 + lkv];
+      }
+      dkt[dj] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(qt, db, dkt[dj], 0, 0, 0);
+    }
   }
 }
 
-template <bool DROP>
-__global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
+template <bool DROP, int D>
+__global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ dout, const float* __restrict__ lse,
     short* __restrict__ dv, int64_t seq, float scale, int causal,
@@ -603,8 +611,8 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     int64_t g_sh, int64_t g_ss,
     const unsigned int* __restrict__ mask, int64_t mask_w,
     float inv_keep) {
-  __shared__ short ldsQ[64][72];
-  __shared__ short ldsDO[64][72];
+  __shared__ short ldsQ[64][D + 8];
+  __shared__ short ldsDO[64][D + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -621,13 +629,13 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
 
   const int64_t mykv = kv0 + lkv;
   const int64_t kvrow = mykv < seq ? mykv : seq - 1;
-  bf16x8 kfrag[4];
+  bf16x8 kfrag[D / 16];
 #pragma unroll
-  for (int c = 0; c < 4; ++c)
+  for (int c = 0; c < D / 16; ++c)
     kfrag[c] = *reinterpret_cast<const bf16x8*>(
         kp + kvrow * in_ss + hi * 8 + 16 * c);
 
-  f32x16 dvt0 = {}, dvt1 = {};
+  f32x16 dvt[D / 32] = {};
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
 
@@ -638,10 +646,17 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int half = 0; half < 2; ++half) {
       int64_t qr = q0 + stage_row + half * 32;
       if (qr >= seq) qr = seq - 1;
-      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row + half * 32][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
-      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row + half * 32][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+#pragma unroll
+      for (int seg = 0; seg < D; seg += 64) {
+        *reinterpret_cast<bf16x8*>(
+            &ldsQ[stage_row + half * 32][stage_seg + seg]) =
+            *reinterpret_cast<const bf16x8*>(
+                qp + qr * in_ss + stage_seg + seg);
+        *reinterpret_cast<bf16x8*>(
+            &ldsDO[stage_row + half * 32][stage_seg + seg]) =
+            *reinterpret_cast<const bf16x8*>(
+                dop + qr * do_ss + stage_seg + seg);
+      }
     }
     __syncthreads();
   };
@@ -661,9 +676,9 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
-      dv_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
-                          lsep, kfrag, dvt0, dvt1, lkv, hi, maskrow,
-                          mask_w, inv_keep);
+      dv_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
+                             causal, lsep, kfrag, dvt, lkv, hi, maskrow,
+                             mask_w, inv_keep);
     }
   }
   q0 = diag_end > q0 ? diag_end : q0;  // 64-aligned when < seq (diag is
@@ -672,9 +687,9 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     stage_q64(q0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      dv_tile<false, DROP>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale,
-                           causal, lsep, kfrag, dvt0, dvt1, lkv, hi,
-                           maskrow, mask_w, inv_keep);
+      dv_tile<false, DROP, D>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv,
+                              scale, causal, lsep, kfrag, dvt, lkv, hi,
+                              maskrow, mask_w, inv_keep);
   }
   for (; q0 < seq; q0 += 64) {
     stage_q64(q0);
@@ -683,23 +698,23 @@ __global__ __launch_bounds__(256, 4) void attn_bwd_dv_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= seq) break;
-      dv_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
-                          lsep, kfrag, dvt0, dvt1, lkv, hi, maskrow,
-                          mask_w, inv_keep);
+      dv_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
+                             causal, lsep, kfrag, dvt, lkv, hi, maskrow,
+                             mask_w, inv_keep);
     }
   }
   if (!active || mykv >= seq) return;
   short* dvp = dv + (bh / heads) * g_sb + (bh % heads) * g_sh +
                mykv * g_ss;
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    dvp[crow(r, hi)] = (short)f2bf(dvt0[r]);
-    dvp[32 + crow(r, hi)] = (short)f2bf(dvt1[r]);
-  }
+  for (int dj = 0; dj < D / 32; ++dj)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      dvp[dj * 32 + crow(r, hi)] = (short)f2bf(dvt[dj][r]);
 }
 
-template <bool DROP>
-__global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
+template <bool DROP, int D>
+__global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -709,8 +724,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     int64_t g_sh, int64_t g_ss,
     const unsigned int* __restrict__ mask, int64_t mask_w,
     float inv_keep) {
-  __shared__ short ldsQ[64][72];
-  __shared__ short ldsDO[64][72];
+  __shared__ short ldsQ[64][D + 8];
+  __shared__ short ldsDO[64][D + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -729,16 +744,16 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
 
   const int64_t mykv = kv0 + lkv;
   const int64_t kvrow = mykv < seq ? mykv : seq - 1;
-  bf16x8 kfrag[4], vfrag[4];
+  bf16x8 kfrag[D / 16], vfrag[D / 16];
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < D / 16; ++c) {
     kfrag[c] = *reinterpret_cast<const bf16x8*>(
         kp + kvrow * in_ss + hi * 8 + 16 * c);
     vfrag[c] = *reinterpret_cast<const bf16x8*>(
         vp + kvrow * in_ss + hi * 8 + 16 * c);
   }
 
-  f32x16 dkt0 = {}, dkt1 = {};
+  f32x16 dkt[D / 32] = {};
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
 
@@ -748,10 +763,17 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int half = 0; half < 2; ++half) {
       int64_t qr = q0 + stage_row + half * 32;
       if (qr >= seq) qr = seq - 1;
-      *reinterpret_cast<bf16x8*>(&ldsQ[stage_row + half * 32][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(qp + qr * in_ss + stage_seg);
-      *reinterpret_cast<bf16x8*>(&ldsDO[stage_row + half * 32][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(dop + qr * do_ss + stage_seg);
+#pragma unroll
+      for (int seg = 0; seg < D; seg += 64) {
+        *reinterpret_cast<bf16x8*>(
+            &ldsQ[stage_row + half * 32][stage_seg + seg]) =
+            *reinterpret_cast<const bf16x8*>(
+                qp + qr * in_ss + stage_seg + seg);
+        *reinterpret_cast<bf16x8*>(
+            &ldsDO[stage_row + half * 32][stage_seg + seg]) =
+            *reinterpret_cast<const bf16x8*>(
+                dop + qr * do_ss + stage_seg + seg);
+      }
     }
     __syncthreads();
   };
@@ -769,9 +791,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
-      dk_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
-                          lsep, dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi,
-                          maskrow, mask_w, inv_keep);
+      dk_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
+                             causal, lsep, dltp, kfrag, vfrag, dkt, lkv,
+                             hi, maskrow, mask_w, inv_keep);
     }
   }
   q0 = diag_end > q0 ? diag_end : q0;
@@ -779,9 +801,9 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     stage_q64(q0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      dk_tile<false, DROP>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv, scale,
-                           causal, lsep, dltp, kfrag, vfrag, dkt0, dkt1,
-                           lkv, hi, maskrow, mask_w, inv_keep);
+      dk_tile<false, DROP, D>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv,
+                              scale, causal, lsep, dltp, kfrag, vfrag,
+                              dkt, lkv, hi, maskrow, mask_w, inv_keep);
   }
   for (; q0 < seq; q0 += 64) {
     stage_q64(q0);
@@ -790,19 +812,19 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t q0s = q0 + sub;
       if (q0s >= seq) break;
-      dk_tile<true, DROP>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale, causal,
-                          lsep, dltp, kfrag, vfrag, dkt0, dkt1, lkv, hi,
-                          maskrow, mask_w, inv_keep);
+      dk_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
+                             causal, lsep, dltp, kfrag, vfrag, dkt, lkv,
+                             hi, maskrow, mask_w, inv_keep);
     }
   }
   if (!active || mykv >= seq) return;
   short* dkp = dk + (bh / heads) * g_sb + (bh % heads) * g_sh +
                mykv * g_ss;
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    dkp[crow(r, hi)] = (short)f2bf(dkt0[r]);
-    dkp[32 + crow(r, hi)] = (short)f2bf(dkt1[r]);
-  }
+  for (int dj = 0; dj < D / 32; ++dj)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      dkp[dj * 32 + crow(r, hi)] = (short)f2bf(dkt[dj][r]);
 }
 
 // ============================================================================
@@ -816,18 +838,20 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dk_kernel(
 // one 32-key sub-tile (of the 64-row staged chunk) of the dQ loop.
 // DROP: dS = P o (dP_d o M/(1-p) - D) — P stays undropped, the V^T dO
 // term is masked+rescaled with the forward's published keep bits.
-template <bool MASKED, bool DROP>
-DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
+template <bool MASKED, bool DROP, int D>
+DEVI void dq_tile(const short (&ldsK)[64][D + 8],
+                  const short (&ldsVr)[64][D + 8],
                   int sub, int64_t kv0, int64_t seq, int64_t myq,
                   float scale, int causal, float mylse, float mydelta,
-                  const bf16x8 (&qfrag)[4], const bf16x8 (&dofrag)[4],
-                  f32x16& dqt0, f32x16& dqt1, int lq, int hi,
+                  const bf16x8 (&qfrag)[D / 16],
+                  const bf16x8 (&dofrag)[D / 16],
+                  f32x16 (&dqt)[D / 32], int lq, int hi,
                   const unsigned int* maskrow, int64_t mask_w,
                   float inv_keep) {
   f32x16 st = {};
   f32x16 dpt = {};
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < D / 16; ++c) {
     bf16x8 kf = *reinterpret_cast<const bf16x8*>(
         &ldsK[sub + lq][hi * 8 + 16 * c]);
     bf16x8 vf = *reinterpret_cast<const bf16x8*>(
@@ -858,21 +882,23 @@ DEVI void dq_tile(const short (&ldsK)[64][72], const short (&ldsVr)[64][72],
   bf16x8 db1 = assemble_pfrag(&ds[8]);
 #pragma unroll
   for (int kc = 0; kc < 2; ++kc) {
-    bf16x8 kt0, kt1;
-#pragma unroll
-    for (int j = 0; j < 8; ++j) {
-      const int kr = sub + kc * 16 + hi * 8 + j;
-      kt0[j] = ldsK[kr][lq];
-      kt1[j] = ldsK[kr][32 + lq];
-    }
     bf16x8 db = kc == 0 ? db0 : db1;
-    dqt0 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt0, db, dqt0, 0, 0, 0);
-    dqt1 = __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt1, db, dqt1, 0, 0, 0);
+#pragma unroll
+    for (int dj = 0; dj < D / 32; ++dj) {
+      bf16x8 kt;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int kr = sub + kc * 16 + hi * 8 + j;
+        kt[j] = ldsK[kr][dj * 32 + lq];
+      }
+      dqt[dj] =
+          __builtin_amdgcn_mfma_f32_32x32x16_bf16(kt, db, dqt[dj], 0, 0, 0);
+    }
   }
 }
 
-template <bool DROP>
-__global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
+template <bool DROP, int D>
+__global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
     const float* __restrict__ lse, const float* __restrict__ delta,
@@ -882,8 +908,8 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
     int64_t o_ss, const unsigned int* __restrict__ mask, int64_t mask_w,
     float inv_keep) {
-  __shared__ short ldsK[64][72];
-  __shared__ short ldsVr[64][72];
+  __shared__ short ldsK[64][D + 8];
+  __shared__ short ldsVr[64][D + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -906,10 +932,10 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
   const float mylse = lse[bh * seq + qrow];
 
   const short* op_ = out + (bh / heads) * o_sb + (bh % heads) * o_sh;
-  bf16x8 qfrag[4], dofrag[4];
+  bf16x8 qfrag[D / 16], dofrag[D / 16];
   float dsum = 0.f;
 #pragma unroll
-  for (int c = 0; c < 4; ++c) {
+  for (int c = 0; c < D / 16; ++c) {
     qfrag[c] = *reinterpret_cast<const bf16x8*>(
         qp + qrow * in_ss + hi * 8 + 16 * c);
     dofrag[c] = *reinterpret_cast<const bf16x8*>(
@@ -927,7 +953,7 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
   if (myq < seq && hi == 0)
     const_cast<float*>(delta)[bh * seq + myq] = mydelta;
 
-  f32x16 dqt0 = {}, dqt1 = {};
+  f32x16 dqt[D / 32] = {};
 
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
@@ -943,10 +969,17 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     for (int half = 0; half < 2; ++half) {
       int64_t kr = kv0 + stage_row + half * 32;
       if (kr >= seq) kr = seq - 1;   // dS there is 0
-      *reinterpret_cast<bf16x8*>(&ldsK[stage_row + half * 32][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(kp + kr * in_ss + stage_seg);
-      *reinterpret_cast<bf16x8*>(&ldsVr[stage_row + half * 32][stage_seg]) =
-          *reinterpret_cast<const bf16x8*>(vp + kr * in_ss + stage_seg);
+#pragma unroll
+      for (int seg = 0; seg < D; seg += 64) {
+        *reinterpret_cast<bf16x8*>(
+            &ldsK[stage_row + half * 32][stage_seg + seg]) =
+            *reinterpret_cast<const bf16x8*>(
+                kp + kr * in_ss + stage_seg + seg);
+        *reinterpret_cast<bf16x8*>(
+            &ldsVr[stage_row + half * 32][stage_seg + seg]) =
+            *reinterpret_cast<const bf16x8*>(
+                vp + kr * in_ss + stage_seg + seg);
+      }
     }
     __syncthreads();
   };
@@ -958,9 +991,10 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     stage_kv64(kv0);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      dq_tile<false, DROP>(ldsK, ldsVr, sub, kv0 + sub, seq, myq, scale,
-                           causal, mylse, mydelta, qfrag, dofrag, dqt0,
-                           dqt1, lq, hi, maskrow, mask_w, inv_keep);
+      dq_tile<false, DROP, D>(ldsK, ldsVr, sub, kv0 + sub, seq, myq,
+                              scale, causal, mylse, mydelta, qfrag,
+                              dofrag, dqt, lq, hi, maskrow, mask_w,
+                              inv_keep);
   }
   for (; kv0 < blk_kv_end; kv0 += 64) {
     stage_kv64(kv0);
@@ -971,19 +1005,19 @@ __global__ __launch_bounds__(256, 3) void attn_bwd_dq_kernel(
     for (int sub = 0; sub < 64; sub += 32) {
       const int64_t kvs = kv0 + sub;
       if (kvs >= wave_kv_end) break;
-      dq_tile<true, DROP>(ldsK, ldsVr, sub, kvs, seq, myq, scale, causal,
-                          mylse, mydelta, qfrag, dofrag, dqt0, dqt1, lq,
-                          hi, maskrow, mask_w, inv_keep);
+      dq_tile<true, DROP, D>(ldsK, ldsVr, sub, kvs, seq, myq, scale,
+                             causal, mylse, mydelta, qfrag, dofrag, dqt,
+                             lq, hi, maskrow, mask_w, inv_keep);
     }
   }
 
   if (!active || myq >= seq) return;
   short* dqp = dq + (bh / heads) * g_sb + (bh % heads) * g_sh + myq * g_ss;
 #pragma unroll
-  for (int r = 0; r < 16; ++r) {
-    dqp[crow(r, hi)] = (short)f2bf(dqt0[r]);
-    dqp[32 + crow(r, hi)] = (short)f2bf(dqt1[r]);
-  }
+  for (int dj = 0; dj < D / 32; ++dj)
+#pragma unroll
+    for (int r = 0; r < 16; ++r)
+      dqp[dj * 32 + crow(r, hi)] = (short)f2bf(dqt[dj][r]);
 }
 
 }  // namespace
@@ -992,7 +1026,8 @@ extern "C" {
 
 void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
                   float* lse, int64_t batch_heads, int64_t seq, float scale,
-                  bool causal, int64_t heads, const int64_t* in_strides,
+                  bool causal, int64_t heads, int64_t head_dim,
+                  const int64_t* in_strides,
                   const int64_t* o_strides, unsigned int* drop_mask,
                   int64_t mask_w, unsigned long long seed, int drop_thresh,
                   float inv_keep, hipStream_t stream) {
@@ -1003,12 +1038,20 @@ void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
       lse, seq, scale, causal ? 1 : 0, heads, in_strides[0],             \
       in_strides[1], in_strides[2], o_strides[0], o_strides[1],          \
       o_strides[2], drop_mask, mask_w, seed, drop_thresh, inv_keep
-  if (drop_mask != nullptr)
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<true>), grid,
+  if (head_dim == 128) {
+    if (drop_mask != nullptr)
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<true, 128>),
+                         grid, dim3(256), 0, stream, FWD_ARGS);
+    else
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<false, 128>),
+                         grid, dim3(256), 0, stream, FWD_ARGS);
+  } else if (drop_mask != nullptr) {
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<true, 64>), grid,
                        dim3(256), 0, stream, FWD_ARGS);
-  else
-    hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<false>), grid,
+  } else {
+    hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_fwd_kernel<false, 64>), grid,
                        dim3(256), 0, stream, FWD_ARGS);
+  }
 #undef FWD_ARGS
 }
 
@@ -1020,10 +1063,12 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   const int64_t* o_strides, const int64_t* do_strides,
                   const int64_t* g_strides, int split_dkdv,
                   const unsigned int* drop_mask, int64_t mask_w,
-                  float inv_keep, hipStream_t stream) {
+                  float inv_keep, int64_t head_dim, hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
   dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
-  if (split_dkdv || drop_mask != nullptr) {
+  // head_dim 128 always runs the split kernels (the combined dkdv
+  // kernel is d=64-only)
+  if (split_dkdv || drop_mask != nullptr || head_dim == 128) {
     // order: dV (needs no delta) -> dQ (computes + publishes delta from
     // the dO/O rows it already loads) -> dK (consumes delta).  The prep
     // pass disappears.  Dropout always runs the split kernels (the
@@ -1052,20 +1097,39 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
       in_strides[0], in_strides[1], in_strides[2], do_strides[0],        \
       do_strides[1], do_strides[2], g_strides[0], g_strides[1],          \
       g_strides[2], drop_mask, mask_w, inv_keep
-    if (drop_mask != nullptr) {
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<true>), grid,
-                         dim3(256), 0, stream, DV_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true>), grid,
-                         dim3(256), 0, stream, DQ_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<true>), grid,
-                         dim3(256), 0, stream, DK_ARGS);
+    if (head_dim == 128) {
+      if (drop_mask != nullptr) {
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, 128>),
+                           grid, dim3(256), 0, stream, DV_ARGS);
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, 128>),
+                           grid, dim3(256), 0, stream, DQ_ARGS);
+        hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, 128>),
+                           grid, dim3(256), 0, stream, DK_ARGS);
+      } else {
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, 128>), grid,
+            dim3(256), 0, stream, DV_ARGS);
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 128>), grid,
+            dim3(256), 0, stream, DQ_ARGS);
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, 128>), grid,
+            dim3(256), 0, stream, DK_ARGS);
+      }
+    } else if (drop_mask != nullptr) {
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, 64>),
+                         grid, dim3(256), 0, stream, DV_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, 64>),
+                         grid, dim3(256), 0, stream, DQ_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, 64>),
+                         grid, dim3(256), 0, stream, DK_ARGS);
     } else {
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<false>), grid,
-                         dim3(256), 0, stream, DV_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false>), grid,
-                         dim3(256), 0, stream, DQ_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<false>), grid,
-                         dim3(256), 0, stream, DK_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, 64>),
+                         grid, dim3(256), 0, stream, DV_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 64>),
+                         grid, dim3(256), 0, stream, DQ_ARGS);
+      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, 64>),
+                         grid, dim3(256), 0, stream, DK_ARGS);
     }
 #undef DV_ARGS
 #undef DQ_ARGS
@@ -1094,7 +1158,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      in_strides[2], do_strides[0], do_strides[1],
                      do_strides[2], g_strides[0], g_strides[1],
                      g_strides[2]);
-  hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false>), grid,
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 64>), grid,
                      dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),

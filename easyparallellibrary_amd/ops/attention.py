@@ -1,4 +1,4 @@
-"""Fused flash attention (hand-written CDNA4 kernels, head_dim 64, bf16).
+"""Fused flash attention (hand-written CDNA4 kernels, head_dim 64/128, bf16).
 
 No reference counterpart (the reference's attention is plain TF ops);
 this is MI355X-native hot-path work — design notes in docs/kernels.md.
@@ -32,7 +32,8 @@ def _warn_fallback(reason):
 
 
 def _kernel_ok(t):
-    return (t.dim() == 4 and t.shape[-1] == 64 and t.stride(-1) == 1
+    return (t.dim() == 4 and t.shape[-1] in (64, 128)
+            and t.stride(-1) == 1
             and t.stride(1) % 8 == 0 and t.stride(2) % 8 == 0)
 
 
@@ -183,7 +184,7 @@ def qkv_flash_attention(qkv, causal=False, scale=None, dropout_p=0.0):
 
 def qkv_native_ok(qkv):
     return (use_native(qkv) and qkv.dtype == torch.bfloat16
-            and qkv.dim() == 5 and qkv.shape[-1] == 64
+            and qkv.dim() == 5 and qkv.shape[-1] in (64, 128)
             and qkv.is_contiguous())
 
 
@@ -195,7 +196,7 @@ def flash_attention(q, k, v, causal=False, scale=None, allow_native=True,
     if scale is None:
         scale = q.shape[-1] ** -0.5
     if allow_native and use_native(q):
-        if q.dtype == torch.bfloat16 and q.shape[-1] == 64:
+        if q.dtype == torch.bfloat16 and q.shape[-1] in (64, 128):
             return _FlashAttention.apply(q, k, v, causal, scale, dropout_p)
         _warn_fallback("dtype={} head_dim={}".format(q.dtype, q.shape[-1]))
     return F.scaled_dot_product_attention(q, k, v, is_causal=causal,

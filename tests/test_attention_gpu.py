@@ -236,3 +236,68 @@ def test_selfattention_dropout_stays_native():
     assert found, "native fused-qkv node not in graph"
     y.sum().backward()
     assert torch.isfinite(x.grad.float()).all()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("causal", [False, True])
+@pytest.mark.parametrize("seq", [128, 256, 200])
+def test_flash_attention_hd128(causal, seq):
+    """head_dim 128 (template<int D> instantiation): fwd + bwd vs fp32
+    reference."""
+    from easyparallellibrary_amd.ops.attention import flash_attention
+    torch.manual_seed(8)
+    b, h, d = 2, 3, 128
+    q = torch.randn(b, h, seq, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q, requires_grad=True)
+    v = torch.randn_like(q, requires_grad=True)
+    scale = d ** -0.5
+    out = flash_attention(q, k, v, causal=causal, scale=scale)
+    assert "FlashAttention" in type(out.grad_fn).__name__  # native ran
+    dout = torch.randn_like(q)
+    out.backward(dout)
+    qf = q.detach().float().requires_grad_(True)
+    kf = k.detach().float().requires_grad_(True)
+    vf = v.detach().float().requires_grad_(True)
+    ref = ref_attention(qf, kf, vf, causal, scale)
+    ref.backward(dout.float())
+    torch.cuda.synchronize()
+    assert (out.float() - ref).abs().max().item() < 4e-2
+    for got, want, name in ((q.grad, qf.grad, "dq"),
+                            (k.grad, kf.grad, "dk"),
+                            (v.grad, vf.grad, "dv")):
+        err = (got.float() - want).abs().max().item()
+        rel = err / want.abs().max().clamp_min(1e-6).item()
+        assert err < 0.15 or rel < 5e-2, (name, err, rel)
+
+
+@pytest.mark.gpu
+def test_flash_attention_hd128_dropout():
+    """hd128 + in-kernel dropout: exact vs reference using the published
+    mask."""
+    from easyparallellibrary_amd.ops.attention import _FlashAttention
+    torch.manual_seed(9)
+    b, h, s, d, p = 2, 2, 128, 128, 0.25
+    scale = d ** -0.5
+    q = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16,
+                    requires_grad=True)
+    k = torch.randn_like(q)
+    v = torch.randn_like(q)
+    out = _FlashAttention.apply(q, k, v, False, scale, p)
+    mask_words = out.grad_fn.saved_tensors[5].reshape(
+        b * h, s, (s + 31) // 32)
+    bits = torch.zeros(b * h, s, s, device="cuda")
+    for w in range((s + 31) // 32):
+        word = mask_words[:, :, w].unsqueeze(-1)
+        shifts = torch.arange(32, device="cuda")
+        bits[:, :, w * 32:(w + 1) * 32] = ((word >> shifts) & 1).float()
+    thresh = int(round(p * 256))
+    inv_keep = 256.0 / (256 - thresh)
+    sc = torch.bmm(q.detach().float().reshape(b * h, s, d),
+                   k.detach().float().reshape(b * h, s, d)
+                   .transpose(1, 2)) * scale
+    P = sc.softmax(-1) * bits * inv_keep
+    ref = torch.bmm(P, v.detach().float().reshape(b * h, s, d))
+    assert torch.allclose(out.float().reshape(b * h, s, d), ref,
+                          atol=4e-2), (out.float().reshape(-1)
+                                       - ref.reshape(-1)).abs().max()
