@@ -215,25 +215,30 @@ def test_flash_attention_dropout_exact(causal, p):
 def test_selfattention_dropout_stays_native():
     """SelfAttention with dropout>0 must stay on the native kernels in
     training mode (dropout silently left the native path)."""
+    import easyparallellibrary_amd.ops.attention as A
     from easyparallellibrary_amd.models.transformer import SelfAttention
     m = SelfAttention(256, 4, causal=True, dropout=0.1).to(
         "cuda", torch.bfloat16)
     m.train()
     x = torch.randn(2, 128, 256, device="cuda", dtype=torch.bfloat16,
                     requires_grad=True)
-    y = m(x)
-    # native path: the fused qkv Function appears in the autograd graph
-    seen, stack, found = set(), [y.grad_fn], False
-    while stack:
-        node = stack.pop()
-        if node is None or id(node) in seen:
-            continue
-        seen.add(id(node))
-        if "QKVFlashAttention" in type(node).__name__:
-            found = True
-            break
-        stack.extend(n for n, _ in getattr(node, "next_functions", ()))
-    assert found, "native fused-qkv node not in graph"
+    qkv_probe = m.qkv(x).reshape(2, 128, 3, 4, 64)
+    assert A.qkv_native_ok(qkv_probe), (
+        qkv_probe.is_contiguous(), qkv_probe.dtype, qkv_probe.shape)
+    called = {}
+    orig = A.qkv_flash_attention
+
+    def spy(qkv, causal=False, scale=None, dropout_p=0.0):
+        called["dropout_p"] = dropout_p
+        return orig(qkv, causal=causal, scale=scale, dropout_p=dropout_p)
+
+    A.qkv_flash_attention = spy
+    try:
+        y = m(x)
+    finally:
+        A.qkv_flash_attention = orig
+    assert called.get("dropout_p") == 0.1, (
+        called or "native fused-qkv path not taken")
     y.sum().backward()
     assert torch.isfinite(x.grad.float()).all()
 
