@@ -43,6 +43,8 @@ def parse_args():
     p.add_argument("--hidden", type=int, default=0,
                    help="override hidden size (moe/moe_pp smoke runs)")
     p.add_argument("--device", default=None)
+    p.add_argument("--dropout", type=float, default=0.0,
+                   help="attention dropout (runs in the native kernels)")
     return p.parse_args()
 
 
@@ -62,7 +64,8 @@ def build_bert_bench(args, epl, world, on_gpu, dtype):
         "zero.level": zero,
     }))
     vocab = 30528
-    model = bert.build_bert(args.model or "bert-large", vocab_size=vocab,
+    model = bert.build_bert(args.model or "bert-large", dropout=args.dropout,
+                            vocab_size=vocab,
                             max_pos=max(512, args.seq_len), num_stages=pp)
     loss_fn = ParallelCrossEntropy()
     engine = epl.Engine(model, loss_fn=loss_fn, optimizer="adamw", lr=1e-4,
@@ -98,7 +101,8 @@ def build_gpt2_bench(args, epl, world, on_gpu, dtype):
         "offload.level": offload,
     }))
     vocab = 50264
-    model = gpt2.build_gpt2(args.model or "gpt2-xl", vocab_size=vocab,
+    model = gpt2.build_gpt2(args.model or "gpt2-xl", dropout=args.dropout,
+                            vocab_size=vocab,
                             max_pos=seq, num_stages=pp)
     engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
                         optimizer="adamw", lr=1e-4, dtype=dtype)

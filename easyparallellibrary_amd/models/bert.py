@@ -26,11 +26,13 @@ BERT_CONFIGS = {
 class BertCore(nn.Module):
     """Single-module BERT (used when num_stages == 1)."""
 
-    def __init__(self, layers, hidden, heads, ffn, vocab_size, max_pos):
+    def __init__(self, layers, hidden, heads, ffn, vocab_size, max_pos,
+                 dropout=0.0):
         super().__init__()
         self.embeddings = Embeddings(vocab_size, hidden, max_pos)
         self.blocks = nn.ModuleList(
-            Block(hidden, heads, ffn, causal=False, pre_ln=False)
+            Block(hidden, heads, ffn, causal=False, pre_ln=False,
+                  dropout=dropout)
             for _ in range(layers))
         self.head = LMHead(hidden, vocab_size)
 
@@ -66,14 +68,15 @@ class _Stage(nn.Module):
 
 
 def build_bert(config="bert-large", vocab_size=30528, max_pos=512,
-               num_stages=1):
+               num_stages=1, dropout=0.0):
     """Build BERT under epl annotations.  vocab defaults to 30528
     (30522 rounded up to /8 for the fused CE kernel's vectorized path)."""
     cfg = BERT_CONFIGS[config] if isinstance(config, str) else dict(config)
     L, H, A, F = cfg["layers"], cfg["hidden"], cfg["heads"], cfg["ffn"]
     if num_stages <= 1:
         with epl.replicate(device_count=1, name="stage_0"):
-            model = BertCore(L, H, A, F, vocab_size, max_pos)
+            model = BertCore(L, H, A, F, vocab_size, max_pos,
+                             dropout=dropout)
         return init_weights(model)
     per = (L + num_stages - 1) // num_stages
     stages = []
@@ -84,7 +87,8 @@ def build_bert(config="bert-large", vocab_size=30528, max_pos=512,
             if s == 0:
                 mods.append(Embeddings(vocab_size, H, max_pos))
             n = min(per, L - layer_idx)
-            mods.extend(Block(H, A, F) for _ in range(n))
+            mods.extend(Block(H, A, F, dropout=dropout)
+                        for _ in range(n))
             layer_idx += n
             if s == num_stages - 1:
                 mods.append(LMHead(H, vocab_size))

@@ -19,12 +19,14 @@ GPT2_CONFIGS = {
 
 
 class GPT2Core(nn.Module):
-    def __init__(self, layers, hidden, heads, ffn, vocab_size, max_pos):
+    def __init__(self, layers, hidden, heads, ffn, vocab_size, max_pos,
+                 dropout=0.0):
         super().__init__()
         self.embeddings = Embeddings(vocab_size, hidden, max_pos,
                                      use_ln=False)
         self.blocks = nn.ModuleList(
-            Block(hidden, heads, ffn, causal=True, pre_ln=True)
+            Block(hidden, heads, ffn, causal=True, pre_ln=True,
+                  dropout=dropout)
             for _ in range(layers))
         self.head = LMHead(hidden, vocab_size, use_ln=True)
 
@@ -36,13 +38,14 @@ class GPT2Core(nn.Module):
 
 
 def build_gpt2(config="gpt2-xl", vocab_size=50264, max_pos=1024,
-               num_stages=1):
+               num_stages=1, dropout=0.0):
     """vocab 50257 rounded up to /8 for the vectorized CE kernel."""
     cfg = GPT2_CONFIGS[config] if isinstance(config, str) else dict(config)
     L, H, A, F = cfg["layers"], cfg["hidden"], cfg["heads"], cfg["ffn"]
     if num_stages <= 1:
         with epl.replicate(device_count=1, name="stage_0"):
-            model = GPT2Core(L, H, A, F, vocab_size, max_pos)
+            model = GPT2Core(L, H, A, F, vocab_size, max_pos,
+                             dropout=dropout)
         return init_weights(model)
     per = (L + num_stages - 1) // num_stages
     stages = []
@@ -53,7 +56,8 @@ def build_gpt2(config="gpt2-xl", vocab_size=50264, max_pos=1024,
             if s == 0:
                 mods.append(Embeddings(vocab_size, H, max_pos, use_ln=False))
             n = min(per, L - layer_idx)
-            mods.extend(Block(H, A, F, causal=True, pre_ln=True)
+            mods.extend(Block(H, A, F, causal=True, pre_ln=True,
+                              dropout=dropout)
                         for _ in range(n))
             layer_idx += n
             if s == num_stages - 1:

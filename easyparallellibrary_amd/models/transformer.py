@@ -4,8 +4,8 @@ equivalents the benchmarks run).
 
 Hot ops ride the hand-written CDNA4 kernels: FusedLayerNorm,
 FusedBiasGelu (FFN first Linear runs bias-free), and the fused-qkv flash
-attention (default for non-causal; causal routes to torch SDPA until the
-staged 64-row variants are measured — see _NATIVE_ATTN below).
+attention — DEFAULT for causal and non-causal, with in-kernel philox
+dropout (see _NATIVE_ATTN below).
 """
 
 import os
@@ -121,11 +121,12 @@ class Block(nn.Module):
     """Transformer block; pre_ln=True for GPT-style, False for BERT."""
 
     def __init__(self, hidden, num_heads, ffn_hidden, causal=False,
-                 pre_ln=False):
+                 pre_ln=False, dropout=0.0):
         super().__init__()
         self.pre_ln = pre_ln
         self.ln1 = FusedLayerNorm(hidden)
-        self.attn = SelfAttention(hidden, num_heads, causal=causal)
+        self.attn = SelfAttention(hidden, num_heads, causal=causal,
+                                  dropout=dropout)
         self.ln2 = FusedLayerNorm(hidden)
         self.mlp = MLP(hidden, ffn_hidden)
 
