@@ -86,3 +86,26 @@ def test_checkpoint_roundtrip_gpu(tmp_path):
     # backward is not bit-deterministic across runs
     assert all(abs(a - b) < 5e-2 for a, b in zip(cont, resumed)), (
         cont, resumed)
+
+
+def test_gpt2_tiny_gc_native_attention_trains():
+    """Gradient checkpointing + native attention: recompute must not
+    re-trigger saved-tensor unpacks (regression: ctx.saved_tensors
+    accessed twice in the Function backward broke under
+    torch.utils.checkpoint)."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models import gpt2
+    from easyparallellibrary_amd.ops.distributed_losses import (
+        ParallelCrossEntropy)
+    epl.init(epl.Config({"gradient_checkpoint.type": "auto"}))
+    torch.manual_seed(0)
+    model = gpt2.build_gpt2(dict(layers=3, hidden=256, heads=4, ffn=1024),
+                            vocab_size=1024, max_pos=128)
+    engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
+                        optimizer="adamw", lr=3e-4, dtype=torch.bfloat16)
+    ids = torch.randint(0, 1024, (4, 128), device=engine.device)
+    tgt = torch.randint(0, 1024, (4 * 128,), device=engine.device)
+    losses = [float(engine.train_step(ids, tgt)) for _ in range(4)]
+    torch.cuda.synchronize()
+    assert all(torch.isfinite(torch.tensor(losses)))
+    assert losses[-1] < losses[0], losses
