@@ -221,11 +221,15 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_fwd_kernel(
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
   const int lq = lane & 31;
-  const int64_t bh = blockIdx.y;
+  // causal: qb rides the Y axis with bh on X — dispatch is x-fastest,
+  // so EVERY batch-head of the longest q-block launches before the
+  // next-longest (global longest-first packing).  The old per-row
+  // reversal still launched one longest block in the LAST row — a
+  // ~1-block drain tail that left causal at ~35% packing efficiency
+  // at s4096 (profiles/r02_attention_ab.txt wave-residency probe).
+  const int64_t bh = causal ? (int64_t)blockIdx.x : (int64_t)blockIdx.y;
   const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
-  // causal: launch the LONGEST q-blocks first (work grows with q0) so
-  // the dispatch tail is short blocks, not 128-kv-tile ones
-  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
+  const int64_t qb = causal ? (int64_t)gridDim.y - 1 - blockIdx.y
                             : (int64_t)blockIdx.x;
   const int64_t q0_blk = qb * 128;
   const int64_t q0 = q0_blk + wave * 32;
@@ -617,8 +621,11 @@ __global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
   const int lkv = lane & 31;
-  const int64_t bh = blockIdx.y;
-  const int64_t kv0_blk = (int64_t)blockIdx.x * 128;
+  // causal: kv0 rides the Y axis (ascending = longest q-range first),
+  // bh on X — global longest-first packing like the fwd kernel
+  const int64_t bh = causal ? (int64_t)blockIdx.x : (int64_t)blockIdx.y;
+  const int64_t kv0_blk =
+      (causal ? (int64_t)blockIdx.y : (int64_t)blockIdx.x) * 128;
   const int64_t kv0 = kv0_blk + wave * 32;
   const bool active = kv0 < seq;
   const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
@@ -730,8 +737,9 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
   const int lkv = lane & 31;
-  const int64_t bh = blockIdx.y;
-  const int64_t kv0_blk = (int64_t)blockIdx.x * 128;
+  const int64_t bh = causal ? (int64_t)blockIdx.x : (int64_t)blockIdx.y;
+  const int64_t kv0_blk =
+      (causal ? (int64_t)blockIdx.y : (int64_t)blockIdx.x) * 128;
   const int64_t kv0 = kv0_blk + wave * 32;
   const bool active = kv0 < seq;
   const int64_t boff = (bh / heads) * in_sb + (bh % heads) * in_sh;
@@ -914,9 +922,9 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
   const int lq = lane & 31;
-  const int64_t bh = blockIdx.y;
-  // causal: longest q-blocks first (see attn_fwd_kernel)
-  const int64_t qb = causal ? (int64_t)gridDim.x - 1 - blockIdx.x
+  // causal: global longest-first packing (see attn_fwd_kernel)
+  const int64_t bh = causal ? (int64_t)blockIdx.x : (int64_t)blockIdx.y;
+  const int64_t qb = causal ? (int64_t)gridDim.y - 1 - blockIdx.y
                             : (int64_t)blockIdx.x;
   const int64_t q0_blk = qb * 128;
   const int64_t q0 = q0_blk + wave * 32;
@@ -1031,7 +1039,12 @@ void epl_attn_fwd(const void* q, const void* k, const void* v, void* out,
                   const int64_t* o_strides, unsigned int* drop_mask,
                   int64_t mask_w, unsigned long long seed, int drop_thresh,
                   float inv_keep, hipStream_t stream) {
-  dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
+  // causal: (x=bh, y=qblocks) — x-fastest dispatch launches every
+  // batch-head of the longest q-block first (kernels read the mapping
+  // from the causal flag)
+  const unsigned nqb = (unsigned)((seq + 127) / 128);
+  dim3 grid = causal ? dim3((unsigned)batch_heads, nqb)
+                     : dim3(nqb, (unsigned)batch_heads);
 #define FWD_ARGS                                                         \
   reinterpret_cast<const short*>(q), reinterpret_cast<const short*>(k),  \
       reinterpret_cast<const short*>(v), reinterpret_cast<short*>(out),  \
@@ -1065,10 +1078,12 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   const unsigned int* drop_mask, int64_t mask_w,
                   float inv_keep, int64_t head_dim, hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
-  dim3 grid((unsigned)((seq + 127) / 128), (unsigned)batch_heads);
-  // head_dim 128 always runs the split kernels (the combined dkdv
-  // kernel is d=64-only)
-  if (split_dkdv || drop_mask != nullptr || head_dim == 128) {
+  const unsigned nqb = (unsigned)((seq + 127) / 128);
+  dim3 grid = causal ? dim3((unsigned)batch_heads, nqb)
+                     : dim3(nqb, (unsigned)batch_heads);
+  // head_dim 128 and causal always run the split kernels (the combined
+  // dkdv kernel is d=64-only and keeps the legacy x-major grid)
+  if (split_dkdv || drop_mask != nullptr || head_dim == 128 || causal) {
     // order: dV (needs no delta) -> dQ (computes + publishes delta from
     // the dO/O rows it already loads) -> dK (consumes delta).  The prep
     // pass disappears.  Dropout always runs the split kernels (the
