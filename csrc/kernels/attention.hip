@@ -215,8 +215,8 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_fwd_kernel(
     int64_t o_sb, int64_t o_sh, int64_t o_ss,
     unsigned int* __restrict__ mask, int64_t mask_w,
     unsigned long long seed, int drop_thresh, float inv_keep) {
-  __shared__ short ldsV[D][72];        // V^T: [d][kv]
-  __shared__ short ldsK[64][D + 8];    // K: [kv][d]
+  __shared__ short ldsV[2][D][72];     // V^T: [d][kv], double-buffered
+  __shared__ short ldsK[2][64][D + 8];  // K: [kv][d]
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -258,22 +258,32 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_fwd_kernel(
   const int stage_kv = threadIdx.x & 63;
   const int stage_d0 = (threadIdx.x >> 6) * 8;
 
-  auto stage_tile = [&](int64_t kv0) {
-    __syncthreads();
+  // Single-barrier double-buffered staging: the NEXT tile's global
+  // loads issue right after the barrier and their latency hides under
+  // the CURRENT tile's MFMA/softmax; the ds_write that consumes them
+  // (start of the next iteration) is where the vmcnt wait lands.  One
+  // __syncthreads per 64-kv tile instead of two (PMC r2: 38% of
+  // wave-cycles were parked at barriers).
+  bf16x8 pv[D / 32], pk[D / 32];
+  auto load_tile = [&](int64_t kv0) {
     int64_t vrow = kv0 + stage_kv;
     if (vrow >= seq) vrow = seq - 1;   // masked columns never contribute
 #pragma unroll
     for (int h2 = 0; h2 < D / 32; ++h2) {
       const int sd = stage_d0 + h2 * 32;
-      bf16x8 vv = *reinterpret_cast<const bf16x8*>(
-          vp + vrow * in_ss + sd);
+      pv[h2] = *reinterpret_cast<const bf16x8*>(vp + vrow * in_ss + sd);
+      pk[h2] = *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
+    }
+  };
+  auto write_tile = [&](int b) {
+#pragma unroll
+    for (int h2 = 0; h2 < D / 32; ++h2) {
+      const int sd = stage_d0 + h2 * 32;
 #pragma unroll
       for (int j = 0; j < 8; ++j)
-        ldsV[sd + j][stage_kv] = vv[j];
-      *reinterpret_cast<bf16x8*>(&ldsK[stage_kv][sd]) =
-          *reinterpret_cast<const bf16x8*>(kp + vrow * in_ss + sd);
+        ldsV[b][sd + j][stage_kv] = pv[h2][j];
+      *reinterpret_cast<bf16x8*>(&ldsK[b][stage_kv][sd]) = pk[h2];
     }
-    __syncthreads();
   };
 
   unsigned int* maskrow =
@@ -282,18 +292,25 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_fwd_kernel(
   const unsigned int seed1 = (unsigned int)(seed >> 32);
   const unsigned int bh32 = (unsigned int)bh;
 
+  int buf = 0;
+  if (blk_kv_end > 0) load_tile(0);
   int64_t kv0 = 0;
   for (; kv0 < bulk_end; kv0 += 64) {
-    stage_tile(kv0);
+    write_tile(buf);
+    __syncthreads();
+    if (kv0 + 64 < blk_kv_end) load_tile(kv0 + 64);
 #pragma unroll
     for (int sub = 0; sub < 64; sub += 32)
-      fwd_tile<false, DROP, D>(ldsK, ldsV, sub, kv0 + sub, seq, myq,
-                               scale, causal, qfrag, ot, m, l, lq, hi,
-                               maskrow, mask_w, seed0, seed1, bh32,
+      fwd_tile<false, DROP, D>(ldsK[buf], ldsV[buf], sub, kv0 + sub, seq,
+                               myq, scale, causal, qfrag, ot, m, l, lq,
+                               hi, maskrow, mask_w, seed0, seed1, bh32,
                                drop_thresh, inv_keep);
+    buf ^= 1;
   }
   for (; kv0 < blk_kv_end; kv0 += 64) {
-    stage_tile(kv0);
+    write_tile(buf);
+    __syncthreads();
+    if (kv0 + 64 < blk_kv_end) load_tile(kv0 + 64);
     if (active) {
       const int64_t wave_kv_end = causal
           ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
@@ -301,12 +318,13 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_fwd_kernel(
       for (int sub = 0; sub < 64; sub += 32) {
         const int64_t kvs = kv0 + sub;
         if (kvs >= wave_kv_end) break;
-        fwd_tile<true, DROP, D>(ldsK, ldsV, sub, kvs, seq, myq, scale,
-                                causal, qfrag, ot, m, l, lq, hi,
+        fwd_tile<true, DROP, D>(ldsK[buf], ldsV[buf], sub, kvs, seq, myq,
+                                scale, causal, qfrag, ot, m, l, lq, hi,
                                 maskrow, mask_w, seed0, seed1, bh32,
                                 drop_thresh, inv_keep);
       }
     }
+    buf ^= 1;
   }
 
   if (!active || myq >= seq) return;
