@@ -55,7 +55,7 @@ void epl_attn_bwd(const void*, const void*, const void*, const void*,
                   int64_t, int64_t, float, bool, int64_t, const int64_t*,
                   const int64_t*, const int64_t*, const int64_t*, int,
                   const unsigned int*, int64_t, float, int64_t,
-                  hipStream_t);
+                  const float*, hipStream_t);
 }
 
 namespace {
@@ -322,7 +322,8 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
               at::Tensor dout, at::Tensor lse, at::Tensor delta_ws,
               at::Tensor dq, at::Tensor dk, at::Tensor dv, double scale,
               bool causal, bool split_dkdv,
-              c10::optional<at::Tensor> drop_mask, double inv_keep) {
+              c10::optional<at::Tensor> drop_mask, double inv_keep,
+              c10::optional<at::Tensor> dlse) {
   int64_t in_s[3], o_s[3], do_s[3], g_s[3], tmp[3];
   attn_strides(q, in_s, "q");
   attn_strides(out, o_s, "out");
@@ -344,12 +345,18 @@ void attn_bwd(at::Tensor q, at::Tensor k, at::Tensor v, at::Tensor out,
     check(*drop_mask, at::kInt, "drop_mask");
     mptr = reinterpret_cast<const unsigned int*>(drop_mask->data_ptr());
   }
+  const float* dlse_ptr = nullptr;
+  if (dlse.has_value()) {
+    check(*dlse, at::kFloat, "dlse");
+    TORCH_CHECK(dlse->numel() == bh * seq, "dlse size mismatch");
+    dlse_ptr = dlse->data_ptr<float>();
+  }
   epl_attn_bwd(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(),
                dout.data_ptr(), lse.data_ptr<float>(),
                delta_ws.data_ptr<float>(), dq.data_ptr(), dk.data_ptr(),
                dv.data_ptr(), bh, seq, (float)scale, causal, heads, in_s,
                o_s, do_s, g_s, split_dkdv ? 1 : 0, mptr, mask_w,
-               (float)inv_keep, q.size(3), cur_stream());
+               (float)inv_keep, q.size(3), dlse_ptr, cur_stream());
 }
 
 void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {

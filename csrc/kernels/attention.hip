@@ -933,7 +933,7 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
     int64_t in_ss, int64_t do_sb, int64_t do_sh, int64_t do_ss,
     int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
     int64_t o_ss, const unsigned int* __restrict__ mask, int64_t mask_w,
-    float inv_keep) {
+    float inv_keep, const float* __restrict__ dlse) {
   __shared__ short ldsK[64][D + 8];
   __shared__ short ldsVr[64][D + 8];
   const int lane = threadIdx.x & 63;
@@ -974,7 +974,12 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
   }
   // D_i = rowsum(dO * O): this lane covers 32 of the 64 d-elements, the
   // partner lane (^32) the rest
-  const float mydelta = dsum + __shfl_xor(dsum, 32, 64);
+  // dlse (ring-attention block merge): lse = logsumexp(S) has
+  // dlse_i/dS_ij = P_ij, so an incoming lse gradient folds into the
+  // per-row constant: dS = P o (dP - (D - dlse)).  The published delta
+  // is pre-adjusted so the dK kernel needs no change.
+  float mydelta = dsum + __shfl_xor(dsum, 32, 64);
+  if (dlse != nullptr) mydelta -= dlse[bh * seq + qrow];
   // publish delta for the dK kernel that follows on the same stream
   if (myq < seq && hi == 0)
     const_cast<float*>(delta)[bh * seq + myq] = mydelta;
@@ -1094,14 +1099,16 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                   const int64_t* o_strides, const int64_t* do_strides,
                   const int64_t* g_strides, int split_dkdv,
                   const unsigned int* drop_mask, int64_t mask_w,
-                  float inv_keep, int64_t head_dim, hipStream_t stream) {
+                  float inv_keep, int64_t head_dim, const float* dlse,
+                  hipStream_t stream) {
   const int64_t rows = batch_heads * seq;
   const unsigned nqb = (unsigned)((seq + 127) / 128);
   dim3 grid = causal ? dim3((unsigned)batch_heads, nqb)
                      : dim3(nqb, (unsigned)batch_heads);
   // head_dim 128 and causal always run the split kernels (the combined
   // dkdv kernel is d=64-only and keeps the legacy x-major grid)
-  if (split_dkdv || drop_mask != nullptr || head_dim == 128 || causal) {
+  if (split_dkdv || drop_mask != nullptr || head_dim == 128 || causal ||
+      dlse != nullptr) {
     // order: dV (needs no delta) -> dQ (computes + publishes delta from
     // the dO/O rows it already loads) -> dK (consumes delta).  The prep
     // pass disappears.  Dropout always runs the split kernels (the
@@ -1121,7 +1128,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
       seq, scale, causal ? 1 : 0, heads, in_strides[0], in_strides[1],   \
       in_strides[2], do_strides[0], do_strides[1], do_strides[2],        \
       g_strides[0], g_strides[1], g_strides[2], o_strides[0],            \
-      o_strides[1], o_strides[2], drop_mask, mask_w, inv_keep
+      o_strides[1], o_strides[2], drop_mask, mask_w, inv_keep, dlse
 #define DK_ARGS                                                          \
   reinterpret_cast<const short*>(q), reinterpret_cast<const short*>(k),  \
       reinterpret_cast<const short*>(v),                                 \
@@ -1203,7 +1210,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      in_strides[2], do_strides[0], do_strides[1],
                      do_strides[2], g_strides[0], g_strides[1],
                      g_strides[2], o_strides[0], o_strides[1],
-                     o_strides[2], drop_mask, mask_w, inv_keep);
+                     o_strides[2], drop_mask, mask_w, inv_keep, dlse);
 }
 
 }  // extern "C"

@@ -117,7 +117,7 @@ class _FlashAttention(torch.autograd.Function):
         split = os.environ.get("EPL_ATTN_BWD_SPLIT", "1") == "1"
         native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
                               ctx.scale, ctx.causal, split, mask,
-                              ctx.inv_keep)
+                              ctx.inv_keep, None)
         return dq, dk, dv, None, None, None
 
 
@@ -168,8 +168,70 @@ class _QKVFlashAttention(torch.autograd.Function):
         delta = torch.empty_like(lse)
         native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
                               ctx.scale, ctx.causal, True, mask,
-                              ctx.inv_keep)
+                              ctx.inv_keep, None)
         return dqkv, None, None, None
+
+
+class _FlashAttentionLse(torch.autograd.Function):
+    """Flash attention that ALSO returns the per-row lse (log-sum-exp of
+    the scaled scores) as a differentiable output — the building block
+    of the native ring-attention path, whose block merge weights depend
+    on lse.  The incoming lse gradient folds into the backward's
+    per-row delta (dS = P o (dP - (D - dlse)))."""
+
+    @staticmethod
+    def forward(ctx, q, k, v, causal, scale):
+        if not (_kernel_ok(q) and _kernel_ok(k) and _kernel_ok(v)
+                and q.stride() == k.stride() == v.stride()):
+            q, k, v = q.contiguous(), k.contiguous(), v.contiguous()
+        b, h, seq, d = q.shape
+        out = torch.empty(b, seq, h, d, dtype=q.dtype,
+                          device=q.device).permute(0, 2, 1, 3)
+        lse = torch.empty(b * h * seq, dtype=torch.float32,
+                          device=q.device)
+        native_ext().attn_fwd(q, k, v, out, lse, scale, causal, None,
+                              0, 0, 1.0)
+        ctx.save_for_backward(q, k, v, out, lse)
+        ctx.causal = causal
+        ctx.scale = scale
+        return out, lse.view(b, h, seq)
+
+    @staticmethod
+    def backward(ctx, dout, dlse):
+        q, k, v, out, lse = ctx.saved_tensors
+        if not _kernel_ok(dout):
+            dout = dout.contiguous()
+
+        def grad_like(t):
+            return torch.empty_strided(t.shape, t.stride(), dtype=t.dtype,
+                                       device=t.device)
+        dq, dk, dv = grad_like(q), grad_like(q), grad_like(q)
+        delta = torch.empty_like(lse)
+        dlse = (dlse.reshape(-1).contiguous().float()
+                if dlse is not None else None)
+        native_ext().attn_bwd(q, k, v, out, dout, lse, delta, dq, dk, dv,
+                              ctx.scale, ctx.causal, True, None, 1.0,
+                              dlse)
+        return dq, dk, dv, None, None
+
+
+def flash_attention_with_lse(q, k, v, causal=False, scale=None):
+    """(out, lse) — lse is [B, H, S] fp32 and differentiable.  Native
+    kernel on GPU bf16 d 64/128; fp32 torch fallback elsewhere (the
+    CPU tests of the ring path run through it)."""
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if use_native(q) and q.dtype == torch.bfloat16             and q.shape[-1] in (64, 128):
+        return _FlashAttentionLse.apply(q, k, v, causal, scale)
+    s = (q.float() @ k.float().transpose(-1, -2)) * scale
+    if causal:
+        seq = q.shape[-2]
+        maskt = torch.triu(torch.ones(seq, seq, device=q.device,
+                                      dtype=torch.bool), 1)
+        s = s.masked_fill(maskt, float("-inf"))
+    lse = torch.logsumexp(s, dim=-1)
+    out = ((s - lse.unsqueeze(-1)).exp() @ v.float()).to(q.dtype)
+    return out, lse
 
 
 def qkv_flash_attention(qkv, causal=False, scale=None, dropout_p=0.0):

@@ -10,17 +10,24 @@ partials against each K/V block as it arrives.  W-1 point-to-point
 hops overlap naturally with the partial-attention compute, and the
 sequence length per rank is bounded only by memory.
 
-The block math is the same online softmax the flash kernels use
-(running max m, running sum l, rescaled accumulator o); here it is
-expressed in differentiable torch ops so autograd derives the backward
-through the ring exchanges (_RingShift: send-next/recv-prev, whose
-adjoint is the opposite rotation).  On GPU the per-block matmuls ride
-hipBLASLt and the exchanges ride RCCL over xGMI; fusing the block loop
-into the flash kernels is queued for a later round.
+Each arriving K/V block is processed by the NATIVE flash kernels
+(ops/attention.flash_attention_with_lse: the forward also returns the
+per-row log-sum-exp as a differentiable output, and the backward folds
+the incoming lse gradient into its per-row delta), and the partial
+(out, lse) pairs merge with a streaming differentiable logaddexp —
+autograd derives the backward through both the merge and the ring
+exchanges (_RingShift: send-next/recv-prev, whose adjoint is the
+opposite rotation).  Fully-masked causal blocks are skipped on the
+compute side while the ring p2p stays in lockstep.  On CPU (or
+non-bf16/d!=64,128) the same structure runs through the fp32 torch
+fallback of flash_attention_with_lse, which is what the CPU exact-match
+tests exercise.
 """
 
 import torch
 import torch.nn as nn
+
+from easyparallellibrary_amd.ops.attention import flash_attention_with_lse
 
 
 class _RingShift(torch.autograd.Function):
@@ -54,57 +61,53 @@ def ring_shift(x, comm):
     return _RingShift.apply(x, comm)
 
 
-def _block_attention(q, k, v, scale, mask):
-    """Partial attention of q against one K/V block; returns the
-    un-normalized (m, l, o) online-softmax triple (all fp32)."""
-    s = (q.float() @ k.float().transpose(-1, -2)) * scale
-    if mask is not None:
-        s = s.masked_fill(mask, float("-inf"))
-    m = s.amax(dim=-1, keepdim=True)
-    # fully masked rows: keep m finite so exp() stays 0 without NaN
-    m = torch.where(torch.isfinite(m), m, torch.zeros_like(m))
-    p = (s - m).exp()
-    l = p.sum(dim=-1, keepdim=True)
-    o = p @ v.float()
-    return m, l, o
-
-
 def ring_attention(q, k, v, comm, causal=False, scale=None):
     """q/k/v: [b, h, s_local, d] — each rank's sequence block.  Returns
-    [b, h, s_local, d].  Requires equal block sizes on every rank."""
+    [b, h, s_local, d].  Requires equal block sizes on every rank.
+
+    Per-block compute = flash_attention_with_lse (native CDNA4 kernels
+    on GPU); partials merge with a streaming differentiable logaddexp:
+      lse' = logaddexp(lse, lse_b)
+      o'   = o * exp(lse - lse') + o_b * exp(lse_b - lse')
+    Under causal masking, block src == rank runs the kernel's causal
+    path (the diagonal aligns because block sizes are equal), earlier
+    blocks run unmasked, and later (fully-masked) blocks are SKIPPED on
+    the compute side — the ring_shift p2p still runs on every rank so
+    the ring stays in lockstep."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
     w = comm.size if comm is not None else 1
     rank = comm.rank if comm is not None else 0
-    sl = q.shape[-2]
 
-    m = l = o = None
+    lse = o = dead = None
     kb, vb = k, v
     for step in range(w):
         src = (rank - step) % w            # whose K/V block we hold now
-        mask = None
-        if causal:
-            qi = torch.arange(rank * sl, (rank + 1) * sl,
-                              device=q.device).unsqueeze(-1)
-            ki = torch.arange(src * sl, (src + 1) * sl,
-                              device=q.device).unsqueeze(0)
-            mask = ki > qi                  # future keys masked
-            # fully masked blocks still run (zero contribution) so the
-            # ring stays in lockstep on every rank
-        mb, lb, ob = _block_attention(q, kb, vb, scale, mask)
-        if m is None:
-            m, l, o = mb, lb, ob
+        if not (causal and src > rank):    # future blocks: skip compute
+            ob, lseb = flash_attention_with_lse(
+                q, kb, vb, causal=(causal and src == rank), scale=scale)
+            ob = ob.float()
+            if o is None:
+                o, lse = ob, lseb
+            else:
+                lse_new = torch.logaddexp(lse, lseb)
+                o = (o * (lse - lse_new).exp().unsqueeze(-1)
+                     + ob * (lseb - lse_new).exp().unsqueeze(-1))
+                lse = lse_new
         else:
-            m_new = torch.maximum(m, mb)
-            a = (m - m_new).exp()
-            b_ = (mb - m_new).exp()
-            l = l * a + lb * b_
-            o = o * a + ob * b_
-            m = m_new
+            # zero-valued contribution keeps the skipped block in the
+            # autograd graph, so every rank runs the SAME chain of
+            # ring-shift backwards (the backward p2p must stay in
+            # lockstep even for blocks that contribute nothing here)
+            z = (kb.sum() + vb.sum()) * 0
+            dead = z if dead is None else dead + z
         if step < w - 1:
             kb = ring_shift(kb, comm)
             vb = ring_shift(vb, comm)
-    return (o / l.clamp_min(1e-20)).to(q.dtype)
+    out = o.to(q.dtype)
+    if dead is not None:
+        out = out + dead.to(out.dtype)
+    return out
 
 
 class RingSelfAttention(nn.Module):
