@@ -221,7 +221,10 @@ def flash_attention_with_lse(q, k, v, causal=False, scale=None):
     CPU tests of the ring path run through it)."""
     if scale is None:
         scale = q.shape[-1] ** -0.5
-    if use_native(q) and q.dtype == torch.bfloat16             and q.shape[-1] in (64, 128):
+    if (use_native(q) and q.dtype == torch.bfloat16
+            and q.shape[-1] in (64, 128) and q.shape == k.shape):
+        # the kernels assume equal q/kv lengths (the ring contract);
+        # unequal blocks take the fp32 fallback below
         return _FlashAttentionLse.apply(q, k, v, causal, scale)
     s = (q.float() @ k.float().transpose(-1, -2)) * scale
     if causal:

@@ -325,14 +325,17 @@ def test_flash_attention_with_lse_block_merge(d):
                     requires_grad=True)
     k = torch.randn_like(q, requires_grad=True)
     v = torch.randn_like(q, requires_grad=True)
-    dout = torch.randn_like(q)
+    # q is ONE ring block (the kernel takes equal q/kv lengths, exactly
+    # the ring-attention contract); kv rotates through 4 blocks
+    qb = q[:, :, :128].detach().clone().requires_grad_(True)
+    dout = torch.randn(b, h, 128, d, device="cuda",
+                       dtype=torch.bfloat16)
 
-    # blocked: 4 kv blocks merged
     o = lse = None
     for i in range(4):
         kb = k[:, :, i * 128:(i + 1) * 128].contiguous()
         vb = v[:, :, i * 128:(i + 1) * 128].contiguous()
-        ob, lseb = flash_attention_with_lse(q, kb, vb, causal=False,
+        ob, lseb = flash_attention_with_lse(qb, kb, vb, causal=False,
                                             scale=scale)
         ob = ob.float()
         if o is None:
@@ -343,10 +346,10 @@ def test_flash_attention_with_lse_block_merge(d):
                  + ob * (lseb - lse_new).exp().unsqueeze(-1))
             lse = lse_new
     o.backward(dout.float())
-    gq, gk, gv = q.grad.clone(), k.grad.clone(), v.grad.clone()
+    gq, gk, gv = qb.grad.clone(), k.grad.clone(), v.grad.clone()
 
-    # reference: fp32 torch on the full kv
-    qf = q.detach().float().requires_grad_(True)
+    # reference: fp32 torch, q block vs the full kv
+    qf = qb.detach().float().requires_grad_(True)
     kf = k.detach().float().requires_grad_(True)
     vf = v.detach().float().requires_grad_(True)
     ref = ref_attention(qf, kf, vf, False, scale)
