@@ -58,6 +58,16 @@ class Layout:
     A *replica* needs ``sum(device_counts)`` ranks; the number of replicas is
     ``world_size // per_replica`` (reference AutoLayout, epl/cluster.py:146-159).
 
+    Deliberately NOT ported: the reference's AwareRowLayout host
+    reordering (epl/cluster.py:169-241) permutes worker rows so
+    cross-HOST pipeline edges minimize inter-node traffic.  This
+    framework targets one 8-GPU MI355X node where every pair of GPUs is
+    one xGMI hop, so host-aware reordering has nothing to optimize;
+    multi-node jobs should order ranks host-contiguously at launch
+    (the launcher already assigns ranks per worker in order), which
+    yields the same placement AwareRowLayout computes for the
+    homogeneous case.
+
     Ranks are laid out replica-major: replica ``r`` owns the contiguous range
     ``[r*per_replica, (r+1)*per_replica)``; inside a replica, taskgraphs are
     packed in order.  This keeps each replica's pipeline chain on adjacent
