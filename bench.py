@@ -39,9 +39,9 @@ def parse_args():
     p.add_argument("--gc", default=None, choices=[None, "", "auto"])
     p.add_argument("--offload", default=None, choices=[None, "", "v0"])
     p.add_argument("--layers", type=int, default=0,
-                   help="override model depth (moe/moe_pp smoke runs)")
+                   help="override model depth (smoke runs)")
     p.add_argument("--hidden", type=int, default=0,
-                   help="override hidden size (moe/moe_pp smoke runs)")
+                   help="override hidden size (smoke runs)")
     p.add_argument("--device", default=None)
     p.add_argument("--dropout", type=float, default=0.0,
                    help="attention dropout (runs in the native kernels)")
@@ -64,7 +64,13 @@ def build_bert_bench(args, epl, world, on_gpu, dtype):
         "zero.level": zero,
     }))
     vocab = 30528
-    model = bert.build_bert(args.model or "bert-large", dropout=args.dropout,
+    cfg = args.model or "bert-large"
+    if args.layers or args.hidden:   # CPU/world-N smoke of this exact path
+        h = args.hidden or 256
+        cfg = dict(layers=args.layers or 2, hidden=h,
+                   heads=max(2, h // 64), ffn=4 * h)
+        vocab = 2048
+    model = bert.build_bert(cfg, dropout=args.dropout,
                             vocab_size=vocab,
                             max_pos=max(512, args.seq_len), num_stages=pp)
     loss_fn = ParallelCrossEntropy()
@@ -79,7 +85,9 @@ def build_bert_bench(args, epl, world, on_gpu, dtype):
         par += "_pp{}".format(pp)
     if zero:
         par += "_zero_{}".format(zero)
-    meta = {"model": args.model or "bert-large",
+    meta = {"model": (args.model or "bert-large")
+            if not (args.layers or args.hidden) else
+            "bert-smoke-{}L".format(args.layers or 2),
             "global_batch": batch * engine.num_replicas,
             "seq_len": args.seq_len, "parallelism": par}
     return engine, (ids, tgt), batch * engine.num_replicas, meta
@@ -101,7 +109,13 @@ def build_gpt2_bench(args, epl, world, on_gpu, dtype):
         "offload.level": offload,
     }))
     vocab = 50264
-    model = gpt2.build_gpt2(args.model or "gpt2-xl", dropout=args.dropout,
+    cfg = args.model or "gpt2-xl"
+    if args.layers or args.hidden:   # CPU/world-N smoke of this exact path
+        h = args.hidden or 256
+        cfg = dict(layers=args.layers or 2, hidden=h,
+                   heads=max(2, h // 64), ffn=4 * h)
+        vocab = 2048
+    model = gpt2.build_gpt2(cfg, dropout=args.dropout,
                             vocab_size=vocab,
                             max_pos=seq, num_stages=pp)
     engine = epl.Engine(model, loss_fn=ParallelCrossEntropy(),
@@ -116,7 +130,9 @@ def build_gpt2_bench(args, epl, world, on_gpu, dtype):
         par += "_gc"
     if offload:
         par += "_offload"
-    meta = {"model": args.model or "gpt2-xl",
+    meta = {"model": (args.model or "gpt2-xl")
+            if not (args.layers or args.hidden) else
+            "gpt2-smoke-{}L".format(args.layers or 2),
             "global_batch": batch * engine.num_replicas,
             "seq_len": seq, "parallelism": par}
     return engine, (ids, tgt), batch * engine.num_replicas, meta
