@@ -68,3 +68,18 @@ tot_fl = sum(f for f, _ in allr)
 tot_t = sum(t for _, t in allr)
 print(f"\nAGGREGATE over all hot GEMM shapes: {tot_fl/tot_t/1e12:.0f} TF "
       f"= {100*tot_fl/tot_t/1e12/PEAK:.1f}% of 2.5 PF dense bf16 peak")
+
+
+# wgrad-TN probe: is the slow TN layout hipBLASLt's kernel choice, or
+# would an explicit transpose-copy + NN GEMM beat it?
+print("\n== wgrad TN vs copy+NN probe ==")
+for (tag, m, k, n) in (("bert proj", 65536, 1024, 1024),
+                       ("bert qkv", 65536, 1024, 3072),
+                       ("gpt2 fc2", 16384, 6400, 1600)):
+    x = torch.randn(m, k, device="cuda", dtype=torch.bfloat16)
+    dy = torch.randn(m, n, device="cuda", dtype=torch.bfloat16)
+    fl = 2.0 * m * k * n
+    t_tn = bench(lambda: dy.t() @ x)
+    t_cp = bench(lambda: dy.t().contiguous() @ x)
+    print(f"{tag:12s} TN {fl/t_tn/1e12:6.0f} TF | transpose-copy+NN "
+          f"{fl/t_cp/1e12:6.0f} TF")
