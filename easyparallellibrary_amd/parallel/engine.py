@@ -753,6 +753,12 @@ class Engine:
     # ---- checkpoint ----------------------------------------------------------
     def save_checkpoint(self, path, save_optimizer=True):
         from easyparallellibrary_amd.runtime import saver
+        # mixed-width stages replicate params across positions with no
+        # per-step sync; surface any nondeterministic-kernel drift at
+        # checkpoint time (advisor r1) — warn-only, collective
+        if any(getattr(tg, "replicated_io", False)
+               for tg in self.plan.taskgraphs):
+            self.check_param_consistency()
         saver.save_checkpoint(self, path, save_optimizer=save_optimizer)
 
     def load_checkpoint(self, path, load_optimizer=True, assign_map=None,
