@@ -24,13 +24,18 @@ class _FusedBiasGelu(torch.autograd.Function):
     def forward(ctx, x, bias):
         x = x.contiguous()
         cols = x.shape[-1]
+        # under autocast the activation is bf16 while the bias PARAM
+        # stays fp32 — the kernel derives its dtype from x, so cast the
+        # bias (param-dtype grads are restored in backward)
+        ctx.param_dtype = bias.dtype
+        kbias = bias.contiguous().to(x.dtype)
         if use_native(x) and x.dtype in (torch.bfloat16, torch.float32) \
                 and cols % 8 == 0:
             out = torch.empty_like(x)
-            native_ext().bias_gelu_fwd(out, x, bias.contiguous())
+            native_ext().bias_gelu_fwd(out, x, kbias)
         else:
-            out = _gelu_tanh((x.float() + bias.float())).to(x.dtype)
-        ctx.save_for_backward(x, bias)
+            out = _gelu_tanh((x.float() + kbias.float())).to(x.dtype)
+        ctx.save_for_backward(x, kbias)
         return out
 
     @staticmethod
@@ -42,7 +47,7 @@ class _FusedBiasGelu(torch.autograd.Function):
                 and cols % 8 == 0:
             dx = torch.empty_like(x)
             dbias = torch.zeros(cols, dtype=torch.float32, device=x.device)
-            native_ext().bias_gelu_bwd(dx, dbias, dy, x, bias.contiguous())
+            native_ext().bias_gelu_bwd(dx, dbias, dy, x, bias)
         else:
             xb = x.float() + bias.float()
             t = torch.tanh(0.7978845608028654 * (xb + 0.044715 * xb ** 3))
@@ -50,7 +55,7 @@ class _FusedBiasGelu(torch.autograd.Function):
                 0.7978845608028654 * (1 + 3 * 0.044715 * xb * xb)
             dx = (dy.float() * dgelu).to(x.dtype)
             dbias = dx.float().reshape(-1, cols).sum(dim=0)
-        return dx, dbias.to(bias.dtype)
+        return dx, dbias.to(ctx.param_dtype)
 
 
 class FusedBiasGelu(nn.Module):

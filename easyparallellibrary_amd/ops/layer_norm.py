@@ -35,15 +35,19 @@ class _FusedLayerNorm(torch.autograd.Function):
         rows = x.numel() // cols
         has_res = residual is not None
         if has_res:
-            residual = residual.contiguous()
+            # the kernel derives its dtype from x; autocast can hand us
+            # a bf16 x with an fp32 residual stream (or vice versa)
+            residual = residual.contiguous().to(x.dtype)
+        ctx.param_dtype = gamma.dtype
+        gamma = gamma.contiguous().to(x.dtype)
+        beta = beta.contiguous().to(x.dtype)
         if _native_ok(x):
             mean = torch.empty(rows, dtype=torch.float32, device=x.device)
             rstd = torch.empty(rows, dtype=torch.float32, device=x.device)
             out = torch.empty_like(x)
             s = torch.empty_like(x) if has_res else None
-            native_ext().layer_norm_fwd(out, x, residual, s,
-                                        gamma.contiguous(),
-                                        beta.contiguous(), mean, rstd, eps)
+            native_ext().layer_norm_fwd(out, x, residual, s, gamma,
+                                        beta, mean, rstd, eps)
             norm_in = s if has_res else x
         else:
             norm_in = (x + residual) if has_res else x
@@ -93,8 +97,10 @@ class _FusedLayerNorm(torch.autograd.Function):
             dx = dx + ds
         if ctx.has_res:
             # d(x) == d(residual) — the add distributes the gradient
-            return dx, dx, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
-        return dx, None, dgamma.to(gamma.dtype), dbeta.to(gamma.dtype), None
+            return (dx, dx, dgamma.to(ctx.param_dtype),
+                    dbeta.to(ctx.param_dtype), None)
+        return (dx, None, dgamma.to(ctx.param_dtype),
+                dbeta.to(ctx.param_dtype), None)
 
 
 class FusedLayerNorm(nn.Module):
