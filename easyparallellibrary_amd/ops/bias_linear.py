@@ -62,7 +62,11 @@ class FusedBiasLinear(nn.Linear):
     (falls back to standard autograd on CPU / unsupported shapes)."""
 
     def forward(self, x):
+        # x.dtype == weight.dtype guards the AMP-O1 case (bf16 autocast
+        # activations over fp32 params): the custom backward's GEMMs
+        # would mix dtypes — let autocast's own F.linear handle it
         if (self.bias is not None and use_native(x)
-                and self.out_features % 8 == 0):
+                and self.out_features % 8 == 0
+                and x.dtype == self.weight.dtype):
             return _LinearFusedBiasGrad.apply(x, self.weight, self.bias)
         return F.linear(x, self.weight, self.bias)
