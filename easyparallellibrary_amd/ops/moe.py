@@ -28,13 +28,14 @@ class _BatchedExpertLinear(torch.autograd.Function):
     """Per-expert batched GEMM with a hand-written backward that issues
     ONLY plain (NN, contiguous) torch.bmm calls.
 
-    torch.bmm's own backward runs hipBLASLt strided-batched GEMMs on
-    TRANSPOSED VIEWS, and that configuration memory-faults on
+    torch.bmm's own autograd backward memory-faults on
     ROCm 7.0.x/gfx950 (still reproduced on 7.0.2 —
     gpurun_out/r2_moe_bmm2.txt, repro tests/moe_bisect_gpu.py bmm2).
-    Materializing the transposes and calling NN bmm dodges the faulting
-    kernel while keeping one grouped GEMM launch per matmul instead of
-    the per-expert 2D loop."""
+    Issuing the backward GEMMs OURSELVES as forward bmm calls (with
+    transposed-view operands — probed fault-free and faster than
+    transpose copies, tests/moe_bmm_layout_probe_gpu.py) dodges it
+    while keeping one grouped GEMM launch per matmul instead of the
+    per-expert 2D loop."""
 
     @staticmethod
     def forward(ctx, x, w):
@@ -45,8 +46,13 @@ class _BatchedExpertLinear(torch.autograd.Function):
     def backward(ctx, go):
         x, w = ctx.saved_tensors
         go = go.contiguous()
-        gx = torch.bmm(go, w.transpose(1, 2).contiguous())
-        gw = torch.bmm(x.transpose(1, 2).contiguous(), go)
+        # transposed VIEWS: the r2 layout probe
+        # (tests/moe_bmm_layout_probe_gpu.py) shows forward bmm with
+        # NT/TN view operands is fault-free on this stack and ~2x the
+        # transpose-copy path (the fault lives only in torch.bmm's own
+        # autograd backward); the copies cost ~8.5 ms/step at 12 layers
+        gx = torch.bmm(go, w.transpose(1, 2))
+        gw = torch.bmm(x.transpose(1, 2), go)
         return gx, gw
 
 
