@@ -25,8 +25,7 @@ def _moe_bmm():
 
 
 class _BatchedExpertLinear(torch.autograd.Function):
-    """Per-expert batched GEMM with a hand-written backward that issues
-    ONLY plain (NN, contiguous) torch.bmm calls.
+    """Per-expert batched GEMM with a hand-written backward.
 
     torch.bmm's own autograd backward memory-faults on
     ROCm 7.0.x/gfx950 (still reproduced on 7.0.2 —
@@ -155,10 +154,8 @@ class ExpertParallelMLP(nn.Module):
         d = d.transpose(0, 1).reshape(self.local_experts,
                                       self.world * capacity, self.hidden)
         # expert FFN: batched GEMM (one hipBLASLt grouped call per
-        # matmul) by default; EPL_MOE_BMM=0 falls back to the per-expert
-        # 2D-GEMM loop that worked around a bf16 batched-GEMM BACKWARD
-        # memory fault on ROCm 7.0/gfx950 (tests/moe_bisect_gpu.py bmm2;
-        # re-tested fixed on the round-2 ROCm 7.0.2 stack)
+        # matmul, hand-written backward above) by default; EPL_MOE_BMM=0
+        # falls back to the per-expert 2D-GEMM loop
         if _moe_bmm():
             he = F.gelu(_BatchedExpertLinear.apply(d.contiguous(), self.w1))
             h = _BatchedExpertLinear.apply(he, self.w2)
