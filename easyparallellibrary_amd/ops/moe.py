@@ -45,12 +45,16 @@ class _BatchedExpertLinear(torch.autograd.Function):
     def backward(ctx, go):
         x, w = ctx.saved_tensors
         go = go.contiguous()
-        # transposed VIEWS: the r2 layout probe
-        # (tests/moe_bmm_layout_probe_gpu.py) shows forward bmm with
-        # NT/TN view operands is fault-free on this stack and ~2x the
-        # transpose-copy path (the fault lives only in torch.bmm's own
-        # autograd backward); the copies cost ~8.5 ms/step at 12 layers
-        gx = torch.bmm(go, w.transpose(1, 2))
+        # transposed VIEWS where the layout is safe (probed per-GEMM in
+        # tests/moe_bmm_layout_probe_gpu.py): the ONE faulting hipBLASLt
+        # config on this stack is NT with a narrow inner dim
+        # ([E,C,1024] @ [E,1024,4096]-view) — exactly gx of the second
+        # expert linear; that operand alone gets a contiguous copy
+        # (~175 us/layer) while the other three GEMMs ride views
+        wt = w.transpose(1, 2)
+        if wt.shape[-1] > wt.shape[-2]:   # widening NT: the faulting one
+            wt = wt.contiguous()
+        gx = torch.bmm(go, wt)
         gw = torch.bmm(x.transpose(1, 2), go)
         return gx, gw
 
