@@ -127,6 +127,11 @@ class ExpertParallelMLP(nn.Module):
         capacity = max(
             1, int(self.capacity_factor * n_tokens * self.top_k /
                    self.num_experts))
+        # dispatch tensor: [num_experts, capacity, hidden]
+        dispatched = x.new_zeros(self.num_experts, capacity, self.hidden)
+        combine_idx = []  # (expert, slot, token, weight)
+        slot_count = torch.zeros(self.num_experts, dtype=torch.long,
+                                 device=x.device)
         flat_e = topi.reshape(-1)
         flat_t = (torch.arange(n_tokens, device=x.device)
                   .repeat_interleave(self.top_k))
@@ -138,24 +143,13 @@ class ExpertParallelMLP(nn.Module):
         counts = torch.bincount(fe, minlength=self.num_experts)
         seg_start = torch.nn.functional.pad(counts.cumsum(0), (1, 0))[:-1]
         pos_in_e = torch.arange(fe.numel(), device=x.device) - seg_start[fe]
-        # SYNC-FREE capacity handling: no boolean select (nonzero forces
-        # a device->host size sync per layer — the op profile showed the
-        # step partly CPU-bound on it).  Over-capacity assignments write
-        # a TRASH SLOT (index `capacity`) that is sliced off before the
-        # a2a — the slice's autograd pads zero grads there, so the
-        # duplicated trash writes backprop exactly zero — and their
-        # combine weight is zeroed by the `valid` mask.
-        valid = pos_in_e < capacity
-        pos_c = pos_in_e.clamp_max_(capacity)
-        fw = fw * valid
-        # dispatch tensor: [num_experts, capacity(+trash), hidden]
-        dispatched = x.new_zeros(self.num_experts, capacity + 1,
-                                 self.hidden)
-        dispatched[fe, pos_c] = x[ft]
+        keep = pos_in_e < capacity
+        fe, ft, fw, pos_in_e = fe[keep], ft[keep], fw[keep], pos_in_e[keep]
+        dispatched[fe, pos_in_e] = x[ft]
 
         # all-to-all: [world, local_experts*capacity, hidden]
-        d = dispatched[:, :capacity].reshape(
-            self.world, self.local_experts * capacity, self.hidden)
+        d = dispatched.reshape(self.world,
+                               self.local_experts * capacity, self.hidden)
         if self.comm is not None and self.world > 1:
             d = functional.all_to_all(d.contiguous(), self.comm,
                                       compress=self._wire_compression())
@@ -184,7 +178,5 @@ class ExpertParallelMLP(nn.Module):
         h = h.reshape(self.num_experts, capacity, self.hidden)
 
         out = x.new_zeros(n_tokens, self.hidden)
-        # gather with the clamped slot (trash gathers are weighted 0)
-        pos_g = pos_c.clamp_max(capacity - 1)
-        out.index_add_(0, ft, h[fe, pos_g] * fw.unsqueeze(-1).to(h.dtype))
+        out.index_add_(0, ft, h[fe, pos_in_e] * fw.unsqueeze(-1).to(h.dtype))
         return out.reshape(orig_shape)
