@@ -45,6 +45,19 @@ void epl_colsum(void*, const void*, float*, int64_t, int64_t, int64_t,
                 bool, hipStream_t);
 void run_mfma_probe(const unsigned short*, const unsigned short*, float*,
                     hipStream_t);
+void epl_moe_dispatch_fwd(void*, const void*, const int64_t*,
+                          const int64_t*, const int64_t*, int64_t, int64_t,
+                          int64_t, hipStream_t);
+void epl_moe_dispatch_bwd(void*, const void*, const int64_t*,
+                          const int64_t*, const int64_t*, int64_t, int64_t,
+                          int64_t, int64_t, hipStream_t);
+void epl_moe_combine_fwd(void*, const void*, const float*, const int64_t*,
+                         const int64_t*, const int64_t*, int64_t, int64_t,
+                         int64_t, int64_t, hipStream_t);
+void epl_moe_combine_bwd(void*, float*, const void*, const void*,
+                         const float*, const int64_t*, const int64_t*,
+                         const int64_t*, int64_t, int64_t, int64_t,
+                         hipStream_t);
 void epl_attn_fwd(const void*, const void*, const void*, void*, float*,
                   int64_t, int64_t, float, bool, int64_t, int64_t,
                   const int64_t*,
@@ -368,6 +381,69 @@ void mfma_probe(at::Tensor A, at::Tensor B, at::Tensor D) {
                  D.data_ptr<float>(), cur_stream());
 }
 
+static void check_idx(const at::Tensor& t, const char* name) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+              t.scalar_type() == at::kLong, name,
+              " must be a contiguous int64 device tensor");
+}
+
+void moe_dispatch_fwd(at::Tensor disp, at::Tensor x, at::Tensor fe,
+                      at::Tensor pos, at::Tensor ft, int64_t cap) {
+  check(disp, at::kBFloat16, "disp");
+  check(x, at::kBFloat16, "x");
+  check_idx(fe, "fe"); check_idx(pos, "pos"); check_idx(ft, "ft");
+  const int64_t hidden = x.size(-1);
+  TORCH_CHECK(hidden % 8 == 0, "hidden % 8");
+  epl_moe_dispatch_fwd(disp.data_ptr(), x.data_ptr(),
+                       fe.data_ptr<int64_t>(), pos.data_ptr<int64_t>(),
+                       ft.data_ptr<int64_t>(), fe.numel(), hidden, cap,
+                       cur_stream());
+}
+
+void moe_dispatch_bwd(at::Tensor dx, at::Tensor ddisp, at::Tensor fe,
+                      at::Tensor pos, at::Tensor inv, int64_t k,
+                      int64_t cap) {
+  check(dx, at::kBFloat16, "dx");
+  check(ddisp, at::kBFloat16, "ddisp");
+  check_idx(fe, "fe"); check_idx(pos, "pos"); check_idx(inv, "inv");
+  const int64_t hidden = dx.size(-1);
+  epl_moe_dispatch_bwd(dx.data_ptr(), ddisp.data_ptr(),
+                       fe.data_ptr<int64_t>(), pos.data_ptr<int64_t>(),
+                       inv.data_ptr<int64_t>(), dx.numel() / hidden, k,
+                       hidden, cap, cur_stream());
+}
+
+void moe_combine_fwd(at::Tensor out, at::Tensor h, at::Tensor fw,
+                     at::Tensor fe, at::Tensor pos, at::Tensor inv,
+                     int64_t k, int64_t cap) {
+  check(out, at::kBFloat16, "out");
+  check(h, at::kBFloat16, "h");
+  check(fw, at::kFloat, "fw");
+  check_idx(fe, "fe"); check_idx(pos, "pos"); check_idx(inv, "inv");
+  const int64_t hidden = out.size(-1);
+  epl_moe_combine_fwd(out.data_ptr(), h.data_ptr(), fw.data_ptr<float>(),
+                      fe.data_ptr<int64_t>(), pos.data_ptr<int64_t>(),
+                      inv.data_ptr<int64_t>(), out.numel() / hidden, k,
+                      hidden, cap, cur_stream());
+}
+
+void moe_combine_bwd(at::Tensor dh, at::Tensor dfw, at::Tensor dout,
+                     at::Tensor h, at::Tensor fw, at::Tensor fe,
+                     at::Tensor pos, at::Tensor ft, int64_t cap) {
+  check(dh, at::kBFloat16, "dh");
+  check(dout, at::kBFloat16, "dout");
+  check(h, at::kBFloat16, "h");
+  check(fw, at::kFloat, "fw");
+  check(dfw, at::kFloat, "dfw");
+  check_idx(fe, "fe"); check_idx(pos, "pos"); check_idx(ft, "ft");
+  const int64_t hidden = dh.size(-1);
+  epl_moe_combine_bwd(dh.data_ptr(), dfw.data_ptr<float>(),
+                      dout.data_ptr(), h.data_ptr(), fw.data_ptr<float>(),
+                      fe.data_ptr<int64_t>(), pos.data_ptr<int64_t>(),
+                      ft.data_ptr<int64_t>(), fe.numel(), hidden, cap,
+                      cur_stream());
+}
+
 void colsum(at::Tensor dy, at::Tensor db, at::Tensor partial) {
   const bool bf16 = is_bf16(dy);
   TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 2,
@@ -408,6 +484,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("bf16_to_f32", &bf16_to_f32);
   m.def("sqnorm", &sqnorm);
   m.def("mfma_probe", &mfma_probe);
+  m.def("moe_dispatch_fwd", &moe_dispatch_fwd);
+  m.def("moe_dispatch_bwd", &moe_dispatch_bwd);
+  m.def("moe_combine_fwd", &moe_combine_fwd);
+  m.def("moe_combine_bwd", &moe_combine_bwd);
   m.def("attn_fwd", &attn_fwd);
   m.def("attn_bwd", &attn_bwd);
 }

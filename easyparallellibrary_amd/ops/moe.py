@@ -17,11 +17,68 @@ import torch.nn as nn
 import torch.nn.functional as F
 
 from easyparallellibrary_amd.comm import functional
+from easyparallellibrary_amd.ops.dispatch import native_ext, use_native
 
 
 def _moe_bmm():
     import os
     return os.environ.get("EPL_MOE_BMM", "1") == "1"
+
+
+def _moe_native_dispatch():
+    import os
+    return os.environ.get("EPL_MOE_NATIVE_DISPATCH", "1") == "1"
+
+
+class _MoEDispatch(torch.autograd.Function):
+    """Fused token dispatch (csrc/kernels/moe.hip): one wave copies one
+    hidden row per kept assignment; over-capacity entries are skipped
+    INLINE (no boolean compaction, no nonzero device->host sync).
+    Backward gathers each token's k slot-gradients (dropped -> zero)."""
+
+    @staticmethod
+    def forward(ctx, x, fe, pos, ft, inv, k, capacity, num_experts):
+        disp = x.new_zeros(num_experts, capacity, x.shape[-1])
+        native_ext().moe_dispatch_fwd(disp, x, fe, pos, ft, capacity)
+        ctx.save_for_backward(fe, pos, inv)
+        ctx.k = k
+        ctx.cap = capacity
+        ctx.n = x.shape[0]
+        return disp
+
+    @staticmethod
+    def backward(ctx, ddisp):
+        fe, pos, inv = ctx.saved_tensors
+        dx = ddisp.new_empty(ctx.n, ddisp.shape[-1])
+        native_ext().moe_dispatch_bwd(dx, ddisp.contiguous(), fe, pos,
+                                      inv, ctx.k, ctx.cap)
+        return dx, None, None, None, None, None, None, None
+
+
+class _MoECombine(torch.autograd.Function):
+    """Fused weighted combine: out[t] = sum_j fw * h[slot_j(t)] with
+    fp32 accumulation; backward writes dh rows (unique slots, no
+    atomics) and per-assignment dfw dot products."""
+
+    @staticmethod
+    def forward(ctx, h, fw, fe, pos, ft, inv, k, capacity, n_tokens):
+        h = h.contiguous()
+        fwf = fw.detach().float().contiguous()
+        out = h.new_empty(n_tokens, h.shape[-1])
+        native_ext().moe_combine_fwd(out, h, fwf, fe, pos, inv, k,
+                                     capacity)
+        ctx.save_for_backward(h, fwf, fe, pos, ft)
+        ctx.cap = capacity
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        h, fwf, fe, pos, ft = ctx.saved_tensors
+        dh = torch.zeros_like(h)
+        dfw = torch.empty_like(fwf)
+        native_ext().moe_combine_bwd(dh, dfw, dout.contiguous(), h, fwf,
+                                     fe, pos, ft, ctx.cap)
+        return dh, dfw, None, None, None, None, None, None, None
 
 
 class _BatchedExpertLinear(torch.autograd.Function):
@@ -127,11 +184,6 @@ class ExpertParallelMLP(nn.Module):
         capacity = max(
             1, int(self.capacity_factor * n_tokens * self.top_k /
                    self.num_experts))
-        # dispatch tensor: [num_experts, capacity, hidden]
-        dispatched = x.new_zeros(self.num_experts, capacity, self.hidden)
-        combine_idx = []  # (expert, slot, token, weight)
-        slot_count = torch.zeros(self.num_experts, dtype=torch.long,
-                                 device=x.device)
         flat_e = topi.reshape(-1)
         flat_t = (torch.arange(n_tokens, device=x.device)
                   .repeat_interleave(self.top_k))
@@ -143,9 +195,26 @@ class ExpertParallelMLP(nn.Module):
         counts = torch.bincount(fe, minlength=self.num_experts)
         seg_start = torch.nn.functional.pad(counts.cumsum(0), (1, 0))[:-1]
         pos_in_e = torch.arange(fe.numel(), device=x.device) - seg_start[fe]
-        keep = pos_in_e < capacity
-        fe, ft, fw, pos_in_e = fe[keep], ft[keep], fw[keep], pos_in_e[keep]
-        dispatched[fe, pos_in_e] = x[ft]
+        native = (_moe_native_dispatch() and use_native(x)
+                  and x.dtype == torch.bfloat16 and self.hidden % 8 == 0)
+        inv = None
+        if native:
+            # fused dispatch kernel: full assignment list, in-kernel
+            # capacity skip — no boolean select / nonzero sync
+            m = fe.numel()
+            inv = torch.empty_like(order)
+            inv[order] = torch.arange(m, device=x.device)
+            dispatched = _MoEDispatch.apply(
+                x, fe, pos_in_e, ft, inv, self.top_k, capacity,
+                self.num_experts)
+        else:
+            keep = pos_in_e < capacity
+            fe, ft, fw, pos_in_e = (fe[keep], ft[keep], fw[keep],
+                                    pos_in_e[keep])
+            # dispatch tensor: [num_experts, capacity, hidden]
+            dispatched = x.new_zeros(self.num_experts, capacity,
+                                     self.hidden)
+            dispatched[fe, pos_in_e] = x[ft]
 
         # all-to-all: [world, local_experts*capacity, hidden]
         d = dispatched.reshape(self.world,
@@ -177,6 +246,11 @@ class ExpertParallelMLP(nn.Module):
                                       compress=self._wire_compression())
         h = h.reshape(self.num_experts, capacity, self.hidden)
 
-        out = x.new_zeros(n_tokens, self.hidden)
-        out.index_add_(0, ft, h[fe, pos_in_e] * fw.unsqueeze(-1).to(h.dtype))
+        if native:
+            out = _MoECombine.apply(h, fw, fe, pos_in_e, ft, inv,
+                                    self.top_k, capacity, n_tokens)
+        else:
+            out = x.new_zeros(n_tokens, self.hidden)
+            out.index_add_(0, ft,
+                           h[fe, pos_in_e] * fw.unsqueeze(-1).to(h.dtype))
         return out.reshape(orig_shape)

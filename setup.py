@@ -20,6 +20,7 @@ ext = CUDAExtension(
         "csrc/bindings.hip",
         "csrc/comm/rccl_comm.hip",
         "csrc/kernels/kernels.hip",
+        "csrc/kernels/moe.hip",
         "csrc/kernels/mfma_probe.hip",
         "csrc/kernels/attention.hip",
     ],
