@@ -361,3 +361,37 @@ def test_flash_attention_with_lse_block_merge(d):
         err = (got.float() - want).abs().max().item()
         rel = err / want.abs().max().clamp_min(1e-6).item()
         assert err < 0.15 or rel < 6e-2, (name, err, rel)
+
+
+@pytest.mark.gpu
+def test_dropout_mask_identical_under_checkpoint():
+    """Gradient-checkpoint recompute must regenerate the SAME philox
+    dropout mask (preserve_rng_state restores the CPU generator our
+    per-call seed draws from) — otherwise the backward would silently
+    use a different mask than the forward."""
+    from torch.utils.checkpoint import checkpoint
+    from easyparallellibrary_amd.ops.attention import _FlashAttention
+    b, h, s, d, p = 2, 2, 256, 64, 0.3
+    scale = d ** -0.5
+
+    def run(use_ckpt):
+        torch.manual_seed(123)
+        q = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        k = torch.randn_like(q, requires_grad=True)
+        v = torch.randn_like(q, requires_grad=True)
+        torch.manual_seed(77)   # the seed the dropout draw consumes
+        fn = lambda a, b_, c: _FlashAttention.apply(a, b_, c, False,
+                                                    scale, p)
+        out = (checkpoint(fn, q, k, v, use_reentrant=False)
+               if use_ckpt else fn(q, k, v))
+        out.float().pow(2).mean().backward()
+        return out.detach().float(), q.grad.float(), v.grad.float()
+
+    o1, gq1, gv1 = run(False)
+    o2, gq2, gv2 = run(True)
+    torch.cuda.synchronize()
+    # bitwise-identical: same mask in forward AND recompute
+    assert torch.equal(o1, o2)
+    assert torch.equal(gq1, gq2), (gq1 - gq2).abs().max()
+    assert torch.equal(gv1, gv2), (gv1 - gv2).abs().max()
