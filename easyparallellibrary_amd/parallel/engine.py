@@ -327,9 +327,16 @@ class Engine:
             self.optimizer = CPUOffloadAdamW(self.flat_groups, **okw)
             n = self.optimizer.attach_reducers(
                 self.reducers, [info["fg"] for info in self._group_infos])
+            # eager CPU apply during backward needs the grad scale final
+            # before backward ends: amp off, no global-norm clip
+            self._offload_eager = bool(
+                n and n == len(self.reducers) and not self.amp.enabled
+                and not self.config.optimizer.max_grad_norm)
             if n:
                 logger.info("offload: per-bucket D2H overlap on %d "
-                            "reducer(s)", n)
+                            "reducer(s)%s", n,
+                            " + eager CPU apply" if self._offload_eager
+                            else "")
         else:
             self.optimizer = OPTIMIZERS[optimizer](self.flat_groups, **okw)
         for sp, bcomm, granks in _pending_sparse:
@@ -444,6 +451,13 @@ class Engine:
             # grad scale is known before backward (amp off, no manual
             # accumulation under pipeline): buckets apply as they reduce
             self.optimizer.begin_eager(float(self.num_micro_batch))
+        if getattr(self, "_offload_eager", False) and not (
+                self._gc_wrapped
+                and self.config.gradient_checkpoint.check_gradients
+                and self.global_step == 0):
+            # same precondition as PBO: scale final before backward
+            self.optimizer.begin_step(
+                float(self.num_micro_batch) * (self._accum_count + 1))
         if self.pipeline is not None:
             loss = self.pipeline.run(inputs, targets)
         else:

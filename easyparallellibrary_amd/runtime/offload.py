@@ -61,6 +61,13 @@ class CPUOffloadAdamW:
         # per-bucket D2H overlap state (attach_reducers)
         self._events = {}       # group idx -> [(lo, hi, event)]
         self._n_buckets = {}    # group idx -> bucket count
+        # eager mode: a worker thread consumes bucket D2H events DURING
+        # backward, runs the CPU AdamW chunk and kicks the H2D of the
+        # updated params — by step() time most of the optimizer is done
+        self._eager_exec = None
+        self._eager_futs = []
+        self._eager_active = False
+        self._eager_scale = 1.0
 
     def attach_reducers(self, reducers, groups_of_reducers):
         """Overlap grad D2H with the tail of backward: as each bucket's
@@ -99,8 +106,45 @@ class CPUOffloadAdamW:
                                 non_blocking=True)
                 ev = torch.cuda.Event()
                 ev.record(self._d2h)
-            self._events.setdefault(gi, []).append((lo, hi, ev))
+            if self._eager_active:
+                self._eager_futs.append(self._eager_exec.submit(
+                    self._eager_task, gi, lo, hi, ev))
+            else:
+                self._events.setdefault(gi, []).append((lo, hi, ev))
         return cb
+
+    # ---- eager CPU apply during backward --------------------------------
+    def begin_step(self, grad_scale):
+        """Arm the eager pipeline for this step: every reduced bucket's
+        CPU AdamW chunk + H2D runs on a worker thread as soon as its
+        D2H lands, overlapping the remaining backward.  The engine
+        gates this to amp-off / no-grad-clip runs (the scale must be
+        final before backward ends).  Safe vs backward param reads for
+        the same reason as PreferBackwardOptimizer: a parameter whose
+        gradient bucket is complete is not read again this step."""
+        if self._d2h is None or not self._n_buckets:
+            return False
+        if self._eager_exec is None:
+            from concurrent.futures import ThreadPoolExecutor
+            self._eager_exec = ThreadPoolExecutor(
+                max_workers=1, thread_name_prefix="epl-offload")
+        self.step_count += 1
+        self._eager_scale = float(grad_scale)
+        self._eager_futs = []
+        self._eager_active = True
+        return True
+
+    def _eager_task(self, gi, lo, hi, ev):
+        ev.synchronize()   # the bucket's fp32 grads are on the host
+        g = self.groups[gi]
+        self._chunk_update(g.state["master_cpu"][lo:hi],
+                           g.state["grad_cpu"][lo:hi],
+                           g.state["exp_avg"][lo:hi],
+                           g.state["exp_avg_sq"][lo:hi],
+                           self._eager_scale)
+        with torch.cuda.stream(self._h2d):
+            g.param_arena[lo:hi].copy_(g.state["master_cpu"][lo:hi],
+                                       non_blocking=True)
 
     def _chunk_update(self, w, gc, m, v, grad_scale):
         """One AdamW update on a CPU arena slice."""
@@ -126,6 +170,23 @@ class CPUOffloadAdamW:
         w.add_(update, alpha=-self.lr)
 
     def step(self, grad_scale=1.0):
+        if self._eager_active:
+            self._eager_active = False
+            for f in self._eager_futs:
+                f.result()   # propagate worker exceptions
+            self._eager_futs = []
+            self._events.clear()
+            # groups whose reducer was not attached (none in practice —
+            # attach gates match begin_step's) would be silently
+            # skipped; guard loudly instead
+            covered = set(self._n_buckets)
+            for gi in range(len(self.groups)):
+                if gi not in covered:
+                    raise RuntimeError(
+                        "offload eager step: group {} had no bucket "
+                        "callbacks".format(gi))
+            torch.cuda.current_stream().wait_stream(self._h2d)
+            return
         self.step_count += 1
         for gi, g in enumerate(self.groups):
             gc = g.state["grad_cpu"]
