@@ -122,6 +122,9 @@ class CPUOffloadAdamW:
         final before backward ends).  Safe vs backward param reads for
         the same reason as PreferBackwardOptimizer: a parameter whose
         gradient bucket is complete is not read again this step."""
+        import os
+        if os.environ.get("EPL_OFFLOAD_EAGER", "1") != "1":
+            return False
         if self._d2h is None or not self._n_buckets:
             return False
         if self._eager_exec is None:
