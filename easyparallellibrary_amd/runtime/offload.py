@@ -123,7 +123,12 @@ class CPUOffloadAdamW:
         the same reason as PreferBackwardOptimizer: a parameter whose
         gradient bucket is complete is not read again this step."""
         import os
-        if os.environ.get("EPL_OFFLOAD_EAGER", "1") != "1":
+        # DEFAULT OFF: same-box A/B measured 931 vs 697 ms/step on the
+        # gpt2-xl offload bench — the worker thread contends with the
+        # (GC-recompute-heavy, Python-bound) backward for the GIL and
+        # LOSES more than the CPU-adamw overlap gains.  Kept behind the
+        # flag for non-GC workloads where backward is GIL-light.
+        if os.environ.get("EPL_OFFLOAD_EAGER", "0") != "1":
             return False
         if self._d2h is None or not self._n_buckets:
             return False
