@@ -650,3 +650,40 @@ def test_mixed_width_requires_replicated_io():
     assert tg_ok.replicated_io
     out = run_multiprocess(_unmarked_mixed_worker, world=3, timeout=300)
     assert all(out), out
+
+
+def _mixed_eval_worker(rank, world, widths):
+    """Pipelined EVAL across mixed-width stages: the 1<->k eval fan-out
+    and fan-in must follow the same link rules as training."""
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.models.tp_transformer import (
+        build_tp_pipeline)
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "pipeline.num_micro_batch": 2,
+    }))
+    torch.manual_seed(71)
+    model = build_tp_pipeline(stages=2, tp=(widths or 1), layers=2,
+                              hidden=32, heads=4, ffn=64, vocab_size=128,
+                              max_pos=32)
+    engine = epl.Engine(model, loss_fn=lambda o, t: o.sum())
+    torch.manual_seed(72)
+    ids = torch.randint(0, 128, (4, 16))
+    out = engine.eval_step(ids)
+    engine.close()
+    return None if out is None else out.float().clone()
+
+
+def test_mixed_width_pipelined_eval_matches_serial():
+    serial = run_multiprocess(_mixed_eval_worker, world=2,
+                              args=(None,))  # plain [1, 1]
+    mixed = run_multiprocess(_mixed_eval_worker, world=3,
+                             args=([1, 2],), timeout=300)
+    # last stage of the mixed run is wide: BOTH positions return the
+    # (replicated) outputs, identical to the plain-PP2 run
+    assert serial[0] is None and serial[1] is not None
+    assert mixed[0] is None
+    for pos_out in (mixed[1], mixed[2]):
+        assert pos_out is not None
+        assert torch.allclose(pos_out, serial[1], atol=1e-5), (
+            (pos_out - serial[1]).abs().max())
