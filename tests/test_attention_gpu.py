@@ -395,3 +395,41 @@ def test_dropout_mask_identical_under_checkpoint():
     assert torch.equal(o1, o2)
     assert torch.equal(gq1, gq2), (gq1 - gq2).abs().max()
     assert torch.equal(gv1, gv2), (gv1 - gv2).abs().max()
+
+
+@pytest.mark.gpu
+def test_flash_attention_fuzz_shapes():
+    """Randomized shape sweep across both head dims, causal, odd seqs
+    and dropout-off paths vs the fp32 reference."""
+    import random
+    rng = random.Random(99)
+    from easyparallellibrary_amd.ops.attention import flash_attention
+    for trial in range(14):
+        b = rng.choice([1, 2, 3, 5])
+        h = rng.choice([1, 2, 4, 7])
+        s = rng.choice([33, 64, 96, 130, 257, 384, 511, 640])
+        d = rng.choice([64, 128])
+        causal = rng.random() < 0.5
+        torch.manual_seed(trial)
+        q = torch.randn(b, h, s, d, device="cuda", dtype=torch.bfloat16,
+                        requires_grad=True)
+        k = torch.randn_like(q, requires_grad=True)
+        v = torch.randn_like(q, requires_grad=True)
+        scale = d ** -0.5
+        out = flash_attention(q, k, v, causal=causal, scale=scale)
+        dout = torch.randn_like(q)
+        out.backward(dout)
+        qf = q.detach().float().requires_grad_(True)
+        kf = k.detach().float().requires_grad_(True)
+        vf = v.detach().float().requires_grad_(True)
+        ref = ref_attention(qf, kf, vf, causal, scale)
+        ref.backward(dout.float())
+        torch.cuda.synchronize()
+        shape = (b, h, s, d, causal)
+        assert (out.float() - ref).abs().max().item() < 4e-2, shape
+        for got, want, nm in ((q.grad, qf.grad, "dq"),
+                              (k.grad, kf.grad, "dk"),
+                              (v.grad, vf.grad, "dv")):
+            err = (got.float() - want).abs().max().item()
+            rel = err / want.abs().max().clamp_min(1e-6).item()
+            assert err < 0.15 or rel < 6e-2, (shape, nm, err, rel)
