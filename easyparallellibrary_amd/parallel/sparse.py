@@ -154,6 +154,14 @@ class SparseGradHandler:
         for p in self.params:
             p.grad = None
 
+    def refresh_master(self):
+        """Re-derive the fp32 masters from (possibly freshly loaded)
+        params — mirrors FlatParamGroup.refresh_master's role in the
+        checkpoint-load ordering (masters refresh BEFORE an optimizer
+        load so a loaded exact master wins)."""
+        for p in self.params:
+            self.state[id(p)]["master"].copy_(p.data.to(torch.float32))
+
     # ---- checkpoint ---------------------------------------------------------
     def state_dict(self):
         return {

@@ -160,6 +160,8 @@ class ShardingLoader:
         # the optimizer load would round the master through bf16.
         for fg in engine.flat_groups:
             fg.refresh_master()
+        for h in getattr(engine, "sparse_handlers", []):
+            h.refresh_master()
         if load_optimizer:
             opt_path = os.path.join(
                 self.path, "opt_rank{}.pt".format(engine.rank))

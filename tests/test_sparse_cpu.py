@@ -133,3 +133,36 @@ def test_sparse_checkpoint_roundtrip(tmp_path):
     out = run_multiprocess(_worker_ckpt, world=1, args=(str(tmp_path),))
     l_next, l_resume = out[0]
     assert abs(l_next - l_resume) < 1e-6, (l_next, l_resume)
+
+
+def _worker_ckpt_no_opt(rank, world, path):
+    """load_optimizer=False must refresh the sparse fp32 master from
+    the loaded params (regression: a stale master overwrote the loaded
+    embedding on the first post-load step)."""
+    import easyparallellibrary_amd as epl
+    epl.init()
+    torch.manual_seed(21)
+    with epl.replicate(device_count=1):
+        model = TinyLM(sparse=True)
+    engine = epl.Engine(model, loss_fn=_ce, optimizer="adamw", lr=1e-2)
+    ids = torch.randint(0, 50, (6, 5))
+    tgt = torch.randint(0, 50, (6,))
+    for _ in range(3):
+        engine.train_step(ids, tgt)
+    engine.save_checkpoint(path, save_optimizer=False)
+    saved_w = model.emb.weight.detach().clone()
+
+    torch.manual_seed(99)  # very different init
+    with epl.replicate(device_count=1):
+        model2 = TinyLM(sparse=True)
+    engine2 = epl.Engine(model2, loss_fn=_ce, optimizer="adamw", lr=0.0)
+    engine2.load_checkpoint(path, load_optimizer=False)
+    assert torch.equal(model2.emb.weight.detach(), saved_w)
+    engine2.train_step(ids, tgt)   # lr 0: params must NOT move
+    return float((model2.emb.weight.detach() - saved_w).abs().max())
+
+
+def test_sparse_master_refresh_on_partial_load(tmp_path):
+    out = run_multiprocess(_worker_ckpt_no_opt, world=1,
+                           args=(str(tmp_path),))
+    assert out[0] == 0.0, out[0]
