@@ -759,6 +759,9 @@ class Engine:
         must not be used afterwards.  COLLECTIVE-free."""
         for r in self.reducers:
             r.remove_hooks()
+        ex = getattr(self.optimizer, "_eager_exec", None)
+        if ex is not None:
+            ex.shutdown(wait=True)
         from easyparallellibrary_amd.comm.backend import destroy_namespace
         n = destroy_namespace(self._ns + "_")
         logger.info("engine %s closed: %d communicator(s) destroyed",
