@@ -623,8 +623,9 @@ DEVI void dk_tile(const short (&ldsQ)[64][D + 8],
   }
 }
 
-template <bool DROP, int D>
-__global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
+template <bool DROP, bool DBUF, int D>
+__global__ __launch_bounds__(256, D == 64 ? (DBUF ? 3 : 4) : 2)
+void attn_bwd_dv_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ dout, const float* __restrict__ lse,
     short* __restrict__ dv, int64_t seq, float scale, int causal,
@@ -633,8 +634,8 @@ __global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
     int64_t g_sh, int64_t g_ss,
     const unsigned int* __restrict__ mask, int64_t mask_w,
     float inv_keep) {
-  __shared__ short ldsQ[64][D + 8];
-  __shared__ short ldsDO[64][D + 8];
+  __shared__ short ldsQ[DBUF ? 2 : 1][64][D + 8];
+  __shared__ short ldsDO[DBUF ? 2 : 1][64][D + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -664,7 +665,40 @@ __global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
 
-  // stage 64 q rows (two 32-row halves) per barrier pair
+  // DBUF=false: stage 64 q rows straight to LDS per barrier pair.
+  // DBUF=true: software pipeline — this chunk's rows were prefetched
+  // into registers during the previous chunk's MFMAs; commit them to
+  // the alternate LDS buffer, one barrier, then issue the next chunk's
+  // global loads so they hide under this chunk's compute (the fwd v7
+  // staging pattern applied to the q/dO loop).
+  bf16x8 rq[2][D / 64], rdo[2][D / 64];
+  auto prefetch = [&](int64_t q0) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int64_t qr = q0 + stage_row + half * 32;
+      if (qr >= seq) qr = seq - 1;
+#pragma unroll
+      for (int s = 0; s < D / 64; ++s) {
+        rq[half][s] = *reinterpret_cast<const bf16x8*>(
+            qp + qr * in_ss + stage_seg + s * 64);
+        rdo[half][s] = *reinterpret_cast<const bf16x8*>(
+            dop + qr * do_ss + stage_seg + s * 64);
+      }
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half)
+#pragma unroll
+      for (int s = 0; s < D / 64; ++s) {
+        *reinterpret_cast<bf16x8*>(
+            &ldsQ[buf][stage_row + half * 32][stage_seg + s * 64]) =
+            rq[half][s];
+        *reinterpret_cast<bf16x8*>(
+            &ldsDO[buf][stage_row + half * 32][stage_seg + s * 64]) =
+            rdo[half][s];
+      }
+  };
   auto stage_q64 = [&](int64_t q0) {
     __syncthreads();
 #pragma unroll
@@ -674,11 +708,11 @@ __global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
 #pragma unroll
       for (int seg = 0; seg < D; seg += 64) {
         *reinterpret_cast<bf16x8*>(
-            &ldsQ[stage_row + half * 32][stage_seg + seg]) =
+            &ldsQ[0][stage_row + half * 32][stage_seg + seg]) =
             *reinterpret_cast<const bf16x8*>(
                 qp + qr * in_ss + stage_seg + seg);
         *reinterpret_cast<bf16x8*>(
-            &ldsDO[stage_row + half * 32][stage_seg + seg]) =
+            &ldsDO[0][stage_row + half * 32][stage_seg + seg]) =
             *reinterpret_cast<const bf16x8*>(
                 dop + qr * do_ss + stage_seg + seg);
       }
@@ -686,46 +720,79 @@ __global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
     __syncthreads();
   };
 
-  // phases (64-row chunks): masked causal diagonal -> branch-free bulk
-  // -> masked tail
+  // unified 64-row chunk loop; the phase test (masked causal diagonal /
+  // branch-free bulk / masked seq tail) is block-uniform per chunk
   const unsigned int* maskrow =
       DROP ? mask + bh * seq * mask_w : (const unsigned int*)nullptr;
   const int64_t diag_end =
       causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
   const int64_t bulk_end = seq & ~(int64_t)63;
-  int64_t q0 = causal ? kv0_blk : 0;
-  for (; q0 < diag_end; q0 += 64) {
-    stage_q64(q0);
-    if (!active) continue;
+  if (!DBUF) {
+    // two-barrier direct staging, phase-split loops (the measured r2
+    // baseline — kept verbatim so the fallback stays spill-free)
+    int64_t q0 = causal ? kv0_blk : 0;
+    for (; q0 < diag_end; q0 += 64) {
+      stage_q64(q0);
+      if (!active) continue;
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32) {
-      const int64_t q0s = q0 + sub;
-      if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
-      dv_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
-                             causal, lsep, kfrag, dvt, lkv, hi, maskrow,
-                             mask_w, inv_keep);
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t q0s = q0 + sub;
+        if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
+        dv_tile<true, DROP, D>(ldsQ[0], ldsDO[0], sub, q0s, seq, mykv,
+                               scale, causal, lsep, kfrag, dvt, lkv, hi,
+                               maskrow, mask_w, inv_keep);
+      }
     }
-  }
-  q0 = diag_end > q0 ? diag_end : q0;  // 64-aligned when < seq (diag is
-                                       // 2x64 or clipped by seq)
-  for (; q0 + 63 < bulk_end; q0 += 64) {
-    stage_q64(q0);
+    q0 = diag_end > q0 ? diag_end : q0;
+    for (; q0 + 63 < bulk_end; q0 += 64) {
+      stage_q64(q0);
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32)
-      dv_tile<false, DROP, D>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv,
-                              scale, causal, lsep, kfrag, dvt, lkv, hi,
-                              maskrow, mask_w, inv_keep);
-  }
-  for (; q0 < seq; q0 += 64) {
-    stage_q64(q0);
-    if (!active) continue;
+      for (int sub = 0; sub < 64; sub += 32)
+        dv_tile<false, DROP, D>(ldsQ[0], ldsDO[0], sub, q0 + sub, seq,
+                                mykv, scale, causal, lsep, kfrag, dvt,
+                                lkv, hi, maskrow, mask_w, inv_keep);
+    }
+    for (; q0 < seq; q0 += 64) {
+      stage_q64(q0);
+      if (!active) continue;
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32) {
-      const int64_t q0s = q0 + sub;
-      if (q0s >= seq) break;
-      dv_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
-                             causal, lsep, kfrag, dvt, lkv, hi, maskrow,
-                             mask_w, inv_keep);
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t q0s = q0 + sub;
+        if (q0s >= seq) break;
+        dv_tile<true, DROP, D>(ldsQ[0], ldsDO[0], sub, q0s, seq, mykv,
+                               scale, causal, lsep, kfrag, dvt, lkv, hi,
+                               maskrow, mask_w, inv_keep);
+      }
+    }
+  } else {
+    const int64_t q_begin = causal ? kv0_blk : 0;
+    int buf = 0;
+    prefetch(q_begin);
+    for (int64_t q0 = q_begin; q0 < seq; q0 += 64) {
+      commit(buf);
+      __syncthreads();
+      if (q0 + 64 < seq) prefetch(q0 + 64);
+      const int rb = buf;
+      buf ^= 1;
+      const bool plain = q0 >= diag_end && q0 + 63 < bulk_end;
+      if (plain) {
+#pragma unroll
+        for (int sub = 0; sub < 64; sub += 32)
+          dv_tile<false, DROP, D>(ldsQ[rb], ldsDO[rb], sub, q0 + sub,
+                                  seq, mykv, scale, causal, lsep, kfrag,
+                                  dvt, lkv, hi, maskrow, mask_w,
+                                  inv_keep);
+      } else if (active) {
+#pragma unroll
+        for (int sub = 0; sub < 64; sub += 32) {
+          const int64_t q0s = q0 + sub;
+          if (q0s >= seq) break;
+          if (causal && q0s + 31 < kv0) continue;
+          dv_tile<true, DROP, D>(ldsQ[rb], ldsDO[rb], sub, q0s, seq,
+                                 mykv, scale, causal, lsep, kfrag, dvt,
+                                 lkv, hi, maskrow, mask_w, inv_keep);
+        }
+      }
     }
   }
   if (!active || mykv >= seq) return;
@@ -738,7 +805,7 @@ __global__ __launch_bounds__(256, D == 64 ? 4 : 2) void attn_bwd_dv_kernel(
       dvp[dj * 32 + crow(r, hi)] = (short)f2bf(dvt[dj][r]);
 }
 
-template <bool DROP, int D>
+template <bool DROP, bool DBUF, int D>
 __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -749,8 +816,8 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
     int64_t g_sh, int64_t g_ss,
     const unsigned int* __restrict__ mask, int64_t mask_w,
     float inv_keep) {
-  __shared__ short ldsQ[64][D + 8];
-  __shared__ short ldsDO[64][D + 8];
+  __shared__ short ldsQ[DBUF ? 2 : 1][64][D + 8];
+  __shared__ short ldsDO[DBUF ? 2 : 1][64][D + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -783,6 +850,35 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
   const int stage_row = threadIdx.x >> 3;
   const int stage_seg = (threadIdx.x & 7) * 8;
 
+  // staging: same DBUF pipeline as the dV kernel (see comment there)
+  bf16x8 rq[2][D / 64], rdo[2][D / 64];
+  auto prefetch = [&](int64_t q0) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int64_t qr = q0 + stage_row + half * 32;
+      if (qr >= seq) qr = seq - 1;
+#pragma unroll
+      for (int s = 0; s < D / 64; ++s) {
+        rq[half][s] = *reinterpret_cast<const bf16x8*>(
+            qp + qr * in_ss + stage_seg + s * 64);
+        rdo[half][s] = *reinterpret_cast<const bf16x8*>(
+            dop + qr * do_ss + stage_seg + s * 64);
+      }
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half)
+#pragma unroll
+      for (int s = 0; s < D / 64; ++s) {
+        *reinterpret_cast<bf16x8*>(
+            &ldsQ[buf][stage_row + half * 32][stage_seg + s * 64]) =
+            rq[half][s];
+        *reinterpret_cast<bf16x8*>(
+            &ldsDO[buf][stage_row + half * 32][stage_seg + s * 64]) =
+            rdo[half][s];
+      }
+  };
   auto stage_q64 = [&](int64_t q0) {
     __syncthreads();
 #pragma unroll
@@ -792,11 +888,11 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
 #pragma unroll
       for (int seg = 0; seg < D; seg += 64) {
         *reinterpret_cast<bf16x8*>(
-            &ldsQ[stage_row + half * 32][stage_seg + seg]) =
+            &ldsQ[0][stage_row + half * 32][stage_seg + seg]) =
             *reinterpret_cast<const bf16x8*>(
                 qp + qr * in_ss + stage_seg + seg);
         *reinterpret_cast<bf16x8*>(
-            &ldsDO[stage_row + half * 32][stage_seg + seg]) =
+            &ldsDO[0][stage_row + half * 32][stage_seg + seg]) =
             *reinterpret_cast<const bf16x8*>(
                 dop + qr * do_ss + stage_seg + seg);
       }
@@ -809,38 +905,72 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dk_kernel(
   const int64_t diag_end =
       causal ? (kv0_blk + 128 < seq ? kv0_blk + 128 : seq) : 0;
   const int64_t bulk_end = seq & ~(int64_t)63;
-  int64_t q0 = causal ? kv0_blk : 0;
-  for (; q0 < diag_end; q0 += 64) {
-    stage_q64(q0);
-    if (!active) continue;
+  if (!DBUF) {
+    int64_t q0 = causal ? kv0_blk : 0;
+    for (; q0 < diag_end; q0 += 64) {
+      stage_q64(q0);
+      if (!active) continue;
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32) {
-      const int64_t q0s = q0 + sub;
-      if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
-      dk_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
-                             causal, lsep, dltp, kfrag, vfrag, dkt, lkv,
-                             hi, maskrow, mask_w, inv_keep);
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t q0s = q0 + sub;
+        if (q0s >= diag_end || (causal && q0s + 31 < kv0)) continue;
+        dk_tile<true, DROP, D>(ldsQ[0], ldsDO[0], sub, q0s, seq, mykv,
+                               scale, causal, lsep, dltp, kfrag, vfrag,
+                               dkt, lkv, hi, maskrow, mask_w, inv_keep);
+      }
     }
-  }
-  q0 = diag_end > q0 ? diag_end : q0;
-  for (; q0 + 63 < bulk_end; q0 += 64) {
-    stage_q64(q0);
+    q0 = diag_end > q0 ? diag_end : q0;
+    for (; q0 + 63 < bulk_end; q0 += 64) {
+      stage_q64(q0);
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32)
-      dk_tile<false, DROP, D>(ldsQ, ldsDO, sub, q0 + sub, seq, mykv,
-                              scale, causal, lsep, dltp, kfrag, vfrag,
-                              dkt, lkv, hi, maskrow, mask_w, inv_keep);
-  }
-  for (; q0 < seq; q0 += 64) {
-    stage_q64(q0);
-    if (!active) continue;
+      for (int sub = 0; sub < 64; sub += 32)
+        dk_tile<false, DROP, D>(ldsQ[0], ldsDO[0], sub, q0 + sub, seq,
+                                mykv, scale, causal, lsep, dltp, kfrag,
+                                vfrag, dkt, lkv, hi, maskrow, mask_w,
+                                inv_keep);
+    }
+    for (; q0 < seq; q0 += 64) {
+      stage_q64(q0);
+      if (!active) continue;
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32) {
-      const int64_t q0s = q0 + sub;
-      if (q0s >= seq) break;
-      dk_tile<true, DROP, D>(ldsQ, ldsDO, sub, q0s, seq, mykv, scale,
-                             causal, lsep, dltp, kfrag, vfrag, dkt, lkv,
-                             hi, maskrow, mask_w, inv_keep);
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t q0s = q0 + sub;
+        if (q0s >= seq) break;
+        dk_tile<true, DROP, D>(ldsQ[0], ldsDO[0], sub, q0s, seq, mykv,
+                               scale, causal, lsep, dltp, kfrag, vfrag,
+                               dkt, lkv, hi, maskrow, mask_w, inv_keep);
+      }
+    }
+  } else {
+    const int64_t q_begin = causal ? kv0_blk : 0;
+    int buf = 0;
+    prefetch(q_begin);
+    for (int64_t q0 = q_begin; q0 < seq; q0 += 64) {
+      commit(buf);
+      __syncthreads();
+      if (q0 + 64 < seq) prefetch(q0 + 64);
+      const int rb = buf;
+      buf ^= 1;
+      const bool plain = q0 >= diag_end && q0 + 63 < bulk_end;
+      if (plain) {
+#pragma unroll
+        for (int sub = 0; sub < 64; sub += 32)
+          dk_tile<false, DROP, D>(ldsQ[rb], ldsDO[rb], sub, q0 + sub,
+                                  seq, mykv, scale, causal, lsep, dltp,
+                                  kfrag, vfrag, dkt, lkv, hi, maskrow,
+                                  mask_w, inv_keep);
+      } else if (active) {
+#pragma unroll
+        for (int sub = 0; sub < 64; sub += 32) {
+          const int64_t q0s = q0 + sub;
+          if (q0s >= seq) break;
+          if (causal && q0s + 31 < kv0) continue;
+          dk_tile<true, DROP, D>(ldsQ[rb], ldsDO[rb], sub, q0s, seq,
+                                 mykv, scale, causal, lsep, dltp, kfrag,
+                                 vfrag, dkt, lkv, hi, maskrow, mask_w,
+                                 inv_keep);
+        }
+      }
     }
   }
   if (!active || mykv >= seq) return;
@@ -1137,39 +1267,81 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
       in_strides[0], in_strides[1], in_strides[2], do_strides[0],        \
       do_strides[1], do_strides[2], g_strides[0], g_strides[1],          \
       g_strides[2], drop_mask, mask_w, inv_keep
+    // d64 dV/dK staging mode (EPL_ATTN_BWD_DBUF): 0 = two-barrier
+    // direct staging; 1 = dV double-buffered single-barrier pipeline
+    // (spill-free at 3 waves/SIMD vs 4 direct); 2 = dK pipelined too
+    // (spills 12-40 VGPRs at 3 waves but still wins on long q loops).
+    // Default -1 = auto: non-causal runs mode 2 (same-box A/B in
+    // profiles/r02_dbuf_ab.txt: bert s512 -7.6%, s4096 -9.9%), causal
+    // keeps direct staging (its per-block q loop is half as long and
+    // mode 2 measured +2% there).  d128 keeps direct staging (doubled
+    // LDS would cost its 2-waves occupancy).
+    static const int dbuf_mode = [] {
+      const char* e = getenv("EPL_ATTN_BWD_DBUF");
+      return e == nullptr ? -1 : atoi(e);
+    }();
+    const int mode = dbuf_mode < 0 ? (causal ? 0 : 2) : dbuf_mode;
+    const bool dbuf = mode >= 1;
+    const bool dbuf_k = mode >= 2;
     if (head_dim == 128) {
       if (drop_mask != nullptr) {
-        hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, 128>),
-                           grid, dim3(256), 0, stream, DV_ARGS);
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, false, 128>), grid,
+            dim3(256), 0, stream, DV_ARGS);
         hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, 128>),
                            grid, dim3(256), 0, stream, DQ_ARGS);
-        hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, 128>),
-                           grid, dim3(256), 0, stream, DK_ARGS);
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, false, 128>), grid,
+            dim3(256), 0, stream, DK_ARGS);
       } else {
         hipLaunchKernelGGL(
-            HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, 128>), grid,
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, false, 128>), grid,
             dim3(256), 0, stream, DV_ARGS);
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 128>), grid,
             dim3(256), 0, stream, DQ_ARGS);
         hipLaunchKernelGGL(
-            HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, 128>), grid,
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, false, 128>), grid,
             dim3(256), 0, stream, DK_ARGS);
       }
     } else if (drop_mask != nullptr) {
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, 64>),
-                         grid, dim3(256), 0, stream, DV_ARGS);
+      if (dbuf)
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, true, 64>), grid,
+            dim3(256), 0, stream, DV_ARGS);
+      else
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, false, 64>), grid,
+            dim3(256), 0, stream, DV_ARGS);
       hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, 64>),
                          grid, dim3(256), 0, stream, DQ_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, 64>),
-                         grid, dim3(256), 0, stream, DK_ARGS);
+      if (dbuf_k)
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, true, 64>), grid,
+            dim3(256), 0, stream, DK_ARGS);
+      else
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, false, 64>), grid,
+            dim3(256), 0, stream, DK_ARGS);
     } else {
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, 64>),
-                         grid, dim3(256), 0, stream, DV_ARGS);
+      if (dbuf)
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, true, 64>), grid,
+            dim3(256), 0, stream, DV_ARGS);
+      else
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, false, 64>), grid,
+            dim3(256), 0, stream, DV_ARGS);
       hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 64>),
                          grid, dim3(256), 0, stream, DQ_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, 64>),
-                         grid, dim3(256), 0, stream, DK_ARGS);
+      if (dbuf_k)
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, true, 64>), grid,
+            dim3(256), 0, stream, DK_ARGS);
+      else
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, false, 64>), grid,
+            dim3(256), 0, stream, DK_ARGS);
     }
 #undef DV_ARGS
 #undef DQ_ARGS
