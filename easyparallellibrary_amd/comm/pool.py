@@ -31,9 +31,15 @@ class CommunicationPool:
 
     def batch_all_reduce(self, buckets, op="sum", async_op=True):
         """All-reduce a list of contiguous buckets, round-robined over the
-        pool (reference: collective_communicator.py:93-123)."""
-        for b in buckets:
-            self.next_comm().all_reduce(b, op=op, async_op=async_op)
+        pool (reference: collective_communicator.py:93-123).
+
+        Mapping is positional (bucket i -> comm i mod K), NOT the stateful
+        next_comm() cursor: every rank must issue bucket i on the same
+        communicator, and cursor state could desync if call counts ever
+        differed across ranks (e.g. a rank-gated extra collective)."""
+        for i, b in enumerate(buckets):
+            self.comms[i % len(self.comms)].all_reduce(
+                b, op=op, async_op=async_op)
 
     def join(self):
         for c in self.comms:

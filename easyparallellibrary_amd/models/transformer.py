@@ -28,10 +28,12 @@ from easyparallellibrary_amd.ops.bias_linear import FusedBiasLinear
 from easyparallellibrary_amd.ops.layer_norm import FusedLayerNorm
 
 # Linear bias grads via the fused colsum kernel.  The kernel wins per-op
-# (vs torch's ~2.2 TB/s reduce) but the custom backward's dw/dx matmul
-# call pattern misses the TunableOp-tuned GEMM entries and measures 592
-# vs 612 samples/s end-to-end, so the DEFAULT stays on torch autograd
-# until the table is retuned with this path on (NOTES.md).
+# (vs torch's ~2.2 TB/s reduce) but the path loses end-to-end: r1 592 vs
+# 612, and the r2 retune experiment (profiles/r02_fbg_retune_ab.txt)
+# showed the TunableOp table already covered the custom backward's GEMM
+# shapes (tuning added zero entries) yet it still measured 593 vs 643 —
+# the regression is intrinsic to the custom Function's GEMM dispatch,
+# so the DEFAULT stays on torch autograd.
 _FBG = os.environ.get("EPL_FUSED_BIAS_GRAD", "0") == "1"
 _Linear = FusedBiasLinear if _FBG else nn.Linear
 
