@@ -1053,7 +1053,7 @@ DEVI void dq_tile(const short (&ldsK)[64][D + 8],
   }
 }
 
-template <bool DROP, int D>
+template <bool DROP, bool DBUF, int D>
 __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
     const short* __restrict__ q, const short* __restrict__ k,
     const short* __restrict__ v, const short* __restrict__ dout,
@@ -1064,8 +1064,8 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
     int64_t g_sb, int64_t g_sh, int64_t g_ss, int64_t o_sb, int64_t o_sh,
     int64_t o_ss, const unsigned int* __restrict__ mask, int64_t mask_w,
     float inv_keep, const float* __restrict__ dlse) {
-  __shared__ short ldsK[64][D + 8];
-  __shared__ short ldsVr[64][D + 8];
+  __shared__ short ldsK[DBUF ? 2 : 1][64][D + 8];
+  __shared__ short ldsVr[DBUF ? 2 : 1][64][D + 8];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
   const int hi = lane >> 5;
@@ -1133,11 +1133,11 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
 #pragma unroll
       for (int seg = 0; seg < D; seg += 64) {
         *reinterpret_cast<bf16x8*>(
-            &ldsK[stage_row + half * 32][stage_seg + seg]) =
+            &ldsK[0][stage_row + half * 32][stage_seg + seg]) =
             *reinterpret_cast<const bf16x8*>(
                 kp + kr * in_ss + stage_seg + seg);
         *reinterpret_cast<bf16x8*>(
-            &ldsVr[stage_row + half * 32][stage_seg + seg]) =
+            &ldsVr[0][stage_row + half * 32][stage_seg + seg]) =
             *reinterpret_cast<const bf16x8*>(
                 vp + kr * in_ss + stage_seg + seg);
       }
@@ -1145,30 +1145,95 @@ __global__ __launch_bounds__(256, D == 64 ? 3 : 2) void attn_bwd_dq_kernel(
     __syncthreads();
   };
 
+  // DBUF: same register-double-buffered single-barrier pipeline as the
+  // dV/dK kernels (next kv chunk prefetched into VGPRs during this
+  // chunk's MFMAs; see attn_bwd_dv_kernel).
+  bf16x8 rk[2][D / 64], rv[2][D / 64];
+  auto prefetch = [&](int64_t kv0) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half) {
+      int64_t kr = kv0 + stage_row + half * 32;
+      if (kr >= seq) kr = seq - 1;
+#pragma unroll
+      for (int s = 0; s < D / 64; ++s) {
+        rk[half][s] = *reinterpret_cast<const bf16x8*>(
+            kp + kr * in_ss + stage_seg + s * 64);
+        rv[half][s] = *reinterpret_cast<const bf16x8*>(
+            vp + kr * in_ss + stage_seg + s * 64);
+      }
+    }
+  };
+  auto commit = [&](int buf) {
+#pragma unroll
+    for (int half = 0; half < 2; ++half)
+#pragma unroll
+      for (int s = 0; s < D / 64; ++s) {
+        *reinterpret_cast<bf16x8*>(
+            &ldsK[buf][stage_row + half * 32][stage_seg + s * 64]) =
+            rk[half][s];
+        *reinterpret_cast<bf16x8*>(
+            &ldsVr[buf][stage_row + half * 32][stage_seg + s * 64]) =
+            rv[half][s];
+      }
+  };
+
   const unsigned int* maskrow =
       DROP ? mask + bh * seq * mask_w : (const unsigned int*)nullptr;
-  int64_t kv0 = 0;
-  for (; kv0 < bulk_end; kv0 += 64) {
-    stage_kv64(kv0);
+  if (!DBUF) {
+    int64_t kv0 = 0;
+    for (; kv0 < bulk_end; kv0 += 64) {
+      stage_kv64(kv0);
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32)
-      dq_tile<false, DROP, D>(ldsK, ldsVr, sub, kv0 + sub, seq, myq,
-                              scale, causal, mylse, mydelta, qfrag,
-                              dofrag, dqt, lq, hi, maskrow, mask_w,
-                              inv_keep);
-  }
-  for (; kv0 < blk_kv_end; kv0 += 64) {
-    stage_kv64(kv0);
-    const int64_t wave_kv_end = causal
-        ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
-    if (!active) continue;
+      for (int sub = 0; sub < 64; sub += 32)
+        dq_tile<false, DROP, D>(ldsK[0], ldsVr[0], sub, kv0 + sub, seq,
+                                myq, scale, causal, mylse, mydelta,
+                                qfrag, dofrag, dqt, lq, hi, maskrow,
+                                mask_w, inv_keep);
+    }
+    for (; kv0 < blk_kv_end; kv0 += 64) {
+      stage_kv64(kv0);
+      const int64_t wave_kv_end = causal
+          ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
+      if (!active) continue;
 #pragma unroll
-    for (int sub = 0; sub < 64; sub += 32) {
-      const int64_t kvs = kv0 + sub;
-      if (kvs >= wave_kv_end) break;
-      dq_tile<true, DROP, D>(ldsK, ldsVr, sub, kvs, seq, myq, scale,
-                             causal, mylse, mydelta, qfrag, dofrag, dqt,
-                             lq, hi, maskrow, mask_w, inv_keep);
+      for (int sub = 0; sub < 64; sub += 32) {
+        const int64_t kvs = kv0 + sub;
+        if (kvs >= wave_kv_end) break;
+        dq_tile<true, DROP, D>(ldsK[0], ldsVr[0], sub, kvs, seq, myq,
+                               scale, causal, mylse, mydelta, qfrag,
+                               dofrag, dqt, lq, hi, maskrow, mask_w,
+                               inv_keep);
+      }
+    }
+  } else {
+    int buf = 0;
+    prefetch(0);
+    for (int64_t kv0 = 0; kv0 < blk_kv_end; kv0 += 64) {
+      commit(buf);
+      __syncthreads();
+      if (kv0 + 64 < blk_kv_end) prefetch(kv0 + 64);
+      const int rb = buf;
+      buf ^= 1;
+      if (kv0 < bulk_end) {
+#pragma unroll
+        for (int sub = 0; sub < 64; sub += 32)
+          dq_tile<false, DROP, D>(ldsK[rb], ldsVr[rb], sub, kv0 + sub,
+                                  seq, myq, scale, causal, mylse,
+                                  mydelta, qfrag, dofrag, dqt, lq, hi,
+                                  maskrow, mask_w, inv_keep);
+      } else if (active) {
+        const int64_t wave_kv_end = causal
+            ? (q0 + 32 < seq ? q0 + 32 : seq) : seq;
+#pragma unroll
+        for (int sub = 0; sub < 64; sub += 32) {
+          const int64_t kvs = kv0 + sub;
+          if (kvs >= wave_kv_end) break;
+          dq_tile<true, DROP, D>(ldsK[rb], ldsVr[rb], sub, kvs, seq,
+                                 myq, scale, causal, mylse, mydelta,
+                                 qfrag, dofrag, dqt, lq, hi, maskrow,
+                                 mask_w, inv_keep);
+        }
+      }
     }
   }
 
@@ -1267,10 +1332,11 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
       in_strides[0], in_strides[1], in_strides[2], do_strides[0],        \
       do_strides[1], do_strides[2], g_strides[0], g_strides[1],          \
       g_strides[2], drop_mask, mask_w, inv_keep
-    // d64 dV/dK staging mode (EPL_ATTN_BWD_DBUF): 0 = two-barrier
-    // direct staging; 1 = dV double-buffered single-barrier pipeline
-    // (spill-free at 3 waves/SIMD vs 4 direct); 2 = dK pipelined too
-    // (spills 12-40 VGPRs at 3 waves but still wins on long q loops).
+    // d64 staging mode (EPL_ATTN_BWD_DBUF): 0 = two-barrier direct
+    // staging; 1 = dV double-buffered single-barrier pipeline
+    // (spill-free); 2 = dK pipelined too (spills 12-40 VGPRs at 3
+    // waves but still wins on long q loops); 3 = dQ pipelined too
+    // (spills 68 B/lane at 3 waves).
     // Default -1 = auto: non-causal runs mode 2 (same-box A/B in
     // profiles/r02_dbuf_ab.txt: bert s512 -7.6%, s4096 -9.9%), causal
     // keeps direct staging (its per-block q loop is half as long and
@@ -1283,13 +1349,15 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
     const int mode = dbuf_mode < 0 ? (causal ? 0 : 2) : dbuf_mode;
     const bool dbuf = mode >= 1;
     const bool dbuf_k = mode >= 2;
+    const bool dbuf_q = mode >= 3;
     if (head_dim == 128) {
       if (drop_mask != nullptr) {
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, false, 128>), grid,
             dim3(256), 0, stream, DV_ARGS);
-        hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, 128>),
-                           grid, dim3(256), 0, stream, DQ_ARGS);
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, false, 128>), grid,
+            dim3(256), 0, stream, DQ_ARGS);
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, false, 128>), grid,
             dim3(256), 0, stream, DK_ARGS);
@@ -1298,7 +1366,7 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
             HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, false, 128>), grid,
             dim3(256), 0, stream, DV_ARGS);
         hipLaunchKernelGGL(
-            HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 128>), grid,
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, false, 128>), grid,
             dim3(256), 0, stream, DQ_ARGS);
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, false, 128>), grid,
@@ -1313,8 +1381,14 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dv_kernel<true, false, 64>), grid,
             dim3(256), 0, stream, DV_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, 64>),
-                         grid, dim3(256), 0, stream, DQ_ARGS);
+      if (dbuf_q)
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, true, 64>), grid,
+            dim3(256), 0, stream, DQ_ARGS);
+      else
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<true, false, 64>), grid,
+            dim3(256), 0, stream, DQ_ARGS);
       if (dbuf_k)
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dk_kernel<true, true, 64>), grid,
@@ -1332,8 +1406,14 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dv_kernel<false, false, 64>), grid,
             dim3(256), 0, stream, DV_ARGS);
-      hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 64>),
-                         grid, dim3(256), 0, stream, DQ_ARGS);
+      if (dbuf_q)
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, true, 64>), grid,
+            dim3(256), 0, stream, DQ_ARGS);
+      else
+        hipLaunchKernelGGL(
+            HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, false, 64>), grid,
+            dim3(256), 0, stream, DQ_ARGS);
       if (dbuf_k)
         hipLaunchKernelGGL(
             HIP_KERNEL_NAME(attn_bwd_dk_kernel<false, true, 64>), grid,
@@ -1370,8 +1450,8 @@ void epl_attn_bwd(const void* q, const void* k, const void* v,
                      in_strides[2], do_strides[0], do_strides[1],
                      do_strides[2], g_strides[0], g_strides[1],
                      g_strides[2]);
-  hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, 64>), grid,
-                     dim3(256), 0, stream,
+  hipLaunchKernelGGL(HIP_KERNEL_NAME(attn_bwd_dq_kernel<false, false, 64>),
+                     grid, dim3(256), 0, stream,
                      reinterpret_cast<const short*>(q),
                      reinterpret_cast<const short*>(k),
                      reinterpret_cast<const short*>(v),
