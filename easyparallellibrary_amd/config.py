@@ -218,6 +218,12 @@ class KernelConfig(BaseConfig):
         # use fused HIP ops when running on GPU; on a GPU box the extension
         # missing is an error (no silent eager fallback)
         "fused": True,
+        # capture the simple-path forward+backward into a hipGraph and
+        # replay it (runtime/hipgraph.py).  Wins where the step is
+        # host-launch-bound (small batch / many small ops, e.g. the MoE
+        # bench); engine falls back to eager with a logged reason when
+        # the step is not capture-safe (multi-rank, dropout, GC, ...).
+        "hip_graph": False,
     }
 
 

@@ -390,6 +390,10 @@ class Engine:
 
         self.global_step = 0
         self._accum_count = 0
+        # hipGraph-captured simple step (runtime/hipgraph.py): None =
+        # requested, eligibility checked lazily at the first step (GC
+        # wrapping happens after __init__); False = off/ineligible
+        self._hipgraph = None if self.config.kernel.hip_graph else False
         logger.info(
             "Engine ready: world=%d stages=%d replicas=%d per_replica=%d "
             "dtype=%s groups=%d", self.world_size, self.num_stages,
@@ -460,6 +464,21 @@ class Engine:
                 float(self.num_micro_batch) * (self._accum_count + 1))
         if self.pipeline is not None:
             loss = self.pipeline.run(inputs, targets)
+        elif self._hipgraph is not False:
+            if self._hipgraph is None:   # lazy first-step eligibility
+                from easyparallellibrary_amd.runtime import hipgraph
+                ok, reason = hipgraph.eligible(self)
+                if ok:
+                    self._hipgraph = hipgraph.HipGraphStep(self)
+                else:
+                    logger.info("kernel.hip_graph requested but the step "
+                                "is not capture-safe: %s — running eager",
+                                reason)
+                    self._hipgraph = False
+            if self._hipgraph is not False:
+                loss = self._hipgraph.run(inputs, targets)
+            else:
+                loss = self._train_step_simple(inputs, targets)
         else:
             loss = self._train_step_simple(inputs, targets)
         if (self._gc_wrapped

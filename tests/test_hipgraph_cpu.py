@@ -1,0 +1,52 @@
+"""hipGraph step gating (runtime/hipgraph.py): on CPU / ineligible
+setups the engine must fall back to the eager path and keep training."""
+
+import torch
+import torch.nn as nn
+
+
+def _mlp():
+    torch.manual_seed(3)
+    return nn.Sequential(nn.Linear(16, 32), nn.ReLU(), nn.Linear(32, 4))
+
+
+def test_hip_graph_requested_on_cpu_falls_back_to_eager():
+    import easyparallellibrary_amd as epl
+
+    epl.init(epl.Config({"kernel.hip_graph": True}))
+    with epl.replicate(device_count=1):
+        model = _mlp()
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    torch.manual_seed(7)
+    x, y = torch.randn(8, 16), torch.randn(8, 4)
+    losses = [float(engine.train_step(x, y)) for _ in range(4)]
+    assert engine._hipgraph is False  # ineligible: no GPU
+    assert losses[-1] < losses[0]
+
+
+def test_eligibility_reasons():
+    import easyparallellibrary_amd as epl
+    from easyparallellibrary_amd.runtime import hipgraph
+
+    epl.init(epl.Config({"kernel.hip_graph": True}))
+    with epl.replicate(device_count=1):
+        model = nn.Sequential(nn.Linear(8, 8), nn.Dropout(0.1))
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    ok, reason = hipgraph.eligible(engine)
+    assert not ok
+    # on CPU the GPU gate fires first; the dropout detector is still
+    # exercised directly
+    assert hipgraph._find_dropout(model) is not None
+
+
+def test_hip_graph_off_by_default():
+    import easyparallellibrary_amd as epl
+
+    epl.init()
+    with epl.replicate(device_count=1):
+        model = _mlp()
+    engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
+                        lr=1e-2)
+    assert engine._hipgraph is False
