@@ -191,8 +191,14 @@ class ExpertParallelMLP(nn.Module):
         # slot assignment per expert (ordered, capacity-dropped)
         order = torch.argsort(flat_e, stable=True)
         fe, ft, fw = flat_e[order], flat_t[order], flat_w[order]
-        # position within expert via segmented arange
-        counts = torch.bincount(fe, minlength=self.num_experts)
+        # position within expert via segmented arange.  scatter_add of
+        # ones instead of bincount: bincount infers its output size on
+        # the host (device sync), which both stalls the step and is
+        # hipGraph-capture-unsupported; integer atomics make the
+        # scatter_add exact and order-independent.
+        counts = torch.zeros(self.num_experts, dtype=torch.long,
+                             device=x.device)
+        counts.scatter_add_(0, fe, torch.ones_like(fe))
         seg_start = torch.nn.functional.pad(counts.cumsum(0), (1, 0))[:-1]
         pos_in_e = torch.arange(fe.numel(), device=x.device) - seg_start[fe]
         native = (_moe_native_dispatch() and use_native(x)

@@ -83,11 +83,12 @@ class HipGraphStep:
 
     First ``WARMUP`` calls run the eager step on a side stream (per the
     torch CUDA-graphs recipe: materializes autograd state and hipBLASLt
-    workspaces at the capture shapes); the next call captures, then
-    every call replays.
+    workspaces at the capture shapes); the next call captures AND
+    replays, then every call replays.  WARMUP=2 keeps the one-time
+    capture inside a typical 3-step bench warmup.
     """
 
-    WARMUP = 3
+    WARMUP = 2
 
     def __init__(self, engine):
         self.engine = engine
@@ -133,7 +134,23 @@ class HipGraphStep:
             if self.calls < self.WARMUP:
                 self.calls += 1
                 return self._warmup_step(inputs, targets)
-            self._capture(inputs, targets)
+            try:
+                self._capture(inputs, targets)
+            except Exception as exc:
+                # a model op turned out not to be capture-safe (e.g. an
+                # op that syncs): recover the stream, disable graphing
+                # for this engine, and run this and all later steps
+                # eagerly — nothing executed during the failed capture,
+                # so the eager rerun IS the step
+                self.graph = None
+                torch.cuda.synchronize()
+                for r in self.engine.reducers:
+                    r.reset()   # hooks may have half-fired mid-capture
+                logger.warning(
+                    "hipGraph capture failed (%s) — falling back to the "
+                    "eager step for this engine", exc)
+                self.engine._hipgraph = False
+                return self.engine._train_step_simple(inputs, targets)
         if (inputs.shape != self.static_in.shape
                 or targets.shape != self.static_tgt.shape):
             raise RuntimeError(
