@@ -178,7 +178,18 @@ def build_moe_bench(args, epl, world, on_gpu, dtype):
         ParallelCrossEntropy)
     batch = args.batch or 8
     seq = args.seq_len if args.seq_len != 512 else 1024
-    epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+    # the world-1 MoE step is host-launch-bound (~19.6 ms of kernels in
+    # a 38 ms step, profiles/r2_final_moe_stats.txt): capture it into a
+    # hipGraph (runtime/hipgraph.py).  Ineligible setups (world>1,
+    # --dropout) fall back to eager inside the engine.  The DP-scaling
+    # configs deliberately stay eager so the driver's N=1 baseline and
+    # N>1 ranks run the same step; opt in there with
+    # EPL_KERNEL_HIP_GRAPH=1 (bert_dp same-box A/B: 639.7 -> 684.0).
+    epl.init(epl.Config({
+        "cluster.colocate_split_and_replicate": True,
+        "kernel.hip_graph":
+            os.environ.get("EPL_KERNEL_HIP_GRAPH", "1") == "1",
+    }))
     vocab = 32000
     layers = args.layers or 12
     hidden = args.hidden or 1024
@@ -307,13 +318,6 @@ def _enable_tunableop():
 def main():
     args = parse_args()
     _enable_tunableop()
-    # hipGraph-captured step where capture-safe (runtime/hipgraph.py):
-    # the engine itself gates on world size / dropout / GC / offload /
-    # ZeRO and falls back to eager with a logged reason, and a failed
-    # capture also falls back, so this is safe to request everywhere.
-    # Same-box A/B (bert_dp, 1xMI355X): eager 639.7 vs graphed 684.0
-    # samples/s.  Opt out with EPL_KERNEL_HIP_GRAPH=0.
-    os.environ.setdefault("EPL_KERNEL_HIP_GRAPH", "1")
     import easyparallellibrary_amd as epl
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
