@@ -103,10 +103,15 @@ def build_gpt2_bench(args, epl, world, on_gpu, dtype):
     nmb = args.micro_batch or (4 if pp > 1 else 1)
     gc = args.gc if args.gc is not None else "auto"
     offload = args.offload if args.offload is not None else ""
+    # world-1 gpt2 steps are launch-heavy (GC recompute doubles the
+    # kernel count): hipGraph-capture where safe; world>1 / --dropout /
+    # offload fall back to eager inside the engine
     epl.init(epl.Config({
         "pipeline.num_micro_batch": nmb,
         "gradient_checkpoint.type": gc,
         "offload.level": offload,
+        "kernel.hip_graph":
+            os.environ.get("EPL_KERNEL_HIP_GRAPH", "1") == "1",
     }))
     vocab = 50264
     cfg = args.model or "gpt2-xl"

@@ -40,10 +40,18 @@ class CheckpointWrapper(nn.Module):
     def __init__(self, inner):
         super().__init__()
         self.inner = inner
+        # preserve/restore RNG state around the recompute ONLY when the
+        # wrapped block actually consumes randomness (dropout): the
+        # get/set_rng_state round-trip is host work per block per step,
+        # and it blocks hipGraph capture (runtime/hipgraph.py) — for the
+        # dropout-free blocks the recompute is deterministic without it
+        from easyparallellibrary_amd.runtime.hipgraph import _find_dropout
+        self.preserve_rng = _find_dropout(inner) is not None
 
     def forward(self, *args, **kwargs):
         if self.enabled and torch.is_grad_enabled() and self.training:
             return checkpoint(self.inner, *args, use_reentrant=False,
+                              preserve_rng_state=self.preserve_rng,
                               **kwargs)
         return self.inner(*args, **kwargs)
 

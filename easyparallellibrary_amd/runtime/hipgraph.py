@@ -62,8 +62,11 @@ def eligible(engine):
         return False, "num_micro_batch {}".format(engine.num_micro_batch)
     if engine.amp.enabled and engine.dtype != torch.bfloat16:
         return False, "AMP loss scaler"
-    if getattr(engine, "_gc_wrapped", False):
-        return False, "gradient checkpointing"
+    # gradient checkpointing IS capture-safe here: the non-reentrant
+    # recompute runs inside the captured backward and records its
+    # kernels, and dropout-free wrappers skip the RNG save/restore
+    # (CheckpointWrapper.preserve_rng) — the dropout gate below blocks
+    # the only RNG-bearing case
     if engine.config.zero.level:
         return False, "ZeRO"
     if engine.config.offload.level:
