@@ -50,3 +50,22 @@ def test_hip_graph_off_by_default():
     engine = epl.Engine(model, loss_fn=nn.MSELoss(), optimizer="adamw",
                         lr=1e-2)
     assert engine._hipgraph is False
+
+
+def test_checkpoint_wrapper_rng_preservation_flag():
+    """Dropout-bearing checkpoint blocks keep the RNG save/restore;
+    clean blocks skip it (host overhead + capture blocker)."""
+    import easyparallellibrary_amd as epl
+    epl.init()
+    from easyparallellibrary_amd.runtime.gc import CheckpointWrapper
+    w_drop = CheckpointWrapper(nn.Sequential(nn.Linear(4, 4),
+                                             nn.Dropout(0.1)))
+    w_clean = CheckpointWrapper(nn.Linear(4, 4))
+    assert w_drop.preserve_rng is True
+    assert w_clean.preserve_rng is False
+    # and the clean wrapper still recomputes correctly
+    x = torch.randn(3, 4, requires_grad=True)
+    w_clean.train()
+    y = w_clean(x).sum()
+    y.backward()
+    assert x.grad is not None
