@@ -143,6 +143,7 @@ print("worker ok", engine.rank)
 
 @pytest.mark.parametrize("script", ["train_bert_dp.py",
                                     "train_bert_pipeline.py",
+                                    "train_bert_zero.py",
                                     "train_moe.py",
                                     "train_long_context.py"])
 def test_examples_tiny_cpu(script):
