@@ -16,7 +16,13 @@ from easyparallellibrary_amd.ops.distributed_losses import ParallelCrossEntropy
 # EPL_EXAMPLE_TINY=1 shrinks everything for a CPU smoke run
 TINY = os.environ.get("EPL_EXAMPLE_TINY", "0") == "1"
 
-epl.init(epl.Config({"cluster.colocate_split_and_replicate": True}))
+epl.init(epl.Config({
+    "cluster.colocate_split_and_replicate": True,
+    # small-batch MoE steps are host-launch-bound: capture fwd+bwd into
+    # a hipGraph and replay (single-rank only; multi-rank runs fall
+    # back to eager inside the engine with a logged reason)
+    "kernel.hip_graph": True,
+}))
 world = int(os.environ.get("WORLD_SIZE", "1"))
 if TINY:
     model = build_moe_transformer(world=world, layers=2, hidden=64, heads=2,
