@@ -14,10 +14,12 @@ training path only —
 - single-rank world: every communicator is a LocalCommunicator, so the
   gradient-reducer hooks enqueue no collectives and the captured graph
   is pure compute (capturing RCCL ops is deliberately out of scope);
-- no AMP loss scaler (bf16 runs scale-free), no gradient checkpointing,
-  no ZeRO / CPU offload / PBO eager apply / sparse-grad handlers — each
-  of those runs host-side logic between backward kernels that a replay
-  would skip;
+- no AMP loss scaler (bf16 runs scale-free), no ZeRO / CPU offload /
+  PBO eager apply / sparse-grad handlers — each of those runs host-side
+  logic between backward kernels that a replay would skip.  Gradient
+  checkpointing IS allowed: the non-reentrant recompute records its
+  kernels inside the captured backward, and dropout-free wrappers skip
+  the RNG save/restore;
 - no dropout anywhere: torch's graph-safe philox covers torch ops, but
   the in-kernel attention dropout takes its seed as a HOST argument,
   which a replay would freeze;
