@@ -38,7 +38,9 @@ dims = (dict(layers=2, hidden=64, heads=4, ffn=128, vocab_size=512,
              max_pos=64) if TINY
         else dict(layers=24, hidden=1024, heads=16, ffn=4096,
                   vocab_size=32000, max_pos=1024))
-model = build_tp_pipeline(stages=2, tp=tp, **dims)
+# a 2-stage pipeline needs >= 2 ranks; at world 1 run the same model
+# as one serial stage so the script stays runnable everywhere
+model = build_tp_pipeline(stages=2 if world > 1 else 1, tp=tp, **dims)
 V = dims["vocab_size"]
 
 

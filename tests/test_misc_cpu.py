@@ -145,7 +145,8 @@ print("worker ok", engine.rank)
                                     "train_bert_pipeline.py",
                                     "train_bert_zero.py",
                                     "train_moe.py",
-                                    "train_long_context.py"])
+                                    "train_long_context.py",
+                                    "train_tp_pipeline.py"])
 def test_examples_tiny_cpu(script):
     """Every example runs end-to-end in tiny mode on CPU."""
     import subprocess
